@@ -1,0 +1,62 @@
+# mlsl_amd build: C++17 core + CDNA4 (gfx950) HIP kernels + RCCL, built with
+# hipcc into an in-tree shared library the Python ctypes binding loads.
+# No MPI, no CUDA, no external deps beyond ROCm.
+
+HIPCC      ?= hipcc
+GPU_ARCH   ?= gfx950
+BUILD      := build
+LIB        := mlsl_amd/libmlsl_amd.so
+SELFTEST   := $(BUILD)/schedule_selftest
+
+CXXFLAGS   := -O3 -std=c++17 -fPIC -Wall -Wextra -Wno-unused-parameter \
+              --offload-arch=$(GPU_ARCH) -I/opt/rocm/include
+LDFLAGS    := -shared -fPIC -L/opt/rocm/lib -lrccl -lamdhip64 -pthread
+
+CSRC := \
+    mlsl_amd/csrc/core/log.cpp \
+    mlsl_amd/csrc/core/config.cpp \
+    mlsl_amd/csrc/comm/schedule.cpp \
+    mlsl_amd/csrc/comm/bootstrap.cpp \
+    mlsl_amd/csrc/comm/mesh.cpp \
+    mlsl_amd/csrc/comm/group.cpp \
+    mlsl_amd/csrc/comm/request.cpp \
+    mlsl_amd/csrc/comm/engine.cpp \
+    mlsl_amd/csrc/comm/context.cpp \
+    mlsl_amd/csrc/comm/device_comm.cpp \
+    mlsl_amd/csrc/dl/environment.cpp \
+    mlsl_amd/csrc/dl/session.cpp \
+    mlsl_amd/csrc/bind/c_api.cpp
+
+HIPSRC := mlsl_amd/csrc/hip/kernels.hip
+
+OBJS := $(CSRC:%.cpp=$(BUILD)/%.o) $(HIPSRC:%.hip=$(BUILD)/%.o)
+
+.PHONY: all lib selftest clean test
+
+all: lib selftest
+
+lib: $(LIB)
+
+$(LIB): $(OBJS)
+	$(HIPCC) $(OBJS) $(LDFLAGS) -o $@
+
+$(BUILD)/%.o: %.cpp
+	@mkdir -p $(dir $@)
+	$(HIPCC) $(CXXFLAGS) -c $< -o $@
+
+$(BUILD)/%.o: %.hip
+	@mkdir -p $(dir $@)
+	$(HIPCC) $(CXXFLAGS) -x hip -c $< -o $@
+
+selftest: $(SELFTEST)
+
+$(SELFTEST): $(BUILD)/mlsl_amd/csrc/tests/schedule_selftest.o $(BUILD)/mlsl_amd/csrc/comm/schedule.o $(BUILD)/mlsl_amd/csrc/core/log.o
+	@mkdir -p $(dir $@)
+	$(HIPCC) $^ -L/opt/rocm/lib -lamdhip64 -pthread -o $@
+
+test: all
+	$(SELFTEST)
+	python -m pytest tests/ -x -q -m "not gpu"
+
+clean:
+	rm -rf $(BUILD) $(LIB)

@@ -1,0 +1,592 @@
+// Flat C API over the C++ core (reference analog: src/c_bind.cpp with its
+// TRY_CATCH_RETURN convention; ~95 functions). Every call catches C++
+// exceptions and returns MLSL_FAILURE with the message retrievable via
+// mlsl_last_error().
+#include "../include/mlsl/c_api.h"
+
+#include <cstring>
+#include <string>
+
+#include "../dl/session.hpp"
+#include "../include/mlsl/mlsl.hpp"
+
+using namespace mlsl;
+
+namespace {
+thread_local std::string g_last_error;
+}
+
+extern "C" {
+
+const char* mlsl_last_error(void) { return g_last_error.c_str(); }
+
+#define C_TRY try {
+#define C_CATCH                                                                \
+    return MLSL_SUCCESS;                                                       \
+    } catch (const std::exception& e) {                                        \
+        g_last_error = e.what();                                               \
+        return MLSL_FAILURE;                                                   \
+    } catch (...) {                                                            \
+        g_last_error = "unknown error";                                        \
+        return MLSL_FAILURE;                                                   \
+    }
+
+/* ---- environment ---- */
+
+int mlsl_init(int rank, int size) {
+    C_TRY Environment::GetEnv().Init(rank, size);
+    C_CATCH
+}
+
+int mlsl_finalize(void) {
+    C_TRY Environment::GetEnv().Finalize();
+    C_CATCH
+}
+
+int mlsl_initialized(int* out) {
+    C_TRY* out = Environment::GetEnv().IsInitialized() ? 1 : 0;
+    C_CATCH
+}
+
+int mlsl_get_version(int* out) {
+    C_TRY* out = Environment::GetVersion();
+    C_CATCH
+}
+
+int mlsl_rank(size_t* out) {
+    C_TRY* out = Environment::GetEnv().GetProcessIdx();
+    C_CATCH
+}
+
+int mlsl_world_size(size_t* out) {
+    C_TRY* out = Environment::GetEnv().GetProcessCount();
+    C_CATCH
+}
+
+int mlsl_alloc(size_t sz, size_t align, void** out) {
+    C_TRY* out = Environment::GetEnv().Alloc(sz, align);
+    C_CATCH
+}
+
+int mlsl_dealloc(void* ptr) {
+    C_TRY Environment::GetEnv().Free(ptr);
+    C_CATCH
+}
+
+int mlsl_wait(mlsl_request req, void** result) {
+    C_TRY void* r = Environment::GetEnv().Wait(static_cast<CommRequest*>(req));
+    if (result) *result = r;
+    C_CATCH
+}
+
+int mlsl_test(mlsl_request req, int* done, void** result) {
+    C_TRY void* r = nullptr;
+    bool d = Environment::GetEnv().Test(static_cast<CommRequest*>(req), &r);
+    if (done) *done = d ? 1 : 0;
+    if (result) *result = r;
+    C_CATCH
+}
+
+int mlsl_set_quant_params(size_t block_elems) {
+    C_TRY QuantParams p;
+    p.block_elems = block_elems;
+    Environment::GetEnv().SetQuantizationParams(p);
+    C_CATCH
+}
+
+/* ---- distribution ---- */
+
+int mlsl_distribution_create(size_t d, size_t m, mlsl_distribution* out) {
+    C_TRY* out = Environment::GetEnv().CreateDistribution(d, m);
+    C_CATCH
+}
+
+int mlsl_distribution_create_with_colors(int dc, int mc, mlsl_distribution* out) {
+    C_TRY* out = Environment::GetEnv().CreateDistributionWithColors(dc, mc);
+    C_CATCH
+}
+
+int mlsl_distribution_free(mlsl_distribution d) {
+    C_TRY Environment::GetEnv().DeleteDistribution(static_cast<Distribution*>(d));
+    C_CATCH
+}
+
+int mlsl_distribution_process_idx(mlsl_distribution d, mlsl_group g, size_t* out) {
+    C_TRY* out = static_cast<Distribution*>(d)->GetProcessIdx(static_cast<GroupKind>(g));
+    C_CATCH
+}
+
+int mlsl_distribution_process_count(mlsl_distribution d, mlsl_group g, size_t* out) {
+    C_TRY* out = static_cast<Distribution*>(d)->GetProcessCount(static_cast<GroupKind>(g));
+    C_CATCH
+}
+
+int mlsl_distribution_barrier(mlsl_distribution d, mlsl_group g) {
+    C_TRY static_cast<Distribution*>(d)->Barrier(static_cast<GroupKind>(g));
+    C_CATCH
+}
+
+#define DIST(d) static_cast<Distribution*>(d)
+#define DT(x) static_cast<DataType>(x)
+#define ROP(x) static_cast<ReduceOp>(x)
+#define GK(x) static_cast<GroupKind>(x)
+
+int mlsl_distribution_bcast(mlsl_distribution d, void* buf, size_t count, mlsl_data_type dt,
+                            size_t root, mlsl_group g, mlsl_request* out) {
+    C_TRY* out = DIST(d)->Bcast(buf, count, DT(dt), root, GK(g));
+    C_CATCH
+}
+
+int mlsl_distribution_reduce(mlsl_distribution d, const void* sbuf, void* rbuf, size_t count,
+                             mlsl_data_type dt, mlsl_reduction op, size_t root, mlsl_group g,
+                             mlsl_request* out) {
+    C_TRY* out = DIST(d)->Reduce(sbuf, rbuf, count, DT(dt), ROP(op), root, GK(g));
+    C_CATCH
+}
+
+int mlsl_distribution_all_reduce(mlsl_distribution d, const void* sbuf, void* rbuf,
+                                 size_t count, mlsl_data_type dt, mlsl_reduction op,
+                                 mlsl_group g, mlsl_request* out) {
+    C_TRY* out = DIST(d)->AllReduce(sbuf, rbuf, count, DT(dt), ROP(op), GK(g));
+    C_CATCH
+}
+
+int mlsl_distribution_all_to_all(mlsl_distribution d, const void* sbuf, size_t send_count,
+                                 void* rbuf, mlsl_data_type dt, mlsl_group g,
+                                 mlsl_request* out) {
+    C_TRY* out = DIST(d)->AlltoAll(sbuf, send_count, rbuf, DT(dt), GK(g));
+    C_CATCH
+}
+
+int mlsl_distribution_all_to_allv(mlsl_distribution d, const void* sbuf, const size_t* scnt,
+                                  const size_t* soff, void* rbuf, const size_t* rcnt,
+                                  const size_t* roff, mlsl_data_type dt, mlsl_group g,
+                                  mlsl_request* out) {
+    C_TRY* out = DIST(d)->AlltoAllv(sbuf, scnt, soff, rbuf, rcnt, roff, DT(dt), GK(g));
+    C_CATCH
+}
+
+int mlsl_distribution_gather(mlsl_distribution d, const void* sbuf, size_t send_count,
+                             void* rbuf, mlsl_data_type dt, size_t root, mlsl_group g,
+                             mlsl_request* out) {
+    C_TRY* out = DIST(d)->Gather(sbuf, send_count, rbuf, DT(dt), root, GK(g));
+    C_CATCH
+}
+
+int mlsl_distribution_all_gather(mlsl_distribution d, const void* sbuf, size_t send_count,
+                                 void* rbuf, mlsl_data_type dt, mlsl_group g,
+                                 mlsl_request* out) {
+    C_TRY* out = DIST(d)->AllGather(sbuf, send_count, rbuf, DT(dt), GK(g));
+    C_CATCH
+}
+
+int mlsl_distribution_all_gatherv(mlsl_distribution d, const void* sbuf, size_t send_count,
+                                  void* rbuf, const size_t* rcnt, mlsl_data_type dt,
+                                  mlsl_group g, mlsl_request* out) {
+    C_TRY* out = DIST(d)->AllGatherv(sbuf, send_count, rbuf, rcnt, DT(dt), GK(g));
+    C_CATCH
+}
+
+int mlsl_distribution_scatter(mlsl_distribution d, const void* sbuf, void* rbuf,
+                              size_t recv_count, mlsl_data_type dt, size_t root, mlsl_group g,
+                              mlsl_request* out) {
+    C_TRY* out = DIST(d)->Scatter(sbuf, rbuf, recv_count, DT(dt), root, GK(g));
+    C_CATCH
+}
+
+int mlsl_distribution_reduce_scatter(mlsl_distribution d, const void* sbuf, void* rbuf,
+                                     size_t recv_count, mlsl_data_type dt, mlsl_reduction op,
+                                     mlsl_group g, mlsl_request* out) {
+    C_TRY* out = DIST(d)->ReduceScatter(sbuf, rbuf, recv_count, DT(dt), ROP(op), GK(g));
+    C_CATCH
+}
+
+/* ---- session / planner ---- */
+
+#define SES(s) static_cast<Session*>(s)
+#define OP(o) static_cast<Operation*>(o)
+#define ACT(a) static_cast<Activation*>(a)
+#define PS(p) static_cast<ParameterSet*>(p)
+#define REG(i) static_cast<OperationRegInfo*>(i)
+#define STATS(x) static_cast<Statistics*>(x)
+#define CBI(b) static_cast<CommBlockInfo*>(b)
+
+int mlsl_session_create(mlsl_phase phase, mlsl_session* out) {
+    C_TRY* out = Environment::GetEnv().CreateSession(static_cast<PhaseKind>(phase));
+    C_CATCH
+}
+
+int mlsl_session_free(mlsl_session s) {
+    C_TRY Environment::GetEnv().DeleteSession(SES(s));
+    C_CATCH
+}
+
+int mlsl_session_set_global_minibatch_size(mlsl_session s, size_t mb) {
+    C_TRY SES(s)->SetGlobalMinibatchSize(mb);
+    C_CATCH
+}
+
+int mlsl_session_get_global_minibatch_size(mlsl_session s, size_t* out) {
+    C_TRY* out = SES(s)->GetGlobalMinibatchSize();
+    C_CATCH
+}
+
+int mlsl_session_create_op_reg_info(mlsl_session s, mlsl_op_type ot, mlsl_op_reg_info* out) {
+    C_TRY* out = SES(s)->CreateOperationRegInfo(static_cast<OpKind>(ot));
+    C_CATCH
+}
+
+int mlsl_session_delete_op_reg_info(mlsl_session s, mlsl_op_reg_info info) {
+    C_TRY SES(s)->DeleteOperationRegInfo(REG(info));
+    C_CATCH
+}
+
+int mlsl_session_add_operation(mlsl_session s, mlsl_op_reg_info info, mlsl_distribution d,
+                               size_t* out_idx) {
+    C_TRY* out_idx = SES(s)->AddOperation(REG(info), DIST(d));
+    C_CATCH
+}
+
+int mlsl_session_remove_operations(mlsl_session s) {
+    C_TRY SES(s)->RemoveOperations();
+    C_CATCH
+}
+
+int mlsl_session_get_operation_count(mlsl_session s, size_t* out) {
+    C_TRY* out = SES(s)->GetOperationCount();
+    C_CATCH
+}
+
+int mlsl_session_get_operation(mlsl_session s, size_t idx, mlsl_operation* out) {
+    C_TRY* out = SES(s)->GetOperation(idx);
+    C_CATCH
+}
+
+int mlsl_session_commit(mlsl_session s) {
+    C_TRY SES(s)->Commit();
+    C_CATCH
+}
+
+int mlsl_session_get_stats(mlsl_session s, mlsl_statistics* out) {
+    C_TRY* out = SES(s)->GetStats();
+    C_CATCH
+}
+
+/* ---- op reg info ---- */
+
+int mlsl_op_reg_info_set_name(mlsl_op_reg_info i, const char* name) {
+    C_TRY REG(i)->SetName(name);
+    C_CATCH
+}
+
+int mlsl_op_reg_info_add_input(mlsl_op_reg_info i, size_t c, size_t s, mlsl_data_type dt,
+                               size_t* out_idx) {
+    C_TRY size_t r = REG(i)->AddInput(c, s, DT(dt));
+    if (out_idx) *out_idx = r;
+    C_CATCH
+}
+
+int mlsl_op_reg_info_add_output(mlsl_op_reg_info i, size_t c, size_t s, mlsl_data_type dt,
+                                size_t* out_idx) {
+    C_TRY size_t r = REG(i)->AddOutput(c, s, DT(dt));
+    if (out_idx) *out_idx = r;
+    C_CATCH
+}
+
+int mlsl_op_reg_info_add_parameter_set(mlsl_op_reg_info i, size_t kc, size_t ks,
+                                       mlsl_data_type dt, int dist_update,
+                                       mlsl_compression comp, size_t* out_idx) {
+    C_TRY size_t r = REG(i)->AddParameterSet(kc, ks, DT(dt), dist_update != 0,
+                                             static_cast<Compression>(comp));
+    if (out_idx) *out_idx = r;
+    C_CATCH
+}
+
+int mlsl_op_reg_info_validate(mlsl_op_reg_info i, mlsl_distribution d) {
+    C_TRY REG(i)->Validate(DIST(d));
+    C_CATCH
+}
+
+/* ---- operation ---- */
+
+int mlsl_operation_set_distribution(mlsl_operation o, mlsl_distribution d) {
+    C_TRY OP(o)->SetDistribution(DIST(d));
+    C_CATCH
+}
+
+int mlsl_operation_get_distribution(mlsl_operation o, mlsl_distribution* out) {
+    C_TRY* out = OP(o)->GetDistribution();
+    C_CATCH
+}
+
+int mlsl_operation_set_prev(mlsl_operation o, mlsl_operation prev, size_t a, size_t b) {
+    C_TRY OP(o)->SetPrev(OP(prev), a, b);
+    C_CATCH
+}
+
+int mlsl_operation_set_next(mlsl_operation o, mlsl_operation next, size_t a, size_t b) {
+    C_TRY OP(o)->SetNext(OP(next), a, b);
+    C_CATCH
+}
+
+int mlsl_operation_get_name(mlsl_operation o, const char** out) {
+    C_TRY* out = OP(o)->GetName();
+    C_CATCH
+}
+
+int mlsl_operation_get_global_minibatch_size(mlsl_operation o, size_t* out) {
+    C_TRY* out = OP(o)->GetGlobalMinibatchSize();
+    C_CATCH
+}
+
+int mlsl_operation_get_local_minibatch_size(mlsl_operation o, size_t* out) {
+    C_TRY* out = OP(o)->GetLocalMinibatchSize();
+    C_CATCH
+}
+
+int mlsl_operation_get_global_minibatch_offset(mlsl_operation o, size_t* out) {
+    C_TRY* out = OP(o)->GetGlobalMinibatchOffset();
+    C_CATCH
+}
+
+int mlsl_operation_get_input_count(mlsl_operation o, size_t* out) {
+    C_TRY* out = OP(o)->GetInputCount();
+    C_CATCH
+}
+
+int mlsl_operation_get_input(mlsl_operation o, size_t idx, mlsl_activation* out) {
+    C_TRY* out = OP(o)->GetInput(idx);
+    C_CATCH
+}
+
+int mlsl_operation_get_output_count(mlsl_operation o, size_t* out) {
+    C_TRY* out = OP(o)->GetOutputCount();
+    C_CATCH
+}
+
+int mlsl_operation_get_output(mlsl_operation o, size_t idx, mlsl_activation* out) {
+    C_TRY* out = OP(o)->GetOutput(idx);
+    C_CATCH
+}
+
+int mlsl_operation_get_parameter_set_count(mlsl_operation o, size_t* out) {
+    C_TRY* out = OP(o)->GetParameterSetCount();
+    C_CATCH
+}
+
+int mlsl_operation_get_parameter_set(mlsl_operation o, size_t idx, mlsl_parameter_set* out) {
+    C_TRY* out = OP(o)->GetParameterSet(idx);
+    C_CATCH
+}
+
+/* ---- activation ---- */
+
+int mlsl_activation_get_global_fm_count(mlsl_activation a, size_t* out) {
+    C_TRY* out = ACT(a)->GetGlobalFmCount();
+    C_CATCH
+}
+
+int mlsl_activation_get_global_fm_offset(mlsl_activation a, size_t* out) {
+    C_TRY* out = ACT(a)->GetGlobalFmOffset();
+    C_CATCH
+}
+
+int mlsl_activation_get_local_fm_count(mlsl_activation a, size_t* out) {
+    C_TRY* out = ACT(a)->GetLocalFmCount();
+    C_CATCH
+}
+
+int mlsl_activation_get_fm_size(mlsl_activation a, size_t* out) {
+    C_TRY* out = ACT(a)->GetFmSize();
+    C_CATCH
+}
+
+int mlsl_activation_get_data_type(mlsl_activation a, mlsl_data_type* out) {
+    C_TRY* out = static_cast<mlsl_data_type>(ACT(a)->GetDataType());
+    C_CATCH
+}
+
+int mlsl_activation_get_comm_buf_size(mlsl_activation a, size_t* out) {
+    C_TRY* out = ACT(a)->GetCommBufSize();
+    C_CATCH
+}
+
+int mlsl_activation_get_pack_block_count(mlsl_activation a, size_t* out) {
+    C_TRY* out = ACT(a)->GetPackBlockCount();
+    C_CATCH
+}
+
+int mlsl_activation_get_unpack_block_count(mlsl_activation a, size_t* out) {
+    C_TRY* out = ACT(a)->GetUnpackBlockCount();
+    C_CATCH
+}
+
+int mlsl_activation_get_pack_block(mlsl_activation a, size_t idx, mlsl_comm_block_info* out) {
+    C_TRY* out = const_cast<CommBlockInfo*>(ACT(a)->GetPackBlock(idx));
+    C_CATCH
+}
+
+int mlsl_activation_get_unpack_block(mlsl_activation a, size_t idx, mlsl_comm_block_info* out) {
+    C_TRY* out = const_cast<CommBlockInfo*>(ACT(a)->GetUnpackBlock(idx));
+    C_CATCH
+}
+
+int mlsl_activation_start_comm(mlsl_activation a, void* buf) {
+    C_TRY ACT(a)->StartComm(buf);
+    C_CATCH
+}
+
+int mlsl_activation_wait_comm(mlsl_activation a, void** out) {
+    C_TRY void* r = ACT(a)->WaitComm();
+    if (out) *out = r;
+    C_CATCH
+}
+
+/* ---- comm block info ---- */
+
+int mlsl_comm_block_info_get_mb_offset(mlsl_comm_block_info b, size_t* out) {
+    C_TRY* out = CBI(b)->GetMbOffset();
+    C_CATCH
+}
+int mlsl_comm_block_info_get_mb_count(mlsl_comm_block_info b, size_t* out) {
+    C_TRY* out = CBI(b)->GetMbCount();
+    C_CATCH
+}
+int mlsl_comm_block_info_get_fm_offset(mlsl_comm_block_info b, size_t* out) {
+    C_TRY* out = CBI(b)->GetFmOffset();
+    C_CATCH
+}
+int mlsl_comm_block_info_get_fm_count(mlsl_comm_block_info b, size_t* out) {
+    C_TRY* out = CBI(b)->GetFmCount();
+    C_CATCH
+}
+int mlsl_comm_block_info_get_fm_size(mlsl_comm_block_info b, size_t* out) {
+    C_TRY* out = CBI(b)->GetFmSize();
+    C_CATCH
+}
+int mlsl_comm_block_info_get_data_type(mlsl_comm_block_info b, mlsl_data_type* out) {
+    C_TRY* out = static_cast<mlsl_data_type>(CBI(b)->GetDataType());
+    C_CATCH
+}
+int mlsl_comm_block_info_get_buf_offset(mlsl_comm_block_info b, size_t* out) {
+    C_TRY* out = CBI(b)->GetBufOffset();
+    C_CATCH
+}
+
+/* ---- parameter set ---- */
+
+int mlsl_parameter_set_get_global_kernel_count(mlsl_parameter_set p, size_t* out) {
+    C_TRY* out = PS(p)->GetGlobalKernelCount();
+    C_CATCH
+}
+int mlsl_parameter_set_get_global_kernel_offset(mlsl_parameter_set p, size_t* out) {
+    C_TRY* out = PS(p)->GetGlobalKernelOffset();
+    C_CATCH
+}
+int mlsl_parameter_set_get_local_kernel_count(mlsl_parameter_set p, size_t* out) {
+    C_TRY* out = PS(p)->GetLocalKernelCount();
+    C_CATCH
+}
+int mlsl_parameter_set_get_owned_kernel_count(mlsl_parameter_set p, size_t* out) {
+    C_TRY* out = PS(p)->GetOwnedKernelCount();
+    C_CATCH
+}
+int mlsl_parameter_set_get_owned_kernel_offset(mlsl_parameter_set p, size_t* out) {
+    C_TRY* out = PS(p)->GetOwnedKernelOffset();
+    C_CATCH
+}
+int mlsl_parameter_set_get_kernel_size(mlsl_parameter_set p, size_t* out) {
+    C_TRY* out = PS(p)->GetKernelSize();
+    C_CATCH
+}
+int mlsl_parameter_set_get_data_type(mlsl_parameter_set p, mlsl_data_type* out) {
+    C_TRY* out = static_cast<mlsl_data_type>(PS(p)->GetDataType());
+    C_CATCH
+}
+int mlsl_parameter_set_is_distributed_update(mlsl_parameter_set p, int* out) {
+    C_TRY* out = PS(p)->IsDistributedUpdate() ? 1 : 0;
+    C_CATCH
+}
+int mlsl_parameter_set_start_gradient_comm(mlsl_parameter_set p, void* buf) {
+    C_TRY PS(p)->StartGradientComm(buf);
+    C_CATCH
+}
+int mlsl_parameter_set_wait_gradient_comm(mlsl_parameter_set p, void** out) {
+    C_TRY void* r = PS(p)->WaitGradientComm();
+    if (out) *out = r;
+    C_CATCH
+}
+int mlsl_parameter_set_test_gradient_comm(mlsl_parameter_set p, int* done, void** out) {
+    C_TRY void* r = nullptr;
+    bool d = PS(p)->TestGradientComm(&r);
+    if (done) *done = d ? 1 : 0;
+    if (out) *out = r;
+    C_CATCH
+}
+int mlsl_parameter_set_start_increment_comm(mlsl_parameter_set p, void* buf) {
+    C_TRY PS(p)->StartIncrementComm(buf);
+    C_CATCH
+}
+int mlsl_parameter_set_wait_increment_comm(mlsl_parameter_set p, void** out) {
+    C_TRY void* r = PS(p)->WaitIncrementComm();
+    if (out) *out = r;
+    C_CATCH
+}
+
+/* ---- statistics ---- */
+
+int mlsl_statistics_start(mlsl_statistics st) {
+    C_TRY STATS(st)->Start();
+    C_CATCH
+}
+int mlsl_statistics_stop(mlsl_statistics st) {
+    C_TRY STATS(st)->Stop();
+    C_CATCH
+}
+int mlsl_statistics_reset(mlsl_statistics st) {
+    C_TRY STATS(st)->Reset();
+    C_CATCH
+}
+int mlsl_statistics_is_enabled(mlsl_statistics st, int* out) {
+    C_TRY* out = STATS(st)->IsEnabled() ? 1 : 0;
+    C_CATCH
+}
+int mlsl_statistics_print(mlsl_statistics st) {
+    C_TRY STATS(st)->Print();
+    C_CATCH
+}
+int mlsl_statistics_get_isolation_comm_cycles(mlsl_statistics st, size_t op,
+                                              unsigned long long* out) {
+    C_TRY* out = STATS(st)->GetIsolationCommCycles(op);
+    C_CATCH
+}
+int mlsl_statistics_get_comm_size(mlsl_statistics st, size_t op, size_t* out) {
+    C_TRY* out = STATS(st)->GetCommSize(op);
+    C_CATCH
+}
+int mlsl_statistics_get_comm_cycles(mlsl_statistics st, size_t op, unsigned long long* out) {
+    C_TRY* out = STATS(st)->GetCommCycles(op);
+    C_CATCH
+}
+int mlsl_statistics_get_compute_cycles(mlsl_statistics st, size_t op, unsigned long long* out) {
+    C_TRY* out = STATS(st)->GetComputeCycles(op);
+    C_CATCH
+}
+int mlsl_statistics_get_total_isolation_comm_cycles(mlsl_statistics st, unsigned long long* out) {
+    C_TRY* out = STATS(st)->GetTotalIsolationCommCycles();
+    C_CATCH
+}
+int mlsl_statistics_get_total_comm_size(mlsl_statistics st, size_t* out) {
+    C_TRY* out = STATS(st)->GetTotalCommSize();
+    C_CATCH
+}
+int mlsl_statistics_get_total_comm_cycles(mlsl_statistics st, unsigned long long* out) {
+    C_TRY* out = STATS(st)->GetTotalCommCycles();
+    C_CATCH
+}
+int mlsl_statistics_get_total_compute_cycles(mlsl_statistics st, unsigned long long* out) {
+    C_TRY* out = STATS(st)->GetTotalComputeCycles();
+    C_CATCH
+}
+
+}  // extern "C"

@@ -1,0 +1,436 @@
+// RCCL/HIP device transport: per-group communicators (one per channel,
+// each on its own HIP stream), fused RCCL collectives as the baseline path,
+// and schedule-driven ncclSend/ncclRecv + local-reduce HIP kernel pipelines
+// as the custom path (MLSL_ALLREDUCE_ALGO=ring|rhd).
+//
+// Re-designs the reference process-mode backend (src/comm_ep.cpp): endpoint
+// servers -> channel streams; shm heap + ReplaceIn/Out staging -> HBM
+// buffers (hipMalloc) used directly; MPI_I* -> RCCL enqueues; MPI_Test
+// polling loop -> hipEventQuery from the progress thread.
+#include "device_comm.hpp"
+
+#include <hip/hip_runtime.h>
+#include <rccl/rccl.h>
+
+#include <cstdlib>
+#include <unordered_map>
+#include <vector>
+
+#include "../core/config.hpp"
+#include "../core/log.hpp"
+#include "../hip/kernels.hpp"
+#include "context.hpp"
+#include "device_state.hpp"
+#include "group.hpp"
+#include "request.hpp"
+
+namespace mlsl {
+
+#define HIP_CHECKD(cmd)                                                       \
+    do {                                                                      \
+        hipError_t e_ = (cmd);                                                \
+        if (e_ != hipSuccess)                                                 \
+            MLSL_THROW(std::string("HIP error: ") + hipGetErrorString(e_));   \
+    } while (0)
+
+#define NCCL_CHECK(cmd)                                                       \
+    do {                                                                      \
+        ncclResult_t r_ = (cmd);                                              \
+        if (r_ != ncclSuccess)                                                \
+            MLSL_THROW(std::string("RCCL error: ") + ncclGetErrorString(r_)); \
+    } while (0)
+
+DeviceReqState::~DeviceReqState() {
+    for (hipEvent_t e : events)
+        if (e) (void)hipEventDestroy(e);
+    if (tmp_dev) (void)hipFree(tmp_dev);
+}
+
+namespace {
+
+ncclDataType_t ToNccl(DataType dt) {
+    switch (dt) {
+        case DataType::F32: return ncclFloat32;
+        case DataType::F64: return ncclFloat64;
+        case DataType::U8: return ncclUint8;
+        case DataType::BF16: return ncclBfloat16;
+        case DataType::F16: return ncclFloat16;
+        case DataType::I32: return ncclInt32;
+        case DataType::I64: return ncclInt64;
+    }
+    return ncclFloat32;
+}
+
+ncclRedOp_t ToNcclOp(ReduceOp op) {
+    switch (op) {
+        case ReduceOp::SUM: return ncclSum;
+        case ReduceOp::MIN: return ncclMin;
+        case ReduceOp::MAX: return ncclMax;
+    }
+    return ncclSum;
+}
+
+// Per-group device communicators: `channels` RCCL comms, each with its own
+// stream — chunk c of a request runs on channel c % channels, so chunks of
+// one large message progress on independent RCCL rings over distinct xGMI
+// link schedules (the endpoint-server fan-out analog).
+struct GroupComms {
+    std::vector<ncclComm_t> comms;
+    std::vector<hipStream_t> streams;
+};
+
+class HipRuntime;
+HipRuntime* g_runtime = nullptr;
+
+class HipRuntime : public DeviceRuntime {
+  public:
+    HipRuntime(int device_id) : device_id_(device_id) {
+        HIP_CHECKD(hipSetDevice(device_id_));
+    }
+
+    ~HipRuntime() override {
+        for (auto& kv : group_comms_) {
+            for (auto c : kv.second.comms) ncclCommDestroy(c);
+            for (auto s : kv.second.streams) (void)hipStreamDestroy(s);
+        }
+        g_runtime = nullptr;
+    }
+
+    int DeviceId() const override { return device_id_; }
+
+    void* AllocDevice(size_t bytes) override {
+        void* p = nullptr;
+        HIP_CHECKD(hipMalloc(&p, bytes));
+        return p;
+    }
+
+    void FreeDevice(void* p) override { (void)hipFree(p); }
+
+    void Synchronize() override { HIP_CHECKD(hipDeviceSynchronize()); }
+
+    std::string Name() const override {
+        hipDeviceProp_t prop;
+        if (hipGetDeviceProperties(&prop, device_id_) == hipSuccess)
+            return std::string("hip:") + prop.gcnArchName;
+        return "hip:?";
+    }
+
+    // Collective over the WORLD: unique-id exchange runs on the bootstrap;
+    // only group members init comms.
+    void EnsureGroupComms(ProcessGroup* g) override {
+        if (group_comms_.count(g->Uid())) return;
+        Context& ctx = Context::Get();
+        const size_t nch = GlobalConfig().num_channels;
+        GroupComms gc;
+        for (size_t ch = 0; ch < nch; ++ch) {
+            ncclUniqueId id{};
+            if (g->IsMember() && g->MyIdx() == 0) NCCL_CHECK(ncclGetUniqueId(&id));
+            std::vector<ncclUniqueId> all(static_cast<size_t>(ctx.Size()));
+            ctx.Boot()->Allgather(&id, sizeof(ncclUniqueId), all.data());
+            if (g->IsMember() && g->Size() > 1) {
+                ncclUniqueId gid = all[static_cast<size_t>(g->WorldRank(0))];
+                ncclComm_t comm;
+                NCCL_CHECK(ncclCommInitRank(&comm, g->Size(), gid, g->MyIdx()));
+                hipStream_t s;
+                HIP_CHECKD(hipStreamCreateWithFlags(&s, hipStreamNonBlocking));
+                gc.comms.push_back(comm);
+                gc.streams.push_back(s);
+            }
+        }
+        group_comms_.emplace(g->Uid(), std::move(gc));
+        MLSL_LOG(DEBUG, "device comms ready for group uid=%d size=%d channels=%zu",
+                 g->Uid(), g->Size(), nch);
+    }
+
+    GroupComms& For(ProcessGroup* g) {
+        auto it = group_comms_.find(g->Uid());
+        MLSL_CHECK(it != group_comms_.end(), "group has no device comms");
+        return it->second;
+    }
+
+  private:
+    int device_id_;
+    std::unordered_map<int, GroupComms> group_comms_;
+};
+
+}  // namespace
+
+DeviceRuntime* CreateDeviceRuntime() {
+    int n = 0;
+    if (hipGetDeviceCount(&n) != hipSuccess || n <= 0) return nullptr;
+    int local_rank = 0;
+    if (const char* e = std::getenv("MLSL_LOCAL_RANK")) local_rank = std::atoi(e);
+    else if (const char* e2 = std::getenv("LOCAL_RANK")) local_rank = std::atoi(e2);
+    else local_rank = Bootstrap::EnvRank();
+    auto* rt = new HipRuntime(local_rank % n);
+    g_runtime = rt;
+    return rt;
+}
+
+void DeviceSetupRequest(CommRequest* req, DeviceReqState& st) {
+    // Persistent scratch for schedule-path chunks + quant/barrier staging.
+    size_t tmp = req->GetTmpBytes();
+    if (req->Spec().op == CollOp::BARRIER) tmp = std::max<size_t>(tmp, 16);
+    st.tmp_bytes = tmp;
+    if (tmp) HIP_CHECKD(hipMalloc(&st.tmp_dev, tmp));
+}
+
+namespace {
+
+// Issue one chunk through the fused RCCL op set.
+void IssueFused(CommRequest* req, ChunkExec& ce, ncclComm_t comm, hipStream_t s,
+                DeviceReqState& st, size_t tmp_off) {
+    const OpSpec& spec = req->Spec();
+    ProcessGroup* g = req->Group();
+    const size_t es = DtypeSize(req->Dtype());
+    const uint8_t* sbase = req->SendBuf() + ce.elem_off * es;
+    uint8_t* rbase = req->RecvBuf() + ce.elem_off * es;
+    const size_t cnt = [&]() -> size_t {
+        // chunk-local element count for splittable ops
+        switch (spec.op) {
+            case CollOp::ALLREDUCE:
+            case CollOp::BCAST:
+            case CollOp::REDUCE: {
+                // Recover from the schedule result size.
+                return ce.sch.result.bytes / es;
+            }
+            default:
+                return spec.count;
+        }
+    }();
+    const ncclDataType_t ndt = ToNccl(req->Dtype());
+    const ncclRedOp_t nop = ToNcclOp(spec.rop);
+    const int gsize = g->Size();
+
+    switch (spec.op) {
+        case CollOp::ALLREDUCE:
+            NCCL_CHECK(ncclAllReduce(sbase, rbase, cnt, ndt, nop, comm, s));
+            break;
+        case CollOp::REDUCE:
+            NCCL_CHECK(ncclReduce(sbase, rbase, cnt, ndt, nop, spec.root, comm, s));
+            break;
+        case CollOp::BCAST:
+            NCCL_CHECK(ncclBroadcast(g->MyIdx() == spec.root ? rbase : rbase, rbase,
+                                     cnt, ndt, spec.root, comm, s));
+            break;
+        case CollOp::REDUCE_SCATTER:
+            NCCL_CHECK(ncclReduceScatter(sbase, rbase, spec.count, ndt, nop, comm, s));
+            break;
+        case CollOp::ALLGATHER:
+            NCCL_CHECK(ncclAllGather(sbase, rbase, spec.count, ndt, comm, s));
+            break;
+        case CollOp::ALLGATHERV: {
+            // grouped bcast-equivalent: every rank sends its block to all.
+            std::vector<size_t> offs(gsize + 1, 0);
+            for (int i = 0; i < gsize; ++i) offs[i + 1] = offs[i] + spec.recv_counts[i] * es;
+            NCCL_CHECK(ncclGroupStart());
+            for (int i = 0; i < gsize; ++i) {
+                if (i == g->MyIdx()) continue;
+                NCCL_CHECK(ncclSend(sbase, spec.recv_counts[g->MyIdx()], ndt, i, comm, s));
+                NCCL_CHECK(ncclRecv(rbase + offs[i], spec.recv_counts[i], ndt, i, comm, s));
+            }
+            NCCL_CHECK(ncclGroupEnd());
+            HIP_CHECKD(hipMemcpyAsync(rbase + offs[g->MyIdx()], sbase,
+                                      spec.recv_counts[g->MyIdx()] * es,
+                                      hipMemcpyDeviceToDevice, s));
+            break;
+        }
+        case CollOp::GATHER: {
+            NCCL_CHECK(ncclGroupStart());
+            if (g->MyIdx() == spec.root) {
+                for (int i = 0; i < gsize; ++i) {
+                    if (i == spec.root) continue;
+                    NCCL_CHECK(ncclRecv(rbase + i * spec.count * es, spec.count, ndt, i, comm, s));
+                }
+            } else {
+                NCCL_CHECK(ncclSend(sbase, spec.count, ndt, spec.root, comm, s));
+            }
+            NCCL_CHECK(ncclGroupEnd());
+            if (g->MyIdx() == spec.root)
+                HIP_CHECKD(hipMemcpyAsync(rbase + spec.root * spec.count * es, sbase,
+                                          spec.count * es, hipMemcpyDeviceToDevice, s));
+            break;
+        }
+        case CollOp::SCATTER: {
+            NCCL_CHECK(ncclGroupStart());
+            if (g->MyIdx() == spec.root) {
+                for (int i = 0; i < gsize; ++i) {
+                    if (i == spec.root) continue;
+                    NCCL_CHECK(ncclSend(sbase + i * spec.count * es, spec.count, ndt, i, comm, s));
+                }
+            } else {
+                NCCL_CHECK(ncclRecv(rbase, spec.count, ndt, spec.root, comm, s));
+            }
+            NCCL_CHECK(ncclGroupEnd());
+            if (g->MyIdx() == spec.root)
+                HIP_CHECKD(hipMemcpyAsync(rbase, sbase + spec.root * spec.count * es,
+                                          spec.count * es, hipMemcpyDeviceToDevice, s));
+            break;
+        }
+        case CollOp::ALLTOALL: {
+            NCCL_CHECK(ncclGroupStart());
+            for (int i = 0; i < gsize; ++i) {
+                if (i == g->MyIdx()) continue;
+                NCCL_CHECK(ncclSend(sbase + i * spec.count * es, spec.count, ndt, i, comm, s));
+                NCCL_CHECK(ncclRecv(rbase + i * spec.count * es, spec.count, ndt, i, comm, s));
+            }
+            NCCL_CHECK(ncclGroupEnd());
+            HIP_CHECKD(hipMemcpyAsync(rbase + g->MyIdx() * spec.count * es,
+                                      sbase + g->MyIdx() * spec.count * es, spec.count * es,
+                                      hipMemcpyDeviceToDevice, s));
+            break;
+        }
+        case CollOp::ALLTOALLV: {
+            NCCL_CHECK(ncclGroupStart());
+            for (int i = 0; i < gsize; ++i) {
+                if (i == g->MyIdx()) continue;
+                NCCL_CHECK(ncclSend(sbase + spec.send_offs[i] * es, spec.send_counts[i], ndt, i, comm, s));
+                NCCL_CHECK(ncclRecv(rbase + spec.recv_offs[i] * es, spec.recv_counts[i], ndt, i, comm, s));
+            }
+            NCCL_CHECK(ncclGroupEnd());
+            HIP_CHECKD(hipMemcpyAsync(rbase + spec.recv_offs[g->MyIdx()] * es,
+                                      sbase + spec.send_offs[g->MyIdx()] * es,
+                                      spec.send_counts[g->MyIdx()] * es,
+                                      hipMemcpyDeviceToDevice, s));
+            break;
+        }
+        case CollOp::BARRIER:
+            NCCL_CHECK(ncclAllReduce(st.tmp_dev, st.tmp_dev, 1, ncclFloat32, ncclSum, comm, s));
+            break;
+        case CollOp::SRLIST: {
+            NCCL_CHECK(ncclGroupStart());
+            for (const auto& p : spec.pairs) {
+                if (p.send_count)
+                    NCCL_CHECK(ncclSend(sbase + p.send_off * es, p.send_count, ndt, p.peer, comm, s));
+                if (p.recv_count)
+                    NCCL_CHECK(ncclRecv(rbase + p.recv_off * es, p.recv_count, ndt, p.peer, comm, s));
+            }
+            NCCL_CHECK(ncclGroupEnd());
+            break;
+        }
+    }
+    (void)tmp_off;
+}
+
+// Issue one chunk by walking its schedule: per phase a grouped
+// ncclSend/ncclRecv, then the local reduce kernel / async copy. This is the
+// hand-written collective path (ring/RHD over explicit p2p) — RCCL supplies
+// transport, OUR schedule supplies the algorithm and OUR kernels the math.
+void IssueSchedule(CommRequest* req, ChunkExec& ce, ncclComm_t comm, hipStream_t s,
+                   uint8_t* tmp_dev) {
+    ProcessGroup* g = req->Group();
+    const size_t es = DtypeSize(req->Dtype());
+    const uint8_t* sbase = req->SendBuf() + ce.elem_off * es;
+    uint8_t* rbase = req->RecvBuf() + ce.elem_off * es;
+    const ncclDataType_t ndt = ncclUint8;  // byte-addressed transfers
+
+    auto ptr = [&](const BufRef& b) -> uint8_t* {
+        switch (b.space) {
+            case Space::SEND: return const_cast<uint8_t*>(sbase) + b.off;
+            case Space::RECV: return rbase + b.off;
+            case Space::TMP: return tmp_dev + b.off;
+        }
+        return nullptr;
+    };
+
+    for (int phase = 0; phase < ce.sch.num_phases; ++phase) {
+        bool grouped = false;
+        for (const auto& st : ce.sch.steps) {
+            if (st.phase != phase) continue;
+            if ((st.send_peer >= 0 && st.send.bytes) || (st.recv_peer >= 0 && st.recv.bytes)) {
+                if (!grouped) {
+                    NCCL_CHECK(ncclGroupStart());
+                    grouped = true;
+                }
+                if (st.send_peer >= 0 && st.send.bytes)
+                    NCCL_CHECK(ncclSend(ptr(st.send), st.send.bytes, ndt, st.send_peer, comm, s));
+                if (st.recv_peer >= 0 && st.recv.bytes)
+                    NCCL_CHECK(ncclRecv(ptr(st.recv), st.recv.bytes, ndt, st.recv_peer, comm, s));
+            }
+        }
+        if (grouped) NCCL_CHECK(ncclGroupEnd());
+        for (const auto& st : ce.sch.steps) {
+            if (st.phase != phase || st.local == Step::LocalOp::NONE) continue;
+            uint8_t* d = ptr(st.local_dst);
+            uint8_t* src = ptr(st.local_src);
+            if (st.local == Step::LocalOp::COPY) {
+                if (d != src)
+                    HIP_CHECKD(hipMemcpyAsync(d, src, st.local_src.bytes,
+                                              hipMemcpyDeviceToDevice, s));
+            } else {
+                LaunchReduce(d, src, st.local_dst.bytes / es, req->Dtype(), ce.sch.rop, s);
+            }
+        }
+        (void)g;
+    }
+}
+
+}  // namespace
+
+bool DeviceAdvanceRequest(CommRequest* req, DeviceReqState& st) {
+    auto* rt = static_cast<HipRuntime*>(Context::Get().Device());
+    if (!st.issued) {
+        GroupComms& gc = rt->For(req->Group());
+        auto& chunks = req->Chunks();
+        const Config& cfg = GlobalConfig();
+        const bool use_schedule =
+            (cfg.allreduce_algo == AllReduceAlgo::RING || cfg.allreduce_algo == AllReduceAlgo::RHD) &&
+            req->Spec().op == CollOp::ALLREDUCE && req->Group()->Size() > 1;
+
+        // Single-rank groups: local copies only, still through the stream
+        // so completion semantics are uniform.
+        if (gc.comms.empty()) {
+            const size_t es = DtypeSize(req->Dtype());
+            hipStream_t s0;
+            HIP_CHECKD(hipStreamCreateWithFlags(&s0, hipStreamNonBlocking));
+            for (auto& ce : chunks) {
+                const uint8_t* sbase = req->SendBuf() + ce.elem_off * es;
+                uint8_t* rbase = req->RecvBuf() + ce.elem_off * es;
+                if (ce.sch.result.bytes && sbase != rbase)
+                    HIP_CHECKD(hipMemcpyAsync(rbase + ce.sch.result.off, sbase,
+                                              ce.sch.result.bytes, hipMemcpyDeviceToDevice, s0));
+            }
+            if (st.events.empty()) {
+                hipEvent_t e;
+                HIP_CHECKD(hipEventCreateWithFlags(&e, hipEventDisableTiming));
+                st.events.push_back(e);
+            }
+            HIP_CHECKD(hipEventRecord(st.events[0], s0));
+            HIP_CHECKD(hipStreamDestroy(s0));  // event keeps the work
+            st.issued = true;
+            return hipEventQuery(st.events[0]) == hipSuccess;
+        }
+
+        // Events per used channel.
+        const size_t nch = gc.comms.size();
+        const size_t used = std::min(chunks.size(), nch);
+        while (st.events.size() < used) {
+            hipEvent_t e;
+            HIP_CHECKD(hipEventCreateWithFlags(&e, hipEventDisableTiming));
+            st.events.push_back(e);
+        }
+        // tmp partitioning per chunk
+        size_t tmp_off = 0;
+        for (auto& ce : chunks) {
+            const size_t ch = ce.chunk_idx % nch;
+            if (use_schedule) {
+                IssueSchedule(req, ce, gc.comms[ch], gc.streams[ch],
+                              static_cast<uint8_t*>(st.tmp_dev) + tmp_off);
+            } else {
+                IssueFused(req, ce, gc.comms[ch], gc.streams[ch], st, tmp_off);
+            }
+            tmp_off += ce.sch.tmp_bytes;
+        }
+        for (size_t ch = 0; ch < used; ++ch)
+            HIP_CHECKD(hipEventRecord(st.events[ch], gc.streams[ch]));
+        st.issued = true;
+    }
+    for (hipEvent_t e : st.events) {
+        hipError_t q = hipEventQuery(e);
+        if (q == hipErrorNotReady) return false;
+        HIP_CHECKD(q);
+    }
+    return true;
+}
+
+}  // namespace mlsl

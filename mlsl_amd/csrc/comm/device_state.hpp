@@ -1,0 +1,21 @@
+// Private device-side execution state of a CommRequest (HIP types; included
+// only by .cpp files compiled with hipcc).
+#pragma once
+
+#include <hip/hip_runtime.h>
+
+#include <vector>
+
+namespace mlsl {
+
+struct DeviceReqState {
+    bool issued = false;
+    // One completion event per channel used by this request.
+    std::vector<hipEvent_t> events;
+    // Persistent device scratch (allocated at Setup in device mode).
+    void* tmp_dev = nullptr;
+    size_t tmp_bytes = 0;
+    ~DeviceReqState();
+};
+
+}  // namespace mlsl

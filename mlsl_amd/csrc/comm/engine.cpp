@@ -1,0 +1,184 @@
+#include "engine.hpp"
+
+#include <algorithm>
+#include <chrono>
+
+#include "../core/log.hpp"
+#include "../core/types.hpp"
+#include "mesh.hpp"
+#include "request.hpp"
+
+namespace mlsl {
+
+Engine::Engine(Mesh* mesh, ProgressMode mode, bool device_mode)
+    : mesh_(mesh), mode_(mode), device_mode_(device_mode) {
+    if (mode_ == ProgressMode::THREAD) {
+        thread_ = std::thread([this]() { Loop(); });
+    }
+}
+
+Engine::~Engine() {
+    stop_.store(true, std::memory_order_release);
+    if (thread_.joinable()) thread_.join();
+}
+
+void Engine::Submit(CommRequest* req) {
+    req->start_seqno_ = NextSeqno();
+    if (mode_ == ProgressMode::INLINE) {
+        // "Thread mode" analog: drive the request to completion on the
+        // calling thread (device mode still returns early: streams are
+        // asynchronous and AdvanceDevice only enqueues + polls).
+        req->state_.store(ReqState::ACTIVE, std::memory_order_release);
+        try {
+            if (device_mode_) {
+                // Issue now; completion is polled in Wait/Test.
+                if (req->AdvanceDevice()) req->MarkDone();
+                else active_.push_back(req);
+            } else {
+                while (!req->AdvanceHost(mesh_)) {
+                    if (mesh_) mesh_->Progress();
+                }
+                req->MarkDone();
+            }
+        } catch (const std::exception& e) {
+            req->MarkFailed(e.what());
+        }
+        return;
+    }
+    // Progress-thread mode: hand off through the SPSC ring (overflow to the
+    // locked deque if the ring is full or there are competing producers).
+    if (!ring_.Push(req)) {
+        std::lock_guard<std::mutex> lk(inbox_mu_);
+        inbox_overflow_.push_back(req);
+    }
+}
+
+void Engine::DrainInbox() {
+    while (CommRequest* r = ring_.Pop()) {
+        r->state_.store(ReqState::ACTIVE, std::memory_order_release);
+        active_.push_back(r);
+    }
+    if (!inbox_overflow_.empty()) {
+        std::lock_guard<std::mutex> lk(inbox_mu_);
+        for (CommRequest* r : inbox_overflow_) {
+            r->state_.store(ReqState::ACTIVE, std::memory_order_release);
+            active_.push_back(r);
+        }
+        inbox_overflow_.clear();
+    }
+    // Newest-first priority above the size threshold (reference
+    // MLSL_MSG_PRIORITY head-first scan, eplib/allreduce_pr.c:69-81 +
+    // env.h:59): large fresh gradients overtake older bulk transfers.
+    const Config& cfg = GlobalConfig();
+    if (cfg.msg_priority && active_.size() > 1) {
+        std::stable_sort(active_.begin(), active_.end(),
+                         [&](CommRequest* a, CommRequest* b) {
+                             const bool pa = a->MessageBytes() >= cfg.msg_priority_threshold;
+                             const bool pb = b->MessageBytes() >= cfg.msg_priority_threshold;
+                             if (pa != pb) return pa > pb;
+                             if (pa) return a->StartSeqno() > b->StartSeqno();
+                             return a->StartSeqno() < b->StartSeqno();
+                         });
+    }
+}
+
+bool Engine::AdvanceOne(CommRequest* req) {
+    try {
+        const bool done = device_mode_ ? req->AdvanceDevice() : req->AdvanceHost(mesh_);
+        if (done) req->MarkDone();
+        return done;
+    } catch (const std::exception& e) {
+        MLSL_LOG(ERROR, "request %s failed in progress engine: %s",
+                 CollOpName(req->Spec().op), e.what());
+        req->MarkFailed(e.what());
+        return true;
+    }
+}
+
+void Engine::ProgressAll() {
+    DrainInbox();
+    if (mesh_) mesh_->Progress();
+    bool any_done = false;
+    for (size_t i = 0; i < active_.size();) {
+        if (AdvanceOne(active_[i])) {
+            active_.erase(active_.begin() + static_cast<long>(i));
+            any_done = true;
+        } else {
+            ++i;
+        }
+    }
+    if (any_done) NotifyDone();
+}
+
+void Engine::Loop() {
+    int idle_spins = 0;
+    while (!stop_.load(std::memory_order_acquire)) {
+        const size_t before = active_.size();
+        try {
+            ProgressAll();
+        } catch (const std::exception& e) {
+            // Transport-level failure (not attributable to one request):
+            // fail everything in flight rather than terminating the process.
+            MLSL_LOG(ERROR, "progress engine transport failure: %s", e.what());
+            for (CommRequest* r : active_) r->MarkFailed(e.what());
+            active_.clear();
+            NotifyDone();
+        }
+        if (active_.empty() && before == 0) {
+            if (++idle_spins > 256) {
+                std::this_thread::sleep_for(std::chrono::microseconds(50));
+            }
+        } else {
+            idle_spins = 0;
+        }
+    }
+    // Drain what we can so Finalize doesn't strand requests.
+    ProgressAll();
+}
+
+void Engine::NotifyDone() {
+    std::lock_guard<std::mutex> lk(done_mu_);
+    done_cv_.notify_all();
+}
+
+void Engine::WaitFor(CommRequest* req) {
+    if (mode_ == ProgressMode::INLINE) {
+        // Inline: caller drives progress until done.
+        while (true) {
+            ReqState st = req->State();
+            if (st == ReqState::DONE || st == ReqState::FAILED || st == ReqState::IDLE)
+                return;
+            // Only device requests can still be pending here.
+            auto it = std::find(active_.begin(), active_.end(), req);
+            if (it != active_.end() && AdvanceOne(req))
+                active_.erase(std::find(active_.begin(), active_.end(), req));
+        }
+    }
+    // Spin briefly (hot path: sub-ms collectives), then block on the condvar.
+    for (int i = 0; i < 4096; ++i) {
+        ReqState st = req->State();
+        if (st == ReqState::DONE || st == ReqState::FAILED || st == ReqState::IDLE) return;
+    }
+    std::unique_lock<std::mutex> lk(done_mu_);
+    done_cv_.wait(lk, [&]() {
+        ReqState st = req->State();
+        return st == ReqState::DONE || st == ReqState::FAILED || st == ReqState::IDLE;
+    });
+}
+
+bool Engine::TestFor(CommRequest* req) {
+    if (mode_ == ProgressMode::INLINE) {
+        auto it = std::find(active_.begin(), active_.end(), req);
+        if (it != active_.end()) {
+            if (AdvanceOne(req)) {
+                active_.erase(std::find(active_.begin(), active_.end(), req));
+                return true;
+            }
+            return false;
+        }
+    }
+    ReqState st = req->State();
+    return st == ReqState::DONE || st == ReqState::FAILED || st == ReqState::IDLE;
+}
+
+}  // namespace mlsl

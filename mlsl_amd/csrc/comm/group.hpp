@@ -1,0 +1,42 @@
+// Process groups: ordered world-rank subsets with group-consistent ids.
+// Reference analog: ProcessGroup + CreateProcessGroup(color)
+// (src/comm.hpp:33-46, src/comm_ep.cpp:1821-1827). Color-split ordering
+// matches MPI_Comm_split with key = world rank.
+#pragma once
+
+#include <cstdint>
+#include <vector>
+
+namespace mlsl {
+
+class DeviceComm;
+
+class ProcessGroup {
+  public:
+    ProcessGroup(int uid, std::vector<int> ranks, int my_world_rank);
+    ~ProcessGroup();
+
+    int Uid() const { return uid_; }
+    int Size() const { return static_cast<int>(ranks_.size()); }
+    int MyIdx() const { return my_idx_; }          // -1 if not a member
+    bool IsMember() const { return my_idx_ >= 0; }
+    int WorldRank(int group_idx) const { return ranks_[group_idx]; }
+    const std::vector<int>& Ranks() const { return ranks_; }
+
+    uint32_t NextFlow() { return flow_seq_++; }
+
+    // Device-side communicators (RCCL comms + streams per channel), created
+    // lazily by the device transport. Owned here so persistent requests can
+    // share them.
+    DeviceComm* Device() const { return device_; }
+    void SetDevice(DeviceComm* d) { device_ = d; }
+
+  private:
+    int uid_;
+    std::vector<int> ranks_;
+    int my_idx_ = -1;
+    uint32_t flow_seq_ = 0;
+    DeviceComm* device_ = nullptr;
+};
+
+}  // namespace mlsl

@@ -1,0 +1,395 @@
+#include "request.hpp"
+
+#include <algorithm>
+#include <cstring>
+
+#include "../core/config.hpp"
+#include "../core/log.hpp"
+#include "context.hpp"
+#include "device_comm.hpp"
+#include "device_state.hpp"
+#include "engine.hpp"
+#include "group.hpp"
+#include "mesh.hpp"
+
+namespace mlsl {
+
+const char* CollOpName(CollOp op) {
+    switch (op) {
+        case CollOp::ALLREDUCE: return "allreduce";
+        case CollOp::REDUCE: return "reduce";
+        case CollOp::REDUCE_SCATTER: return "reduce_scatter";
+        case CollOp::ALLGATHER: return "allgather";
+        case CollOp::ALLGATHERV: return "allgatherv";
+        case CollOp::BCAST: return "bcast";
+        case CollOp::GATHER: return "gather";
+        case CollOp::SCATTER: return "scatter";
+        case CollOp::ALLTOALL: return "alltoall";
+        case CollOp::ALLTOALLV: return "alltoallv";
+        case CollOp::BARRIER: return "barrier";
+        case CollOp::SRLIST: return "srlist";
+    }
+    return "?";
+}
+
+void ChunkExec::Reset() {
+    cur_phase = sch.steps.empty() ? sch.num_phases : 0;
+    finished = sch.num_phases == 0;
+    state.assign(sch.steps.size(), StepState{});
+}
+
+struct CommRequest::DeviceState : DeviceReqState {};
+
+CommRequest::CommRequest(ProcessGroup* group, DataType dt, CompType ctype)
+    : group_(group), dtype_(dt), ctype_(ctype) {}
+
+CommRequest::~CommRequest() = default;
+
+#define MLSL_ADD_OP()                                                        \
+    MLSL_CHECK(!has_op_, "request already describes an op");                 \
+    has_op_ = true;
+
+void CommRequest::AddAllReduce(size_t count, ReduceOp op) {
+    MLSL_ADD_OP();
+    spec_.op = CollOp::ALLREDUCE;
+    spec_.count = count;
+    spec_.rop = op;
+}
+
+void CommRequest::AddReduce(size_t count, ReduceOp op, int root) {
+    MLSL_ADD_OP();
+    spec_.op = CollOp::REDUCE;
+    spec_.count = count;
+    spec_.rop = op;
+    spec_.root = root;
+}
+
+void CommRequest::AddReduceScatter(size_t recv_count, ReduceOp op) {
+    MLSL_ADD_OP();
+    spec_.op = CollOp::REDUCE_SCATTER;
+    spec_.count = recv_count;
+    spec_.rop = op;
+}
+
+void CommRequest::AddAllGather(size_t send_count) {
+    MLSL_ADD_OP();
+    spec_.op = CollOp::ALLGATHER;
+    spec_.count = send_count;
+}
+
+void CommRequest::AddAllGatherv(size_t send_count, const std::vector<size_t>& recv_counts) {
+    MLSL_ADD_OP();
+    spec_.op = CollOp::ALLGATHERV;
+    spec_.count = send_count;
+    spec_.recv_counts = recv_counts;
+}
+
+void CommRequest::AddBcast(size_t count, int root) {
+    MLSL_ADD_OP();
+    spec_.op = CollOp::BCAST;
+    spec_.count = count;
+    spec_.root = root;
+}
+
+void CommRequest::AddGather(size_t send_count, int root) {
+    MLSL_ADD_OP();
+    spec_.op = CollOp::GATHER;
+    spec_.count = send_count;
+    spec_.root = root;
+}
+
+void CommRequest::AddScatter(size_t recv_count, int root) {
+    MLSL_ADD_OP();
+    spec_.op = CollOp::SCATTER;
+    spec_.count = recv_count;
+    spec_.root = root;
+}
+
+void CommRequest::AddAlltoAll(size_t send_count) {
+    MLSL_ADD_OP();
+    spec_.op = CollOp::ALLTOALL;
+    spec_.count = send_count;
+}
+
+void CommRequest::AddAlltoAllv(const std::vector<size_t>& scnt, const std::vector<size_t>& soff,
+                               const std::vector<size_t>& rcnt, const std::vector<size_t>& roff) {
+    MLSL_ADD_OP();
+    spec_.op = CollOp::ALLTOALLV;
+    spec_.send_counts = scnt;
+    spec_.send_offs = soff;
+    spec_.recv_counts = rcnt;
+    spec_.recv_offs = roff;
+}
+
+void CommRequest::AddBarrier() {
+    MLSL_ADD_OP();
+    spec_.op = CollOp::BARRIER;
+}
+
+void CommRequest::AddSendRecvList(const std::vector<SRPair>& pairs) {
+    MLSL_ADD_OP();
+    spec_.op = CollOp::SRLIST;
+    spec_.pairs = pairs;
+}
+
+size_t CommRequest::MessageBytes() const {
+    const size_t es = DtypeSize(dtype_);
+    switch (spec_.op) {
+        case CollOp::ALLREDUCE:
+        case CollOp::REDUCE:
+        case CollOp::BCAST:
+            return spec_.count * es;
+        case CollOp::REDUCE_SCATTER:
+        case CollOp::ALLGATHER:
+        case CollOp::GATHER:
+        case CollOp::SCATTER:
+        case CollOp::ALLTOALL:
+            return spec_.count * es * static_cast<size_t>(group_->Size());
+        case CollOp::ALLGATHERV: {
+            size_t t = 0;
+            for (size_t c : spec_.recv_counts) t += c;
+            return t * es;
+        }
+        case CollOp::ALLTOALLV: {
+            size_t t = 0;
+            for (size_t c : spec_.send_counts) t += c;
+            return t * es;
+        }
+        default:
+            return 0;
+    }
+}
+
+void CommRequest::BuildChunks() {
+    const Config& cfg = GlobalConfig();
+    const int gr = group_->MyIdx(), gs = group_->Size();
+    MLSL_CHECK(group_->IsMember(), "request on a group this rank is not in");
+    const size_t es = DtypeSize(dtype_);
+
+    // Chunk fan-out (element-splittable ops only).
+    size_t n_chunks = 1;
+    const bool splittable = spec_.op == CollOp::ALLREDUCE || spec_.op == CollOp::BCAST ||
+                            spec_.op == CollOp::REDUCE;
+    if (splittable && gs > 1) {
+        n_chunks = cfg.num_channels;
+        if (MessageBytes() >= cfg.large_msg_mb * (1024 * 1024) && cfg.large_msg_chunks > 1)
+            n_chunks *= cfg.large_msg_chunks;
+        // Keep chunks >= 4 KB so per-message overhead stays amortized.
+        const size_t min_chunk_elems = std::max<size_t>(1, 4096 / es);
+        n_chunks = std::max<size_t>(1, std::min(n_chunks, spec_.count / std::max<size_t>(1, min_chunk_elems)));
+        n_chunks = std::min<size_t>(n_chunks, 4096);
+        if (n_chunks == 0) n_chunks = 1;
+    }
+
+    AllReduceAlgo algo = cfg.allreduce_algo;
+    if (algo == AllReduceAlgo::AUTO) {
+        const bool pow2 = (gs & (gs - 1)) == 0;
+        algo = (pow2 && MessageBytes() <= 65536) ? AllReduceAlgo::RHD : AllReduceAlgo::RING;
+    }
+
+    chunks_.clear();
+    chunks_.resize(n_chunks);
+    total_tmp_bytes_ = 0;
+    for (size_t c = 0; c < n_chunks; ++c) {
+        ChunkExec& ce = chunks_[c];
+        ce.chunk_idx = c;
+        size_t cnt = spec_.count, off = 0;
+        if (n_chunks > 1) {
+            off = SegOffset(spec_.count, n_chunks, c);
+            cnt = SegCount(spec_.count, n_chunks, c);
+        }
+        ce.elem_off = off;
+        switch (spec_.op) {
+            case CollOp::ALLREDUCE:
+                ce.sch = (algo == AllReduceAlgo::RHD && (gs & (gs - 1)) == 0)
+                             ? BuildAllReduceRHD(gr, gs, cnt, dtype_, spec_.rop)
+                             : BuildAllReduceRing(gr, gs, cnt, dtype_, spec_.rop);
+                break;
+            case CollOp::REDUCE:
+                ce.sch = BuildReduce(gr, gs, cnt, dtype_, spec_.rop, spec_.root);
+                break;
+            case CollOp::BCAST:
+                ce.sch = BuildBcast(gr, gs, cnt, dtype_, spec_.root);
+                break;
+            case CollOp::REDUCE_SCATTER:
+                ce.sch = BuildReduceScatter(gr, gs, spec_.count, dtype_, spec_.rop);
+                break;
+            case CollOp::ALLGATHER:
+                ce.sch = BuildAllGather(gr, gs, spec_.count, dtype_);
+                break;
+            case CollOp::ALLGATHERV:
+                ce.sch = BuildAllGatherv(gr, gs, spec_.recv_counts, dtype_);
+                break;
+            case CollOp::GATHER:
+                ce.sch = BuildGather(gr, gs, spec_.count, dtype_, spec_.root);
+                break;
+            case CollOp::SCATTER:
+                ce.sch = BuildScatter(gr, gs, spec_.count, dtype_, spec_.root);
+                break;
+            case CollOp::ALLTOALL:
+                ce.sch = BuildAlltoAll(gr, gs, spec_.count, dtype_);
+                break;
+            case CollOp::ALLTOALLV:
+                ce.sch = BuildAlltoAllv(gr, gs, spec_.send_counts, spec_.send_offs,
+                                        spec_.recv_counts, spec_.recv_offs, dtype_);
+                break;
+            case CollOp::BARRIER:
+                ce.sch = BuildBarrier(gr, gs);
+                break;
+            case CollOp::SRLIST:
+                ce.sch = BuildSendRecvList(gr, gs, spec_.pairs, dtype_);
+                break;
+        }
+        total_tmp_bytes_ += ce.sch.tmp_bytes;
+    }
+}
+
+void CommRequest::Setup() {
+    MLSL_CHECK(has_op_, "Setup() before describing the op");
+    BuildChunks();
+    Context& ctx = Context::Get();
+    if (ctx.DeviceMode()) {
+        dev_ = std::make_unique<DeviceState>();
+        DeviceSetupRequest(this, *dev_);
+    } else {
+        for (auto& ce : chunks_) ce.tmp.resize(ce.sch.tmp_bytes);
+    }
+    setup_done_ = true;
+}
+
+uint64_t CommRequest::MakeTag(size_t chunk, int phase) const {
+    return (static_cast<uint64_t>(group_->Uid() & 0xFFFFF) << 44) |
+           (static_cast<uint64_t>(flow_ & 0xFFFFFF) << 20) |
+           (static_cast<uint64_t>(chunk & 0xFFF) << 8) |
+           (static_cast<uint64_t>(phase) & 0xFF);
+}
+
+void CommRequest::Start(const void* sbuf, void* rbuf) {
+    MLSL_CHECK(setup_done_, "Start() before Setup()");
+    ReqState st = state_.load(std::memory_order_acquire);
+    MLSL_CHECK(st == ReqState::IDLE || st == ReqState::DONE,
+               "Start() while request in flight");
+    sbuf_ = static_cast<const uint8_t*>(sbuf);
+    rbuf_ = static_cast<uint8_t*>(rbuf);
+    flow_ = group_->NextFlow();
+    for (auto& ce : chunks_) ce.Reset();
+    if (dev_) dev_->issued = false;
+    error_.clear();
+
+    Context& ctx = Context::Get();
+    if (GlobalConfig().check_pointers) {
+        // Pointer checker (reference src/pointer_checker.*): every buffer
+        // handed to a collective must be a known registered allocation.
+        const size_t bytes = MessageBytes();
+        MLSL_CHECK(ctx.CheckBuffer(sbuf_, bytes) && ctx.CheckBuffer(rbuf_, 0),
+                   "collective buffer not from Environment::Alloc "
+                   "(MLSL_CHECK_POINTERS=1)");
+    }
+    state_.store(ReqState::QUEUED, std::memory_order_release);
+    ctx.GetEngine()->Submit(this);
+}
+
+void* CommRequest::Wait() {
+    Context::Get().GetEngine()->WaitFor(this);
+    if (state_.load(std::memory_order_acquire) == ReqState::FAILED)
+        MLSL_THROW("request failed: " + error_);
+    state_.store(ReqState::IDLE, std::memory_order_release);
+    if (chunks_.empty()) return rbuf_;
+    const BufRef& res = chunks_[0].sch.result;
+    switch (res.space) {
+        case Space::RECV: return rbuf_ + res.off;
+        case Space::SEND: return const_cast<uint8_t*>(sbuf_) + res.off;
+        case Space::TMP: return nullptr;
+    }
+    return rbuf_;
+}
+
+bool CommRequest::Test() {
+    ReqState st = state_.load(std::memory_order_acquire);
+    if (st == ReqState::FAILED) MLSL_THROW("request failed: " + error_);
+    if (st == ReqState::DONE || st == ReqState::IDLE) return true;
+    return Context::Get().GetEngine()->TestFor(this);
+}
+
+void CommRequest::MarkDone() {
+    state_.store(ReqState::DONE, std::memory_order_release);
+}
+
+void CommRequest::MarkFailed(const std::string& what) {
+    error_ = what;
+    state_.store(ReqState::FAILED, std::memory_order_release);
+}
+
+// ---------------------------------------------------------------------------
+// Host executor: advance every chunk's phase program against the mesh.
+
+bool CommRequest::AdvanceHost(Mesh* mesh) {
+    const size_t es = DtypeSize(dtype_);
+    const size_t msg_es = es;  // chunk offsets are in elements of dtype_
+    bool all_done = true;
+
+    for (auto& ce : chunks_) {
+        if (ce.finished) continue;
+        const uint8_t* sbase = sbuf_ + ce.elem_off * msg_es;
+        uint8_t* rbase = rbuf_ + ce.elem_off * msg_es;
+        auto ptr = [&](const BufRef& b) -> uint8_t* {
+            switch (b.space) {
+                case Space::SEND: return const_cast<uint8_t*>(sbase) + b.off;
+                case Space::RECV: return rbase + b.off;
+                case Space::TMP: return ce.tmp.data() + b.off;
+            }
+            return nullptr;
+        };
+
+        while (!ce.finished) {
+            bool phase_done = true;
+            for (size_t i = 0; i < ce.sch.steps.size(); ++i) {
+                const Step& st = ce.sch.steps[i];
+                if (st.phase != ce.cur_phase) continue;
+                auto& ss = ce.state[i];
+                // Post receive first (so early sends always land).
+                if (st.recv_peer >= 0 && st.recv.bytes > 0 && !ss.recv_posted) {
+                    mesh->PostRecv(group_->WorldRank(st.recv_peer),
+                                   MakeTag(ce.chunk_idx, st.phase), ptr(st.recv),
+                                   st.recv.bytes, &ss.recv_done);
+                    ss.recv_posted = true;
+                }
+                if (st.send_peer >= 0 && st.send.bytes > 0 && !ss.send_started) {
+                    if (mesh->StartSend(group_->WorldRank(st.send_peer),
+                                        MakeTag(ce.chunk_idx, st.phase), ptr(st.send),
+                                        st.send.bytes, &ss.send_done))
+                        ss.send_started = true;
+                }
+                const bool send_ok =
+                    st.send_peer < 0 || st.send.bytes == 0 || (ss.send_started && ss.send_done);
+                const bool recv_ok = st.recv_peer < 0 || st.recv.bytes == 0 || ss.recv_done;
+                if (send_ok && recv_ok) {
+                    if (st.local != Step::LocalOp::NONE && !ss.local_done) {
+                        uint8_t* d = ptr(st.local_dst);
+                        uint8_t* s = ptr(st.local_src);
+                        if (st.local == Step::LocalOp::COPY) {
+                            if (d != s) std::memmove(d, s, st.local_src.bytes);
+                        } else {
+                            HostReduce(d, s, st.local_dst.bytes / es, dtype_, ce.sch.rop);
+                        }
+                        ss.local_done = true;
+                    }
+                } else {
+                    phase_done = false;
+                }
+            }
+            if (!phase_done) break;
+            ce.cur_phase++;
+            if (ce.cur_phase >= ce.sch.num_phases) ce.finished = true;
+        }
+        if (!ce.finished) all_done = false;
+    }
+    return all_done;
+}
+
+bool CommRequest::AdvanceDevice() {
+    MLSL_CHECK(dev_ != nullptr, "device state missing");
+    return DeviceAdvanceRequest(this, *dev_);
+}
+
+}  // namespace mlsl

@@ -1,0 +1,446 @@
+// CDNA4 (gfx950) kernels for the communication hot path: local reductions
+// for the schedule-driven collectives, int8 block quantization with error
+// feedback, and pack/unpack between layer layouts and comm buffers.
+//
+// Design per the MI355X rules: 64-wide wavefronts, 256-thread blocks,
+// 16-byte vectorized accesses where alignment permits, grid capped at
+// ~8 blocks/CU with grid-stride loops (memory-bound ops: HBM3E is the
+// roofline, not VALU).
+#include <hip/hip_bf16.h>
+#include <hip/hip_fp16.h>
+#include <hip/hip_runtime.h>
+
+#include "../core/log.hpp"
+#include "kernels.hpp"
+
+namespace mlsl {
+
+#define HIP_CHECK(cmd)                                                        \
+    do {                                                                      \
+        hipError_t e_ = (cmd);                                                \
+        if (e_ != hipSuccess)                                                 \
+            MLSL_THROW(std::string("HIP error: ") + hipGetErrorString(e_));   \
+    } while (0)
+
+namespace {
+
+constexpr int kBlock = 256;
+// 256 CUs × 8 blocks/CU: enough residency to cover HBM latency without
+// oversubscribing the launch path.
+constexpr int kMaxGrid = 2048;
+
+inline int GridFor(size_t work_items) {
+    size_t g = (work_items + kBlock - 1) / kBlock;
+    if (g > kMaxGrid) g = kMaxGrid;
+    if (g == 0) g = 1;
+    return static_cast<int>(g);
+}
+
+template <typename T, ReduceOp OP>
+__device__ __forceinline__ T Apply(T a, T b) {
+    if constexpr (OP == ReduceOp::SUM) return a + b;
+    if constexpr (OP == ReduceOp::MIN) return a < b ? a : b;
+    return a > b ? a : b;
+}
+
+// ---- f32: float4-vectorized in-place reduce ----
+template <ReduceOp OP>
+__global__ void ReduceF32Kernel(float* __restrict__ dst,
+                                const float* __restrict__ src, size_t n) {
+    const size_t tid = blockIdx.x * blockDim.x + threadIdx.x;
+    const size_t stride = gridDim.x * blockDim.x;
+    const size_t n4 = n / 4;
+    const float4* s4 = reinterpret_cast<const float4*>(src);
+    float4* d4 = reinterpret_cast<float4*>(dst);
+    for (size_t i = tid; i < n4; i += stride) {
+        float4 a = d4[i], b = s4[i];
+        a.x = Apply<float, OP>(a.x, b.x);
+        a.y = Apply<float, OP>(a.y, b.y);
+        a.z = Apply<float, OP>(a.z, b.z);
+        a.w = Apply<float, OP>(a.w, b.w);
+        d4[i] = a;
+    }
+    for (size_t i = n4 * 4 + tid; i < n; i += stride)
+        dst[i] = Apply<float, OP>(dst[i], src[i]);
+}
+
+// ---- generic scalar fallback (f64/i32/i64/u8) ----
+template <typename T, ReduceOp OP>
+__global__ void ReduceScalarKernel(T* __restrict__ dst, const T* __restrict__ src,
+                                   size_t n) {
+    const size_t tid = blockIdx.x * blockDim.x + threadIdx.x;
+    const size_t stride = gridDim.x * blockDim.x;
+    for (size_t i = tid; i < n; i += stride)
+        dst[i] = Apply<T, OP>(dst[i], src[i]);
+}
+
+// ---- bf16: 8-element (16 B) vectorized via short4 pairs ----
+using ushort4_t = __attribute__((ext_vector_type(4))) unsigned short;
+
+template <ReduceOp OP>
+__global__ void ReduceBf16Kernel(unsigned short* __restrict__ dst,
+                                 const unsigned short* __restrict__ src, size_t n) {
+    const size_t tid = blockIdx.x * blockDim.x + threadIdx.x;
+    const size_t stride = gridDim.x * blockDim.x;
+    const size_t n4 = n / 4;
+    const ushort4_t* s4 = reinterpret_cast<const ushort4_t*>(src);
+    ushort4_t* d4 = reinterpret_cast<ushort4_t*>(dst);
+    for (size_t i = tid; i < n4; i += stride) {
+        ushort4_t a = d4[i], b = s4[i];
+        ushort4_t r;
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+            float fa = __bfloat162float(__hip_bfloat16_raw{a[j]});
+            float fb = __bfloat162float(__hip_bfloat16_raw{b[j]});
+            __hip_bfloat16 hr = __float2bfloat16(Apply<float, OP>(fa, fb));
+            r[j] = reinterpret_cast<unsigned short&>(hr);
+        }
+        d4[i] = r;
+    }
+    for (size_t i = n4 * 4 + tid; i < n; i += stride) {
+        float fa = __bfloat162float(__hip_bfloat16_raw{dst[i]});
+        float fb = __bfloat162float(__hip_bfloat16_raw{src[i]});
+        __hip_bfloat16 hr = __float2bfloat16(Apply<float, OP>(fa, fb));
+        dst[i] = reinterpret_cast<unsigned short&>(hr);
+    }
+}
+
+template <ReduceOp OP>
+__global__ void ReduceF16Kernel(__half* __restrict__ dst,
+                                const __half* __restrict__ src, size_t n) {
+    const size_t tid = blockIdx.x * blockDim.x + threadIdx.x;
+    const size_t stride = gridDim.x * blockDim.x;
+    for (size_t i = tid; i < n; i += stride) {
+        float fa = __half2float(dst[i]);
+        float fb = __half2float(src[i]);
+        dst[i] = __float2half(Apply<float, OP>(fa, fb));
+    }
+}
+
+// Runtime op -> compile-time instantiation dispatch.
+#define MLSL_LAUNCH_BY_OP(KER, DST, SRC, N, OP, STREAM)                               \
+    do {                                                                              \
+        const int grid_ = GridFor((N) / 4 + 1);                                       \
+        switch (OP) {                                                                 \
+            case ReduceOp::SUM:                                                       \
+                KER<ReduceOp::SUM><<<dim3(grid_), dim3(kBlock), 0, (STREAM)>>>(       \
+                    (DST), (SRC), (N));                                               \
+                break;                                                                \
+            case ReduceOp::MIN:                                                       \
+                KER<ReduceOp::MIN><<<dim3(grid_), dim3(kBlock), 0, (STREAM)>>>(       \
+                    (DST), (SRC), (N));                                               \
+                break;                                                                \
+            case ReduceOp::MAX:                                                       \
+                KER<ReduceOp::MAX><<<dim3(grid_), dim3(kBlock), 0, (STREAM)>>>(       \
+                    (DST), (SRC), (N));                                               \
+                break;                                                                \
+        }                                                                             \
+    } while (0)
+
+template <typename T>
+void LaunchScalarByOp(T* dst, const T* src, size_t n, ReduceOp op, hipStream_t stream) {
+    const int grid = GridFor(n);
+    switch (op) {
+        case ReduceOp::SUM:
+            hipLaunchKernelGGL((ReduceScalarKernel<T, ReduceOp::SUM>), dim3(grid), dim3(kBlock), 0, stream, dst, src, n);
+            break;
+        case ReduceOp::MIN:
+            hipLaunchKernelGGL((ReduceScalarKernel<T, ReduceOp::MIN>), dim3(grid), dim3(kBlock), 0, stream, dst, src, n);
+            break;
+        case ReduceOp::MAX:
+            hipLaunchKernelGGL((ReduceScalarKernel<T, ReduceOp::MAX>), dim3(grid), dim3(kBlock), 0, stream, dst, src, n);
+            break;
+    }
+}
+
+}  // namespace
+
+void LaunchReduce(void* dst, const void* src, size_t count, DataType dt,
+                  ReduceOp op, hipStream_t stream) {
+    switch (dt) {
+        case DataType::F32:
+            MLSL_LAUNCH_BY_OP(ReduceF32Kernel, static_cast<float*>(dst),
+                              static_cast<const float*>(src), count, op, stream);
+            break;
+        case DataType::BF16:
+            MLSL_LAUNCH_BY_OP(ReduceBf16Kernel, static_cast<unsigned short*>(dst),
+                              static_cast<const unsigned short*>(src), count, op, stream);
+            break;
+        case DataType::F16:
+            MLSL_LAUNCH_BY_OP(ReduceF16Kernel, static_cast<__half*>(dst),
+                              static_cast<const __half*>(src), count, op, stream);
+            break;
+        case DataType::F64:
+            LaunchScalarByOp(static_cast<double*>(dst), static_cast<const double*>(src), count, op, stream);
+            break;
+        case DataType::U8:
+            LaunchScalarByOp(static_cast<uint8_t*>(dst), static_cast<const uint8_t*>(src), count, op, stream);
+            break;
+        case DataType::I32:
+            LaunchScalarByOp(static_cast<int32_t*>(dst), static_cast<const int32_t*>(src), count, op, stream);
+            break;
+        case DataType::I64:
+            LaunchScalarByOp(static_cast<int64_t*>(dst), static_cast<const int64_t*>(src), count, op, stream);
+            break;
+    }
+    HIP_CHECK(hipGetLastError());
+}
+
+void LaunchReduceOut(void* dst, const void* a, const void* b, size_t count,
+                     DataType dt, ReduceOp op, hipStream_t stream) {
+    // dst = a; dst += b  (two passes is fine: memory-bound and rarely used;
+    // the in-place variant is the hot one).
+    HIP_CHECK(hipMemcpyAsync(dst, a, count * DtypeSize(dt), hipMemcpyDeviceToDevice, stream));
+    LaunchReduce(dst, b, count, dt, op, stream);
+}
+
+// ---------------------------------------------------------------------------
+// int8 block quantization with error feedback.
+// Wire block: [float scale][float reserved][int8 x block_elems], 8-byte header.
+
+namespace {
+
+__device__ __forceinline__ float LoadAsF32(const float* p, size_t i) { return p[i]; }
+__device__ __forceinline__ float LoadAsF32(const unsigned short* p, size_t i) {
+    return __bfloat162float(__hip_bfloat16_raw{p[i]});
+}
+__device__ __forceinline__ void StoreFromF32(float* p, size_t i, float v) { p[i] = v; }
+__device__ __forceinline__ void StoreFromF32(unsigned short* p, size_t i, float v) {
+    __hip_bfloat16 h = __float2bfloat16(v);
+    p[i] = reinterpret_cast<unsigned short&>(h);
+}
+
+// One workgroup per quant block: LDS max-reduction for the scale, then
+// quantize + update the error-feedback residual.
+template <typename T, bool USE_ERR>
+__global__ void QuantizeKernel(const T* __restrict__ in, T* __restrict__ err,
+                               uint8_t* __restrict__ wire, size_t count,
+                               size_t block_elems) {
+    const size_t blk = blockIdx.x;
+    const size_t base = blk * block_elems;
+    const size_t n = min(block_elems, count - base);
+    uint8_t* wblock = wire + blk * (block_elems + 8);
+    float* hdr = reinterpret_cast<float*>(wblock);
+    int8_t* payload = reinterpret_cast<int8_t*>(wblock + 8);
+
+    __shared__ float smax[kBlock / 64];
+    float m = 0.f;
+    for (size_t i = threadIdx.x; i < n; i += blockDim.x) {
+        float v = LoadAsF32(in, base + i);
+        if (USE_ERR) v += LoadAsF32(err, base + i);
+        m = fmaxf(m, fabsf(v));
+    }
+    // wave reduce
+    for (int off = 32; off > 0; off >>= 1)
+        m = fmaxf(m, __shfl_down(m, off, 64));
+    if ((threadIdx.x & 63) == 0) smax[threadIdx.x >> 6] = m;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        float mm = 0.f;
+        for (int w = 0; w < static_cast<int>(blockDim.x) / 64; ++w) mm = fmaxf(mm, smax[w]);
+        hdr[0] = mm > 0.f ? mm / 127.f : 1.f;
+        hdr[1] = 0.f;
+    }
+    __syncthreads();
+    const float scale = hdr[0];
+    const float inv = 1.f / scale;
+    for (size_t i = threadIdx.x; i < n; i += blockDim.x) {
+        float v = LoadAsF32(in, base + i);
+        if (USE_ERR) v += LoadAsF32(err, base + i);
+        float q = nearbyintf(v * inv);
+        q = fminf(127.f, fmaxf(-127.f, q));
+        payload[i] = static_cast<int8_t>(q);
+        if (USE_ERR) StoreFromF32(err, base + i, v - q * scale);
+    }
+    // Zero the tail so compressed-domain accumulation stays clean.
+    for (size_t i = n + threadIdx.x; i < block_elems; i += blockDim.x) payload[i] = 0;
+}
+
+template <typename T>
+__global__ void DequantizeKernel(const uint8_t* __restrict__ wire, T* __restrict__ out,
+                                 size_t count, size_t block_elems) {
+    const size_t blk = blockIdx.x;
+    const size_t base = blk * block_elems;
+    const size_t n = min(block_elems, count - base);
+    const uint8_t* wblock = wire + blk * (block_elems + 8);
+    const float scale = reinterpret_cast<const float*>(wblock)[0];
+    const int8_t* payload = reinterpret_cast<const int8_t*>(wblock + 8);
+    for (size_t i = threadIdx.x; i < n; i += blockDim.x)
+        StoreFromF32(out, base + i, static_cast<float>(payload[i]) * scale);
+}
+
+// acc_wire += wire in the compressed domain: dequant both, sum, requant with
+// a fresh scale (the reference's external reduce_sum hook, quant/quant.c:89).
+__global__ void QuantAccumKernel(uint8_t* __restrict__ acc, const uint8_t* __restrict__ in,
+                                 size_t count, size_t block_elems) {
+    const size_t blk = blockIdx.x;
+    const size_t base = blk * block_elems;
+    const size_t n = min(block_elems, count - base);
+    uint8_t* ablock = acc + blk * (block_elems + 8);
+    const uint8_t* iblock = in + blk * (block_elems + 8);
+    float* ahdr = reinterpret_cast<float*>(ablock);
+    const float as = ahdr[0];
+    const float is = reinterpret_cast<const float*>(iblock)[0];
+    int8_t* ap = reinterpret_cast<int8_t*>(ablock + 8);
+    const int8_t* ip = reinterpret_cast<const int8_t*>(iblock + 8);
+
+    __shared__ float smax[kBlock / 64];
+    float m = 0.f;
+    for (size_t i = threadIdx.x; i < n; i += blockDim.x) {
+        float v = static_cast<float>(ap[i]) * as + static_cast<float>(ip[i]) * is;
+        m = fmaxf(m, fabsf(v));
+    }
+    for (int off = 32; off > 0; off >>= 1)
+        m = fmaxf(m, __shfl_down(m, off, 64));
+    if ((threadIdx.x & 63) == 0) smax[threadIdx.x >> 6] = m;
+    __syncthreads();
+    __shared__ float new_scale;
+    if (threadIdx.x == 0) {
+        float mm = 0.f;
+        for (int w = 0; w < static_cast<int>(blockDim.x) / 64; ++w) mm = fmaxf(mm, smax[w]);
+        new_scale = mm > 0.f ? mm / 127.f : 1.f;
+    }
+    __syncthreads();
+    const float inv = 1.f / new_scale;
+    for (size_t i = threadIdx.x; i < n; i += blockDim.x) {
+        float v = static_cast<float>(ap[i]) * as + static_cast<float>(ip[i]) * is;
+        float q = nearbyintf(v * inv);
+        ap[i] = static_cast<int8_t>(fminf(127.f, fmaxf(-127.f, q)));
+    }
+    if (threadIdx.x == 0) ahdr[0] = new_scale;
+}
+
+}  // namespace
+
+void LaunchQuantize(const void* in, void* err, void* wire, size_t count,
+                    size_t block_elems, DataType dt, bool use_err,
+                    hipStream_t stream) {
+    const size_t nblocks = (count + block_elems - 1) / block_elems;
+    dim3 grid(static_cast<uint32_t>(nblocks));
+    if (dt == DataType::F32) {
+        if (use_err)
+            hipLaunchKernelGGL((QuantizeKernel<float, true>), grid, dim3(kBlock), 0, stream,
+                               static_cast<const float*>(in), static_cast<float*>(err),
+                               static_cast<uint8_t*>(wire), count, block_elems);
+        else
+            hipLaunchKernelGGL((QuantizeKernel<float, false>), grid, dim3(kBlock), 0, stream,
+                               static_cast<const float*>(in), static_cast<float*>(err),
+                               static_cast<uint8_t*>(wire), count, block_elems);
+    } else if (dt == DataType::BF16) {
+        if (use_err)
+            hipLaunchKernelGGL((QuantizeKernel<unsigned short, true>), grid, dim3(kBlock), 0, stream,
+                               static_cast<const unsigned short*>(in),
+                               static_cast<unsigned short*>(err),
+                               static_cast<uint8_t*>(wire), count, block_elems);
+        else
+            hipLaunchKernelGGL((QuantizeKernel<unsigned short, false>), grid, dim3(kBlock), 0, stream,
+                               static_cast<const unsigned short*>(in),
+                               static_cast<unsigned short*>(err),
+                               static_cast<uint8_t*>(wire), count, block_elems);
+    } else {
+        MLSL_THROW("quantization supports f32/bf16 only");
+    }
+    HIP_CHECK(hipGetLastError());
+}
+
+void LaunchDequantize(const void* wire, void* out, size_t count,
+                      size_t block_elems, DataType dt, hipStream_t stream) {
+    const size_t nblocks = (count + block_elems - 1) / block_elems;
+    dim3 grid(static_cast<uint32_t>(nblocks));
+    if (dt == DataType::F32) {
+        hipLaunchKernelGGL((DequantizeKernel<float>), grid, dim3(kBlock), 0, stream,
+                           static_cast<const uint8_t*>(wire), static_cast<float*>(out),
+                           count, block_elems);
+    } else if (dt == DataType::BF16) {
+        hipLaunchKernelGGL((DequantizeKernel<unsigned short>), grid, dim3(kBlock), 0, stream,
+                           static_cast<const uint8_t*>(wire),
+                           static_cast<unsigned short*>(out), count, block_elems);
+    } else {
+        MLSL_THROW("dequantization supports f32/bf16 only");
+    }
+    HIP_CHECK(hipGetLastError());
+}
+
+void LaunchQuantAccum(void* acc_wire, const void* wire, size_t count,
+                      size_t block_elems, hipStream_t stream) {
+    const size_t nblocks = (count + block_elems - 1) / block_elems;
+    hipLaunchKernelGGL(QuantAccumKernel, dim3(static_cast<uint32_t>(nblocks)),
+                       dim3(kBlock), 0, stream, static_cast<uint8_t*>(acc_wire),
+                       static_cast<const uint8_t*>(wire), count, block_elems);
+    HIP_CHECK(hipGetLastError());
+}
+
+// ---------------------------------------------------------------------------
+// Pack/unpack between [mb][fm][fmSize] layer layout and comm-buffer blocks.
+
+namespace {
+
+template <typename T, bool PACK>
+__global__ void PackKernel(const T* __restrict__ src, T* __restrict__ dst,
+                           PackBlockDesc d) {
+    const size_t total = d.mb_count * d.fm_count * d.fm_size;
+    const size_t tid = blockIdx.x * blockDim.x + threadIdx.x;
+    const size_t stride = gridDim.x * blockDim.x;
+    for (size_t i = tid; i < total; i += stride) {
+        const size_t k = i % d.fm_size;
+        const size_t fm = (i / d.fm_size) % d.fm_count;
+        const size_t mb = i / (d.fm_size * d.fm_count);
+        const size_t layer_idx =
+            ((d.mb_offset + mb) * d.local_fm_count + d.fm_offset + fm) * d.fm_size + k;
+        const size_t buf_idx = d.buf_offset + i;
+        if (PACK) dst[buf_idx] = src[layer_idx];
+        else dst[layer_idx] = src[buf_idx];
+    }
+}
+
+}  // namespace
+
+void LaunchPack(const void* src, void* dst, const PackBlockDesc& d, DataType dt,
+                hipStream_t stream) {
+    const size_t total = d.mb_count * d.fm_count * d.fm_size;
+    const int grid = GridFor(total);
+    switch (DtypeSize(dt)) {
+        case 4:
+            hipLaunchKernelGGL((PackKernel<uint32_t, true>), dim3(grid), dim3(kBlock), 0, stream,
+                               static_cast<const uint32_t*>(src), static_cast<uint32_t*>(dst), d);
+            break;
+        case 8:
+            hipLaunchKernelGGL((PackKernel<uint64_t, true>), dim3(grid), dim3(kBlock), 0, stream,
+                               static_cast<const uint64_t*>(src), static_cast<uint64_t*>(dst), d);
+            break;
+        case 2:
+            hipLaunchKernelGGL((PackKernel<uint16_t, true>), dim3(grid), dim3(kBlock), 0, stream,
+                               static_cast<const uint16_t*>(src), static_cast<uint16_t*>(dst), d);
+            break;
+        default:
+            hipLaunchKernelGGL((PackKernel<uint8_t, true>), dim3(grid), dim3(kBlock), 0, stream,
+                               static_cast<const uint8_t*>(src), static_cast<uint8_t*>(dst), d);
+    }
+    HIP_CHECK(hipGetLastError());
+}
+
+void LaunchUnpack(const void* src, void* dst, const PackBlockDesc& d, DataType dt,
+                  hipStream_t stream) {
+    const size_t total = d.mb_count * d.fm_count * d.fm_size;
+    const int grid = GridFor(total);
+    switch (DtypeSize(dt)) {
+        case 4:
+            hipLaunchKernelGGL((PackKernel<uint32_t, false>), dim3(grid), dim3(kBlock), 0, stream,
+                               static_cast<const uint32_t*>(src), static_cast<uint32_t*>(dst), d);
+            break;
+        case 8:
+            hipLaunchKernelGGL((PackKernel<uint64_t, false>), dim3(grid), dim3(kBlock), 0, stream,
+                               static_cast<const uint64_t*>(src), static_cast<uint64_t*>(dst), d);
+            break;
+        case 2:
+            hipLaunchKernelGGL((PackKernel<uint16_t, false>), dim3(grid), dim3(kBlock), 0, stream,
+                               static_cast<const uint16_t*>(src), static_cast<uint16_t*>(dst), d);
+            break;
+        default:
+            hipLaunchKernelGGL((PackKernel<uint8_t, false>), dim3(grid), dim3(kBlock), 0, stream,
+                               static_cast<const uint8_t*>(src), static_cast<uint8_t*>(dst), d);
+    }
+    HIP_CHECK(hipGetLastError());
+}
+
+}  // namespace mlsl

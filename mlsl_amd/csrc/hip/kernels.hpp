@@ -1,0 +1,54 @@
+// Host-callable launchers for the CDNA4 HIP kernels (gfx950).
+#pragma once
+
+#include <hip/hip_runtime.h>
+
+#include "../core/types.hpp"
+
+namespace mlsl {
+
+// dst[i] op= src[i] for count elements, on `stream`.
+// Memory-bound: vectorized 16-byte accesses, grid-stride, XCD-friendly
+// grid sizing (see kernels.hip).
+void LaunchReduce(void* dst, const void* src, size_t count, DataType dt,
+                  ReduceOp op, hipStream_t stream);
+
+// dst = a + b elementwise into a third buffer (out-of-place variant used by
+// fused pipelines).
+void LaunchReduceOut(void* dst, const void* a, const void* b, size_t count,
+                     DataType dt, ReduceOp op, hipStream_t stream);
+
+// --- int8 block quantization with error feedback (quant/quant.c contract,
+//     fused into the allreduce path; see comm/quant.cpp) ---
+// Wire block layout: [float scale][float reserved][int8 x block_elems].
+// in: fp32/bf16 gradients; err: same shape residual carried across steps
+// (error feedback); out: packed wire blocks.
+void LaunchQuantize(const void* in, void* err, void* wire, size_t count,
+                    size_t block_elems, DataType dt, bool use_err,
+                    hipStream_t stream);
+// Dequantize wire blocks into out (fp32/bf16).
+void LaunchDequantize(const void* wire, void* out, size_t count,
+                      size_t block_elems, DataType dt, hipStream_t stream);
+// Compressed-domain accumulate: acc_wire += wire (dequant-sum-requant per
+// block, matching the reference's reduce_sum plugin hook quant/quant.c:89-94).
+void LaunchQuantAccum(void* acc_wire, const void* wire, size_t count,
+                      size_t block_elems, hipStream_t stream);
+
+// Strided pack/unpack between layer layout [mb][fm][fmSize] and a contiguous
+// comm buffer block (CommBlockInfo contract; reference computes the shapes,
+// the consumer does the copy — tests/examples/mlsl_test/mlsl_test.cpp:214-254.
+// Here it is a library kernel so GPU consumers get a fused path).
+struct PackBlockDesc {
+    size_t mb_offset, mb_count;
+    size_t fm_offset, fm_count;
+    size_t fm_size;          // elements per feature map
+    size_t buf_offset;       // element offset in the comm buffer
+    size_t local_fm_count;   // feature maps per sample in the local layout
+    size_t local_mb_count;   // samples in the local layout
+};
+void LaunchPack(const void* src, void* dst, const PackBlockDesc& d, DataType dt,
+                hipStream_t stream);
+void LaunchUnpack(const void* src, void* dst, const PackBlockDesc& d, DataType dt,
+                  hipStream_t stream);
+
+}  // namespace mlsl

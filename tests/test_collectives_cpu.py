@@ -1,0 +1,45 @@
+"""Multi-process CPU (TCP transport) collective tests — the reference's
+`mpiexec -n N ./mlsl_test` matrix re-done natively on 127.0.0.1."""
+import pytest
+
+from tests.mp import run_ranks
+
+
+@pytest.mark.parametrize("world", [1, 2, 4])
+def test_plumbing_allreduce(world):
+    # driver config 1: mlsl_sample AllReduce COUNT=128 fp32
+    run_ranks("plumbing_allreduce", world)
+
+
+@pytest.mark.parametrize("world", [1, 2, 3, 4])
+def test_collectives_sweep(world):
+    run_ranks("collectives_sweep", world)
+
+
+def test_hybrid_grid():
+    run_ranks("hybrid_grid", 4)
+
+
+@pytest.mark.parametrize("world", [2])
+def test_inline_progress_mode(world):
+    run_ranks("collectives_sweep", world, extra_env={"MLSL_PROGRESS": "inline"})
+
+
+@pytest.mark.parametrize("world", [2, 4])
+def test_priority_concurrent(world):
+    run_ranks("priority_and_inline", world,
+              extra_env={"MLSL_MSG_PRIORITY": "1", "MLSL_MSG_PRIORITY_THRESHOLD": "4000"})
+
+
+def test_chunked_channels():
+    # exercise the chunk-over-channels fan-out (endpoint-parallelism analog)
+    run_ranks("collectives_sweep", 2,
+              extra_env={"MLSL_NUM_CHANNELS": "4", "MLSL_LARGE_MSG_SIZE_MB": "0"})
+
+
+def test_rhd_algo():
+    run_ranks("collectives_sweep", 4, extra_env={"MLSL_ALLREDUCE_ALGO": "rhd"})
+
+
+def test_ring_algo():
+    run_ranks("collectives_sweep", 3, extra_env={"MLSL_ALLREDUCE_ALGO": "ring"})

@@ -1,0 +1,243 @@
+"""Worker entry points for multi-process tests (see tests/mp.py).
+
+Each worker asserts analytic expected values; process exit code is the test
+result. Run as: python -m tests.workers <worker_name>
+"""
+import os
+import sys
+
+import numpy as np
+
+
+def _init():
+    import mlsl_amd as mx
+    mx.init()
+    return mx, mx.rank(), mx.world_size()
+
+
+def plumbing_allreduce(*, count=128):
+    """The mlsl_sample check (mlsl_to_oneccl/mlsl_sample.cpp:40-55):
+    AllReduce of `count` fp32, each rank contributes its idx;
+    expected per element = (size-1)*size/2."""
+    mx, rank, size = _init()
+    d = mx.Distribution(size, 1)
+    buf = np.full(count, float(rank), dtype=np.float32)
+    req = d.all_reduce(buf, buf, count, op="sum", group="data")
+    mx.wait(req)
+    want = (size - 1) * size / 2.0
+    assert np.all(buf == want), f"allreduce got {buf[:4]} want {want}"
+    d.barrier("global")
+    mx.finalize()
+
+
+def collectives_sweep():
+    """All 12 collectives with analytic values on the world group."""
+    mx, rank, size = _init()
+    d = mx.Distribution(size, 1)
+    g = "data"
+
+    # all_reduce sum + max, several sizes incl. chunked path
+    for count in (1, 7, 1024, 50000):
+        a = np.arange(count, dtype=np.float32) + rank
+        out = np.zeros(count, dtype=np.float32)
+        mx.wait(d.all_reduce(a, out, count, op="sum", group=g))
+        want = size * np.arange(count, dtype=np.float32) + size * (size - 1) / 2.0
+        assert np.allclose(out, want), f"allreduce count={count}"
+    a = np.full(64, float(rank * 3), dtype=np.float32)
+    mx.wait(d.all_reduce(a, a, 64, op="max", group=g))
+    assert np.all(a == (size - 1) * 3.0), "allreduce max in-place"
+
+    # f64 + i64
+    a = np.arange(100, dtype=np.float64) * (rank + 1)
+    out = np.zeros(100, dtype=np.float64)
+    mx.wait(d.all_reduce(a, out, 100, op="sum", group=g))
+    want = np.arange(100, dtype=np.float64) * (size * (size + 1) / 2.0)
+    assert np.allclose(out, want), "f64 allreduce"
+    a = np.arange(64, dtype=np.int64) + rank
+    out = np.zeros(64, dtype=np.int64)
+    mx.wait(d.all_reduce(a, out, 64, op="sum", group=g))
+    assert np.all(out == size * np.arange(64) + size * (size - 1) // 2), "i64"
+
+    # bcast
+    b = np.zeros(33, dtype=np.float32)
+    if rank == min(1, size - 1):
+        b[:] = np.arange(33) + 5
+    mx.wait(d.bcast(b, 33, root=min(1, size - 1), group=g))
+    assert np.all(b == np.arange(33) + 5), "bcast"
+
+    # reduce
+    a = np.full(17, float(rank + 1), dtype=np.float32)
+    out = np.zeros(17, dtype=np.float32)
+    mx.wait(d.reduce(a, out, 17, op="sum", root=0, group=g))
+    if rank == 0:
+        assert np.all(out == size * (size + 1) / 2.0), "reduce"
+
+    # reduce_scatter: rank r gets sum of segment r
+    per = 13
+    a = np.arange(per * size, dtype=np.float32) + rank
+    out = np.zeros(per, dtype=np.float32)
+    mx.wait(d.reduce_scatter(a, out, per, op="sum", group=g))
+    seg = np.arange(rank * per, (rank + 1) * per, dtype=np.float32)
+    want = size * seg + size * (size - 1) / 2.0
+    assert np.allclose(out, want), "reduce_scatter"
+
+    # all_gather
+    a = np.full(9, float(rank * 100), dtype=np.float32) + np.arange(9, dtype=np.float32)
+    out = np.zeros(9 * size, dtype=np.float32)
+    mx.wait(d.all_gather(a, 9, out, group=g))
+    for j in range(size):
+        assert np.all(out[j * 9:(j + 1) * 9] == j * 100 + np.arange(9)), "all_gather"
+
+    # all_gatherv
+    counts = [3 + 2 * i for i in range(size)]
+    a = np.full(counts[rank], float(rank), dtype=np.float32)
+    out = np.zeros(sum(counts), dtype=np.float32)
+    mx.wait(d.all_gatherv(a, counts[rank], out, counts, group=g))
+    off = 0
+    for j in range(size):
+        assert np.all(out[off:off + counts[j]] == j), "all_gatherv"
+        off += counts[j]
+
+    # gather / scatter
+    a = np.full(5, float(rank), dtype=np.float32)
+    out = np.zeros(5 * size, dtype=np.float32)
+    mx.wait(d.gather(a, 5, out, root=0, group=g))
+    if rank == 0:
+        for j in range(size):
+            assert np.all(out[j * 5:(j + 1) * 5] == j), "gather"
+    sb = np.arange(4 * size, dtype=np.float32) if rank == 0 else np.zeros(4 * size, np.float32)
+    rb = np.zeros(4, dtype=np.float32)
+    mx.wait(d.scatter(sb, rb, 4, root=0, group=g))
+    assert np.all(rb == np.arange(rank * 4, rank * 4 + 4)), "scatter"
+
+    # alltoall
+    per = 6
+    a = np.zeros(per * size, dtype=np.float32)
+    for j in range(size):
+        a[j * per:(j + 1) * per] = rank * 1000 + j * 10 + np.arange(per)
+    out = np.zeros(per * size, dtype=np.float32)
+    mx.wait(d.all_to_all(a, per, out, group=g))
+    for j in range(size):
+        assert np.all(out[j * per:(j + 1) * per] == j * 1000 + rank * 10 + np.arange(per)), "a2a"
+
+    # alltoallv (ragged)
+    scnt = [(rank + j) % size + 1 for j in range(size)]
+    rcnt = [(j + rank) % size + 1 for j in range(size)]
+    soff, acc = [], 0
+    for c in scnt:
+        soff.append(acc)
+        acc += c
+    roff, acc = [], 0
+    for c in rcnt:
+        roff.append(acc)
+        acc += c
+    a = np.zeros(sum(scnt), dtype=np.float32)
+    for j in range(size):
+        a[soff[j]:soff[j] + scnt[j]] = rank * 100 + j
+    out = np.zeros(sum(rcnt), dtype=np.float32)
+    mx.wait(d.all_to_allv(a, scnt, soff, out, rcnt, roff, group=g))
+    for j in range(size):
+        assert np.all(out[roff[j]:roff[j] + rcnt[j]] == j * 100 + rank), "a2av"
+
+    # barrier + test() path
+    d.barrier("global")
+    a = np.full(1000, float(rank), dtype=np.float32)
+    out = np.zeros(1000, dtype=np.float32)
+    req = d.all_reduce(a, out, 1000, op="sum", group=g)
+    import time
+    done, _ = mx.test(req)
+    t0 = time.time()
+    while not done:
+        assert time.time() - t0 < 60, "test() never completed"
+        done, _ = mx.test(req)
+    assert np.all(out == size * (size - 1) / 2.0 + 0 * out), "test-completed allreduce"
+
+    mx.finalize()
+
+
+def hybrid_grid():
+    """Distribution grid math on a data x model grid (world must be 4):
+    checks group indices/counts and per-group collectives."""
+    mx, rank, size = _init()
+    assert size == 4
+    d = mx.Distribution(2, 2)
+    # lId = rank % 4; iM = lId / 2 (data idx), iF = lId % 2 (model idx)
+    assert d.process_count("data") == 2
+    assert d.process_count("model") == 2
+    assert d.process_idx("data") == rank // 2
+    assert d.process_idx("model") == rank % 2
+
+    # model-group allreduce: partners are {0,1} and {2,3}
+    a = np.full(8, float(rank), dtype=np.float32)
+    out = np.zeros(8, dtype=np.float32)
+    mx.wait(d.all_reduce(a, out, 8, op="sum", group="model"))
+    want = {0: 1.0, 1: 1.0, 2: 5.0, 3: 5.0}[rank]
+    assert np.all(out == want), f"model group allreduce got {out[0]} want {want}"
+
+    # data-group allreduce: partners are {0,2} and {1,3}
+    mx.wait(d.all_reduce(a, out, 8, op="sum", group="data"))
+    want = {0: 2.0, 1: 4.0, 2: 2.0, 3: 4.0}[rank]
+    assert np.all(out == want), f"data group allreduce got {out[0]} want {want}"
+
+    # colors variant: same grouping built by hand
+    d2 = mx.Distribution(colors=(rank % 2, rank // 2))
+    assert d2.process_count("data") == 2
+    assert d2.process_count("model") == 2
+    mx.wait(d2.all_reduce(a, out, 8, op="sum", group="model"))
+    want = {0: 1.0, 1: 1.0, 2: 5.0, 3: 5.0}[rank]
+    assert np.all(out == want), "colors model group"
+    mx.finalize()
+
+
+def srlist_ring():
+    """SendRecvList neighbor exchange via the C++ API's SRLIST (exercised
+    through the planner later; here via raw distribution barrier + manual
+    check is skipped on the Python surface for now)."""
+    mx, rank, size = _init()
+    d = mx.Distribution(size, 1)
+    d.barrier("global")
+    mx.finalize()
+
+
+def priority_and_inline():
+    """MLSL_PROGRESS=inline and MLSL_MSG_PRIORITY paths still compute
+    correct results under concurrent requests."""
+    mx, rank, size = _init()
+    d = mx.Distribution(size, 1)
+    reqs = []
+    bufs = []
+    for k in range(8):
+        n = 1000 * (k + 1)
+        a = np.full(n, float(rank + k), dtype=np.float32)
+        out = np.zeros(n, dtype=np.float32)
+        reqs.append(d.all_reduce(a, out, n, op="sum", group="data"))
+        bufs.append((a, out, k))
+    for req, (a, out, k) in zip(reqs, bufs):
+        mx.wait(req)
+        want = size * k + size * (size - 1) / 2.0
+        assert np.all(out == want), f"concurrent req {k}"
+    mx.finalize()
+
+
+WORKERS = {
+    "plumbing_allreduce": plumbing_allreduce,
+    "collectives_sweep": collectives_sweep,
+    "hybrid_grid": hybrid_grid,
+    "srlist_ring": srlist_ring,
+    "priority_and_inline": priority_and_inline,
+}
+
+
+def main():
+    name = sys.argv[1]
+    fn = WORKERS.get(name)
+    if fn is None:
+        # planner workers live in a separate module to keep this one lean
+        from tests import workers_planner
+        fn = workers_planner.WORKERS[name]
+    fn()
+    print(f"OK {name} rank={os.environ.get('RANK')}")
+
+
+if __name__ == "__main__":
+    main()
