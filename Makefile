@@ -25,7 +25,8 @@ CSRC := \
     mlsl_amd/csrc/comm/device_comm.cpp \
     mlsl_amd/csrc/dl/environment.cpp \
     mlsl_amd/csrc/dl/session.cpp \
-    mlsl_amd/csrc/bind/c_api.cpp
+    mlsl_amd/csrc/bind/c_api.cpp \
+    mlsl_amd/csrc/bind/ops_api.cpp
 
 HIPSRC := mlsl_amd/csrc/hip/kernels.hip
 

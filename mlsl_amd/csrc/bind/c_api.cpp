@@ -20,6 +20,9 @@ extern "C" {
 
 const char* mlsl_last_error(void) { return g_last_error.c_str(); }
 
+/* used by ops_api.cpp so all bindings share one error slot */
+void mlsl_set_last_error_impl(const char* msg) { g_last_error = msg ? msg : ""; }
+
 #define C_TRY try {
 #define C_CATCH                                                                \
     return MLSL_SUCCESS;                                                       \

@@ -1,0 +1,296 @@
+"""Planner (Session/Operation/Activation/ParameterSet) workers with analytic
+expected values — the native rebuild of the reference's mlsl_test
+(tests/examples/mlsl_test/mlsl_test.cpp): a 2-layer synthetic CC net run
+over the {model_parts} x {dist_update} matrix, verifying forward activation
+exchange, backward gradient exchange, reduced parameter gradients and
+distributed-update increments.
+
+Layout convention: activations are [mb][fm][fm_size] float32; pack/unpack is
+done with the CommBlockInfo geometry the planner reports.
+"""
+import os
+
+import numpy as np
+
+
+def _init():
+    import mlsl_amd as mx
+    mx.init()
+    return mx, mx.rank(), mx.world_size()
+
+
+def _pack(act, local_mb, buf, local):
+    """Pack local [mb][fm][S] into the comm buffer per pack blocks."""
+    S = act.fm_size
+    lf = act.local_fm_count
+    for i in range(act.pack_block_count):
+        b = act.pack_block(i)
+        src = local.reshape(local_mb, lf, S)[
+            b.mb_offset:b.mb_offset + b.mb_count,
+            b.fm_offset:b.fm_offset + b.fm_count, :]
+        n = b.mb_count * b.fm_count * b.fm_size
+        buf[b.buf_offset:b.buf_offset + n] = src.reshape(-1)
+
+
+def _unpack(act, local_mb, result, local):
+    S = act.fm_size
+    lf = act.local_fm_count
+    for i in range(act.unpack_block_count):
+        b = act.unpack_block(i)
+        n = b.mb_count * b.fm_count * b.fm_size
+        blk = result[b.buf_offset:b.buf_offset + n].reshape(
+            b.mb_count, b.fm_count, b.fm_size)
+        local.reshape(local_mb, lf, S)[
+            b.mb_offset:b.mb_offset + b.mb_count,
+            b.fm_offset:b.fm_offset + b.fm_count, :] = blk
+
+
+def _as_np(ptr, n):
+    import ctypes
+    buf = (ctypes.c_float * n).from_address(ptr)
+    return np.frombuffer(buf, dtype=np.float32)
+
+
+def grid_shapes():
+    """Pure shape math at world 4, grid (2,2) and (4,1): fm/kernel sharding
+    and distributed-update owned counts (reference mlsl_impl.cpp:388-444)."""
+    mx, rank, size = _init()
+    assert size == 4
+    s = mx.Session()
+    s.set_global_minibatch_size(16)
+    d22 = mx.Distribution(2, 2)
+
+    info = s.create_op_reg_info("cc")
+    info.set_name("fc")
+    info.add_input(64, 9, "f32")
+    info.add_output(128, 9, "f32")
+    info.add_parameter_set(64 * 128, 4, "f32", distributed_update=False)
+    info.validate(d22)
+    oi = s.add_operation(info, d22)
+    op = s.operation(oi)
+
+    assert op.global_minibatch_size == 16
+    assert op.local_minibatch_size == 8
+    assert op.global_minibatch_offset == 8 * (rank // 2)
+
+    ia, oa = op.input(0), op.output(0)
+    assert ia.global_fm_count == 64 and ia.local_fm_count == 32
+    assert ia.global_fm_offset == 32 * (rank % 2)
+    # CC output keeps the full fm range locally (partial sums)
+    assert oa.global_fm_count == 128 and oa.local_fm_count == 128
+
+    ps = op.parameter_set(0)
+    assert ps.global_kernel_count == 64 * 128
+    assert ps.local_kernel_count == 64 * 128 // 2
+    assert ps.owned_kernel_count == ps.local_kernel_count
+    assert ps.owned_kernel_offset == 0
+    assert not ps.distributed_update
+
+    info2 = s.create_op_reg_info("cc")
+    info2.add_input(64, 9, "f32")
+    info2.add_output(128, 9, "f32")
+    info2.add_parameter_set(64 * 128, 4, "f32", distributed_update=True)
+    oi2 = s.add_operation(info2, d22)
+    ps2 = s.operation(oi2).parameter_set(0)
+    # local = 4096, dp = 2 -> owned = 2048, offset = 2048 * data_idx
+    assert ps2.owned_kernel_count == 2048
+    assert ps2.local_kernel_count == 4096
+    assert ps2.owned_kernel_offset == 2048 * (rank // 2)
+    assert ps2.distributed_update
+    mx.finalize()
+
+
+def mlsl_net():
+    """The 2-layer net over the reference's run matrix. Parameters via env:
+    MP (model parts), DIST_UPDATE (0/1). World size divisible by MP."""
+    mp = int(os.environ.get("MP", "1"))
+    dist_update = os.environ.get("DIST_UPDATE", "0") == "1"
+    mx, rank, size = _init()
+    dp = size // mp
+    MB = 8 * dp           # global minibatch
+    S = 6                 # fm size (elements per feature map)
+    F0, F1, F2 = 8 * mp, 16 * mp, 8 * mp   # fm counts: in, mid, out
+    KS = 4                # kernel size
+
+    s = mx.Session()
+    s.set_global_minibatch_size(MB)
+    d = mx.Distribution(dp, mp)
+    didx = d.process_idx("data")
+    midx = d.process_idx("model")
+    lmb = MB // dp
+
+    def make_op(name, fin, fout, du):
+        info = s.create_op_reg_info("cc")
+        info.set_name(name)
+        info.add_input(fin, S, "f32")
+        info.add_output(fout, S, "f32")
+        info.add_parameter_set(fin * fout, KS, "f32", distributed_update=du)
+        info.validate(d)
+        return s.operation(s.add_operation(info, d))
+
+    op0 = make_op("fc0", F0, F1, dist_update)
+    op1 = make_op("fc1", F1, F2, dist_update)
+    op0.set_next(op1, 0, 0)   # op0.out[0] -> op1.in[0]
+    s.commit()
+
+    out0, in1 = op0.output(0), op1.input(0)
+    f1_local = in1.local_fm_count          # F1/mp
+    assert f1_local == F1 // mp
+
+    if mp > 1:
+        # case 1: fwd ReduceScatter, bwd AllGather over the model group
+        assert out0.pack_block_count == mp
+        assert in1.unpack_block_count == 1
+        assert out0.comm_buf_size == mp * f1_local * lmb * S * 4
+
+    comm0 = np.zeros(max(out0.comm_buf_size, 4) // 4, dtype=np.float32)
+    comm1 = np.zeros(max(in1.comm_buf_size, 4) // 4, dtype=np.float32)
+
+    ps0 = op0.parameter_set(0)
+    grad0 = np.zeros(ps0.local_kernel_count * KS, dtype=np.float32)
+
+    for it in range(3):  # reference: epochs x minibatches loop
+        # ---- forward: op0 produces partial sums over the model group ----
+        # partial(b, fm, k) from model-rank m = fm*1000 + k + b*7 + it + m
+        if mp > 1:
+            part = np.zeros((lmb, F1, S), dtype=np.float32)
+            b_idx = np.arange(lmb)[:, None, None]
+            fm_idx = np.arange(F1)[None, :, None]
+            k_idx = np.arange(S)[None, None, :]
+            part[:] = fm_idx * 1000 + k_idx + b_idx * 7 + it + midx
+            _pack(out0, lmb, comm0, part)
+            out0.start_comm(comm0)
+
+            ptr = in1.wait_comm()
+            res = _as_np(ptr, f1_local * lmb * S)
+            inp = np.zeros((lmb, f1_local, S), dtype=np.float32)
+            _unpack(in1, lmb, res, inp)
+            gfm = midx * f1_local + np.arange(f1_local)[None, :, None]
+            want = (mp * (gfm * 1000 + k_idx + b_idx * 7 + it)
+                    + mp * (mp - 1) / 2.0)
+            assert np.allclose(inp, want), \
+                f"fwd case1 mismatch it={it} rank={rank}: {inp[0,0,:3]} vs {want[0,0,:3]}"
+
+        # ---- backward: op1 sends input-grads back (AllGather) ----
+        if mp > 1:
+            g = np.zeros((lmb, f1_local, S), dtype=np.float32)
+            gfm = midx * f1_local + np.arange(f1_local)[None, :, None]
+            g[:] = gfm * 2000 + k_idx + b_idx + it
+            _pack(in1, lmb, comm1, g)
+            in1.start_comm(comm1)
+
+            ptr = out0.wait_comm()
+            res = _as_np(ptr, F1 * lmb * S)
+            full = np.zeros((lmb, F1, S), dtype=np.float32)
+            _unpack(out0, lmb, res, full)
+            fm_all = np.arange(F1)[None, :, None]
+            want = fm_all * 2000 + k_idx + b_idx + it
+            assert np.allclose(full, want), f"bwd case1 mismatch it={it} rank={rank}"
+
+        # ---- parameter gradient exchange over the data group ----
+        # grad(j, e) from data-rank dd = (gk0 + j)*10 + e + dd + it
+        lk = ps0.local_kernel_count
+        gk0 = ps0.global_kernel_offset
+        j_idx = np.arange(lk)[:, None]
+        e_idx = np.arange(KS)[None, :]
+        grad0.reshape(lk, KS)[:] = (gk0 + j_idx) * 10 + e_idx + didx + it
+        ps0.start_gradient_comm(grad0)
+        ptr = ps0.wait_gradient_comm()
+
+        if dp > 1:
+            if dist_update:
+                ok, oo = ps0.owned_kernel_count, ps0.owned_kernel_offset
+                shard = _as_np(ptr, ok * KS).reshape(ok, KS)
+                jj = oo + np.arange(ok)[:, None]
+                want = dp * ((gk0 + jj) * 10 + e_idx + it) + dp * (dp - 1) / 2.0
+                assert np.allclose(shard, want), f"grad RS mismatch it={it} rank={rank}"
+                # increments: owner writes inc(j,e) = (gk0+j)*3 + e + it,
+                # AllGather distributes to the full local buffer
+                inc = grad0  # reuse buffer, reference does the same
+                inc.reshape(lk, KS)[oo:oo + ok, :] = (gk0 + jj) * 3 + e_idx + it
+                ps0.start_increment_comm(inc)
+                ptr2 = ps0.wait_increment_comm()
+                full = _as_np(ptr2, lk * KS).reshape(lk, KS)
+                want_full = (gk0 + j_idx) * 3 + e_idx + it
+                assert np.allclose(full, want_full), f"inc AG mismatch it={it} rank={rank}"
+            else:
+                red = _as_np(ptr, lk * KS).reshape(lk, KS)
+                want = dp * ((gk0 + j_idx) * 10 + e_idx + it) + dp * (dp - 1) / 2.0
+                assert np.allclose(red, want), f"grad AR mismatch it={it} rank={rank}"
+        else:
+            red = _as_np(ptr, lk * KS).reshape(lk, KS)
+            assert np.allclose(red, grad0.reshape(lk, KS)), "dp=1 grad passthrough"
+
+    # stats surface
+    st = s.stats
+    if st.enabled:
+        assert st.total_comm_size > 0
+        st.print()
+    d.barrier("global")
+    mx.finalize()
+
+
+def alltoall_transition():
+    """Cases 4/5: DP<->MP layout transition via AlltoAll
+    (reference mlsl_impl.cpp:203-226). op0 is pure-DP (out mp=1, ACT so no
+    reduce), op1 is pure-MP; world = N."""
+    mx, rank, size = _init()
+    N = size
+    if N == 1:
+        mx.finalize()
+        return
+    MB = 4 * N
+    S = 3
+    F = 4 * N
+
+    s = mx.Session()
+    s.set_global_minibatch_size(MB)
+    d_dp = mx.Distribution(N, 1)
+    d_mp = mx.Distribution(1, N)
+
+    info0 = s.create_op_reg_info("act")
+    info0.add_input(F, S, "f32")
+    info0.add_output(F, S, "f32")
+    op0 = s.operation(s.add_operation(info0, d_dp))
+    info1 = s.create_op_reg_info("act")
+    info1.add_input(F, S, "f32")
+    info1.add_output(F, S, "f32")
+    op1 = s.operation(s.add_operation(info1, d_mp))
+    op0.set_next(op1, 0, 0)
+    s.commit()
+
+    out0, in1 = op0.output(0), op1.input(0)
+    lmb0 = MB // N          # op0 local minibatch
+    lmb1 = MB               # op1 local minibatch (dp=1)
+    lf0 = F                 # op0 local fm (mp=1)
+    lf1 = F // N            # op1 local fm
+    assert out0.pack_block_count == N and in1.unpack_block_count == N
+
+    # fwd: value(b_global, fm, k) = b_global*10000 + fm*100 + k
+    local = np.zeros((lmb0, lf0, S), dtype=np.float32)
+    b0 = op0.global_minibatch_offset
+    bg = (b0 + np.arange(lmb0))[:, None, None]
+    fm = np.arange(lf0)[None, :, None]
+    k = np.arange(S)[None, None, :]
+    local[:] = bg * 10000 + fm * 100 + k
+
+    comm0 = np.zeros(out0.comm_buf_size // 4, dtype=np.float32)
+    _pack(out0, lmb0, comm0, local)
+    out0.start_comm(comm0)
+    ptr = in1.wait_comm()
+    res = _as_np(ptr, lmb1 * lf1 * S)
+    got = np.zeros((lmb1, lf1, S), dtype=np.float32)
+    _unpack(in1, lmb1, res, got)
+    # op1 (model-rank r) holds fms [r*lf1, (r+1)*lf1) for the WHOLE minibatch
+    fm1 = (rank * lf1 + np.arange(lf1))[None, :, None]
+    bg1 = np.arange(lmb1)[:, None, None]
+    want = bg1 * 10000 + fm1 * 100 + k
+    assert np.allclose(got, want), f"a2a fwd mismatch rank={rank}"
+    mx.finalize()
+
+
+WORKERS = {
+    "grid_shapes": grid_shapes,
+    "mlsl_net": mlsl_net,
+    "alltoall_transition": alltoall_transition,
+}
