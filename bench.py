@@ -65,9 +65,11 @@ def main():
         local = int(os.environ.get("LOCAL_RANK", rank))
         torch.cuda.set_device(local % torch.cuda.device_count())
         buf = torch.randn(count, dtype=torch.float32, device="cuda")
+        out = torch.empty_like(buf)
     else:
         import numpy as np
         buf = np.random.randn(count).astype(np.float32)
+        out = np.empty_like(buf)
 
     d = mx.Distribution(size, 1)
 
@@ -76,7 +78,9 @@ def main():
             torch.cuda.synchronize()
 
     def one_step():
-        mx.wait(d.all_reduce(buf, buf, count, op="sum", group="data"))
+        # out-of-place: at n_gpus=1 this is a measured 2x256MiB HBM pass,
+        # never a skipped no-op; at n>1 it is the standard allreduce shape.
+        mx.wait(d.all_reduce(buf, out, count, op="sum", group="data"))
 
     for _ in range(args.warmup):
         one_step()

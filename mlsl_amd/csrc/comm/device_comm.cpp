@@ -144,7 +144,12 @@ class HipRuntime : public DeviceRuntime {
 
     GroupComms& For(ProcessGroup* g) {
         auto it = group_comms_.find(g->Uid());
-        MLSL_CHECK(it != group_comms_.end(), "group has no device comms");
+        if (it == group_comms_.end()) {
+            // Single-rank / self groups need no communicators; requests on
+            // them take the local-copy path.
+            MLSL_CHECK(g->Size() <= 1, "multi-rank group has no device comms");
+            it = group_comms_.emplace(g->Uid(), GroupComms{}).first;
+        }
         return it->second;
     }
 
@@ -377,12 +382,16 @@ bool DeviceAdvanceRequest(CommRequest* req, DeviceReqState& st) {
             (cfg.allreduce_algo == AllReduceAlgo::RING || cfg.allreduce_algo == AllReduceAlgo::RHD) &&
             req->Spec().op == CollOp::ALLREDUCE && req->Group()->Size() > 1;
 
-        // Single-rank groups: local copies only, still through the stream
-        // so completion semantics are uniform.
+        // Single-rank groups: local copies only, still through a (persistent)
+        // stream so completion semantics are uniform.
         if (gc.comms.empty()) {
+            if (gc.streams.empty()) {
+                hipStream_t s0;
+                HIP_CHECKD(hipStreamCreateWithFlags(&s0, hipStreamNonBlocking));
+                gc.streams.push_back(s0);
+            }
+            hipStream_t s0 = gc.streams[0];
             const size_t es = DtypeSize(req->Dtype());
-            hipStream_t s0;
-            HIP_CHECKD(hipStreamCreateWithFlags(&s0, hipStreamNonBlocking));
             for (auto& ce : chunks) {
                 const uint8_t* sbase = req->SendBuf() + ce.elem_off * es;
                 uint8_t* rbase = req->RecvBuf() + ce.elem_off * es;
@@ -396,7 +405,6 @@ bool DeviceAdvanceRequest(CommRequest* req, DeviceReqState& st) {
                 st.events.push_back(e);
             }
             HIP_CHECKD(hipEventRecord(st.events[0], s0));
-            HIP_CHECKD(hipStreamDestroy(s0));  // event keeps the work
             st.issued = true;
             return hipEventQuery(st.events[0]) == hipSuccess;
         }
