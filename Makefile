@@ -16,6 +16,7 @@ CSRC := \
     mlsl_amd/csrc/core/log.cpp \
     mlsl_amd/csrc/core/config.cpp \
     mlsl_amd/csrc/comm/schedule.cpp \
+    mlsl_amd/csrc/comm/quant.cpp \
     mlsl_amd/csrc/comm/bootstrap.cpp \
     mlsl_amd/csrc/comm/mesh.cpp \
     mlsl_amd/csrc/comm/group.cpp \

@@ -1,0 +1,72 @@
+#!/usr/bin/env python3
+"""Single-GPU kernel microbenchmarks (run under rocprofv3 for the committed
+profiles/ evidence): local reduce streaming bandwidth, int8 quantize /
+dequantize / compressed-accumulate throughput, pack/unpack."""
+import json
+import sys
+import time
+
+import torch
+
+sys.path.insert(0, __file__.rsplit("/", 2)[0])
+from mlsl_amd import ops  # noqa: E402
+
+
+def timed(fn, iters=20, warmup=3):
+    for _ in range(warmup):
+        fn()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(iters):
+        fn()
+    torch.cuda.synchronize()
+    return (time.perf_counter() - t0) / iters
+
+
+def main():
+    assert torch.cuda.is_available()
+    torch.cuda.set_device(0)
+    n = 1 << 26  # 64M elements = 256 MiB fp32
+    res = {}
+
+    a = torch.randn(n, device="cuda")
+    b = torch.randn(n, device="cuda")
+    dt = timed(lambda: ops.reduce_(a, b, n))
+    res["reduce_f32_256MiB_TBps"] = round(3 * 4 * n / dt / 1e12, 3)
+
+    abf = torch.randn(n, device="cuda", dtype=torch.bfloat16)
+    bbf = torch.randn(n, device="cuda", dtype=torch.bfloat16)
+    dt = timed(lambda: ops.reduce_(abf, bbf, n, dtype="bf16"))
+    res["reduce_bf16_128MiB_TBps"] = round(3 * 2 * n / dt / 1e12, 3)
+
+    wire = torch.empty(ops.wire_bytes(n), device="cuda", dtype=torch.uint8)
+    err = torch.zeros(n, device="cuda")
+    dt = timed(lambda: ops.quantize(a, wire, n, err=err))
+    # reads in+err (8B/elem), writes wire(~1B)+err(4B)
+    res["quantize_f32_TBps"] = round((8 + 1 + 4) * n / dt / 1e12, 3)
+    res["quantize_f32_Gelems_s"] = round(n / dt / 1e9, 2)
+
+    out = torch.empty_like(a)
+    dt = timed(lambda: ops.dequantize(wire, out, n))
+    res["dequantize_f32_TBps"] = round((1 + 4) * n / dt / 1e12, 3)
+
+    wb = torch.empty_like(wire)
+    ops.quantize(b, wb, n)
+    dt = timed(lambda: ops.quant_accum(wire, wb, n))
+    res["quant_accum_TBps"] = round(3 * 1 * n / dt / 1e12, 3)
+
+    # pack: half the fms of a [mb][fm][s] block
+    mb, fm, s = 64, 512, 256
+    src = torch.randn(mb * fm * s, device="cuda")
+    dst = torch.empty(mb * (fm // 2) * s, device="cuda")
+    kw = dict(mb_offset=0, mb_count=mb, fm_offset=fm // 4, fm_count=fm // 2,
+              fm_size=s, buf_offset=0, local_fm_count=fm, local_mb_count=mb,
+              dtype="f32")
+    dt = timed(lambda: ops.pack(src, dst, **kw))
+    res["pack_f32_TBps"] = round(2 * 4 * mb * (fm // 2) * s / dt / 1e12, 3)
+
+    print(json.dumps(res, indent=1))
+
+
+if __name__ == "__main__":
+    main()

@@ -20,6 +20,7 @@
 #include "../core/log.hpp"
 #include "../hip/kernels.hpp"
 #include "context.hpp"
+#include "quant.hpp"
 #include "device_state.hpp"
 #include "group.hpp"
 #include "request.hpp"
@@ -174,10 +175,23 @@ DeviceRuntime* CreateDeviceRuntime() {
 
 void DeviceSetupRequest(CommRequest* req, DeviceReqState& st) {
     // Persistent scratch for schedule-path chunks + quant/barrier staging.
-    size_t tmp = req->GetTmpBytes();
+    size_t tmp = 0;
+    for (auto& ce : req->Chunks()) {
+        if (req->Compressed()) {
+            // [wire][schedule scratch][error-feedback residual]
+            tmp += ce.sch.result.bytes + ce.sch.tmp_bytes +
+                   req->Spec().count * DtypeSize(req->Dtype());
+        } else {
+            tmp += ce.sch.tmp_bytes;
+        }
+    }
     if (req->Spec().op == CollOp::BARRIER) tmp = std::max<size_t>(tmp, 16);
     st.tmp_bytes = tmp;
-    if (tmp) HIP_CHECKD(hipMalloc(&st.tmp_dev, tmp));
+    if (tmp) {
+        HIP_CHECKD(hipMalloc(&st.tmp_dev, tmp));
+        // zero once: error-feedback residuals must start clean
+        HIP_CHECKD(hipMemset(st.tmp_dev, 0, tmp));
+    }
 }
 
 namespace {
@@ -322,11 +336,9 @@ void IssueFused(CommRequest* req, ChunkExec& ce, ncclComm_t comm, hipStream_t s,
 // hand-written collective path (ring/RHD over explicit p2p) — RCCL supplies
 // transport, OUR schedule supplies the algorithm and OUR kernels the math.
 void IssueSchedule(CommRequest* req, ChunkExec& ce, ncclComm_t comm, hipStream_t s,
-                   uint8_t* tmp_dev) {
+                   const uint8_t* sbase, uint8_t* rbase, uint8_t* tmp_dev) {
     ProcessGroup* g = req->Group();
     const size_t es = DtypeSize(req->Dtype());
-    const uint8_t* sbase = req->SendBuf() + ce.elem_off * es;
-    uint8_t* rbase = req->RecvBuf() + ce.elem_off * es;
     const ncclDataType_t ndt = ncclUint8;  // byte-addressed transfers
 
     auto ptr = [&](const BufRef& b) -> uint8_t* {
@@ -362,6 +374,10 @@ void IssueSchedule(CommRequest* req, ChunkExec& ce, ncclComm_t comm, hipStream_t
                 if (d != src)
                     HIP_CHECKD(hipMemcpyAsync(d, src, st.local_src.bytes,
                                               hipMemcpyDeviceToDevice, s));
+            } else if (ce.sch.quant_block > 0) {
+                const size_t blk = ce.sch.quant_block;
+                const size_t units = st.local_dst.bytes / (blk + 8);
+                LaunchQuantAccum(d, src, units * blk, blk, s);
             } else {
                 LaunchReduce(d, src, st.local_dst.bytes / es, req->Dtype(), ce.sch.rop, s);
             }
@@ -382,6 +398,7 @@ bool DeviceAdvanceRequest(CommRequest* req, DeviceReqState& st) {
             (cfg.allreduce_algo == AllReduceAlgo::RING || cfg.allreduce_algo == AllReduceAlgo::RHD) &&
             req->Spec().op == CollOp::ALLREDUCE && req->Group()->Size() > 1;
 
+        const bool compressed = req->Compressed();
         // Single-rank groups: local copies only, still through a (persistent)
         // stream so completion semantics are uniform.
         if (gc.comms.empty()) {
@@ -395,9 +412,20 @@ bool DeviceAdvanceRequest(CommRequest* req, DeviceReqState& st) {
             for (auto& ce : chunks) {
                 const uint8_t* sbase = req->SendBuf() + ce.elem_off * es;
                 uint8_t* rbase = req->RecvBuf() + ce.elem_off * es;
-                if (ce.sch.result.bytes && sbase != rbase)
+                if (compressed) {
+                    // quantize -> dequantize so n=1 keeps quantized-allreduce
+                    // semantics (and exercises error feedback)
+                    uint8_t* wire = static_cast<uint8_t*>(st.tmp_dev);
+                    uint8_t* err = wire + ce.sch.result.bytes + ce.sch.tmp_bytes;
+                    const size_t blk = req->QParams().block_elems;
+                    LaunchQuantize(sbase, err, wire, req->Spec().count, blk,
+                                   req->Dtype(), true, s0);
+                    LaunchDequantize(wire, rbase, req->Spec().count, blk,
+                                     req->Dtype(), s0);
+                } else if (ce.sch.result.bytes && sbase != rbase) {
                     HIP_CHECKD(hipMemcpyAsync(rbase + ce.sch.result.off, sbase,
                                               ce.sch.result.bytes, hipMemcpyDeviceToDevice, s0));
+                }
             }
             if (st.events.empty()) {
                 hipEvent_t e;
@@ -419,15 +447,34 @@ bool DeviceAdvanceRequest(CommRequest* req, DeviceReqState& st) {
         }
         // tmp partitioning per chunk
         size_t tmp_off = 0;
+        const size_t es = DtypeSize(req->Dtype());
         for (auto& ce : chunks) {
             const size_t ch = ce.chunk_idx % nch;
-            if (use_schedule) {
+            uint8_t* tbase = static_cast<uint8_t*>(st.tmp_dev) + tmp_off;
+            if (compressed) {
+                // quantize -> compressed-domain ring -> dequantize, all on
+                // one stream (driver config 5: int8 allreduce of bf16 grads)
+                const size_t wire_b = ce.sch.result.bytes;
+                uint8_t* wire = tbase;
+                uint8_t* scratch = tbase + wire_b;
+                uint8_t* err = scratch + ce.sch.tmp_bytes;
+                const size_t blk = req->QParams().block_elems;
+                LaunchQuantize(req->SendBuf(), err, wire, req->Spec().count, blk,
+                               req->Dtype(), true, gc.streams[ch]);
+                IssueSchedule(req, ce, gc.comms[ch], gc.streams[ch], wire, wire,
+                              scratch);
+                LaunchDequantize(wire, req->RecvBuf(), req->Spec().count, blk,
+                                 req->Dtype(), gc.streams[ch]);
+                tmp_off += wire_b + ce.sch.tmp_bytes + req->Spec().count * es;
+            } else if (use_schedule) {
                 IssueSchedule(req, ce, gc.comms[ch], gc.streams[ch],
-                              static_cast<uint8_t*>(st.tmp_dev) + tmp_off);
+                              req->SendBuf() + ce.elem_off * es,
+                              req->RecvBuf() + ce.elem_off * es, tbase);
+                tmp_off += ce.sch.tmp_bytes;
             } else {
                 IssueFused(req, ce, gc.comms[ch], gc.streams[ch], st, tmp_off);
+                tmp_off += ce.sch.tmp_bytes;
             }
-            tmp_off += ce.sch.tmp_bytes;
         }
         for (size_t ch = 0; ch < used; ++ch)
             HIP_CHECKD(hipEventRecord(st.events[ch], gc.streams[ch]));

@@ -11,6 +11,7 @@
 #include "engine.hpp"
 #include "group.hpp"
 #include "mesh.hpp"
+#include "quant.hpp"
 
 namespace mlsl {
 
@@ -35,6 +36,7 @@ const char* CollOpName(CollOp op) {
 void ChunkExec::Reset() {
     cur_phase = sch.steps.empty() ? sch.num_phases : 0;
     finished = sch.num_phases == 0;
+    prologue_done = false;
     state.assign(sch.steps.size(), StepState{});
 }
 
@@ -132,6 +134,26 @@ void CommRequest::AddSendRecvList(const std::vector<SRPair>& pairs) {
     spec_.pairs = pairs;
 }
 
+void CommRequest::SetCompression(Compression c, const QuantParams& qp) {
+    MLSL_CHECK(!setup_done_, "SetCompression after Setup");
+    comp_ = c;
+    qparams_ = qp;
+}
+
+bool CommRequest::Compressed() const {
+    return comp_ == Compression::QUANT_INT8 && spec_.op == CollOp::ALLREDUCE &&
+           (dtype_ == DataType::F32 || dtype_ == DataType::BF16) &&
+           spec_.rop == ReduceOp::SUM;
+}
+
+size_t CommRequest::WireBytesFor(const ChunkExec& ce) const {
+    const size_t cnt = ce.sch.result.bytes ? ce.sch.result.bytes /
+        (qparams_.block_elems + 8) * qparams_.block_elems : 0;
+    (void)cnt;
+    // result.bytes for compressed chunks is in wire bytes already
+    return ce.sch.result.bytes;
+}
+
 size_t CommRequest::MessageBytes() const {
     const size_t es = DtypeSize(dtype_);
     switch (spec_.op) {
@@ -170,7 +192,7 @@ void CommRequest::BuildChunks() {
     size_t n_chunks = 1;
     const bool splittable = spec_.op == CollOp::ALLREDUCE || spec_.op == CollOp::BCAST ||
                             spec_.op == CollOp::REDUCE;
-    if (splittable && gs > 1) {
+    if (splittable && gs > 1 && !Compressed()) {
         n_chunks = cfg.num_channels;
         if (MessageBytes() >= cfg.large_msg_mb * (1024 * 1024) && cfg.large_msg_chunks > 1)
             n_chunks *= cfg.large_msg_chunks;
@@ -201,9 +223,19 @@ void CommRequest::BuildChunks() {
         ce.elem_off = off;
         switch (spec_.op) {
             case CollOp::ALLREDUCE:
-                ce.sch = (algo == AllReduceAlgo::RHD && (gs & (gs - 1)) == 0)
-                             ? BuildAllReduceRHD(gr, gs, cnt, dtype_, spec_.rop)
-                             : BuildAllReduceRing(gr, gs, cnt, dtype_, spec_.rop);
+                if (Compressed()) {
+                    // Quantized allreduce: ring over int8 wire blocks with
+                    // compressed-domain accumulation (quant/quant.c path
+                    // fused into the collective — reference cqueue.c:1977,
+                    // 2283). Single chunk; the wire lives in TMP.
+                    const size_t blk = qparams_.block_elems;
+                    const size_t nblocks = (cnt + blk - 1) / blk;
+                    ce.sch = BuildAllReduceRingUnits(gr, gs, nblocks, blk + 8, blk);
+                } else {
+                    ce.sch = (algo == AllReduceAlgo::RHD && (gs & (gs - 1)) == 0)
+                                 ? BuildAllReduceRHD(gr, gs, cnt, dtype_, spec_.rop)
+                                 : BuildAllReduceRing(gr, gs, cnt, dtype_, spec_.rop);
+                }
                 break;
             case CollOp::REDUCE:
                 ce.sch = BuildReduce(gr, gs, cnt, dtype_, spec_.rop, spec_.root);
@@ -252,7 +284,16 @@ void CommRequest::Setup() {
         dev_ = std::make_unique<DeviceState>();
         DeviceSetupRequest(this, *dev_);
     } else {
-        for (auto& ce : chunks_) ce.tmp.resize(ce.sch.tmp_bytes);
+        for (auto& ce : chunks_) {
+            size_t t = ce.sch.tmp_bytes;
+            if (Compressed()) {
+                // [wire][schedule scratch][error-feedback residual]
+                const size_t wire = ce.sch.result.bytes;
+                const size_t err = spec_.count * DtypeSize(dtype_);
+                t = wire + ce.sch.tmp_bytes + err;
+            }
+            ce.tmp.assign(t, 0);
+        }
     }
     setup_done_ = true;
 }
@@ -295,6 +336,7 @@ void* CommRequest::Wait() {
         MLSL_THROW("request failed: " + error_);
     state_.store(ReqState::IDLE, std::memory_order_release);
     if (chunks_.empty()) return rbuf_;
+    if (Compressed()) return rbuf_;  // dequantized into the user recv buffer
     const BufRef& res = chunks_[0].sch.result;
     switch (res.space) {
         case Space::RECV: return rbuf_ + res.off;
@@ -328,11 +370,21 @@ bool CommRequest::AdvanceHost(Mesh* mesh) {
     const size_t msg_es = es;  // chunk offsets are in elements of dtype_
     bool all_done = true;
 
+    const bool compressed = Compressed();
     for (auto& ce : chunks_) {
         if (ce.finished) continue;
         const uint8_t* sbase = sbuf_ + ce.elem_off * msg_es;
         uint8_t* rbase = rbuf_ + ce.elem_off * msg_es;
+        uint8_t* wire = ce.tmp.data();                         // compressed only
+        const size_t wire_bytes = compressed ? ce.sch.result.bytes : 0;
         auto ptr = [&](const BufRef& b) -> uint8_t* {
+            if (compressed) {
+                switch (b.space) {
+                    case Space::SEND: return wire + b.off;
+                    case Space::RECV: return wire + b.off;
+                    case Space::TMP: return ce.tmp.data() + wire_bytes + b.off;
+                }
+            }
             switch (b.space) {
                 case Space::SEND: return const_cast<uint8_t*>(sbase) + b.off;
                 case Space::RECV: return rbase + b.off;
@@ -340,6 +392,13 @@ bool CommRequest::AdvanceHost(Mesh* mesh) {
             }
             return nullptr;
         };
+        if (compressed && !ce.prologue_done) {
+            // quantize user send (+ residual) into the wire
+            uint8_t* err = ce.tmp.data() + wire_bytes + ce.sch.tmp_bytes;
+            HostQuantize(sbase, err, wire, spec_.count, qparams_.block_elems,
+                         dtype_, true);
+            ce.prologue_done = true;
+        }
 
         while (!ce.finished) {
             bool phase_done = true;
@@ -369,6 +428,10 @@ bool CommRequest::AdvanceHost(Mesh* mesh) {
                         uint8_t* s = ptr(st.local_src);
                         if (st.local == Step::LocalOp::COPY) {
                             if (d != s) std::memmove(d, s, st.local_src.bytes);
+                        } else if (ce.sch.quant_block > 0) {
+                            const size_t blk = ce.sch.quant_block;
+                            const size_t units = st.local_dst.bytes / (blk + 8);
+                            HostQuantAccum(d, s, units * blk, blk);
                         } else {
                             HostReduce(d, s, st.local_dst.bytes / es, dtype_, ce.sch.rop);
                         }
@@ -380,7 +443,13 @@ bool CommRequest::AdvanceHost(Mesh* mesh) {
             }
             if (!phase_done) break;
             ce.cur_phase++;
-            if (ce.cur_phase >= ce.sch.num_phases) ce.finished = true;
+            if (ce.cur_phase >= ce.sch.num_phases) {
+                ce.finished = true;
+                if (compressed) {
+                    HostDequantize(wire, rbase, spec_.count, qparams_.block_elems,
+                                   dtype_);
+                }
+            }
         }
         if (!ce.finished) all_done = false;
     }

@@ -67,6 +67,7 @@ struct ChunkExec {
     Schedule sch;
     size_t elem_off = 0;      // element offset of this chunk in the message
     size_t chunk_idx = 0;
+    bool prologue_done = false;   // compressed path: quantize ran
     int cur_phase = -1;
     struct StepState {
         bool send_started = false, send_done = false;
@@ -101,6 +102,16 @@ class CommRequest {
                       const std::vector<size_t>& rcnt, const std::vector<size_t>& roff);
     void AddBarrier();
     void AddSendRecvList(const std::vector<SRPair>& pairs);
+    // Gradient compression (reference quant/quant.c contract): int8 block
+    // quantization with error feedback, fused into the allreduce. Call
+    // before Setup(); honored for ALLREDUCE on f32/bf16.
+    void SetCompression(Compression c, const QuantParams& qp);
+    Compression GetCompression() const { return comp_; }
+    const QuantParams& QParams() const { return qparams_; }
+    bool Compressed() const;
+    // Compressed-path layout inside each chunk's tmp:
+    // [wire W][schedule scratch][error-feedback residual].
+    size_t WireBytesFor(const ChunkExec& ce) const;
 
     // ---- lifecycle ----
     void Setup();                         // compile schedules, size scratch
@@ -142,6 +153,8 @@ class CommRequest {
     OpSpec spec_;
     bool has_op_ = false;
     bool setup_done_ = false;
+    Compression comp_ = Compression::NONE;
+    QuantParams qparams_;
 
     std::vector<ChunkExec> chunks_;
     size_t total_tmp_bytes_ = 0;

@@ -48,8 +48,8 @@ int CeilLog2(int n) {
 // group ring; each phase moves one segment to the next neighbor. This is the
 // per-xGMI-link-bound algorithm the chunk-over-channels layer parallelizes
 // (reference analog: GET_EP_PAYLOAD endpoint fan-out, src/comm_ep.cpp:99-115).
-Schedule BuildAllReduceRing(int rank, int size, size_t count, DataType dt, ReduceOp op) {
-    const size_t es = DtypeSize(dt);
+static Schedule RingAllReduceImpl(int rank, int size, size_t count, size_t es,
+                                  DataType dt, ReduceOp op) {
     if (size == 1) return SelfOnly(count * es, dt, op);
 
     Schedule sch;
@@ -94,6 +94,18 @@ Schedule BuildAllReduceRing(int rank, int size, size_t count, DataType dt, Reduc
     }
     sch.tmp_bytes = max_seg;
     sch.result = Ref(Space::RECV, 0, count * es);
+    return sch;
+}
+
+Schedule BuildAllReduceRing(int rank, int size, size_t count, DataType dt, ReduceOp op) {
+    return RingAllReduceImpl(rank, size, count, DtypeSize(dt), dt, op);
+}
+
+Schedule BuildAllReduceRingUnits(int rank, int size, size_t units, size_t unit_bytes,
+                                 size_t quant_block) {
+    Schedule sch = RingAllReduceImpl(rank, size, units, unit_bytes, DataType::U8,
+                                     ReduceOp::SUM);
+    sch.quant_block = quant_block;
     return sch;
 }
 

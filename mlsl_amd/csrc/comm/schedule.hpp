@@ -55,6 +55,10 @@ struct Schedule {
     size_t tmp_bytes = 0;
     DataType dtype = DataType::F32;
     ReduceOp rop = ReduceOp::SUM;
+    // >0: REDUCE local ops are compressed-domain int8 block accumulations
+    // (QuantAccum) over wire blocks of this many elements; buffer refs are
+    // then in wire bytes (unit = quant_block + 8).
+    size_t quant_block = 0;
     // Where the result lives after the final phase (returned by Wait()).
     BufRef result;
 
@@ -73,6 +77,10 @@ size_t SegCount(size_t count, size_t parts, size_t i);
 // SEND and RECV at execution time; builders emit SEND-space reads only in
 // phase 0 positions that are safe, or route through TMP).
 Schedule BuildAllReduceRing(int rank, int size, size_t count, DataType dt, ReduceOp op);
+// Ring allreduce over opaque fixed-size units (used by the quantized path:
+// unit = one int8 wire block of quant_block elements + 8-byte header).
+Schedule BuildAllReduceRingUnits(int rank, int size, size_t units, size_t unit_bytes,
+                                 size_t quant_block);
 Schedule BuildAllReduceRHD(int rank, int size, size_t count, DataType dt, ReduceOp op);
 Schedule BuildReduceScatter(int rank, int size, size_t recv_count, DataType dt, ReduceOp op);
 Schedule BuildAllGather(int rank, int size, size_t send_count, DataType dt);

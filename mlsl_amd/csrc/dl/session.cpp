@@ -333,9 +333,11 @@ ParameterSet::ParameterSet(Operation* op, const RegEntry& re, size_t idx)
 
     if (need_comm_) {
         grad_req_ = std::make_unique<CommRequest>(dg, dt_, CompType::PARAM_GRAD);
-        if (comp_ == Compression::QUANT_INT8) {
-            // Quantized path wires in at the request layer (comm/quant.cpp);
-            // described here so Setup sizes the wire buffers.
+        if (comp_ == Compression::QUANT_INT8 && !distributed_update_) {
+            // int8 gradient compression with error feedback, fused into the
+            // allreduce (reference quant path, cqueue.c:1977-1994).
+            grad_req_->SetCompression(comp_,
+                                      Environment::GetEnv().GetQuantizationParams());
         }
         if (distributed_update_) {
             grad_req_->AddReduceScatter(owned_kernel_count_ * kernel_size_, ReduceOp::SUM);

@@ -43,3 +43,8 @@ def test_rhd_algo():
 
 def test_ring_algo():
     run_ranks("collectives_sweep", 3, extra_env={"MLSL_ALLREDUCE_ALGO": "ring"})
+
+
+@pytest.mark.parametrize("world", [1, 2, 4])
+def test_quantized_allreduce(world):
+    run_ranks("quantized_allreduce", world)

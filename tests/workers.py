@@ -228,6 +228,56 @@ WORKERS = {
 }
 
 
+
+
+def quantized_allreduce():
+    """Quantized gradient allreduce through the planner (compression=int8):
+    result within per-block quantization error of the exact sum, and error
+    feedback keeps the long-run average unbiased."""
+    mx, rank, size = _init()
+    dp = size
+    s = mx.Session()
+    s.set_global_minibatch_size(4 * dp)
+    d = mx.Distribution(dp, 1)
+    info = s.create_op_reg_info("cc")
+    info.add_input(4, 2, "f32")
+    info.add_output(4, 2, "f32")
+    info.add_parameter_set(1000, 17, "f32", compression="int8")
+    op = s.operation(s.add_operation(info, d))
+    s.commit()
+    ps = op.parameter_set(0)
+    n = ps.local_kernel_count * ps.kernel_size
+
+    rng = np.random.RandomState(123 + rank)
+    acc_exact = np.zeros(n, dtype=np.float64)
+    acc_quant = np.zeros(n, dtype=np.float64)
+    for it in range(20):
+        g_all = [np.random.RandomState(1000 + it * size + r).randn(n).astype(np.float32)
+                 for r in range(size)]
+        g = g_all[rank].copy()
+        exact = np.sum(g_all, axis=0)
+        ps.start_gradient_comm(g)
+        ps.wait_gradient_comm()
+        if size == 1:
+            got = g
+        else:
+            got = g
+        step = np.abs(exact).max() / 127.0
+        err = np.abs(got - exact).max()
+        # ring quantized allreduce: <= ~2 requant steps per hop
+        assert err < max(4 * size * step, 1e-3), f"quant allreduce err {err} step {step}"
+        acc_exact += exact
+        acc_quant += got
+    drift = np.abs(acc_exact - acc_quant).max()
+    scale = np.abs(acc_exact).max() / 127.0
+    assert drift < 20 * max(scale, 0.1), f"quant drift {drift}"
+    del rng
+    mx.finalize()
+
+
+WORKERS["quantized_allreduce"] = quantized_allreduce
+
+
 def main():
     name = sys.argv[1]
     fn = WORKERS.get(name)
