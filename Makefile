@@ -15,6 +15,8 @@ LDFLAGS    := -shared -fPIC -L/opt/rocm/lib -lrccl -lamdhip64 -pthread
 CSRC := \
     mlsl_amd/csrc/core/log.cpp \
     mlsl_amd/csrc/core/config.cpp \
+    mlsl_amd/csrc/core/sysinfo.cpp \
+    mlsl_amd/csrc/core/signals.cpp \
     mlsl_amd/csrc/comm/schedule.cpp \
     mlsl_amd/csrc/comm/quant.cpp \
     mlsl_amd/csrc/comm/bootstrap.cpp \
@@ -56,7 +58,21 @@ $(SELFTEST): $(BUILD)/mlsl_amd/csrc/tests/schedule_selftest.o $(BUILD)/mlsl_amd/
 	@mkdir -p $(dir $@)
 	$(HIPCC) $^ -L/opt/rocm/lib -lamdhip64 -pthread -o $@
 
-test: all
+SAMPLES := $(BUILD)/mlsl_sample $(BUILD)/cmlsl_sample
+
+samples: $(SAMPLES)
+
+$(BUILD)/mlsl_sample: samples/mlsl_sample.cpp $(LIB)
+	@mkdir -p $(dir $@)
+	$(HIPCC) -O2 -std=c++17 -Imlsl_amd/csrc/include samples/mlsl_sample.cpp \
+	    -Lmlsl_amd -lmlsl_amd -Wl,-rpath,'$$ORIGIN/../mlsl_amd' -o $@
+
+$(BUILD)/cmlsl_sample: samples/cmlsl_sample.c $(LIB)
+	@mkdir -p $(dir $@)
+	gcc -O2 -Imlsl_amd/csrc/include samples/cmlsl_sample.c \
+	    -Lmlsl_amd -lmlsl_amd -Wl,-rpath,'$$ORIGIN/../mlsl_amd' -o $@
+
+test: all samples
 	$(SELFTEST)
 	python -m pytest tests/ -x -q -m "not gpu"
 

@@ -4,6 +4,8 @@
 #include <cstring>
 
 #include "../core/log.hpp"
+#include "../core/signals.hpp"
+#include "../core/sysinfo.hpp"
 #include "../core/types.hpp"
 #include "device_comm.hpp"
 #include "request.hpp"
@@ -24,6 +26,9 @@ void Context::Init(int rank, int size) {
     Config& cfg = GlobalConfig();
     cfg = Config::FromEnv();  // re-read: tests mutate env between inits
     SetLogLevel(static_cast<LogLevel>(cfg.log_level));
+
+    InstallSignalHandlers();
+    AutoConfig();
 
     boot_ = std::make_unique<Bootstrap>(rank, size);
     rank_ = boot_->Rank();

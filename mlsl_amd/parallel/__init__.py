@@ -1,0 +1,9 @@
+"""Parallelism helpers on top of the comm core.
+
+- GradBucketer: bucketed non-blocking gradient allreduce overlapped with
+  backward compute (the reference's Backward1/Backward2/Update split,
+  tests/examples/mlsl_test/mlsl_test.cpp:464-528, generalized to arbitrary
+  bucket lists — the driver's ResNet-50 overlap config).
+- shard math helpers for distributed update (ZeRO-1 ancestor).
+"""
+from .bucketer import GradBucketer  # noqa: F401

@@ -1,0 +1,39 @@
+"""Run the migration samples (C++ / C / Python) at world 2 over TCP —
+driver config 1, through every language binding."""
+import os
+import subprocess
+
+import pytest
+
+from tests.mp import REPO, free_port, run_ranks
+
+
+def _run_binary(path, world=2):
+    if not os.path.exists(path):
+        subprocess.run(["make", "samples"], cwd=REPO, check=True,
+                       capture_output=True, timeout=600)
+    port = free_port()
+    procs = []
+    for r in range(world):
+        env = dict(os.environ, RANK=str(r), WORLD_SIZE=str(world),
+                   MASTER_ADDR="127.0.0.1", MLSL_PORT=str(port),
+                   MLSL_TRANSPORT="tcp")
+        procs.append(subprocess.Popen([path], env=env, cwd=REPO,
+                                      stdout=subprocess.PIPE,
+                                      stderr=subprocess.STDOUT, text=True))
+    for r, p in enumerate(procs):
+        out, _ = p.communicate(timeout=120)
+        assert p.returncode == 0, f"rank {r}: {out}"
+        assert "PASSED" in out
+
+
+def test_cpp_sample():
+    _run_binary(os.path.join(REPO, "build", "mlsl_sample"))
+
+
+def test_c_sample():
+    _run_binary(os.path.join(REPO, "build", "cmlsl_sample"))
+
+
+def test_python_sample():
+    run_ranks("py_sample", 2)

@@ -278,6 +278,16 @@ def quantized_allreduce():
 WORKERS["quantized_allreduce"] = quantized_allreduce
 
 
+def py_sample():
+    import runpy
+    import sys as _sys
+    _sys.argv = ["mlsl_sample.py"]
+    runpy.run_path("samples/mlsl_sample.py", run_name="__main__")
+
+
+WORKERS["py_sample"] = py_sample
+
+
 def main():
     name = sys.argv[1]
     fn = WORKERS.get(name)
