@@ -10,6 +10,8 @@
 #include <hip/hip_fp16.h>
 #include <hip/hip_runtime.h>
 
+#include <algorithm>
+
 #include "../core/log.hpp"
 #include "kernels.hpp"
 
@@ -43,7 +45,7 @@ __device__ __forceinline__ T Apply(T a, T b) {
     return a > b ? a : b;
 }
 
-// ---- f32: float4-vectorized in-place reduce ----
+// ---- f32: float4-vectorized in-place reduce, 2-deep unroll for MLP ----
 template <ReduceOp OP>
 __global__ void ReduceF32Kernel(float* __restrict__ dst,
                                 const float* __restrict__ src, size_t n) {
@@ -52,7 +54,22 @@ __global__ void ReduceF32Kernel(float* __restrict__ dst,
     const size_t n4 = n / 4;
     const float4* s4 = reinterpret_cast<const float4*>(src);
     float4* d4 = reinterpret_cast<float4*>(dst);
-    for (size_t i = tid; i < n4; i += stride) {
+    size_t i = tid;
+    for (; i + stride < n4; i += 2 * stride) {
+        float4 a0 = d4[i], b0 = s4[i];
+        float4 a1 = d4[i + stride], b1 = s4[i + stride];
+        a0.x = Apply<float, OP>(a0.x, b0.x);
+        a0.y = Apply<float, OP>(a0.y, b0.y);
+        a0.z = Apply<float, OP>(a0.z, b0.z);
+        a0.w = Apply<float, OP>(a0.w, b0.w);
+        a1.x = Apply<float, OP>(a1.x, b1.x);
+        a1.y = Apply<float, OP>(a1.y, b1.y);
+        a1.z = Apply<float, OP>(a1.z, b1.z);
+        a1.w = Apply<float, OP>(a1.w, b1.w);
+        d4[i] = a0;
+        d4[i + stride] = a1;
+    }
+    for (; i < n4; i += stride) {
         float4 a = d4[i], b = s4[i];
         a.x = Apply<float, OP>(a.x, b.x);
         a.y = Apply<float, OP>(a.y, b.y);
@@ -60,8 +77,8 @@ __global__ void ReduceF32Kernel(float* __restrict__ dst,
         a.w = Apply<float, OP>(a.w, b.w);
         d4[i] = a;
     }
-    for (size_t i = n4 * 4 + tid; i < n; i += stride)
-        dst[i] = Apply<float, OP>(dst[i], src[i]);
+    for (size_t j = n4 * 4 + tid; j < n; j += stride)
+        dst[j] = Apply<float, OP>(dst[j], src[j]);
 }
 
 // ---- generic scalar fallback (f64/i32/i64/u8) ----
@@ -210,104 +227,107 @@ __device__ __forceinline__ void StoreFromF32(unsigned short* p, size_t i, float 
     p[i] = reinterpret_cast<unsigned short&>(h);
 }
 
-// One workgroup per quant block: LDS max-reduction for the scale, then
-// quantize + update the error-feedback residual.
+// Wave-per-block quantization: each 64-lane wavefront owns one wire block
+// (4 waves per 256-thread workgroup, grid-strided over blocks). The scale
+// reduction is a pure cross-lane shuffle — no LDS, no __syncthreads — so
+// small blocks stay launch- and HBM-bound, not barrier-bound.
+__device__ __forceinline__ float WaveMax(float m) {
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1)
+        m = fmaxf(m, __shfl_down(m, off, 64));
+    return __shfl(m, 0, 64);
+}
+
 template <typename T, bool USE_ERR>
 __global__ void QuantizeKernel(const T* __restrict__ in, T* __restrict__ err,
                                uint8_t* __restrict__ wire, size_t count,
                                size_t block_elems) {
-    const size_t blk = blockIdx.x;
-    const size_t base = blk * block_elems;
-    const size_t n = min(block_elems, count - base);
-    uint8_t* wblock = wire + blk * (block_elems + 8);
-    float* hdr = reinterpret_cast<float*>(wblock);
-    int8_t* payload = reinterpret_cast<int8_t*>(wblock + 8);
+    const size_t nblocks = (count + block_elems - 1) / block_elems;
+    const int lane = threadIdx.x & 63;
+    const size_t wave = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+    const size_t wstride = (gridDim.x * blockDim.x) >> 6;
+    for (size_t blk = wave; blk < nblocks; blk += wstride) {
+        const size_t base = blk * block_elems;
+        const size_t n = min(block_elems, count - base);
+        uint8_t* wblock = wire + blk * (block_elems + 8);
+        float* hdr = reinterpret_cast<float*>(wblock);
+        int8_t* payload = reinterpret_cast<int8_t*>(wblock + 8);
 
-    __shared__ float smax[kBlock / 64];
-    float m = 0.f;
-    for (size_t i = threadIdx.x; i < n; i += blockDim.x) {
-        float v = LoadAsF32(in, base + i);
-        if (USE_ERR) v += LoadAsF32(err, base + i);
-        m = fmaxf(m, fabsf(v));
+        float m = 0.f;
+        for (size_t i = lane; i < n; i += 64) {
+            float v = LoadAsF32(in, base + i);
+            if (USE_ERR) v += LoadAsF32(err, base + i);
+            m = fmaxf(m, fabsf(v));
+        }
+        m = WaveMax(m);
+        const float scale = m > 0.f ? m / 127.f : 1.f;
+        if (lane == 0) {
+            hdr[0] = scale;
+            hdr[1] = 0.f;
+        }
+        const float inv = 1.f / scale;
+        for (size_t i = lane; i < n; i += 64) {
+            float v = LoadAsF32(in, base + i);
+            if (USE_ERR) v += LoadAsF32(err, base + i);
+            float q = nearbyintf(v * inv);
+            q = fminf(127.f, fmaxf(-127.f, q));
+            payload[i] = static_cast<int8_t>(q);
+            if (USE_ERR) StoreFromF32(err, base + i, v - q * scale);
+        }
+        for (size_t i = n + lane; i < block_elems; i += 64) payload[i] = 0;
     }
-    // wave reduce
-    for (int off = 32; off > 0; off >>= 1)
-        m = fmaxf(m, __shfl_down(m, off, 64));
-    if ((threadIdx.x & 63) == 0) smax[threadIdx.x >> 6] = m;
-    __syncthreads();
-    if (threadIdx.x == 0) {
-        float mm = 0.f;
-        for (int w = 0; w < static_cast<int>(blockDim.x) / 64; ++w) mm = fmaxf(mm, smax[w]);
-        hdr[0] = mm > 0.f ? mm / 127.f : 1.f;
-        hdr[1] = 0.f;
-    }
-    __syncthreads();
-    const float scale = hdr[0];
-    const float inv = 1.f / scale;
-    for (size_t i = threadIdx.x; i < n; i += blockDim.x) {
-        float v = LoadAsF32(in, base + i);
-        if (USE_ERR) v += LoadAsF32(err, base + i);
-        float q = nearbyintf(v * inv);
-        q = fminf(127.f, fmaxf(-127.f, q));
-        payload[i] = static_cast<int8_t>(q);
-        if (USE_ERR) StoreFromF32(err, base + i, v - q * scale);
-    }
-    // Zero the tail so compressed-domain accumulation stays clean.
-    for (size_t i = n + threadIdx.x; i < block_elems; i += blockDim.x) payload[i] = 0;
 }
 
 template <typename T>
 __global__ void DequantizeKernel(const uint8_t* __restrict__ wire, T* __restrict__ out,
                                  size_t count, size_t block_elems) {
-    const size_t blk = blockIdx.x;
-    const size_t base = blk * block_elems;
-    const size_t n = min(block_elems, count - base);
-    const uint8_t* wblock = wire + blk * (block_elems + 8);
-    const float scale = reinterpret_cast<const float*>(wblock)[0];
-    const int8_t* payload = reinterpret_cast<const int8_t*>(wblock + 8);
-    for (size_t i = threadIdx.x; i < n; i += blockDim.x)
-        StoreFromF32(out, base + i, static_cast<float>(payload[i]) * scale);
+    const size_t nblocks = (count + block_elems - 1) / block_elems;
+    const int lane = threadIdx.x & 63;
+    const size_t wave = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+    const size_t wstride = (gridDim.x * blockDim.x) >> 6;
+    for (size_t blk = wave; blk < nblocks; blk += wstride) {
+        const size_t base = blk * block_elems;
+        const size_t n = min(block_elems, count - base);
+        const uint8_t* wblock = wire + blk * (block_elems + 8);
+        const float scale = reinterpret_cast<const float*>(wblock)[0];
+        const int8_t* payload = reinterpret_cast<const int8_t*>(wblock + 8);
+        for (size_t i = lane; i < n; i += 64)
+            StoreFromF32(out, base + i, static_cast<float>(payload[i]) * scale);
+    }
 }
 
 // acc_wire += wire in the compressed domain: dequant both, sum, requant with
 // a fresh scale (the reference's external reduce_sum hook, quant/quant.c:89).
 __global__ void QuantAccumKernel(uint8_t* __restrict__ acc, const uint8_t* __restrict__ in,
                                  size_t count, size_t block_elems) {
-    const size_t blk = blockIdx.x;
-    const size_t base = blk * block_elems;
-    const size_t n = min(block_elems, count - base);
-    uint8_t* ablock = acc + blk * (block_elems + 8);
-    const uint8_t* iblock = in + blk * (block_elems + 8);
-    float* ahdr = reinterpret_cast<float*>(ablock);
-    const float as = ahdr[0];
-    const float is = reinterpret_cast<const float*>(iblock)[0];
-    int8_t* ap = reinterpret_cast<int8_t*>(ablock + 8);
-    const int8_t* ip = reinterpret_cast<const int8_t*>(iblock + 8);
+    const size_t nblocks = (count + block_elems - 1) / block_elems;
+    const int lane = threadIdx.x & 63;
+    const size_t wave = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+    const size_t wstride = (gridDim.x * blockDim.x) >> 6;
+    for (size_t blk = wave; blk < nblocks; blk += wstride) {
+        const size_t base = blk * block_elems;
+        const size_t n = min(block_elems, count - base);
+        uint8_t* ablock = acc + blk * (block_elems + 8);
+        const uint8_t* iblock = in + blk * (block_elems + 8);
+        float* ahdr = reinterpret_cast<float*>(ablock);
+        const float as = ahdr[0];
+        const float is = reinterpret_cast<const float*>(iblock)[0];
+        int8_t* ap = reinterpret_cast<int8_t*>(ablock + 8);
+        const int8_t* ip = reinterpret_cast<const int8_t*>(iblock + 8);
 
-    __shared__ float smax[kBlock / 64];
-    float m = 0.f;
-    for (size_t i = threadIdx.x; i < n; i += blockDim.x) {
-        float v = static_cast<float>(ap[i]) * as + static_cast<float>(ip[i]) * is;
-        m = fmaxf(m, fabsf(v));
+        float m = 0.f;
+        for (size_t i = lane; i < n; i += 64)
+            m = fmaxf(m, fabsf(ap[i] * as + ip[i] * is));
+        m = WaveMax(m);
+        const float ns = m > 0.f ? m / 127.f : 1.f;
+        const float inv = 1.f / ns;
+        for (size_t i = lane; i < n; i += 64) {
+            float v = ap[i] * as + ip[i] * is;
+            float q = nearbyintf(v * inv);
+            ap[i] = static_cast<int8_t>(fminf(127.f, fmaxf(-127.f, q)));
+        }
+        if (lane == 0) ahdr[0] = ns;
     }
-    for (int off = 32; off > 0; off >>= 1)
-        m = fmaxf(m, __shfl_down(m, off, 64));
-    if ((threadIdx.x & 63) == 0) smax[threadIdx.x >> 6] = m;
-    __syncthreads();
-    __shared__ float new_scale;
-    if (threadIdx.x == 0) {
-        float mm = 0.f;
-        for (int w = 0; w < static_cast<int>(blockDim.x) / 64; ++w) mm = fmaxf(mm, smax[w]);
-        new_scale = mm > 0.f ? mm / 127.f : 1.f;
-    }
-    __syncthreads();
-    const float inv = 1.f / new_scale;
-    for (size_t i = threadIdx.x; i < n; i += blockDim.x) {
-        float v = static_cast<float>(ap[i]) * as + static_cast<float>(ip[i]) * is;
-        float q = nearbyintf(v * inv);
-        ap[i] = static_cast<int8_t>(fminf(127.f, fmaxf(-127.f, q)));
-    }
-    if (threadIdx.x == 0) ahdr[0] = new_scale;
 }
 
 }  // namespace
@@ -316,7 +336,8 @@ void LaunchQuantize(const void* in, void* err, void* wire, size_t count,
                     size_t block_elems, DataType dt, bool use_err,
                     hipStream_t stream) {
     const size_t nblocks = (count + block_elems - 1) / block_elems;
-    dim3 grid(static_cast<uint32_t>(nblocks));
+    // 4 waves per 256-thread workgroup, grid-strided over wire blocks.
+    dim3 grid(static_cast<uint32_t>(std::min<size_t>((nblocks + 3) / 4, kMaxGrid)));
     if (dt == DataType::F32) {
         if (use_err)
             hipLaunchKernelGGL((QuantizeKernel<float, true>), grid, dim3(kBlock), 0, stream,
@@ -346,7 +367,7 @@ void LaunchQuantize(const void* in, void* err, void* wire, size_t count,
 void LaunchDequantize(const void* wire, void* out, size_t count,
                       size_t block_elems, DataType dt, hipStream_t stream) {
     const size_t nblocks = (count + block_elems - 1) / block_elems;
-    dim3 grid(static_cast<uint32_t>(nblocks));
+    dim3 grid(static_cast<uint32_t>(std::min<size_t>((nblocks + 3) / 4, kMaxGrid)));
     if (dt == DataType::F32) {
         hipLaunchKernelGGL((DequantizeKernel<float>), grid, dim3(kBlock), 0, stream,
                            static_cast<const uint8_t*>(wire), static_cast<float*>(out),
@@ -364,7 +385,8 @@ void LaunchDequantize(const void* wire, void* out, size_t count,
 void LaunchQuantAccum(void* acc_wire, const void* wire, size_t count,
                       size_t block_elems, hipStream_t stream) {
     const size_t nblocks = (count + block_elems - 1) / block_elems;
-    hipLaunchKernelGGL(QuantAccumKernel, dim3(static_cast<uint32_t>(nblocks)),
+    dim3 grid(static_cast<uint32_t>(std::min<size_t>((nblocks + 3) / 4, kMaxGrid)));
+    hipLaunchKernelGGL(QuantAccumKernel, grid,
                        dim3(kBlock), 0, stream, static_cast<uint8_t*>(acc_wire),
                        static_cast<const uint8_t*>(wire), count, block_elems);
     HIP_CHECK(hipGetLastError());
