@@ -92,6 +92,19 @@ def set_quant_params(block_elems=256):
     check(lib().mlsl_set_quant_params(block_elems))
 
 
+def set_compute_stream(stream_ptr):
+    """Order device collectives after work on this HIP stream (pass
+    torch.cuda.current_stream().cuda_stream). Default: the legacy default
+    stream, which covers plain torch usage."""
+    import ctypes as _c
+    L = lib()
+    if not hasattr(L, "_scs_declared"):
+        L.mlsl_set_compute_stream.argtypes = [_c.c_void_p]
+        L.mlsl_set_compute_stream.restype = _c.c_int
+        L._scs_declared = True
+    check(L.mlsl_set_compute_stream(_c.c_void_p(stream_ptr)))
+
+
 def wait(req):
     """Complete a generic-collective request; returns the result pointer."""
     out = c_void_p()

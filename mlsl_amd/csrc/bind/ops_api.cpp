@@ -4,6 +4,8 @@
 // internally.
 #include <hip/hip_runtime.h>
 
+#include "../comm/context.hpp"
+#include "../comm/device_comm.hpp"
 #include "../core/types.hpp"
 #include "../hip/kernels.hpp"
 #include "../include/mlsl/c_api.h"
@@ -27,6 +29,13 @@ int mlsl_hip_device_count(int* out) {
     if (hipGetDeviceCount(&n) != hipSuccess) n = 0;
     *out = n;
     return MLSL_SUCCESS;
+}
+
+int mlsl_set_compute_stream(void* stream) {
+    OPS_TRY
+    mlsl::DeviceRuntime* rt = mlsl::Context::Get().Device();
+    if (rt) rt->SetComputeStream(stream);
+    OPS_CATCH
 }
 
 int mlsl_hip_synchronize(void) {

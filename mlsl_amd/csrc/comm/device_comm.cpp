@@ -44,6 +44,7 @@ namespace mlsl {
 DeviceReqState::~DeviceReqState() {
     for (hipEvent_t e : events)
         if (e) (void)hipEventDestroy(e);
+    if (dep_event) (void)hipEventDestroy(dep_event);
     if (tmp_dev) (void)hipFree(tmp_dev);
 }
 
@@ -98,6 +99,11 @@ class HipRuntime : public DeviceRuntime {
     }
 
     int DeviceId() const override { return device_id_; }
+
+    void SetComputeStream(void* stream) override {
+        compute_stream_ = static_cast<hipStream_t>(stream);
+    }
+    void* ComputeStream() const override { return compute_stream_; }
 
     void* AllocDevice(size_t bytes) override {
         void* p = nullptr;
@@ -156,6 +162,7 @@ class HipRuntime : public DeviceRuntime {
 
   private:
     int device_id_;
+    hipStream_t compute_stream_ = nullptr;  // null = legacy default stream
     std::unordered_map<int, GroupComms> group_comms_;
 };
 
@@ -392,6 +399,15 @@ bool DeviceAdvanceRequest(CommRequest* req, DeviceReqState& st) {
     auto* rt = static_cast<HipRuntime*>(Context::Get().Device());
     if (!st.issued) {
         GroupComms& gc = rt->For(req->Group());
+        // Order after the caller's compute stream: the producer kernels
+        // (e.g. torch backward on the default stream) must land before the
+        // collective reads the buffers.
+        if (!st.dep_event)
+            HIP_CHECKD(hipEventCreateWithFlags(&st.dep_event, hipEventDisableTiming));
+        HIP_CHECKD(hipEventRecord(st.dep_event,
+                                  static_cast<hipStream_t>(rt->ComputeStream())));
+        for (hipStream_t cs : gc.streams)
+            HIP_CHECKD(hipStreamWaitEvent(cs, st.dep_event, 0));
         auto& chunks = req->Chunks();
         const Config& cfg = GlobalConfig();
         const bool use_schedule =
@@ -408,6 +424,7 @@ bool DeviceAdvanceRequest(CommRequest* req, DeviceReqState& st) {
                 gc.streams.push_back(s0);
             }
             hipStream_t s0 = gc.streams[0];
+            HIP_CHECKD(hipStreamWaitEvent(s0, st.dep_event, 0));
             const size_t es = DtypeSize(req->Dtype());
             for (auto& ce : chunks) {
                 const uint8_t* sbase = req->SendBuf() + ce.elem_off * es;

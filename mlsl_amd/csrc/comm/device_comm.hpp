@@ -25,6 +25,10 @@ class DeviceRuntime {
     // Create the per-group device communicators (collective over group).
     virtual void EnsureGroupComms(ProcessGroup* g) = 0;
     virtual std::string Name() const = 0;
+    // Producer-ordering stream (0 = HIP legacy default stream). Collectives
+    // issued after SetComputeStream order behind work on that stream.
+    virtual void SetComputeStream(void* stream) = 0;
+    virtual void* ComputeStream() const = 0;
 };
 
 // Factory: returns nullptr when no HIP device is visible.

@@ -10,6 +10,10 @@ namespace mlsl {
 
 struct DeviceReqState {
     bool issued = false;
+    // Records the caller's compute stream at Start; channel streams wait on
+    // it so collectives order after the producer kernels (torch default
+    // stream unless mlsl_set_compute_stream was called).
+    hipEvent_t dep_event = nullptr;
     // One completion event per channel used by this request.
     std::vector<hipEvent_t> events;
     // Persistent device scratch (allocated at Setup in device mode).
