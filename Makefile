@@ -9,7 +9,7 @@ LIB        := mlsl_amd/libmlsl_amd.so
 SELFTEST   := $(BUILD)/schedule_selftest
 
 CXXFLAGS   := -O3 -std=c++17 -fPIC -Wall -Wextra -Wno-unused-parameter \
-              --offload-arch=$(GPU_ARCH) -I/opt/rocm/include
+              --offload-arch=$(GPU_ARCH) -I/opt/rocm/include -MMD -MP
 LDFLAGS    := -shared -fPIC -L/opt/rocm/lib -lrccl -lamdhip64 -pthread
 
 CSRC := \
@@ -78,3 +78,6 @@ test: all samples
 
 clean:
 	rm -rf $(BUILD) $(LIB)
+
+-include $(OBJS:.o=.d)
+-include $(BUILD)/mlsl_amd/csrc/tests/schedule_selftest.d

@@ -22,8 +22,15 @@ Engine::~Engine() {
     if (thread_.joinable()) thread_.join();
 }
 
+static uint64_t NowNs() {
+    return std::chrono::duration_cast<std::chrono::nanoseconds>(
+               std::chrono::steady_clock::now().time_since_epoch())
+        .count();
+}
+
 void Engine::Submit(CommRequest* req) {
     req->start_seqno_ = NextSeqno();
+    req->start_ns_ = NowNs();
     if (mode_ == ProgressMode::INLINE) {
         // "Thread mode" analog: drive the request to completion on the
         // calling thread (device mode still returns early: streams are
@@ -85,8 +92,20 @@ void Engine::DrainInbox() {
 bool Engine::AdvanceOne(CommRequest* req) {
     try {
         const bool done = device_mode_ ? req->AdvanceDevice() : req->AdvanceHost(mesh_);
-        if (done) req->MarkDone();
-        return done;
+        if (done) {
+            req->MarkDone();
+            return true;
+        }
+        // Failure detection: a collective stuck past MLSL_TIMEOUT (peer
+        // died, mismatched schedule) fails loudly instead of hanging Wait
+        // forever (reference had none of this — SURVEY.md 5.3).
+        const int tmo = GlobalConfig().timeout_sec;
+        if (tmo > 0 && NowNs() - req->start_ns_ > static_cast<uint64_t>(tmo) * 1000000000ull) {
+            req->MarkFailed(std::string("timeout after ") + std::to_string(tmo) +
+                            "s in " + CollOpName(req->Spec().op));
+            return true;
+        }
+        return false;
     } catch (const std::exception& e) {
         MLSL_LOG(ERROR, "request %s failed in progress engine: %s",
                  CollOpName(req->Spec().op), e.what());

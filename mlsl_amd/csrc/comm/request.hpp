@@ -164,6 +164,7 @@ class CommRequest {
     uint8_t* rbuf_ = nullptr;
     uint32_t flow_ = 0;            // group-consistent sequence at Start
     uint64_t start_seqno_ = 0;     // engine-local, for priority ordering
+    uint64_t start_ns_ = 0;        // steady-clock ns at Submit (watchdog)
     std::atomic<ReqState> state_{ReqState::IDLE};
     std::string error_;
 

@@ -506,9 +506,16 @@ void Statistics::Update(const Event& ev, size_t bytes) {
     const unsigned long long delta = now - last_ts_;
     last_ts_ = now;
     OpStats& os = per_op_[ev.op_idx];
-    if (ev.is_compute) os.compute_cycles += delta;
-    else os.comm_cycles += delta;
+    const int cls = ev.is_param ? (ev.is_input_or_inc ? INC : GRAD)
+                                : (ev.is_input_or_inc ? IA : OA);
+    if (ev.is_compute) {
+        os.compute_cycles += delta;
+    } else {
+        os.comm_cycles += delta;
+        os.ent[cls].cycles += delta;
+    }
     os.comm_bytes += bytes;
+    os.ent[cls].bytes += bytes;
 }
 
 void Statistics::CollectIsolation() {
@@ -578,14 +585,19 @@ void Statistics::Print() {
     // src/mlsl_impl.hpp:40).
     FILE* f = std::fopen("mlsl_stats.log", "a");
     if (!f) return;
-    std::fprintf(f, "# op  comm_bytes  comm_cycles  compute_cycles  isolation_cycles\n");
+    std::fprintf(f, "# op | compute_cyc | comm_cyc | comm_KB | isolation_cyc |"
+                    " IA KB/cyc | OA KB/cyc | GRAD KB/cyc | INC KB/cyc\n");
     for (size_t i = 0; i < per_op_.size(); ++i) {
         const OpStats& os = per_op_[i];
         const char* name = i < session_->GetOperationCount()
                                ? session_->GetOperation(i)->GetName()
                                : "?";
-        std::fprintf(f, "%zu(%s)  %zu  %llu  %llu  %llu\n", i, name, os.comm_bytes,
-                     os.comm_cycles, os.compute_cycles, os.isolation_cycles);
+        std::fprintf(f, "%zu(%s) | %llu | %llu | %zu | %llu", i, name,
+                     os.compute_cycles, os.comm_cycles, os.comm_bytes / 1024,
+                     os.isolation_cycles);
+        for (int c = 0; c < 4; ++c)
+            std::fprintf(f, " | %zu/%llu", os.ent[c].bytes / 1024, os.ent[c].cycles);
+        std::fprintf(f, "\n");
     }
     std::fclose(f);
 }

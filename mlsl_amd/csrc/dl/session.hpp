@@ -239,10 +239,15 @@ class Statistics {
     unsigned long long GetTotalComputeCycles() const;
     size_t GetTotalCommSize() const;
 
+    // per-entity classes in the printed table (reference per-op table of
+    // [KB, Kcycles] per IA/OA/GRAD/INC, mlsl_impl_stats.cpp:97-363)
+    enum EntClass { IA = 0, OA = 1, GRAD = 2, INC = 3 };
+
   private:
     struct OpStats {
         unsigned long long comm_cycles = 0, compute_cycles = 0, isolation_cycles = 0;
         size_t comm_bytes = 0;
+        struct Ent { unsigned long long cycles = 0; size_t bytes = 0; } ent[4];
     };
     void EnsureSize(size_t n);
 
