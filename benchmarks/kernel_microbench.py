@@ -8,7 +8,8 @@ import time
 
 import torch
 
-sys.path.insert(0, __file__.rsplit("/", 2)[0])
+import os
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 from mlsl_amd import ops  # noqa: E402
 
 
