@@ -46,6 +46,8 @@ DeviceReqState::~DeviceReqState() {
         if (e) (void)hipEventDestroy(e);
     if (dep_event) (void)hipEventDestroy(dep_event);
     if (tmp_dev) (void)hipFree(tmp_dev);
+    if (stage_send) (void)hipFree(stage_send);
+    if (stage_recv) (void)hipFree(stage_recv);
 }
 
 namespace {
@@ -202,6 +204,64 @@ void DeviceSetupRequest(CommRequest* req, DeviceReqState& st) {
 }
 
 namespace {
+
+bool IsDevicePtr(const void* p) {
+    if (!p) return true;
+    hipPointerAttribute_t attr;
+    if (hipPointerGetAttributes(&attr, p) != hipSuccess) {
+        (void)hipGetLastError();
+        return false;  // unregistered host memory
+    }
+    return attr.type == hipMemoryTypeDevice || attr.type == hipMemoryTypeManaged;
+}
+
+// Bytes this rank contributes / receives for the whole request.
+void IoBytes(CommRequest* req, size_t* send_b, size_t* recv_b) {
+    const OpSpec& spec = req->Spec();
+    const size_t es = DtypeSize(req->Dtype());
+    const size_t N = static_cast<size_t>(req->Group()->Size());
+    const bool is_root = req->Group()->MyIdx() == spec.root;
+    size_t s = 0, r = 0;
+    switch (spec.op) {
+        case CollOp::ALLREDUCE: case CollOp::BCAST: case CollOp::REDUCE:
+            s = r = spec.count * es;
+            if (spec.op == CollOp::REDUCE && !is_root) r = spec.count * es; // scratch ok
+            break;
+        case CollOp::REDUCE_SCATTER: s = N * spec.count * es; r = spec.count * es; break;
+        case CollOp::ALLGATHER: s = spec.count * es; r = N * spec.count * es; break;
+        case CollOp::ALLGATHERV: {
+            size_t t = 0;
+            for (size_t c : spec.recv_counts) t += c;
+            s = spec.recv_counts[req->Group()->MyIdx()] * es;
+            r = t * es;
+            break;
+        }
+        case CollOp::GATHER: s = spec.count * es; r = is_root ? N * spec.count * es : 0; break;
+        case CollOp::SCATTER: s = is_root ? N * spec.count * es : 0; r = spec.count * es; break;
+        case CollOp::ALLTOALL: s = r = N * spec.count * es; break;
+        case CollOp::ALLTOALLV: {
+            size_t ts = 0, tr = 0;
+            for (size_t i = 0; i < N; ++i) {
+                ts = std::max(ts, (spec.send_offs[i] + spec.send_counts[i]) * es);
+                tr = std::max(tr, (spec.recv_offs[i] + spec.recv_counts[i]) * es);
+            }
+            s = ts; r = tr;
+            break;
+        }
+        case CollOp::SRLIST: {
+            size_t ts = 0, tr = 0;
+            for (const auto& ppp : spec.pairs) {
+                ts = std::max(ts, (ppp.send_off + ppp.send_count) * es);
+                tr = std::max(tr, (ppp.recv_off + ppp.recv_count) * es);
+            }
+            s = ts; r = tr;
+            break;
+        }
+        case CollOp::BARRIER: break;
+    }
+    *send_b = s;
+    *recv_b = r;
+}
 
 // Issue one chunk through the fused RCCL op set.
 void IssueFused(CommRequest* req, ChunkExec& ce, ncclComm_t comm, hipStream_t s,
@@ -399,6 +459,11 @@ bool DeviceAdvanceRequest(CommRequest* req, DeviceReqState& st) {
     auto* rt = static_cast<HipRuntime*>(Context::Get().Device());
     if (!st.issued) {
         GroupComms& gc = rt->For(req->Group());
+        if (gc.comms.empty() && gc.streams.empty()) {
+            hipStream_t s0;
+            HIP_CHECKD(hipStreamCreateWithFlags(&s0, hipStreamNonBlocking));
+            gc.streams.push_back(s0);
+        }
         // Order after the caller's compute stream: the producer kernels
         // (e.g. torch backward on the default stream) must land before the
         // collective reads the buffers.
@@ -408,6 +473,35 @@ bool DeviceAdvanceRequest(CommRequest* req, DeviceReqState& st) {
                                   static_cast<hipStream_t>(rt->ComputeStream())));
         for (hipStream_t cs : gc.streams)
             HIP_CHECKD(hipStreamWaitEvent(cs, st.dep_event, 0));
+
+        // Host-buffer staging (ReplaceIn analog): stage unregistered host
+        // buffers through persistent HBM so RCCL/kernels see device memory.
+        size_t send_b = 0, recv_b = 0;
+        IoBytes(req, &send_b, &recv_b);
+        const bool s_host = send_b > 0 && !IsDevicePtr(req->UserSendBuf());
+        const bool r_host = recv_b > 0 && !IsDevicePtr(req->UserRecvBuf());
+        st.recv_staged = r_host;
+        if (s_host) {
+            if (!st.stage_send || st.stage_send_bytes < send_b) {
+                if (st.stage_send) (void)hipFree(st.stage_send);
+                HIP_CHECKD(hipMalloc(&st.stage_send, send_b));
+                st.stage_send_bytes = send_b;
+            }
+            HIP_CHECKD(hipMemcpyAsync(st.stage_send, req->UserSendBuf(), send_b,
+                                      hipMemcpyHostToDevice, gc.streams[0]));
+            if (gc.streams.size() > 1) {
+                HIP_CHECKD(hipEventRecord(st.dep_event, gc.streams[0]));
+                for (size_t i = 1; i < gc.streams.size(); ++i)
+                    HIP_CHECKD(hipStreamWaitEvent(gc.streams[i], st.dep_event, 0));
+            }
+        }
+        if (r_host && (!st.stage_recv || st.stage_recv_bytes < recv_b)) {
+            if (st.stage_recv) (void)hipFree(st.stage_recv);
+            HIP_CHECKD(hipMalloc(&st.stage_recv, recv_b));
+            st.stage_recv_bytes = recv_b;
+        }
+        req->SetDeviceBuffers(s_host ? static_cast<const uint8_t*>(st.stage_send) : nullptr,
+                              r_host ? static_cast<uint8_t*>(st.stage_recv) : nullptr);
         auto& chunks = req->Chunks();
         const Config& cfg = GlobalConfig();
         const bool use_schedule =
@@ -418,13 +512,7 @@ bool DeviceAdvanceRequest(CommRequest* req, DeviceReqState& st) {
         // Single-rank groups: local copies only, still through a (persistent)
         // stream so completion semantics are uniform.
         if (gc.comms.empty()) {
-            if (gc.streams.empty()) {
-                hipStream_t s0;
-                HIP_CHECKD(hipStreamCreateWithFlags(&s0, hipStreamNonBlocking));
-                gc.streams.push_back(s0);
-            }
             hipStream_t s0 = gc.streams[0];
-            HIP_CHECKD(hipStreamWaitEvent(s0, st.dep_event, 0));
             const size_t es = DtypeSize(req->Dtype());
             for (auto& ce : chunks) {
                 const uint8_t* sbase = req->SendBuf() + ce.elem_off * es;
@@ -441,9 +529,12 @@ bool DeviceAdvanceRequest(CommRequest* req, DeviceReqState& st) {
                                      req->Dtype(), s0);
                 } else if (ce.sch.result.bytes && sbase != rbase) {
                     HIP_CHECKD(hipMemcpyAsync(rbase + ce.sch.result.off, sbase,
-                                              ce.sch.result.bytes, hipMemcpyDeviceToDevice, s0));
+                                              ce.sch.result.bytes, hipMemcpyDefault, s0));
                 }
             }
+            if (st.recv_staged)
+                HIP_CHECKD(hipMemcpyAsync(req->UserRecvBuf(), st.stage_recv, recv_b,
+                                          hipMemcpyDeviceToHost, s0));
             if (st.events.empty()) {
                 hipEvent_t e;
                 HIP_CHECKD(hipEventCreateWithFlags(&e, hipEventDisableTiming));
@@ -495,6 +586,19 @@ bool DeviceAdvanceRequest(CommRequest* req, DeviceReqState& st) {
         }
         for (size_t ch = 0; ch < used; ++ch)
             HIP_CHECKD(hipEventRecord(st.events[ch], gc.streams[ch]));
+        if (st.recv_staged) {
+            // join all channels on stream 0, then stage the result out
+            for (size_t ch = 0; ch < used; ++ch)
+                HIP_CHECKD(hipStreamWaitEvent(gc.streams[0], st.events[ch], 0));
+            HIP_CHECKD(hipMemcpyAsync(req->UserRecvBuf(), st.stage_recv, recv_b,
+                                      hipMemcpyDeviceToHost, gc.streams[0]));
+            while (st.events.size() < used + 1) {
+                hipEvent_t e;
+                HIP_CHECKD(hipEventCreateWithFlags(&e, hipEventDisableTiming));
+                st.events.push_back(e);
+            }
+            HIP_CHECKD(hipEventRecord(st.events[used], gc.streams[0]));
+        }
         st.issued = true;
     }
     for (hipEvent_t e : st.events) {

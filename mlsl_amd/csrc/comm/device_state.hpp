@@ -19,6 +19,13 @@ struct DeviceReqState {
     // Persistent device scratch (allocated at Setup in device mode).
     void* tmp_dev = nullptr;
     size_t tmp_bytes = 0;
+    // Host-buffer staging (reference ReplaceIn/ReplaceOut,
+    // src/comm_ep.cpp:363-566): user buffers that are not device memory are
+    // staged through persistent HBM buffers around the collective.
+    void* stage_send = nullptr;
+    void* stage_recv = nullptr;
+    size_t stage_send_bytes = 0, stage_recv_bytes = 0;
+    bool recv_staged = false;
     ~DeviceReqState();
 };
 

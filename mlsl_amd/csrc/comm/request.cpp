@@ -312,6 +312,8 @@ void CommRequest::Start(const void* sbuf, void* rbuf) {
                "Start() while request in flight");
     sbuf_ = static_cast<const uint8_t*>(sbuf);
     rbuf_ = static_cast<uint8_t*>(rbuf);
+    dev_sbuf_ = nullptr;
+    dev_rbuf_ = nullptr;
     flow_ = group_->NextFlow();
     for (auto& ce : chunks_) ce.Reset();
     if (dev_) dev_->issued = false;

@@ -137,10 +137,18 @@ class CommRequest {
     ReqState State() const { return state_.load(std::memory_order_acquire); }
     uint64_t StartSeqno() const { return start_seqno_; }
 
-    // Device-executor access (device_comm.cpp).
+    // Device-executor access (device_comm.cpp). SendBuf/RecvBuf return the
+    // staging override when host buffers were staged into HBM
+    // (ReplaceIn/Out analog); User*Buf always return the caller's pointers.
     std::vector<ChunkExec>& Chunks() { return chunks_; }
-    const uint8_t* SendBuf() const { return sbuf_; }
-    uint8_t* RecvBuf() const { return rbuf_; }
+    const uint8_t* SendBuf() const { return dev_sbuf_ ? dev_sbuf_ : sbuf_; }
+    uint8_t* RecvBuf() const { return dev_rbuf_ ? dev_rbuf_ : rbuf_; }
+    const uint8_t* UserSendBuf() const { return sbuf_; }
+    uint8_t* UserRecvBuf() const { return rbuf_; }
+    void SetDeviceBuffers(const uint8_t* s, uint8_t* r) {
+        dev_sbuf_ = s;
+        dev_rbuf_ = r;
+    }
 
   private:
     friend class Engine;
@@ -162,6 +170,8 @@ class CommRequest {
     // per-Start state
     const uint8_t* sbuf_ = nullptr;
     uint8_t* rbuf_ = nullptr;
+    const uint8_t* dev_sbuf_ = nullptr;  // staging overrides (device mode)
+    uint8_t* dev_rbuf_ = nullptr;
     uint32_t flow_ = 0;            // group-consistent sequence at Start
     uint64_t start_seqno_ = 0;     // engine-local, for priority ordering
     uint64_t start_ns_ = 0;        // steady-clock ns at Submit (watchdog)

@@ -176,3 +176,21 @@ class TestDevicePaths:
                 net.step(it)
         finally:
             mx.finalize()
+
+
+@requires_gpu
+class TestHostStaging:
+    def test_numpy_buffers_on_device_engine(self):
+        """Host (numpy) buffers ride the GPU path via staging
+        (ReplaceIn/Out analog)."""
+        import numpy as np
+        import mlsl_amd as mx
+        mx.init()
+        try:
+            d = mx.Distribution(1, 1)
+            a = np.random.randn(100_000).astype(np.float32)
+            out = np.zeros_like(a)
+            mx.wait(d.all_reduce(a, out, a.size, op="sum", group="data"))
+            assert np.allclose(out, a)
+        finally:
+            mx.finalize()
