@@ -48,3 +48,8 @@ def test_ring_algo():
 @pytest.mark.parametrize("world", [1, 2, 4])
 def test_quantized_allreduce(world):
     run_ranks("quantized_allreduce", world)
+
+
+@pytest.mark.parametrize("world", [1, 2])
+def test_compat_shim(world):
+    run_ranks("compat_shim", world)

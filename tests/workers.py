@@ -288,6 +288,27 @@ def py_sample():
 WORKERS["py_sample"] = py_sample
 
 
+
+
+def compat_shim():
+    """Reference-style Python surface (mlsl_amd.compat) end to end."""
+    from mlsl_amd.compat import MLSL, DataType, GroupType, ReductionType
+    m = MLSL()
+    m.Init()
+    rank, size = m.GetProcessIdx(), m.GetProcessCount()
+    dist = m.CreateDistribution(size, 1)
+    buf = np.full(128, float(rank), dtype=np.float32)
+    req = dist.AllReduce(buf, buf, 128, DataType.FLOAT, ReductionType.SUM,
+                         GroupType.DATA)
+    m.Wait(req)
+    assert np.all(buf == (size - 1) * size / 2.0)
+    m.DeleteDistribution(dist)
+    m.Finalize()
+
+
+WORKERS["compat_shim"] = compat_shim
+
+
 def main():
     name = sys.argv[1]
     fn = WORKERS.get(name)
