@@ -37,7 +37,7 @@ OBJS := $(CSRC:%.cpp=$(BUILD)/%.o) $(HIPSRC:%.hip=$(BUILD)/%.o)
 
 .PHONY: all lib selftest clean test
 
-all: lib selftest
+all: lib selftest apitest
 
 lib: $(LIB)
 
@@ -57,6 +57,14 @@ selftest: $(SELFTEST)
 $(SELFTEST): $(BUILD)/mlsl_amd/csrc/tests/schedule_selftest.o $(BUILD)/mlsl_amd/csrc/comm/schedule.o $(BUILD)/mlsl_amd/csrc/core/log.o
 	@mkdir -p $(dir $@)
 	$(HIPCC) $^ -L/opt/rocm/lib -lamdhip64 -pthread -o $@
+
+APITEST := $(BUILD)/api_selftest
+
+apitest: $(APITEST)
+
+$(APITEST): $(BUILD)/mlsl_amd/csrc/tests/api_selftest.o $(LIB)
+	@mkdir -p $(dir $@)
+	$(HIPCC) $< -Lmlsl_amd -lmlsl_amd -Wl,-rpath,'$$ORIGIN/../mlsl_amd' -pthread -o $@
 
 SAMPLES := $(BUILD)/mlsl_sample $(BUILD)/cmlsl_sample
 
