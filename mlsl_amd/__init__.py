@@ -17,7 +17,7 @@ Quick start (one process per GPU, torchrun-style env):
 from .api import (  # noqa: F401
     COMPRESSION, DTYPE, DTYPE_SIZE, GROUP, OPTYPE, REDOP,
     Activation, CommBlockInfo, Distribution, Operation, OperationRegInfo,
-    ParameterSet, Session, Statistics,
+    ParameterSet, PersistentRequest, Session, Statistics,
     alloc, free, finalize, init, is_initialized, rank, set_compute_stream,
     set_quant_params,
     test, version, wait, world_size,

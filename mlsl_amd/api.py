@@ -255,6 +255,74 @@ class Distribution:
         return req
 
 
+class PersistentRequest:
+    """Describe-once/run-many collective (persistent request): Setup happens
+    at construction, start/wait/test are allocation-free."""
+
+    def __init__(self, dist, kind, count, dtype="f32", op="sum", group="global",
+                 quantized=False):
+        import ctypes as _c
+        L = lib()
+        if not hasattr(L, "_persist_declared"):
+            P = _c.POINTER
+            L.mlsl_persistent_all_reduce.argtypes = [c_void_p, c_size_t, c_int, c_int,
+                                                     c_int, c_int, P(c_void_p)]
+            L.mlsl_persistent_reduce_scatter.argtypes = [c_void_p, c_size_t, c_int,
+                                                         c_int, c_int, P(c_void_p)]
+            L.mlsl_persistent_all_gather.argtypes = [c_void_p, c_size_t, c_int, c_int,
+                                                     P(c_void_p)]
+            L.mlsl_persistent_all_to_all.argtypes = [c_void_p, c_size_t, c_int, c_int,
+                                                     P(c_void_p)]
+            L.mlsl_request_start.argtypes = [c_void_p, c_void_p, c_void_p]
+            L.mlsl_request_wait.argtypes = [c_void_p, P(c_void_p)]
+            L.mlsl_request_test.argtypes = [c_void_p, P(c_int)]
+            L.mlsl_request_destroy.argtypes = [c_void_p]
+            for n in ("mlsl_persistent_all_reduce", "mlsl_persistent_reduce_scatter",
+                      "mlsl_persistent_all_gather", "mlsl_persistent_all_to_all",
+                      "mlsl_request_start", "mlsl_request_wait", "mlsl_request_test",
+                      "mlsl_request_destroy"):
+                getattr(L, n).restype = c_int
+            L._persist_declared = True
+        h = c_void_p()
+        if kind == "all_reduce":
+            check(L.mlsl_persistent_all_reduce(dist._h, count, DTYPE[dtype], REDOP[op],
+                                               GROUP[group], 1 if quantized else 0,
+                                               ctypes.byref(h)))
+        elif kind == "reduce_scatter":
+            check(L.mlsl_persistent_reduce_scatter(dist._h, count, DTYPE[dtype],
+                                                   REDOP[op], GROUP[group],
+                                                   ctypes.byref(h)))
+        elif kind == "all_gather":
+            check(L.mlsl_persistent_all_gather(dist._h, count, DTYPE[dtype],
+                                               GROUP[group], ctypes.byref(h)))
+        elif kind == "all_to_all":
+            check(L.mlsl_persistent_all_to_all(dist._h, count, DTYPE[dtype],
+                                               GROUP[group], ctypes.byref(h)))
+        else:
+            raise ValueError(kind)
+        self._h = h
+
+    def start(self, sbuf, rbuf):
+        sp, _ = _as_ptr_dtype(sbuf)
+        rp, _ = _as_ptr_dtype(rbuf)
+        check(lib().mlsl_request_start(self._h, sp, rp))
+
+    def wait(self):
+        out = c_void_p()
+        check(lib().mlsl_request_wait(self._h, ctypes.byref(out)))
+        return out.value
+
+    def test(self):
+        done = c_int(0)
+        check(lib().mlsl_request_test(self._h, ctypes.byref(done)))
+        return bool(done.value)
+
+    def destroy(self):
+        if self._h:
+            check(lib().mlsl_request_destroy(self._h))
+            self._h = None
+
+
 class CommBlockInfo:
     def __init__(self, handle):
         self._h = handle

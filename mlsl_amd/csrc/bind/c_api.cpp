@@ -204,6 +204,77 @@ int mlsl_distribution_reduce_scatter(mlsl_distribution d, const void* sbuf, void
     C_CATCH
 }
 
+/* ---- persistent requests ---- */
+
+#include "../comm/group.hpp"
+
+static CommRequest* NewPersistent(mlsl_distribution d, mlsl_group g, mlsl_data_type dt) {
+    Distribution* dist = DIST(d);
+    return new CommRequest(dist->Group(GK(g)), DT(dt), CompType::GENERIC);
+}
+
+int mlsl_persistent_all_reduce(mlsl_distribution d, size_t count, mlsl_data_type dt,
+                               mlsl_reduction op, mlsl_group g, int quantized,
+                               mlsl_request* out) {
+    C_TRY CommRequest* r = NewPersistent(d, g, dt);
+    r->AddAllReduce(count, ROP(op));
+    if (quantized)
+        r->SetCompression(Compression::QUANT_INT8,
+                          Environment::GetEnv().GetQuantizationParams());
+    r->Setup();
+    *out = r;
+    C_CATCH
+}
+
+int mlsl_persistent_reduce_scatter(mlsl_distribution d, size_t recv_count,
+                                   mlsl_data_type dt, mlsl_reduction op, mlsl_group g,
+                                   mlsl_request* out) {
+    C_TRY CommRequest* r = NewPersistent(d, g, dt);
+    r->AddReduceScatter(recv_count, ROP(op));
+    r->Setup();
+    *out = r;
+    C_CATCH
+}
+
+int mlsl_persistent_all_gather(mlsl_distribution d, size_t send_count, mlsl_data_type dt,
+                               mlsl_group g, mlsl_request* out) {
+    C_TRY CommRequest* r = NewPersistent(d, g, dt);
+    r->AddAllGather(send_count);
+    r->Setup();
+    *out = r;
+    C_CATCH
+}
+
+int mlsl_persistent_all_to_all(mlsl_distribution d, size_t send_count, mlsl_data_type dt,
+                               mlsl_group g, mlsl_request* out) {
+    C_TRY CommRequest* r = NewPersistent(d, g, dt);
+    r->AddAlltoAll(send_count);
+    r->Setup();
+    *out = r;
+    C_CATCH
+}
+
+int mlsl_request_start(mlsl_request req, const void* sbuf, void* rbuf) {
+    C_TRY static_cast<CommRequest*>(req)->Start(sbuf, rbuf);
+    C_CATCH
+}
+
+int mlsl_request_wait(mlsl_request req, void** result) {
+    C_TRY void* r = static_cast<CommRequest*>(req)->Wait();
+    if (result) *result = r;
+    C_CATCH
+}
+
+int mlsl_request_test(mlsl_request req, int* done) {
+    C_TRY* done = static_cast<CommRequest*>(req)->Test() ? 1 : 0;
+    C_CATCH
+}
+
+int mlsl_request_destroy(mlsl_request req) {
+    C_TRY delete static_cast<CommRequest*>(req);
+    C_CATCH
+}
+
 /* ---- session / planner ---- */
 
 #define SES(s) static_cast<Session*>(s)

@@ -183,16 +183,18 @@ DeviceRuntime* CreateDeviceRuntime() {
 }
 
 void DeviceSetupRequest(CommRequest* req, DeviceReqState& st) {
-    // Persistent scratch for schedule-path chunks + quant/barrier staging.
+    // Persistent scratch only where the device executor needs it: the
+    // compressed path (wire + residual), the custom schedule path (segment
+    // staging), and the barrier token. The fused RCCL path allocates
+    // nothing — one-shot hot-loop requests must not pay a hipMalloc+memset
+    // per iteration.
     size_t tmp = 0;
-    for (auto& ce : req->Chunks()) {
-        if (req->Compressed()) {
-            // [wire][schedule scratch][error-feedback residual]
+    if (req->Compressed()) {
+        for (auto& ce : req->Chunks())
             tmp += ce.sch.result.bytes + ce.sch.tmp_bytes +
                    req->Spec().count * DtypeSize(req->Dtype());
-        } else {
-            tmp += ce.sch.tmp_bytes;
-        }
+    } else if (req->UsesDeviceSchedule()) {
+        for (auto& ce : req->Chunks()) tmp += ce.sch.tmp_bytes;
     }
     if (req->Spec().op == CollOp::BARRIER) tmp = std::max<size_t>(tmp, 16);
     st.tmp_bytes = tmp;
@@ -503,10 +505,7 @@ bool DeviceAdvanceRequest(CommRequest* req, DeviceReqState& st) {
         req->SetDeviceBuffers(s_host ? static_cast<const uint8_t*>(st.stage_send) : nullptr,
                               r_host ? static_cast<uint8_t*>(st.stage_recv) : nullptr);
         auto& chunks = req->Chunks();
-        const Config& cfg = GlobalConfig();
-        const bool use_schedule =
-            (cfg.allreduce_algo == AllReduceAlgo::RING || cfg.allreduce_algo == AllReduceAlgo::RHD) &&
-            req->Spec().op == CollOp::ALLREDUCE && req->Group()->Size() > 1;
+        const bool use_schedule = req->UsesDeviceSchedule();
 
         const bool compressed = req->Compressed();
         // Single-rank groups: local copies only, still through a (persistent)

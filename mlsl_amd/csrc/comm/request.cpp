@@ -146,6 +146,13 @@ bool CommRequest::Compressed() const {
            spec_.rop == ReduceOp::SUM;
 }
 
+bool CommRequest::UsesDeviceSchedule() const {
+    const Config& cfg = GlobalConfig();
+    return (cfg.allreduce_algo == AllReduceAlgo::RING ||
+            cfg.allreduce_algo == AllReduceAlgo::RHD) &&
+           spec_.op == CollOp::ALLREDUCE && group_->Size() > 1 && !Compressed();
+}
+
 size_t CommRequest::WireBytesFor(const ChunkExec& ce) const {
     const size_t cnt = ce.sch.result.bytes ? ce.sch.result.bytes /
         (qparams_.block_elems + 8) * qparams_.block_elems : 0;

@@ -109,6 +109,9 @@ class CommRequest {
     Compression GetCompression() const { return comp_; }
     const QuantParams& QParams() const { return qparams_; }
     bool Compressed() const;
+    // True when the device executor will walk the schedule (custom ring/RHD
+    // over ncclSend/Recv) instead of the fused RCCL op.
+    bool UsesDeviceSchedule() const;
     // Compressed-path layout inside each chunk's tmp:
     // [wire W][schedule scratch][error-feedback residual].
     size_t WireBytesFor(const ChunkExec& ce) const;

@@ -86,6 +86,24 @@ int mlsl_distribution_reduce_scatter(mlsl_distribution d, const void* sbuf, void
                                      size_t recv_count, mlsl_data_type dt, mlsl_reduction op,
                                      mlsl_group g, mlsl_request* out);
 
+/* persistent requests: describe once, Start/Wait/Test every iteration
+ * (the reference's Session-level persistent-request contract exposed for
+ * generic collectives; hot loops skip per-call planning/allocation). */
+int mlsl_persistent_all_reduce(mlsl_distribution d, size_t count, mlsl_data_type dt,
+                               mlsl_reduction op, mlsl_group g, int quantized,
+                               mlsl_request* out);
+int mlsl_persistent_reduce_scatter(mlsl_distribution d, size_t recv_count,
+                                   mlsl_data_type dt, mlsl_reduction op, mlsl_group g,
+                                   mlsl_request* out);
+int mlsl_persistent_all_gather(mlsl_distribution d, size_t send_count, mlsl_data_type dt,
+                               mlsl_group g, mlsl_request* out);
+int mlsl_persistent_all_to_all(mlsl_distribution d, size_t send_count, mlsl_data_type dt,
+                               mlsl_group g, mlsl_request* out);
+int mlsl_request_start(mlsl_request req, const void* sbuf, void* rbuf);
+int mlsl_request_wait(mlsl_request req, void** result);
+int mlsl_request_test(mlsl_request req, int* done);
+int mlsl_request_destroy(mlsl_request req);
+
 /* session / planner (reference mlsl.hpp:731-795) */
 int mlsl_session_create(mlsl_phase phase, mlsl_session* out);
 int mlsl_session_free(mlsl_session s);
