@@ -77,8 +77,12 @@ def main():
             buf = np.random.randn(count).astype(np.float32)
             out = np.empty_like(buf)
 
+        preq = mx.PersistentRequest(d, "all_reduce", count, dtype="f32",
+                                    op="sum", group="data")
+
         def run():
-            mx.wait(d.all_reduce(buf, out, count, op="sum", group="data"))
+            preq.start(buf, out)
+            preq.wait()
 
         def sync():
             if use_cuda:
@@ -106,6 +110,7 @@ def main():
         if rank == 0:
             print(json.dumps(rec))
             lines.append(rec)
+        preq.destroy()
 
         if tdist is not None:
             tb = buf.clone() if use_cuda else buf.copy()
