@@ -335,12 +335,33 @@ void CommRequest::Start(const void* sbuf, void* rbuf) {
                    "collective buffer not from Environment::Alloc "
                    "(MLSL_CHECK_POINTERS=1)");
     }
+    if (ctx.DeviceMode()) {
+        // Device path: issue inline — RCCL/kernel enqueues are non-blocking
+        // and the STREAMS are the progress engine; Wait/Test poll the
+        // completion events directly. This keeps small-message latency at
+        // enqueue cost (no ring/thread/condvar hop).
+        state_.store(ReqState::ACTIVE, std::memory_order_release);
+        start_seqno_ = ctx.GetEngine()->NextSeqno();
+        if (AdvanceDevice()) MarkDone();
+        return;
+    }
     state_.store(ReqState::QUEUED, std::memory_order_release);
     ctx.GetEngine()->Submit(this);
 }
 
 void* CommRequest::Wait() {
-    Context::Get().GetEngine()->WaitFor(this);
+    Context& ctx = Context::Get();
+    if (ctx.DeviceMode()) {
+        while (state_.load(std::memory_order_acquire) == ReqState::ACTIVE) {
+            try {
+                if (AdvanceDevice()) MarkDone();
+            } catch (const std::exception& e) {
+                MarkFailed(e.what());
+            }
+        }
+    } else {
+        ctx.GetEngine()->WaitFor(this);
+    }
     if (state_.load(std::memory_order_acquire) == ReqState::FAILED)
         MLSL_THROW("request failed: " + error_);
     state_.store(ReqState::IDLE, std::memory_order_release);
@@ -359,7 +380,20 @@ bool CommRequest::Test() {
     ReqState st = state_.load(std::memory_order_acquire);
     if (st == ReqState::FAILED) MLSL_THROW("request failed: " + error_);
     if (st == ReqState::DONE || st == ReqState::IDLE) return true;
-    return Context::Get().GetEngine()->TestFor(this);
+    Context& ctx = Context::Get();
+    if (ctx.DeviceMode()) {
+        try {
+            if (AdvanceDevice()) {
+                MarkDone();
+                return true;
+            }
+        } catch (const std::exception& e) {
+            MarkFailed(e.what());
+            MLSL_THROW("request failed: " + error_);
+        }
+        return false;
+    }
+    return ctx.GetEngine()->TestFor(this);
 }
 
 void CommRequest::MarkDone() {
