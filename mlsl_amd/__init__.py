@@ -18,7 +18,8 @@ from .api import (  # noqa: F401
     COMPRESSION, DTYPE, DTYPE_SIZE, GROUP, OPTYPE, REDOP,
     Activation, CommBlockInfo, Distribution, Operation, OperationRegInfo,
     ParameterSet, PersistentRequest, Session, Statistics,
-    alloc, free, finalize, init, is_initialized, rank, set_compute_stream,
+    alloc, configure, free, finalize, init, is_initialized, rank,
+    set_compute_stream,
     set_quant_params,
     test, version, wait, world_size,
 )

@@ -54,6 +54,17 @@ def finalize():
     check(lib().mlsl_finalize())
 
 
+def configure(config):
+    """Tenant re-split: configure("color=N") — ranks sharing N become their
+    own world (reference Environment::Configure)."""
+    L = lib()
+    if not hasattr(L, "_cfg_declared"):
+        L.mlsl_configure.argtypes = [ctypes.c_char_p]
+        L.mlsl_configure.restype = c_int
+        L._cfg_declared = True
+    check(L.mlsl_configure(config.encode()))
+
+
 def is_initialized():
     v = c_int(0)
     check(lib().mlsl_initialized(ctypes.byref(v)))

@@ -46,6 +46,11 @@ int mlsl_finalize(void) {
     C_CATCH
 }
 
+int mlsl_configure(const char* config) {
+    C_TRY Environment::GetEnv().Configure(config);
+    C_CATCH
+}
+
 int mlsl_initialized(int* out) {
     C_TRY* out = Environment::GetEnv().IsInitialized() ? 1 : 0;
     C_CATCH

@@ -31,8 +31,10 @@ void Context::Init(int rank, int size) {
     AutoConfig();
 
     boot_ = std::make_unique<Bootstrap>(rank, size);
-    rank_ = boot_->Rank();
-    size_ = boot_->Size();
+    boot_rank_ = boot_->Rank();
+    boot_size_ = boot_->Size();
+    rank_ = boot_rank_;
+    size_ = boot_size_;
 
     // Transport selection: device mode when a HIP device is visible unless
     // MLSL_TRANSPORT forces tcp.
@@ -52,12 +54,12 @@ void Context::Init(int rank, int size) {
 
     // World and self groups.
     {
-        std::vector<int> all(static_cast<size_t>(size_));
-        for (int i = 0; i < size_; ++i) all[i] = i;
-        groups_.push_back(std::make_unique<ProcessGroup>(next_group_uid_++, all, rank_));
+        std::vector<int> all(static_cast<size_t>(boot_size_));
+        for (int i = 0; i < boot_size_; ++i) all[i] = i;
+        groups_.push_back(std::make_unique<ProcessGroup>(next_group_uid_++, all, boot_rank_));
         world_ = groups_.back().get();
         groups_.push_back(std::make_unique<ProcessGroup>(
-            next_group_uid_++, std::vector<int>{rank_}, rank_));
+            next_group_uid_++, std::vector<int>{boot_rank_}, boot_rank_));
         self_ = groups_.back().get();
     }
     if (device_mode_) device_->EnsureGroupComms(world_);
@@ -93,21 +95,47 @@ void Context::Finalize() {
     g_initialized = false;
 }
 
+void Context::Configure(int tenant_color) {
+    MLSL_CHECK(initialized_, "Context not initialized");
+    tenant_color_ = tenant_color;
+    // Re-split the world by tenant color; groups and Rank/Size become
+    // tenant-relative (reference Configure color-split semantics).
+    struct TC { int32_t tenant; int32_t color; };
+    std::vector<TC> all(static_cast<size_t>(boot_size_));
+    TC mine{tenant_color_, 0};
+    boot_->Allgather(&mine, sizeof(TC), all.data());
+    std::vector<int> members;
+    for (int i = 0; i < boot_size_; ++i)
+        if (all[i].tenant == tenant_color_) members.push_back(i);
+    groups_.push_back(std::make_unique<ProcessGroup>(next_group_uid_++, members,
+                                                     boot_rank_));
+    world_ = groups_.back().get();
+    rank_ = world_->MyIdx();
+    size_ = world_->Size();
+    if (device_mode_) device_->EnsureGroupComms(world_);
+    MLSL_LOG(INFO, "configured tenant color=%d: rank %d/%d", tenant_color_, rank_,
+             size_);
+}
+
 ProcessGroup* Context::CreateGroup(int color) {
     MLSL_CHECK(initialized_, "Context not initialized");
     // Collective color exchange over the bootstrap (MPI_Comm_split analog,
     // ordered by world rank — reference src/comm_ep.cpp:1821-1827).
-    std::vector<int32_t> colors(static_cast<size_t>(size_));
-    int32_t mine = color;
-    boot_->Allgather(&mine, sizeof(int32_t), colors.data());
+    // Colors are namespaced by tenant so Configure'd sub-worlds cannot
+    // accidentally merge groups.
+    struct TC { int32_t tenant; int32_t color; };
+    std::vector<TC> colors(static_cast<size_t>(boot_size_));
+    TC mine{tenant_color_, color};
+    boot_->Allgather(&mine, sizeof(TC), colors.data());
     std::vector<int> members;
-    for (int i = 0; i < size_; ++i)
-        if (colors[i] == color && color >= 0) members.push_back(i);
+    for (int i = 0; i < boot_size_; ++i)
+        if (colors[i].tenant == tenant_color_ && colors[i].color == color && color >= 0)
+            members.push_back(i);
     if (color < 0) members.clear();
 
     groups_.push_back(std::make_unique<ProcessGroup>(next_group_uid_++,
-                                                     members.empty() ? std::vector<int>{rank_} : members,
-                                                     rank_));
+                                                     members.empty() ? std::vector<int>{boot_rank_} : members,
+                                                     boot_rank_));
     ProcessGroup* g = groups_.back().get();
     // Collective over the WORLD (the unique-id exchange inside runs on the
     // bootstrap): every rank must call, members then init their comms.

@@ -28,8 +28,15 @@ class Context {
     void Init(int rank = -1, int size = -1);
     void Finalize();
 
+    // Multi-tenant re-split (reference Environment::Configure "color=N",
+    // src/mlsl.cpp:620-647): ranks with the same color form their own
+    // world; Rank/Size and new groups are tenant-relative afterwards.
+    void Configure(int tenant_color);
+
     int Rank() const { return rank_; }
     int Size() const { return size_; }
+    int BootRank() const { return boot_rank_; }
+    int BootSize() const { return boot_size_; }
     bool DeviceMode() const { return device_mode_; }
     int DeviceId() const { return device_id_; }
 
@@ -56,7 +63,9 @@ class Context {
     Context() = default;
 
     bool initialized_ = false;
-    int rank_ = 0, size_ = 1;
+    int rank_ = 0, size_ = 1;           // tenant-relative after Configure
+    int boot_rank_ = 0, boot_size_ = 1; // transport-wide
+    int tenant_color_ = 0;
     bool device_mode_ = false;
     int device_id_ = -1;
 

@@ -3,6 +3,7 @@
 // core: ProcessGroup = RCCL communicator set / TCP mesh subset).
 #include "../include/mlsl/mlsl.hpp"
 
+#include <cstring>
 #include <functional>
 #include <memory>
 #include <mutex>
@@ -241,6 +242,13 @@ void Environment::Init(int* argc, char** argv[]) {
 }
 
 void Environment::Init(int rank, int size) { Context::Get().Init(rank, size); }
+
+void Environment::Configure(const char* config) {
+    if (!config) return;
+    const char* p = std::strstr(config, "color=");
+    MLSL_CHECK(p != nullptr, "Configure expects \"color=N\"");
+    Context::Get().Configure(std::atoi(p + 6));
+}
 
 void Environment::Finalize() {
     RequestStorage::Get().Clear();

@@ -42,6 +42,7 @@ const char* mlsl_last_error(void);
 /* environment */
 int mlsl_init(int rank, int size);  /* -1,-1: from env (RANK/WORLD_SIZE) */
 int mlsl_finalize(void);
+int mlsl_configure(const char* config);  /* "color=N" tenant re-split */
 int mlsl_initialized(int* out);
 int mlsl_get_version(int* out);
 int mlsl_rank(size_t* out);

@@ -85,6 +85,9 @@ class Environment {
 
     void Init(int* argc = nullptr, char** argv[] = nullptr);
     void Init(int rank, int size);
+    // Multi-tenant world re-split: Configure("color=N") makes ranks with
+    // the same N their own world (reference mlsl.hpp Configure semantics).
+    void Configure(const char* config = nullptr);
     void Finalize();
     bool IsInitialized() const;
 

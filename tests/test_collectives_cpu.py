@@ -53,3 +53,7 @@ def test_quantized_allreduce(world):
 @pytest.mark.parametrize("world", [1, 2])
 def test_compat_shim(world):
     run_ranks("compat_shim", world)
+
+
+def test_configure_tenants():
+    run_ranks("configure_tenants", 4)

@@ -309,6 +309,29 @@ def compat_shim():
 WORKERS["compat_shim"] = compat_shim
 
 
+def configure_tenants():
+    """Environment.Configure("color=N"): 4 boot ranks split into two
+    2-rank tenant worlds; collectives stay inside each tenant."""
+    import mlsl_amd as mx
+    boot_rank = int(os.environ["RANK"])
+    mx.init()
+    assert mx.world_size() == 4
+    color = boot_rank // 2
+    mx.configure(f"color={color}")
+    assert mx.world_size() == 2, mx.world_size()
+    assert mx.rank() == boot_rank % 2
+    d = mx.Distribution(2, 1)
+    a = np.full(16, float(boot_rank), dtype=np.float32)
+    out = np.zeros(16, dtype=np.float32)
+    mx.wait(d.all_reduce(a, out, 16, op="sum", group="data"))
+    want = {0: 1.0, 1: 1.0, 2: 5.0, 3: 5.0}[boot_rank]
+    assert np.all(out == want), f"tenant allreduce got {out[0]} want {want}"
+    mx.finalize()
+
+
+WORKERS["configure_tenants"] = configure_tenants
+
+
 def main():
     name = sys.argv[1]
     fn = WORKERS.get(name)
