@@ -417,8 +417,31 @@ __global__ void PackKernel(const T* __restrict__ src, T* __restrict__ dst,
 
 }  // namespace
 
+namespace {
+
+// Row-contiguous blocks whose fm rows are 16-byte multiples vectorize to
+// uint4 moves (4x fewer instructions; coalescing unchanged).
+bool Pack16(const PackBlockDesc& d, size_t es, PackBlockDesc* out) {
+    if ((d.fm_size * es) % 16 != 0 || (d.buf_offset * es) % 16 != 0) return false;
+    *out = d;
+    out->fm_size = d.fm_size * es / 16;
+    out->buf_offset = d.buf_offset * es / 16;
+    return true;
+}
+
+}  // namespace
+
 void LaunchPack(const void* src, void* dst, const PackBlockDesc& d, DataType dt,
                 hipStream_t stream) {
+    const size_t es = DtypeSize(dt);
+    PackBlockDesc d16;
+    if (Pack16(d, es, &d16)) {
+        const size_t total16 = d16.mb_count * d16.fm_count * d16.fm_size;
+        PackKernel<uint4, true><<<dim3(GridFor(total16)), dim3(kBlock), 0, stream>>>(
+            static_cast<const uint4*>(src), static_cast<uint4*>(dst), d16);
+        HIP_CHECK(hipGetLastError());
+        return;
+    }
     const size_t total = d.mb_count * d.fm_count * d.fm_size;
     const int grid = GridFor(total);
     switch (DtypeSize(dt)) {
@@ -443,6 +466,15 @@ void LaunchPack(const void* src, void* dst, const PackBlockDesc& d, DataType dt,
 
 void LaunchUnpack(const void* src, void* dst, const PackBlockDesc& d, DataType dt,
                   hipStream_t stream) {
+    const size_t es = DtypeSize(dt);
+    PackBlockDesc d16;
+    if (Pack16(d, es, &d16)) {
+        const size_t total16 = d16.mb_count * d16.fm_count * d16.fm_size;
+        PackKernel<uint4, false><<<dim3(GridFor(total16)), dim3(kBlock), 0, stream>>>(
+            static_cast<const uint4*>(src), static_cast<uint4*>(dst), d16);
+        HIP_CHECK(hipGetLastError());
+        return;
+    }
     const size_t total = d.mb_count * d.fm_count * d.fm_size;
     const int grid = GridFor(total);
     switch (DtypeSize(dt)) {
