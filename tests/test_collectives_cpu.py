@@ -57,3 +57,23 @@ def test_compat_shim(world):
 
 def test_configure_tenants():
     run_ranks("configure_tenants", 4)
+
+
+def test_fault_peer_death():
+    # rank 1 exits deliberately; rank 0 must fail fast, not hang
+    import subprocess, sys, os
+    from tests.mp import REPO, free_port
+    port = free_port()
+    procs = []
+    for r in range(2):
+        env = dict(os.environ, RANK=str(r), WORLD_SIZE="2",
+                   MASTER_ADDR="127.0.0.1", MLSL_PORT=str(port),
+                   MLSL_TRANSPORT="tcp", MLSL_TIMEOUT="10")
+        procs.append(subprocess.Popen(
+            [sys.executable, "-m", "tests.workers", "fault_peer_death"],
+            env=env, cwd=REPO, stdout=subprocess.PIPE,
+            stderr=subprocess.STDOUT, text=True))
+    out0, _ = procs[0].communicate(timeout=60)
+    procs[1].communicate(timeout=10)
+    assert procs[0].returncode == 0, out0
+    assert "expected failure" in out0

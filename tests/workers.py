@@ -332,6 +332,30 @@ def configure_tenants():
 WORKERS["configure_tenants"] = configure_tenants
 
 
+def fault_peer_death():
+    """Failure detection (SURVEY.md 5.3 — absent in the reference): when a
+    peer dies mid-collective the survivor gets a failed request (transport
+    EOF or MLSL_TIMEOUT watchdog), never an infinite hang."""
+    import mlsl_amd as mx
+    from mlsl_amd import MlslError
+    rank = int(os.environ["RANK"])
+    mx.init()
+    d = mx.Distribution(mx.world_size(), 1)
+    if rank == 1:
+        os._exit(0)  # die without participating
+    a = np.ones(200000, dtype=np.float32)
+    out = np.zeros_like(a)
+    try:
+        mx.wait(d.all_reduce(a, out, a.size, op="sum", group="data"))
+    except MlslError as e:
+        print(f"OK got expected failure: {e}", flush=True)
+        os._exit(0)
+    raise AssertionError("allreduce with dead peer did not fail")
+
+
+WORKERS["fault_peer_death"] = fault_peer_death
+
+
 def main():
     name = sys.argv[1]
     fn = WORKERS.get(name)
