@@ -7,3 +7,8 @@
 - shard math helpers for distributed update (ZeRO-1 ancestor).
 """
 from .bucketer import GradBucketer  # noqa: F401
+
+try:  # torch is optional at import time
+    from .ddp import DistributedData  # noqa: F401
+except ImportError:  # pragma: no cover
+    pass

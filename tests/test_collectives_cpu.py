@@ -77,3 +77,8 @@ def test_fault_peer_death():
     procs[1].communicate(timeout=10)
     assert procs[0].returncode == 0, out0
     assert "expected failure" in out0
+
+
+@pytest.mark.parametrize("world", [2])
+def test_ddp_wrapper(world):
+    run_ranks("ddp_wrapper", world, timeout=300)

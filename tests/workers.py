@@ -356,6 +356,45 @@ def fault_peer_death():
 WORKERS["fault_peer_death"] = fault_peer_death
 
 
+def ddp_wrapper():
+    """DistributedData (torch hook-driven bucketed allreduce) matches
+    manual gradient averaging at world 2 on CPU tensors."""
+    import torch
+    import mlsl_amd as mx
+    from mlsl_amd.parallel import DistributedData
+
+    torch.manual_seed(42)  # same init on every rank pre-broadcast
+    mx.init()
+    rank, size = mx.rank(), mx.world_size()
+
+    model = torch.nn.Sequential(
+        torch.nn.Linear(32, 64), torch.nn.ReLU(), torch.nn.Linear(64, 8))
+    ref = torch.nn.Sequential(
+        torch.nn.Linear(32, 64), torch.nn.ReLU(), torch.nn.Linear(64, 8))
+    ref.load_state_dict(model.state_dict())
+
+    dd = DistributedData(model, bucket_mb=1)
+
+    for it in range(3):
+        xs = [torch.randn(16, 32, generator=torch.Generator().manual_seed(100 + it * size + r))
+              for r in range(size)]
+        # distributed: each rank its own shard
+        model.zero_grad()
+        model(xs[rank]).sum().backward()
+        dd.finish_gradients()
+        # reference: average of per-shard grads on the full data
+        ref.zero_grad()
+        for x in xs:
+            ref(x).sum().backward()
+        for p, q in zip(model.parameters(), ref.parameters()):
+            want = q.grad / size
+            assert torch.allclose(p.grad, want, atol=1e-5), f"it={it} grad mismatch"
+    mx.finalize()
+
+
+WORKERS["ddp_wrapper"] = ddp_wrapper
+
+
 def main():
     name = sys.argv[1]
     fn = WORKERS.get(name)
