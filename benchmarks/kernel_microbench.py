@@ -35,6 +35,9 @@ def main():
     dt = timed(lambda: ops.reduce_(a, b, n))
     res["reduce_f32_256MiB_TBps"] = round(3 * 4 * n / dt / 1e12, 3)
 
+    dt = timed(lambda: ops.reduce_nt(a, b, n))
+    res["reduce_f32_nt_TBps"] = round(3 * 4 * n / dt / 1e12, 3)
+
     abf = torch.randn(n, device="cuda", dtype=torch.bfloat16)
     bbf = torch.randn(n, device="cuda", dtype=torch.bfloat16)
     dt = timed(lambda: ops.reduce_(abf, bbf, n, dtype="bf16"))

@@ -15,6 +15,9 @@ void LaunchReduce(void* dst, const void* src, size_t count, DataType dt,
 
 // dst = a + b elementwise into a third buffer (out-of-place variant used by
 // fused pipelines).
+// f32 SUM with nontemporal loads/stores (A/B benchmarking).
+void LaunchReduceNT(void* dst, const void* src, size_t count, hipStream_t stream);
+
 void LaunchReduceOut(void* dst, const void* a, const void* b, size_t count,
                      DataType dt, ReduceOp op, hipStream_t stream);
 

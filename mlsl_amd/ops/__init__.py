@@ -22,6 +22,7 @@ def _lib():
         L.mlsl_hip_device_count.argtypes = [c.POINTER(c.c_int)]
         L.mlsl_hip_synchronize.argtypes = []
         L.mlsl_hip_reduce.argtypes = [c.c_void_p, c.c_void_p, c.c_size_t, c.c_int, c.c_int]
+        L.mlsl_hip_reduce_nt.argtypes = [c.c_void_p, c.c_void_p, c.c_size_t]
         L.mlsl_hip_quantize.argtypes = [c.c_void_p, c.c_void_p, c.c_void_p, c.c_size_t,
                                         c.c_size_t, c.c_int, c.c_int]
         L.mlsl_hip_dequantize.argtypes = [c.c_void_p, c.c_void_p, c.c_size_t,
@@ -31,6 +32,7 @@ def _lib():
         L.mlsl_hip_pack.argtypes = ptypes
         L.mlsl_hip_unpack.argtypes = ptypes
         for n in ("mlsl_hip_device_count", "mlsl_hip_synchronize", "mlsl_hip_reduce",
+                  "mlsl_hip_reduce_nt",
                   "mlsl_hip_quantize", "mlsl_hip_dequantize", "mlsl_hip_quant_accum",
                   "mlsl_hip_pack", "mlsl_hip_unpack"):
             getattr(L, n).restype = c.c_int
@@ -54,6 +56,13 @@ def reduce_(dst, src, count, dtype=None, op="sum"):
     sp, d2 = _as_ptr_dtype(src)
     dt = dtype or d1 or d2
     check(_lib().mlsl_hip_reduce(dp, sp, count, DTYPE[dt], REDOP[op]))
+
+
+def reduce_nt(dst, src, count):
+    """f32 sum with nontemporal loads/stores (benchmark variant)."""
+    dp, _ = _as_ptr_dtype(dst)
+    sp, _ = _as_ptr_dtype(src)
+    check(_lib().mlsl_hip_reduce_nt(dp, sp, count))
 
 
 def wire_bytes(count, block=256):
