@@ -83,21 +83,21 @@ __global__ void ReduceF32Kernel(float* __restrict__ dst,
 
 // NT variant: nontemporal loads/stores (bypass L2 retention) — measured
 // A/B against the default policy on pure streaming reductions.
+using float4_ev = __attribute__((ext_vector_type(4))) float;
+
 template <ReduceOp OP>
 __global__ void ReduceF32NTKernel(float* __restrict__ dst,
                                   const float* __restrict__ src, size_t n) {
     const size_t tid = blockIdx.x * blockDim.x + threadIdx.x;
     const size_t stride = gridDim.x * blockDim.x;
     const size_t n4 = n / 4;
-    const float4* s4 = reinterpret_cast<const float4*>(src);
-    float4* d4 = reinterpret_cast<float4*>(dst);
+    const float4_ev* s4 = reinterpret_cast<const float4_ev*>(src);
+    float4_ev* d4 = reinterpret_cast<float4_ev*>(dst);
     for (size_t i = tid; i < n4; i += stride) {
-        float4 a = __builtin_nontemporal_load(d4 + i);
-        float4 b = __builtin_nontemporal_load(s4 + i);
-        a.x = Apply<float, OP>(a.x, b.x);
-        a.y = Apply<float, OP>(a.y, b.y);
-        a.z = Apply<float, OP>(a.z, b.z);
-        a.w = Apply<float, OP>(a.w, b.w);
+        float4_ev a = __builtin_nontemporal_load(d4 + i);
+        float4_ev b = __builtin_nontemporal_load(s4 + i);
+#pragma unroll
+        for (int j = 0; j < 4; ++j) a[j] = Apply<float, OP>(a[j], b[j]);
         __builtin_nontemporal_store(a, d4 + i);
     }
     for (size_t j = n4 * 4 + tid; j < n; j += stride)
