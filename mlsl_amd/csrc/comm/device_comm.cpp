@@ -81,6 +81,11 @@ ncclRedOp_t ToNcclOp(ReduceOp op) {
 struct GroupComms {
     std::vector<ncclComm_t> comms;
     std::vector<hipStream_t> streams;
+    // High-priority lane for MLSL_MSG_PRIORITY traffic (fresh urgent
+    // gradients overtake queued bulk work — stream-priority analog of the
+    // reference's newest-first server scan).
+    hipStream_t prio_stream = nullptr;
+    ncclComm_t prio_comm = nullptr;
 };
 
 class HipRuntime;
@@ -96,6 +101,8 @@ class HipRuntime : public DeviceRuntime {
         for (auto& kv : group_comms_) {
             for (auto c : kv.second.comms) ncclCommDestroy(c);
             for (auto s : kv.second.streams) (void)hipStreamDestroy(s);
+            if (kv.second.prio_comm) ncclCommDestroy(kv.second.prio_comm);
+            if (kv.second.prio_stream) (void)hipStreamDestroy(kv.second.prio_stream);
         }
         g_runtime = nullptr;
     }
@@ -145,6 +152,18 @@ class HipRuntime : public DeviceRuntime {
                 gc.comms.push_back(comm);
                 gc.streams.push_back(s);
             }
+        }
+        if (GlobalConfig().msg_priority && g->IsMember() && g->Size() > 1) {
+            ncclUniqueId id{};
+            if (g->MyIdx() == 0) NCCL_CHECK(ncclGetUniqueId(&id));
+            std::vector<ncclUniqueId> all(static_cast<size_t>(ctx.Size()));
+            ctx.Boot()->Allgather(&id, sizeof(ncclUniqueId), all.data());
+            ncclUniqueId gid = all[static_cast<size_t>(g->WorldRank(0))];
+            NCCL_CHECK(ncclCommInitRank(&gc.prio_comm, g->Size(), gid, g->MyIdx()));
+            int lo = 0, hi = 0;
+            HIP_CHECKD(hipDeviceGetStreamPriorityRange(&lo, &hi));
+            HIP_CHECKD(hipStreamCreateWithPriority(&gc.prio_stream,
+                                                   hipStreamNonBlocking, hi));
         }
         group_comms_.emplace(g->Uid(), std::move(gc));
         MLSL_LOG(DEBUG, "device comms ready for group uid=%d size=%d channels=%zu",
@@ -552,11 +571,20 @@ bool DeviceAdvanceRequest(CommRequest* req, DeviceReqState& st) {
             HIP_CHECKD(hipEventCreateWithFlags(&e, hipEventDisableTiming));
             st.events.push_back(e);
         }
+        // Priority lane: urgent traffic preempts queued bulk work.
+        const Config& cfgp = GlobalConfig();
+        const bool prio = cfgp.msg_priority && gc.prio_comm &&
+                          req->MessageBytes() >= cfgp.msg_priority_threshold;
+        if (prio) HIP_CHECKD(hipStreamWaitEvent(gc.prio_stream, st.dep_event, 0));
         // tmp partitioning per chunk
         size_t tmp_off = 0;
         const size_t es = DtypeSize(req->Dtype());
         for (auto& ce : chunks) {
             const size_t ch = ce.chunk_idx % nch;
+            ncclComm_t comm_ = prio ? gc.prio_comm : gc.comms[ch];
+            hipStream_t strm_ = prio ? gc.prio_stream : gc.streams[ch];
+            (void)comm_;
+            (void)strm_;
             uint8_t* tbase = static_cast<uint8_t*>(st.tmp_dev) + tmp_off;
             if (compressed) {
                 // quantize -> compressed-domain ring -> dequantize, all on
@@ -567,24 +595,24 @@ bool DeviceAdvanceRequest(CommRequest* req, DeviceReqState& st) {
                 uint8_t* err = scratch + ce.sch.tmp_bytes;
                 const size_t blk = req->QParams().block_elems;
                 LaunchQuantize(req->SendBuf(), err, wire, req->Spec().count, blk,
-                               req->Dtype(), true, gc.streams[ch]);
-                IssueSchedule(req, ce, gc.comms[ch], gc.streams[ch], wire, wire,
-                              scratch);
+                               req->Dtype(), true, strm_);
+                IssueSchedule(req, ce, comm_, strm_, wire, wire, scratch);
                 LaunchDequantize(wire, req->RecvBuf(), req->Spec().count, blk,
-                                 req->Dtype(), gc.streams[ch]);
+                                 req->Dtype(), strm_);
                 tmp_off += wire_b + ce.sch.tmp_bytes + req->Spec().count * es;
             } else if (use_schedule) {
-                IssueSchedule(req, ce, gc.comms[ch], gc.streams[ch],
+                IssueSchedule(req, ce, comm_, strm_,
                               req->SendBuf() + ce.elem_off * es,
                               req->RecvBuf() + ce.elem_off * es, tbase);
                 tmp_off += ce.sch.tmp_bytes;
             } else {
-                IssueFused(req, ce, gc.comms[ch], gc.streams[ch], st, tmp_off);
+                IssueFused(req, ce, comm_, strm_, st, tmp_off);
                 tmp_off += ce.sch.tmp_bytes;
             }
         }
         for (size_t ch = 0; ch < used; ++ch)
-            HIP_CHECKD(hipEventRecord(st.events[ch], gc.streams[ch]));
+            HIP_CHECKD(hipEventRecord(st.events[ch],
+                                      prio ? gc.prio_stream : gc.streams[ch]));
         if (st.recv_staged) {
             // join all channels on stream 0, then stage the result out
             for (size_t ch = 0; ch < used; ++ch)

@@ -335,6 +335,8 @@ void CommRequest::Start(const void* sbuf, void* rbuf) {
                    "collective buffer not from Environment::Alloc "
                    "(MLSL_CHECK_POINTERS=1)");
     }
+    MLSL_LOG(TRACE, "start %s count=%zu dtype=%s group=%d flow=%u",
+             CollOpName(spec_.op), spec_.count, DtypeName(dtype_), group_->Uid(), flow_);
     if (ctx.DeviceMode()) {
         // Device path: issue inline — RCCL/kernel enqueues are non-blocking
         // and the STREAMS are the progress engine; Wait/Test poll the
@@ -345,6 +347,9 @@ void CommRequest::Start(const void* sbuf, void* rbuf) {
         if (AdvanceDevice()) MarkDone();
         return;
     }
+    MLSL_LOG(TRACE, "start %s count=%zu dtype=%s group=%d flow=%u sbuf=%p rbuf=%p",
+             CollOpName(spec_.op), spec_.count, DtypeName(dtype_), group_->Uid(),
+             flow_, static_cast<const void*>(sbuf_), static_cast<void*>(rbuf_));
     state_.store(ReqState::QUEUED, std::memory_order_release);
     ctx.GetEngine()->Submit(this);
 }
