@@ -42,6 +42,7 @@ namespace mlsl {
     } while (0)
 
 DeviceReqState::~DeviceReqState() {
+    if (graph_exec) (void)hipGraphExecDestroy(graph_exec);
     for (hipEvent_t e : events)
         if (e) (void)hipEventDestroy(e);
     if (dep_event) (void)hipEventDestroy(dep_event);
@@ -476,68 +477,125 @@ void IssueSchedule(CommRequest* req, ChunkExec& ce, ncclComm_t comm, hipStream_t
 
 }  // namespace
 
+namespace {
+
+// Capture the eager issue body (already enqueued between Begin/End) is not
+// possible retroactively; instead IssueEager is factored so it can run (a)
+// directly or (b) inside a stream capture. Returns the stream carrying the
+// final dependency.
+bool GraphEligible(CommRequest* req, GroupComms& gc) {
+    // Single effective stream only: every chunk must land on channel 0 and
+    // no host staging (H2D/D2H of *pageable* memory is not capturable).
+    if (req->Compressed()) return true;  // single stream by construction
+    if (gc.comms.size() > 1 && req->Chunks().size() > 1) return false;
+    return true;
+}
+
+bool AllEventsDone(DeviceReqState& st) {
+    for (hipEvent_t e : st.events) {
+        hipError_t q = hipEventQuery(e);
+        if (q == hipErrorNotReady) return false;
+        if (q != hipSuccess) MLSL_THROW(std::string("HIP event error: ") +
+                                        hipGetErrorString(q));
+    }
+    return true;
+}
+
+void EnsureEvents(DeviceReqState& st, size_t n) {
+    while (st.events.size() < n) {
+        hipEvent_t e;
+        HIP_CHECKD(hipEventCreateWithFlags(&e, hipEventDisableTiming));
+        st.events.push_back(e);
+    }
+}
+
+}  // namespace
+
 bool DeviceAdvanceRequest(CommRequest* req, DeviceReqState& st) {
     auto* rt = static_cast<HipRuntime*>(Context::Get().Device());
-    if (!st.issued) {
-        GroupComms& gc = rt->For(req->Group());
-        if (gc.comms.empty() && gc.streams.empty()) {
-            hipStream_t s0;
-            HIP_CHECKD(hipStreamCreateWithFlags(&s0, hipStreamNonBlocking));
-            gc.streams.push_back(s0);
-        }
-        // Order after the caller's compute stream: the producer kernels
-        // (e.g. torch backward on the default stream) must land before the
-        // collective reads the buffers.
-        if (!st.dep_event)
-            HIP_CHECKD(hipEventCreateWithFlags(&st.dep_event, hipEventDisableTiming));
-        HIP_CHECKD(hipEventRecord(st.dep_event,
-                                  static_cast<hipStream_t>(rt->ComputeStream())));
-        for (hipStream_t cs : gc.streams)
-            HIP_CHECKD(hipStreamWaitEvent(cs, st.dep_event, 0));
+    if (st.issued) return AllEventsDone(st);
 
-        // Host-buffer staging (ReplaceIn analog): stage unregistered host
-        // buffers through persistent HBM so RCCL/kernels see device memory.
-        size_t send_b = 0, recv_b = 0;
-        IoBytes(req, &send_b, &recv_b);
-        const bool s_host = send_b > 0 && !IsDevicePtr(req->UserSendBuf());
-        const bool r_host = recv_b > 0 && !IsDevicePtr(req->UserRecvBuf());
-        st.recv_staged = r_host;
-        if (s_host) {
-            if (!st.stage_send || st.stage_send_bytes < send_b) {
-                if (st.stage_send) (void)hipFree(st.stage_send);
-                HIP_CHECKD(hipMalloc(&st.stage_send, send_b));
-                st.stage_send_bytes = send_b;
-            }
-            HIP_CHECKD(hipMemcpyAsync(st.stage_send, req->UserSendBuf(), send_b,
-                                      hipMemcpyHostToDevice, gc.streams[0]));
-            if (gc.streams.size() > 1) {
-                HIP_CHECKD(hipEventRecord(st.dep_event, gc.streams[0]));
-                for (size_t i = 1; i < gc.streams.size(); ++i)
-                    HIP_CHECKD(hipStreamWaitEvent(gc.streams[i], st.dep_event, 0));
-            }
-        }
-        if (r_host && (!st.stage_recv || st.stage_recv_bytes < recv_b)) {
-            if (st.stage_recv) (void)hipFree(st.stage_recv);
-            HIP_CHECKD(hipMalloc(&st.stage_recv, recv_b));
-            st.stage_recv_bytes = recv_b;
-        }
-        req->SetDeviceBuffers(s_host ? static_cast<const uint8_t*>(st.stage_send) : nullptr,
-                              r_host ? static_cast<uint8_t*>(st.stage_recv) : nullptr);
-        auto& chunks = req->Chunks();
-        const bool use_schedule = req->UsesDeviceSchedule();
+    GroupComms& gc = rt->For(req->Group());
+    if (gc.comms.empty() && gc.streams.empty()) {
+        hipStream_t s0;
+        HIP_CHECKD(hipStreamCreateWithFlags(&s0, hipStreamNonBlocking));
+        gc.streams.push_back(s0);
+    }
+    // Order after the caller's compute stream: the producer kernels
+    // (e.g. torch backward on the default stream) must land before the
+    // collective reads the buffers.
+    if (!st.dep_event)
+        HIP_CHECKD(hipEventCreateWithFlags(&st.dep_event, hipEventDisableTiming));
+    HIP_CHECKD(hipEventRecord(st.dep_event,
+                              static_cast<hipStream_t>(rt->ComputeStream())));
+    for (hipStream_t cs : gc.streams)
+        HIP_CHECKD(hipStreamWaitEvent(cs, st.dep_event, 0));
 
-        const bool compressed = req->Compressed();
-        // Single-rank groups: local copies only, still through a (persistent)
-        // stream so completion semantics are uniform.
+    // hipGraph replay (MLSL_USE_GRAPHS): same request + same buffers ->
+    // launch the captured graph instead of re-enqueueing.
+    if (st.graph_exec && req->SendBuf() == st.captured_sbuf &&
+        req->RecvBuf() == st.captured_rbuf) {
+        HIP_CHECKD(hipGraphLaunch(st.graph_exec, gc.streams[0]));
+        EnsureEvents(st, 1);
+        HIP_CHECKD(hipEventRecord(st.events[0], gc.streams[0]));
+        st.issued = true;
+        return AllEventsDone(st);
+    }
+
+    // Host-buffer staging (ReplaceIn analog): stage unregistered host
+    // buffers through persistent HBM so RCCL/kernels see device memory.
+    size_t send_b = 0, recv_b = 0;
+    IoBytes(req, &send_b, &recv_b);
+    const bool s_host = send_b > 0 && !IsDevicePtr(req->UserSendBuf());
+    const bool r_host = recv_b > 0 && !IsDevicePtr(req->UserRecvBuf());
+    st.recv_staged = r_host;
+    if (s_host) {
+        if (!st.stage_send || st.stage_send_bytes < send_b) {
+            if (st.stage_send) (void)hipFree(st.stage_send);
+            HIP_CHECKD(hipMalloc(&st.stage_send, send_b));
+            st.stage_send_bytes = send_b;
+        }
+        HIP_CHECKD(hipMemcpyAsync(st.stage_send, req->UserSendBuf(), send_b,
+                                  hipMemcpyHostToDevice, gc.streams[0]));
+        if (gc.streams.size() > 1) {
+            HIP_CHECKD(hipEventRecord(st.dep_event, gc.streams[0]));
+            for (size_t i = 1; i < gc.streams.size(); ++i)
+                HIP_CHECKD(hipStreamWaitEvent(gc.streams[i], st.dep_event, 0));
+        }
+    }
+    if (r_host && (!st.stage_recv || st.stage_recv_bytes < recv_b)) {
+        if (st.stage_recv) (void)hipFree(st.stage_recv);
+        HIP_CHECKD(hipMalloc(&st.stage_recv, recv_b));
+        st.stage_recv_bytes = recv_b;
+    }
+    req->SetDeviceBuffers(s_host ? static_cast<const uint8_t*>(st.stage_send) : nullptr,
+                          r_host ? static_cast<uint8_t*>(st.stage_recv) : nullptr);
+    auto& chunks = req->Chunks();
+    const bool use_schedule = req->UsesDeviceSchedule();
+    const bool compressed = req->Compressed();
+    const Config& cfg = GlobalConfig();
+    const bool prio = cfg.msg_priority && gc.prio_comm &&
+                      req->MessageBytes() >= cfg.msg_priority_threshold;
+    if (prio) HIP_CHECKD(hipStreamWaitEvent(gc.prio_stream, st.dep_event, 0));
+
+    // Graph capture only when everything lands on ONE capturable stream:
+    // no pageable staging, no priority lane, single channel.
+    const bool try_capture = cfg.use_graphs && !st.graph_failed && !s_host &&
+                             !r_host && !prio && GraphEligible(req, gc);
+
+    const size_t es = DtypeSize(req->Dtype());
+    const size_t nch = gc.comms.empty() ? 1 : gc.comms.size();
+    const size_t used = gc.comms.empty() ? 1 : std::min(chunks.size(), nch);
+
+    auto issue_all = [&]() {
         if (gc.comms.empty()) {
             hipStream_t s0 = gc.streams[0];
-            const size_t es = DtypeSize(req->Dtype());
             for (auto& ce : chunks) {
                 const uint8_t* sbase = req->SendBuf() + ce.elem_off * es;
                 uint8_t* rbase = req->RecvBuf() + ce.elem_off * es;
                 if (compressed) {
-                    // quantize -> dequantize so n=1 keeps quantized-allreduce
-                    // semantics (and exercises error feedback)
+                    // quantize -> dequantize keeps quantized-allreduce
+                    // semantics (and error feedback) at n=1
                     uint8_t* wire = static_cast<uint8_t*>(st.tmp_dev);
                     uint8_t* err = wire + ce.sch.result.bytes + ce.sch.tmp_bytes;
                     const size_t blk = req->QParams().block_elems;
@@ -547,48 +605,24 @@ bool DeviceAdvanceRequest(CommRequest* req, DeviceReqState& st) {
                                      req->Dtype(), s0);
                 } else if (ce.sch.result.bytes && sbase != rbase) {
                     HIP_CHECKD(hipMemcpyAsync(rbase + ce.sch.result.off, sbase,
-                                              ce.sch.result.bytes, hipMemcpyDefault, s0));
+                                              ce.sch.result.bytes, hipMemcpyDefault,
+                                              s0));
                 }
             }
             if (st.recv_staged)
                 HIP_CHECKD(hipMemcpyAsync(req->UserRecvBuf(), st.stage_recv, recv_b,
                                           hipMemcpyDeviceToHost, s0));
-            if (st.events.empty()) {
-                hipEvent_t e;
-                HIP_CHECKD(hipEventCreateWithFlags(&e, hipEventDisableTiming));
-                st.events.push_back(e);
-            }
-            HIP_CHECKD(hipEventRecord(st.events[0], s0));
-            st.issued = true;
-            return hipEventQuery(st.events[0]) == hipSuccess;
+            return;
         }
-
-        // Events per used channel.
-        const size_t nch = gc.comms.size();
-        const size_t used = std::min(chunks.size(), nch);
-        while (st.events.size() < used) {
-            hipEvent_t e;
-            HIP_CHECKD(hipEventCreateWithFlags(&e, hipEventDisableTiming));
-            st.events.push_back(e);
-        }
-        // Priority lane: urgent traffic preempts queued bulk work.
-        const Config& cfgp = GlobalConfig();
-        const bool prio = cfgp.msg_priority && gc.prio_comm &&
-                          req->MessageBytes() >= cfgp.msg_priority_threshold;
-        if (prio) HIP_CHECKD(hipStreamWaitEvent(gc.prio_stream, st.dep_event, 0));
-        // tmp partitioning per chunk
         size_t tmp_off = 0;
-        const size_t es = DtypeSize(req->Dtype());
         for (auto& ce : chunks) {
             const size_t ch = ce.chunk_idx % nch;
             ncclComm_t comm_ = prio ? gc.prio_comm : gc.comms[ch];
             hipStream_t strm_ = prio ? gc.prio_stream : gc.streams[ch];
-            (void)comm_;
-            (void)strm_;
             uint8_t* tbase = static_cast<uint8_t*>(st.tmp_dev) + tmp_off;
             if (compressed) {
-                // quantize -> compressed-domain ring -> dequantize, all on
-                // one stream (driver config 5: int8 allreduce of bf16 grads)
+                // quantize -> compressed-domain ring -> dequantize (driver
+                // config 5: int8 allreduce of bf16/f32 grads)
                 const size_t wire_b = ce.sch.result.bytes;
                 uint8_t* wire = tbase;
                 uint8_t* scratch = tbase + wire_b;
@@ -610,6 +644,51 @@ bool DeviceAdvanceRequest(CommRequest* req, DeviceReqState& st) {
                 tmp_off += ce.sch.tmp_bytes;
             }
         }
+    };
+
+    if (try_capture) {
+        bool body_ok = true;
+        hipError_t ce0 = hipStreamBeginCapture(gc.streams[0],
+                                               hipStreamCaptureModeThreadLocal);
+        if (ce0 == hipSuccess) {
+            try {
+                issue_all();
+            } catch (const std::exception&) {
+                body_ok = false;
+            }
+            hipGraph_t graph = nullptr;
+            hipError_t ce1 = hipStreamEndCapture(gc.streams[0], &graph);
+            if (body_ok && ce1 == hipSuccess && graph &&
+                hipGraphInstantiate(&st.graph_exec, graph, nullptr, nullptr, 0) ==
+                    hipSuccess) {
+                (void)hipGraphDestroy(graph);
+                st.captured_sbuf = req->SendBuf();
+                st.captured_rbuf = req->RecvBuf();
+                HIP_CHECKD(hipGraphLaunch(st.graph_exec, gc.streams[0]));
+            } else {
+                // capture produced nothing executable -> run eagerly
+                (void)hipGetLastError();
+                if (graph) (void)hipGraphDestroy(graph);
+                st.graph_exec = nullptr;
+                st.graph_failed = true;
+                issue_all();
+            }
+        } else {
+            (void)hipGetLastError();
+            st.graph_failed = true;
+            issue_all();
+        }
+    } else {
+        issue_all();
+    }
+
+    // Completion events (outside any graph). Graph path + single-channel
+    // path complete on stream 0; multi-channel records one per channel.
+    if (st.graph_exec || gc.comms.empty()) {
+        EnsureEvents(st, 1);
+        HIP_CHECKD(hipEventRecord(st.events[0], gc.streams[0]));
+    } else {
+        EnsureEvents(st, used);
         for (size_t ch = 0; ch < used; ++ch)
             HIP_CHECKD(hipEventRecord(st.events[ch],
                                       prio ? gc.prio_stream : gc.streams[ch]));
@@ -619,21 +698,12 @@ bool DeviceAdvanceRequest(CommRequest* req, DeviceReqState& st) {
                 HIP_CHECKD(hipStreamWaitEvent(gc.streams[0], st.events[ch], 0));
             HIP_CHECKD(hipMemcpyAsync(req->UserRecvBuf(), st.stage_recv, recv_b,
                                       hipMemcpyDeviceToHost, gc.streams[0]));
-            while (st.events.size() < used + 1) {
-                hipEvent_t e;
-                HIP_CHECKD(hipEventCreateWithFlags(&e, hipEventDisableTiming));
-                st.events.push_back(e);
-            }
+            EnsureEvents(st, used + 1);
             HIP_CHECKD(hipEventRecord(st.events[used], gc.streams[0]));
         }
-        st.issued = true;
     }
-    for (hipEvent_t e : st.events) {
-        hipError_t q = hipEventQuery(e);
-        if (q == hipErrorNotReady) return false;
-        HIP_CHECKD(q);
-    }
-    return true;
+    st.issued = true;
+    return AllEventsDone(st);
 }
 
 }  // namespace mlsl

@@ -26,6 +26,12 @@ struct DeviceReqState {
     void* stage_recv = nullptr;
     size_t stage_send_bytes = 0, stage_recv_bytes = 0;
     bool recv_staged = false;
+    // hipGraph replay (MLSL_USE_GRAPHS): the issue sequence captured once,
+    // replayed on subsequent Starts with the same buffers.
+    hipGraphExec_t graph_exec = nullptr;
+    const void* captured_sbuf = nullptr;
+    void* captured_rbuf = nullptr;
+    bool graph_failed = false;   // capture failed once -> stay eager
     ~DeviceReqState();
 };
 

@@ -49,6 +49,7 @@ Config Config::FromEnv() {
     c.check_pointers = EnvBool("MLSL_CHECK_POINTERS", false);
     if (const char* e = std::getenv("MLSL_TRANSPORT")) c.transport = e;
     c.timeout_sec = static_cast<int>(EnvSize("MLSL_TIMEOUT", 300));
+    c.use_graphs = EnvBool("MLSL_USE_GRAPHS", false);
     return c;
 }
 

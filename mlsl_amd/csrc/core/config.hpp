@@ -39,6 +39,10 @@ struct Config {
     bool check_pointers = false;   // MLSL_CHECK_POINTERS: validate collective bufs
     std::string transport = "auto";  // MLSL_TRANSPORT=auto|tcp|rccl
     int timeout_sec = 300;         // MLSL_TIMEOUT: bootstrap/collective timeout
+    bool use_graphs = false;       // MLSL_USE_GRAPHS: hipGraph replay of
+                                   // persistent device requests (single
+                                   // channel; eager fallback on capture
+                                   // failure)
 
     static Config FromEnv();
     void Dump() const;  // rank-0 dump of effective values (ref comm_ep.cpp:1701)

@@ -194,3 +194,31 @@ class TestHostStaging:
             assert np.allclose(out, a)
         finally:
             mx.finalize()
+
+
+@requires_gpu
+class TestGraphReplay:
+    def test_persistent_graph_replay(self):
+        """MLSL_USE_GRAPHS=1: a persistent request is captured once and
+        replayed; results stay correct as buffer contents change."""
+        import os
+        import mlsl_amd as mx
+        os.environ["MLSL_USE_GRAPHS"] = "1"
+        try:
+            mx.init()
+            d = mx.Distribution(1, 1)
+            n = 1 << 16
+            x = torch.zeros(n, device="cuda")
+            y = torch.zeros(n, device="cuda")
+            preq = mx.PersistentRequest(d, "all_reduce", n, dtype="f32",
+                                        op="sum", group="data")
+            for it in range(6):
+                x.fill_(float(it + 1))
+                preq.start(x, y)
+                preq.wait()
+                torch.cuda.synchronize()
+                assert torch.all(y == it + 1), f"replay iter {it}"
+            preq.destroy()
+            mx.finalize()
+        finally:
+            os.environ.pop("MLSL_USE_GRAPHS", None)
