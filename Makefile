@@ -19,6 +19,7 @@ CSRC := \
     mlsl_amd/csrc/core/signals.cpp \
     mlsl_amd/csrc/comm/schedule.cpp \
     mlsl_amd/csrc/comm/quant.cpp \
+    mlsl_amd/csrc/comm/device_pool.cpp \
     mlsl_amd/csrc/comm/bootstrap.cpp \
     mlsl_amd/csrc/comm/mesh.cpp \
     mlsl_amd/csrc/comm/group.cpp \

@@ -18,6 +18,7 @@
 
 #include "../core/config.hpp"
 #include "../core/log.hpp"
+#include "device_pool.hpp"
 #include "../hip/kernels.hpp"
 #include "context.hpp"
 #include "quant.hpp"
@@ -46,9 +47,17 @@ DeviceReqState::~DeviceReqState() {
     for (hipEvent_t e : events)
         if (e) (void)hipEventDestroy(e);
     if (dep_event) (void)hipEventDestroy(dep_event);
-    if (tmp_dev) (void)hipFree(tmp_dev);
-    if (stage_send) (void)hipFree(stage_send);
-    if (stage_recv) (void)hipFree(stage_recv);
+    // Return scratch to the HBM pool while the runtime is alive; fall back
+    // to hipFree during teardown.
+    DeviceRuntime* rt = Context::Initialized() ? Context::Get().Device() : nullptr;
+    auto rel = [&](void* p) {
+        if (!p) return;
+        if (rt) rt->FreeDevice(p);
+        else (void)hipFree(p);
+    };
+    rel(tmp_dev);
+    rel(stage_send);
+    rel(stage_recv);
 }
 
 namespace {
@@ -96,6 +105,10 @@ class HipRuntime : public DeviceRuntime {
   public:
     HipRuntime(int device_id) : device_id_(device_id) {
         HIP_CHECKD(hipSetDevice(device_id_));
+        // HBM pool: cache up to MLSL_HEAP_SIZE_MB of freed blocks
+        // (0 = unlimited cache; 288 GB per GPU makes caching the default).
+        const size_t heap_mb = GlobalConfig().heap_mb;
+        pool_ = std::make_unique<DevicePool>(heap_mb * 1024 * 1024);
     }
 
     ~HipRuntime() override {
@@ -115,13 +128,9 @@ class HipRuntime : public DeviceRuntime {
     }
     void* ComputeStream() const override { return compute_stream_; }
 
-    void* AllocDevice(size_t bytes) override {
-        void* p = nullptr;
-        HIP_CHECKD(hipMalloc(&p, bytes));
-        return p;
-    }
+    void* AllocDevice(size_t bytes) override { return pool_->Alloc(bytes); }
 
-    void FreeDevice(void* p) override { (void)hipFree(p); }
+    void FreeDevice(void* p) override { pool_->Free(p); }
 
     void Synchronize() override { HIP_CHECKD(hipDeviceSynchronize()); }
 
@@ -185,6 +194,7 @@ class HipRuntime : public DeviceRuntime {
   private:
     int device_id_;
     hipStream_t compute_stream_ = nullptr;  // null = legacy default stream
+    std::unique_ptr<DevicePool> pool_;
     std::unordered_map<int, GroupComms> group_comms_;
 };
 
@@ -219,7 +229,7 @@ void DeviceSetupRequest(CommRequest* req, DeviceReqState& st) {
     if (req->Spec().op == CollOp::BARRIER) tmp = std::max<size_t>(tmp, 16);
     st.tmp_bytes = tmp;
     if (tmp) {
-        HIP_CHECKD(hipMalloc(&st.tmp_dev, tmp));
+        st.tmp_dev = Context::Get().Device()->AllocDevice(tmp);
         // zero once: error-feedback residuals must start clean
         HIP_CHECKD(hipMemset(st.tmp_dev, 0, tmp));
     }
@@ -551,8 +561,8 @@ bool DeviceAdvanceRequest(CommRequest* req, DeviceReqState& st) {
     st.recv_staged = r_host;
     if (s_host) {
         if (!st.stage_send || st.stage_send_bytes < send_b) {
-            if (st.stage_send) (void)hipFree(st.stage_send);
-            HIP_CHECKD(hipMalloc(&st.stage_send, send_b));
+            if (st.stage_send) rt->FreeDevice(st.stage_send);
+            st.stage_send = rt->AllocDevice(send_b);
             st.stage_send_bytes = send_b;
         }
         HIP_CHECKD(hipMemcpyAsync(st.stage_send, req->UserSendBuf(), send_b,
@@ -564,8 +574,8 @@ bool DeviceAdvanceRequest(CommRequest* req, DeviceReqState& st) {
         }
     }
     if (r_host && (!st.stage_recv || st.stage_recv_bytes < recv_b)) {
-        if (st.stage_recv) (void)hipFree(st.stage_recv);
-        HIP_CHECKD(hipMalloc(&st.stage_recv, recv_b));
+        if (st.stage_recv) rt->FreeDevice(st.stage_recv);
+        st.stage_recv = rt->AllocDevice(recv_b);
         st.stage_recv_bytes = recv_b;
     }
     req->SetDeviceBuffers(s_host ? static_cast<const uint8_t*>(st.stage_send) : nullptr,
