@@ -145,6 +145,36 @@ __global__ void ReduceBf16Kernel(unsigned short* __restrict__ dst,
     }
 }
 
+using ushort8_ev = __attribute__((ext_vector_type(8))) unsigned short;
+
+template <ReduceOp OP>
+__global__ void ReduceBf16NTKernel(unsigned short* __restrict__ dst,
+                                   const unsigned short* __restrict__ src, size_t n) {
+    const size_t tid = blockIdx.x * blockDim.x + threadIdx.x;
+    const size_t stride = gridDim.x * blockDim.x;
+    const size_t n8 = n / 8;
+    const ushort8_ev* s8 = reinterpret_cast<const ushort8_ev*>(src);
+    ushort8_ev* d8 = reinterpret_cast<ushort8_ev*>(dst);
+    for (size_t i = tid; i < n8; i += stride) {
+        ushort8_ev a = __builtin_nontemporal_load(d8 + i);
+        ushort8_ev b = __builtin_nontemporal_load(s8 + i);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+            float fa = __bfloat162float(__hip_bfloat16_raw{a[j]});
+            float fb = __bfloat162float(__hip_bfloat16_raw{b[j]});
+            __hip_bfloat16 hr = __float2bfloat16(Apply<float, OP>(fa, fb));
+            a[j] = reinterpret_cast<unsigned short&>(hr);
+        }
+        __builtin_nontemporal_store(a, d8 + i);
+    }
+    for (size_t i = n8 * 8 + tid; i < n; i += stride) {
+        float fa = __bfloat162float(__hip_bfloat16_raw{dst[i]});
+        float fb = __bfloat162float(__hip_bfloat16_raw{src[i]});
+        __hip_bfloat16 hr = __float2bfloat16(Apply<float, OP>(fa, fb));
+        dst[i] = reinterpret_cast<unsigned short&>(hr);
+    }
+}
+
 template <ReduceOp OP>
 __global__ void ReduceF16Kernel(__half* __restrict__ dst,
                                 const __half* __restrict__ src, size_t n) {
@@ -197,14 +227,27 @@ void LaunchScalarByOp(T* dst, const T* src, size_t n, ReduceOp op, hipStream_t s
 
 void LaunchReduce(void* dst, const void* src, size_t count, DataType dt,
                   ReduceOp op, hipStream_t stream) {
+    // Nontemporal path for streams past L2 reach (32 MiB aggregate):
+    // measured +25% on 256 MiB f32 reductions (docs/BENCHMARKS.md).
+    const bool nt = count * DtypeSize(dt) >= (16u << 20);
     switch (dt) {
         case DataType::F32:
-            MLSL_LAUNCH_BY_OP(ReduceF32Kernel, static_cast<float*>(dst),
-                              static_cast<const float*>(src), count, op, stream);
+            if (nt) {
+                MLSL_LAUNCH_BY_OP(ReduceF32NTKernel, static_cast<float*>(dst),
+                                  static_cast<const float*>(src), count, op, stream);
+            } else {
+                MLSL_LAUNCH_BY_OP(ReduceF32Kernel, static_cast<float*>(dst),
+                                  static_cast<const float*>(src), count, op, stream);
+            }
             break;
         case DataType::BF16:
-            MLSL_LAUNCH_BY_OP(ReduceBf16Kernel, static_cast<unsigned short*>(dst),
-                              static_cast<const unsigned short*>(src), count, op, stream);
+            if (nt) {
+                MLSL_LAUNCH_BY_OP(ReduceBf16NTKernel, static_cast<unsigned short*>(dst),
+                                  static_cast<const unsigned short*>(src), count, op, stream);
+            } else {
+                MLSL_LAUNCH_BY_OP(ReduceBf16Kernel, static_cast<unsigned short*>(dst),
+                                  static_cast<const unsigned short*>(src), count, op, stream);
+            }
             break;
         case DataType::F16:
             MLSL_LAUNCH_BY_OP(ReduceF16Kernel, static_cast<__half*>(dst),
