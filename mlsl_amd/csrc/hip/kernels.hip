@@ -318,6 +318,7 @@ __global__ void QuantizeKernel(const T* __restrict__ in, T* __restrict__ err,
     const int lane = threadIdx.x & 63;
     const size_t wave = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
     const size_t wstride = (gridDim.x * blockDim.x) >> 6;
+    const bool vec4 = (block_elems & 3) == 0;
     for (size_t blk = wave; blk < nblocks; blk += wstride) {
         const size_t base = blk * block_elems;
         const size_t n = min(block_elems, count - base);
@@ -326,10 +327,26 @@ __global__ void QuantizeKernel(const T* __restrict__ in, T* __restrict__ err,
         int8_t* payload = reinterpret_cast<int8_t*>(wblock + 8);
 
         float m = 0.f;
-        for (size_t i = lane; i < n; i += 64) {
-            float v = LoadAsF32(in, base + i);
-            if (USE_ERR) v += LoadAsF32(err, base + i);
-            m = fmaxf(m, fabsf(v));
+        if (vec4) {
+            for (size_t i = lane * 4; i + 3 < n; i += 256) {
+#pragma unroll
+                for (int j = 0; j < 4; ++j) {
+                    float v = LoadAsF32(in, base + i + j);
+                    if (USE_ERR) v += LoadAsF32(err, base + i + j);
+                    m = fmaxf(m, fabsf(v));
+                }
+            }
+            for (size_t i = (n & ~size_t(3)) + lane; i < n; i += 64) {
+                float v = LoadAsF32(in, base + i);
+                if (USE_ERR) v += LoadAsF32(err, base + i);
+                m = fmaxf(m, fabsf(v));
+            }
+        } else {
+            for (size_t i = lane; i < n; i += 64) {
+                float v = LoadAsF32(in, base + i);
+                if (USE_ERR) v += LoadAsF32(err, base + i);
+                m = fmaxf(m, fabsf(v));
+            }
         }
         m = WaveMax(m);
         const float scale = m > 0.f ? m / 127.f : 1.f;
@@ -338,13 +355,38 @@ __global__ void QuantizeKernel(const T* __restrict__ in, T* __restrict__ err,
             hdr[1] = 0.f;
         }
         const float inv = 1.f / scale;
-        for (size_t i = lane; i < n; i += 64) {
-            float v = LoadAsF32(in, base + i);
-            if (USE_ERR) v += LoadAsF32(err, base + i);
-            float q = nearbyintf(v * inv);
-            q = fminf(127.f, fmaxf(-127.f, q));
-            payload[i] = static_cast<int8_t>(q);
-            if (USE_ERR) StoreFromF32(err, base + i, v - q * scale);
+        if (vec4) {
+            int32_t* p4 = reinterpret_cast<int32_t*>(payload);
+            for (size_t i = lane * 4; i + 3 < n; i += 256) {
+                int32_t packed = 0;
+#pragma unroll
+                for (int j = 0; j < 4; ++j) {
+                    float v = LoadAsF32(in, base + i + j);
+                    if (USE_ERR) v += LoadAsF32(err, base + i + j);
+                    float q = nearbyintf(v * inv);
+                    q = fminf(127.f, fmaxf(-127.f, q));
+                    packed |= (static_cast<int32_t>(q) & 0xff) << (8 * j);
+                    if (USE_ERR) StoreFromF32(err, base + i + j, v - q * scale);
+                }
+                p4[i >> 2] = packed;
+            }
+            for (size_t i = (n & ~size_t(3)) + lane; i < n; i += 64) {
+                float v = LoadAsF32(in, base + i);
+                if (USE_ERR) v += LoadAsF32(err, base + i);
+                float q = nearbyintf(v * inv);
+                q = fminf(127.f, fmaxf(-127.f, q));
+                payload[i] = static_cast<int8_t>(q);
+                if (USE_ERR) StoreFromF32(err, base + i, v - q * scale);
+            }
+        } else {
+            for (size_t i = lane; i < n; i += 64) {
+                float v = LoadAsF32(in, base + i);
+                if (USE_ERR) v += LoadAsF32(err, base + i);
+                float q = nearbyintf(v * inv);
+                q = fminf(127.f, fmaxf(-127.f, q));
+                payload[i] = static_cast<int8_t>(q);
+                if (USE_ERR) StoreFromF32(err, base + i, v - q * scale);
+            }
         }
         for (size_t i = n + lane; i < block_elems; i += 64) payload[i] = 0;
     }
@@ -357,14 +399,29 @@ __global__ void DequantizeKernel(const uint8_t* __restrict__ wire, T* __restrict
     const int lane = threadIdx.x & 63;
     const size_t wave = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
     const size_t wstride = (gridDim.x * blockDim.x) >> 6;
+    const bool vec4 = (block_elems & 3) == 0;
     for (size_t blk = wave; blk < nblocks; blk += wstride) {
         const size_t base = blk * block_elems;
         const size_t n = min(block_elems, count - base);
         const uint8_t* wblock = wire + blk * (block_elems + 8);
         const float scale = reinterpret_cast<const float*>(wblock)[0];
         const int8_t* payload = reinterpret_cast<const int8_t*>(wblock + 8);
-        for (size_t i = lane; i < n; i += 64)
-            StoreFromF32(out, base + i, static_cast<float>(payload[i]) * scale);
+        if (vec4) {
+            const int32_t* p4 = reinterpret_cast<const int32_t*>(payload);
+            for (size_t i = lane * 4; i + 3 < n; i += 256) {
+                const int32_t packed = p4[i >> 2];
+#pragma unroll
+                for (int j = 0; j < 4; ++j) {
+                    const int8_t q = static_cast<int8_t>((packed >> (8 * j)) & 0xff);
+                    StoreFromF32(out, base + i + j, static_cast<float>(q) * scale);
+                }
+            }
+            for (size_t i = (n & ~size_t(3)) + lane; i < n; i += 64)
+                StoreFromF32(out, base + i, static_cast<float>(payload[i]) * scale);
+        } else {
+            for (size_t i = lane; i < n; i += 64)
+                StoreFromF32(out, base + i, static_cast<float>(payload[i]) * scale);
+        }
     }
 }
 
@@ -376,6 +433,7 @@ __global__ void QuantAccumKernel(uint8_t* __restrict__ acc, const uint8_t* __res
     const int lane = threadIdx.x & 63;
     const size_t wave = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
     const size_t wstride = (gridDim.x * blockDim.x) >> 6;
+    const bool vec4 = (block_elems & 3) == 0;
     for (size_t blk = wave; blk < nblocks; blk += wstride) {
         const size_t base = blk * block_elems;
         const size_t n = min(block_elems, count - base);
@@ -388,15 +446,54 @@ __global__ void QuantAccumKernel(uint8_t* __restrict__ acc, const uint8_t* __res
         const int8_t* ip = reinterpret_cast<const int8_t*>(iblock + 8);
 
         float m = 0.f;
-        for (size_t i = lane; i < n; i += 64)
-            m = fmaxf(m, fabsf(ap[i] * as + ip[i] * is));
+        if (vec4) {
+            const int32_t* a4 = reinterpret_cast<const int32_t*>(ap);
+            const int32_t* i4 = reinterpret_cast<const int32_t*>(ip);
+            for (size_t i = lane * 4; i + 3 < n; i += 256) {
+                const int32_t pa = a4[i >> 2], pb = i4[i >> 2];
+#pragma unroll
+                for (int j = 0; j < 4; ++j) {
+                    const float v = static_cast<int8_t>((pa >> (8 * j)) & 0xff) * as +
+                                    static_cast<int8_t>((pb >> (8 * j)) & 0xff) * is;
+                    m = fmaxf(m, fabsf(v));
+                }
+            }
+            for (size_t i = (n & ~size_t(3)) + lane; i < n; i += 64)
+                m = fmaxf(m, fabsf(ap[i] * as + ip[i] * is));
+        } else {
+            for (size_t i = lane; i < n; i += 64)
+                m = fmaxf(m, fabsf(ap[i] * as + ip[i] * is));
+        }
         m = WaveMax(m);
         const float ns = m > 0.f ? m / 127.f : 1.f;
         const float inv = 1.f / ns;
-        for (size_t i = lane; i < n; i += 64) {
-            float v = ap[i] * as + ip[i] * is;
-            float q = nearbyintf(v * inv);
-            ap[i] = static_cast<int8_t>(fminf(127.f, fmaxf(-127.f, q)));
+        if (vec4) {
+            int32_t* a4 = reinterpret_cast<int32_t*>(ap);
+            const int32_t* i4 = reinterpret_cast<const int32_t*>(ip);
+            for (size_t i = lane * 4; i + 3 < n; i += 256) {
+                const int32_t pa = a4[i >> 2], pb = i4[i >> 2];
+                int32_t packed = 0;
+#pragma unroll
+                for (int j = 0; j < 4; ++j) {
+                    const float v = static_cast<int8_t>((pa >> (8 * j)) & 0xff) * as +
+                                    static_cast<int8_t>((pb >> (8 * j)) & 0xff) * is;
+                    float q = nearbyintf(v * inv);
+                    q = fminf(127.f, fmaxf(-127.f, q));
+                    packed |= (static_cast<int32_t>(q) & 0xff) << (8 * j);
+                }
+                a4[i >> 2] = packed;
+            }
+            for (size_t i = (n & ~size_t(3)) + lane; i < n; i += 64) {
+                const float v = ap[i] * as + ip[i] * is;
+                float q = nearbyintf(v * inv);
+                ap[i] = static_cast<int8_t>(fminf(127.f, fmaxf(-127.f, q)));
+            }
+        } else {
+            for (size_t i = lane; i < n; i += 64) {
+                const float v = ap[i] * as + ip[i] * is;
+                float q = nearbyintf(v * inv);
+                ap[i] = static_cast<int8_t>(fminf(127.f, fmaxf(-127.f, q)));
+            }
         }
         if (lane == 0) ahdr[0] = ns;
     }
