@@ -227,8 +227,8 @@ void LaunchScalarByOp(T* dst, const T* src, size_t n, ReduceOp op, hipStream_t s
 
 void LaunchReduce(void* dst, const void* src, size_t count, DataType dt,
                   ReduceOp op, hipStream_t stream) {
-    // Nontemporal path for streams past L2 reach (32 MiB aggregate):
-    // measured +25% on 256 MiB f32 reductions (docs/BENCHMARKS.md).
+    // Nontemporal path for f32 streams past L2 reach: measured 4.2 -> 5.3
+    // TB/s on 256 MiB (docs/BENCHMARKS.md); bf16 measured slower with NT.
     const bool nt = count * DtypeSize(dt) >= (16u << 20);
     switch (dt) {
         case DataType::F32:
@@ -241,13 +241,9 @@ void LaunchReduce(void* dst, const void* src, size_t count, DataType dt,
             }
             break;
         case DataType::BF16:
-            if (nt) {
-                MLSL_LAUNCH_BY_OP(ReduceBf16NTKernel, static_cast<unsigned short*>(dst),
-                                  static_cast<const unsigned short*>(src), count, op, stream);
-            } else {
-                MLSL_LAUNCH_BY_OP(ReduceBf16Kernel, static_cast<unsigned short*>(dst),
-                                  static_cast<const unsigned short*>(src), count, op, stream);
-            }
+            // NT measured slower for bf16 (5.64 -> 5.01 TB/s): keep default
+            MLSL_LAUNCH_BY_OP(ReduceBf16Kernel, static_cast<unsigned short*>(dst),
+                              static_cast<const unsigned short*>(src), count, op, stream);
             break;
         case DataType::F16:
             MLSL_LAUNCH_BY_OP(ReduceF16Kernel, static_cast<__half*>(dst),
