@@ -395,7 +395,9 @@ __global__ void DequantizeKernel(const uint8_t* __restrict__ wire, T* __restrict
     const int lane = threadIdx.x & 63;
     const size_t wave = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
     const size_t wstride = (gridDim.x * blockDim.x) >> 6;
-    const bool vec4 = (block_elems & 3) == 0;
+    // 16-B stores need a 16-B-aligned output base (sliced tensors may not be)
+    const bool vec4 = (block_elems & 3) == 0 &&
+                      (reinterpret_cast<uintptr_t>(out) & 15) == 0;
     for (size_t blk = wave; blk < nblocks; blk += wstride) {
         const size_t base = blk * block_elems;
         const size_t n = min(block_elems, count - base);
@@ -406,10 +408,20 @@ __global__ void DequantizeKernel(const uint8_t* __restrict__ wire, T* __restrict
             const int32_t* p4 = reinterpret_cast<const int32_t*>(payload);
             for (size_t i = lane * 4; i + 3 < n; i += 256) {
                 const int32_t packed = p4[i >> 2];
+                if constexpr (sizeof(T) == 4) {
+                    // f32: one 16-B store per lane pass
+                    float4_ev v;
 #pragma unroll
-                for (int j = 0; j < 4; ++j) {
-                    const int8_t q = static_cast<int8_t>((packed >> (8 * j)) & 0xff);
-                    StoreFromF32(out, base + i + j, static_cast<float>(q) * scale);
+                    for (int j = 0; j < 4; ++j)
+                        v[j] = static_cast<int8_t>((packed >> (8 * j)) & 0xff) * scale;
+                    *reinterpret_cast<float4_ev*>(
+                        reinterpret_cast<float*>(out) + base + i) = v;
+                } else {
+#pragma unroll
+                    for (int j = 0; j < 4; ++j) {
+                        const int8_t q = static_cast<int8_t>((packed >> (8 * j)) & 0xff);
+                        StoreFromF32(out, base + i + j, static_cast<float>(q) * scale);
+                    }
                 }
             }
             for (size_t i = (n & ~size_t(3)) + lane; i < n; i += 64)
