@@ -27,6 +27,20 @@ def resnet50_buckets(bucket_mb=25, dtype_size=4, total_params=25_557_032):
     return buckets
 
 
+def transformer_buckets(model="llama8b-ish", dtype_size=2):
+    """Gradient bucket lists for transformer-shaped workloads (bf16):
+    per-layer fused buckets (attention + MLP weights) — the bucket sizes a
+    Ulysses/TP trainer would allreduce per layer."""
+    shapes = {
+        # (layers, hidden, intermediate)
+        "gpt2-l": (36, 1280, 5120),
+        "llama8b-ish": (32, 4096, 14336),
+    }[model]
+    L, H, I = shapes
+    per_layer = 4 * H * H + 3 * H * I + 2 * H  # qkv+o proj, gate/up/down, norms
+    return [per_layer for _ in range(L)] + [2 * 32000 * H // 2]  # embeddings
+
+
 class SyntheticNet:
     """Two CC layers on a Distribution grid; one step = forward exchange +
     backward exchange + gradient allreduce + (optional) increment allgather.

@@ -8,6 +8,8 @@
 """
 from .bucketer import GradBucketer  # noqa: F401
 
+from . import seqpar  # noqa: F401
+
 try:  # torch is optional at import time
     from .ddp import DistributedData  # noqa: F401
 except ImportError:  # pragma: no cover

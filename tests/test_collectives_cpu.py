@@ -82,3 +82,8 @@ def test_fault_peer_death():
 @pytest.mark.parametrize("world", [2])
 def test_ddp_wrapper(world):
     run_ranks("ddp_wrapper", world, timeout=300)
+
+
+@pytest.mark.parametrize("world", [2, 4])
+def test_seqpar_reshard(world):
+    run_ranks("seqpar_reshard", world)
