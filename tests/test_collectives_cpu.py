@@ -87,3 +87,12 @@ def test_ddp_wrapper(world):
 @pytest.mark.parametrize("world", [2, 4])
 def test_seqpar_reshard(world):
     run_ranks("seqpar_reshard", world)
+
+
+def test_pointer_checker():
+    run_ranks("pointer_checker", 2, extra_env={"MLSL_CHECK_POINTERS": "1"})
+
+
+@pytest.mark.parametrize("world", [1, 2, 3])
+def test_edge_cases(world):
+    run_ranks("edge_cases", world)

@@ -430,6 +430,54 @@ def seqpar_reshard():
 WORKERS["seqpar_reshard"] = seqpar_reshard
 
 
+def pointer_checker():
+    """MLSL_CHECK_POINTERS=1: collectives accept registered Environment
+    allocations and reject foreign buffers (reference pointer_checker)."""
+    import ctypes
+    import mlsl_amd as mx
+    from mlsl_amd import MlslError
+    mx.init()
+    size = mx.world_size()
+    d = mx.Distribution(size, 1)
+    n = 1024
+    ptr = mx.alloc(n * 4)
+    buf = (ctypes.c_float * n).from_address(ptr)
+    for i in range(n):
+        buf[i] = float(mx.rank())
+    mx.wait(d.all_reduce(ptr, ptr, n, dtype="f32", op="sum", group="data"))
+    assert buf[0] == (size - 1) * size / 2.0
+    # a numpy array is NOT a registered allocation -> must be rejected
+    a = np.ones(16, dtype=np.float32)
+    try:
+        mx.wait(d.all_reduce(a, a, 16, op="sum", group="data"))
+    except MlslError:
+        mx.free(ptr)
+        mx.finalize()
+        return
+    raise AssertionError("unregistered buffer was not rejected")
+
+
+WORKERS["pointer_checker"] = pointer_checker
+
+
+def edge_cases():
+    """Zero-count and single-element collectives terminate correctly."""
+    import mlsl_amd as mx
+    mx.init()
+    size = mx.world_size()
+    d = mx.Distribution(size, 1)
+    a = np.zeros(1, dtype=np.float32)
+    out = np.zeros(1, dtype=np.float32)
+    mx.wait(d.all_reduce(a, out, 0, op="sum", group="data"))
+    a[0] = mx.rank()
+    mx.wait(d.all_reduce(a, out, 1, op="sum", group="data"))
+    assert out[0] == (size - 1) * size / 2.0
+    mx.finalize()
+
+
+WORKERS["edge_cases"] = edge_cases
+
+
 def main():
     name = sys.argv[1]
     fn = WORKERS.get(name)
