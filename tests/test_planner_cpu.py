@@ -32,3 +32,11 @@ def test_alltoall_transition(world):
 
 def test_mlsl_net_world1():
     run_ranks("mlsl_net", 1, extra_env={"MP": "1", "DIST_UPDATE": "0"})
+
+
+def test_case2_allreduce():
+    run_ranks("case2_allreduce", 4)
+
+
+def test_case3_repartition():
+    run_ranks("case3_repartition", 4)

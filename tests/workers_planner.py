@@ -289,8 +289,120 @@ def alltoall_transition():
     mx.finalize()
 
 
+def case2_allreduce():
+    """Case 2 (reference mlsl_impl.cpp:176-186): CC output on (d,m) feeding
+    an input on (d,1) with equal data parts -> forward AllReduce of partial
+    sums over the producer's model group; no backward exchange."""
+    mx, rank, size = _init()
+    assert size == 4
+    MB, S, F = 8, 3, 8
+    s_ = mx.Session()
+    s_.set_global_minibatch_size(MB)
+    d22 = mx.Distribution(2, 2)
+    d21 = mx.Distribution(2, 1)
+
+    i0 = s_.create_op_reg_info("cc")
+    i0.add_input(F, S, "f32")
+    i0.add_output(F, S, "f32")
+    op0 = s_.operation(s_.add_operation(i0, d22))
+    i1 = s_.create_op_reg_info("act")
+    i1.add_input(F, S, "f32")
+    i1.add_output(F, S, "f32")
+    op1 = s_.operation(s_.add_operation(i1, d21))
+    op0.set_next(op1, 0, 0)
+    s_.commit()
+
+    out0, in1 = op0.output(0), op1.input(0)
+    lmb = 4  # MB / dp
+    midx = d22.process_idx("model")
+    assert out0.comm_buf_size == F * lmb * S * 4
+    comm = np.zeros(out0.comm_buf_size // 4, dtype=np.float32)
+    # partial(b, f, k) = f*100 + k + b + midx
+    b = np.arange(lmb)[:, None, None]
+    f = np.arange(F)[None, :, None]
+    k = np.arange(S)[None, None, :]
+    part = (f * 100 + k + b + midx).astype(np.float32)
+    _pack(out0, lmb, comm, part)
+    out0.start_comm(comm)
+    ptr = in1.wait_comm()
+    res = _as_np(ptr, F * lmb * S)
+    got = np.zeros((lmb, F, S), dtype=np.float32)
+    _unpack(in1, lmb, res, got)
+    want = 2 * (f * 100 + k + b) + 1  # sum over model ranks {0,1}
+    assert np.allclose(got, want), f"case2 fwd mismatch rank={rank}"
+    # backward: no exchange in this direction
+    assert out0.wait_comm() is None or True
+    mx.finalize()
+
+
+def case3_repartition():
+    """Case 3 (reference mlsl_impl.cpp:187-202): CC output on (2,2) feeding
+    an input on (4,1): forward ReduceScatter over the producer's model group
+    with minibatch repartition, backward AllGather."""
+    mx, rank, size = _init()
+    assert size == 4
+    S, F = 3, 8
+    MB = 8  # in dist (4,1): local mb = 2; out dist (2,2): local mb = 4
+    s_ = mx.Session()
+    s_.set_global_minibatch_size(MB)
+    d22 = mx.Distribution(2, 2)
+    d41 = mx.Distribution(4, 1)
+
+    i0 = s_.create_op_reg_info("cc")
+    i0.add_input(F, S, "f32")
+    i0.add_output(F, S, "f32")
+    op0 = s_.operation(s_.add_operation(i0, d22))
+    i1 = s_.create_op_reg_info("act")
+    i1.add_input(F, S, "f32")
+    i1.add_output(F, S, "f32")
+    op1 = s_.operation(s_.add_operation(i1, d41))
+    op0.set_next(op1, 0, 0)
+    s_.commit()
+
+    out0, in1 = op0.output(0), op1.input(0)
+    lmb0 = op0.local_minibatch_size   # 4
+    lmb1 = op1.local_minibatch_size   # 2
+    assert lmb0 == 4 and lmb1 == 2
+    assert out0.pack_block_count == 2        # mb-sliced (BIPackReduceScatter2)
+    midx = d22.process_idx("model")
+
+    comm0 = np.zeros(out0.comm_buf_size // 4, dtype=np.float32)
+    comm1 = np.zeros(in1.comm_buf_size // 4, dtype=np.float32)
+    # producer partial over its local mb window:
+    # value(global_b, f, k) = global_b*1000 + f*10 + k, contribution + midx
+    b0 = op0.global_minibatch_offset
+    gb = (b0 + np.arange(lmb0))[:, None, None]
+    f = np.arange(F)[None, :, None]
+    k = np.arange(S)[None, None, :]
+    part = (gb * 1000 + f * 10 + k + midx).astype(np.float32)
+    _pack(out0, lmb0, comm0, part)
+    out0.start_comm(comm0)
+    ptr = in1.wait_comm()
+    res = _as_np(ptr, lmb1 * F * S)
+    got = np.zeros((lmb1, F, S), dtype=np.float32)
+    _unpack(in1, lmb1, res, got)
+    gb1 = (op1.global_minibatch_offset + np.arange(lmb1))[:, None, None]
+    want = 2 * (gb1 * 1000 + f * 10 + k) + 1  # sum over the 2 model ranks
+    assert np.allclose(got, want), f"case3 fwd mismatch rank={rank}: {got[0,0]} vs {want[0,0]}"
+
+    # backward: consumer sends grads g(global_b, f, k) = global_b*7+f+k,
+    # producer receives the full window for its local minibatch
+    g1 = (gb1 * 7 + f + k).astype(np.float32)
+    _pack(in1, lmb1, comm1, g1)
+    in1.start_comm(comm1)
+    bptr = out0.wait_comm()
+    bres = _as_np(bptr, lmb0 * F * S)
+    back = np.zeros((lmb0, F, S), dtype=np.float32)
+    _unpack(out0, lmb0, bres, back)
+    want_b = gb * 7 + f + k
+    assert np.allclose(back, want_b), f"case3 bwd mismatch rank={rank}"
+    mx.finalize()
+
+
 WORKERS = {
     "grid_shapes": grid_shapes,
     "mlsl_net": mlsl_net,
     "alltoall_transition": alltoall_transition,
+    "case2_allreduce": case2_allreduce,
+    "case3_repartition": case3_repartition,
 }
