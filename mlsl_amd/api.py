@@ -255,6 +255,33 @@ class Distribution:
                                               root, GROUP[group], ctypes.byref(req)))
         return req
 
+    def send_recv_list(self, sbuf, rbuf, pairs, dtype=None, group="global"):
+        """Neighbor exchange: pairs = [(peer, send_off, send_count,
+        recv_off, recv_count), ...] in elements (CommOpSRList — declared but
+        unimplemented in the reference; first-class here)."""
+        dt = self._dt(sbuf, rbuf, dtype)
+        sp, _ = _as_ptr_dtype(sbuf)
+        rp, _ = _as_ptr_dtype(rbuf)
+        L = lib()
+        if not hasattr(L, "_srl_declared"):
+            P = ctypes.POINTER
+            L.mlsl_distribution_send_recv_list.argtypes = [
+                c_void_p, c_void_p, c_void_p, P(c_size_t), P(c_size_t),
+                P(c_size_t), P(c_size_t), P(c_size_t), c_size_t, c_int, c_int,
+                P(c_void_p)]
+            L.mlsl_distribution_send_recv_list.restype = c_int
+            L._srl_declared = True
+        peers = _sizes([p[0] for p in pairs])
+        soffs = _sizes([p[1] for p in pairs])
+        scnts = _sizes([p[2] for p in pairs])
+        roffs = _sizes([p[3] for p in pairs])
+        rcnts = _sizes([p[4] for p in pairs])
+        req = c_void_p()
+        check(L.mlsl_distribution_send_recv_list(
+            self._h, sp, rp, peers, soffs, scnts, roffs, rcnts, len(pairs),
+            DTYPE[dt], GROUP[group], ctypes.byref(req)))
+        return req
+
     def reduce_scatter(self, sbuf, rbuf, recv_count, op="sum", dtype=None, group="global"):
         dt = self._dt(sbuf, rbuf, dtype)
         sp, _ = _as_ptr_dtype(sbuf)

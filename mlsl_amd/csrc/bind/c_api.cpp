@@ -7,6 +7,7 @@
 #include <cstring>
 #include <string>
 
+#include "../comm/group.hpp"
 #include "../dl/session.hpp"
 #include "../include/mlsl/mlsl.hpp"
 
@@ -209,9 +210,20 @@ int mlsl_distribution_reduce_scatter(mlsl_distribution d, const void* sbuf, void
     C_CATCH
 }
 
-/* ---- persistent requests ---- */
+int mlsl_distribution_send_recv_list(mlsl_distribution d, const void* sbuf, void* rbuf,
+                                     const size_t* peers, const size_t* soffs,
+                                     const size_t* scnts, const size_t* roffs,
+                                     const size_t* rcnts, size_t npairs,
+                                     mlsl_data_type dt, mlsl_group g, mlsl_request* out) {
+    C_TRY std::vector<SRPair> pairs(npairs);
+    for (size_t i = 0; i < npairs; ++i)
+        pairs[i] = SRPair{static_cast<int>(peers[i]), soffs[i], scnts[i], roffs[i],
+                          rcnts[i]};
+    *out = DIST(d)->SendRecvList(sbuf, rbuf, pairs, DT(dt), GK(g));
+    C_CATCH
+}
 
-#include "../comm/group.hpp"
+/* ---- persistent requests ---- */
 
 static CommRequest* NewPersistent(mlsl_distribution d, mlsl_group g, mlsl_data_type dt) {
     Distribution* dist = DIST(d);

@@ -154,10 +154,7 @@ bool CommRequest::UsesDeviceSchedule() const {
 }
 
 size_t CommRequest::WireBytesFor(const ChunkExec& ce) const {
-    const size_t cnt = ce.sch.result.bytes ? ce.sch.result.bytes /
-        (qparams_.block_elems + 8) * qparams_.block_elems : 0;
-    (void)cnt;
-    // result.bytes for compressed chunks is in wire bytes already
+    // compressed chunks: Schedule::result is already in wire bytes
     return ce.sch.result.bytes;
 }
 
@@ -415,14 +412,13 @@ void CommRequest::MarkFailed(const std::string& what) {
 
 bool CommRequest::AdvanceHost(Mesh* mesh) {
     const size_t es = DtypeSize(dtype_);
-    const size_t msg_es = es;  // chunk offsets are in elements of dtype_
     bool all_done = true;
 
     const bool compressed = Compressed();
     for (auto& ce : chunks_) {
         if (ce.finished) continue;
-        const uint8_t* sbase = sbuf_ + ce.elem_off * msg_es;
-        uint8_t* rbase = rbuf_ + ce.elem_off * msg_es;
+        const uint8_t* sbase = sbuf_ + ce.elem_off * es;
+        uint8_t* rbase = rbuf_ + ce.elem_off * es;
         uint8_t* wire = ce.tmp.data();                         // compressed only
         const size_t wire_bytes = compressed ? ce.sch.result.bytes : 0;
         auto ptr = [&](const BufRef& b) -> uint8_t* {

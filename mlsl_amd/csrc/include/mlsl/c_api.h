@@ -87,6 +87,12 @@ int mlsl_distribution_reduce_scatter(mlsl_distribution d, const void* sbuf, void
                                      size_t recv_count, mlsl_data_type dt, mlsl_reduction op,
                                      mlsl_group g, mlsl_request* out);
 
+int mlsl_distribution_send_recv_list(mlsl_distribution d, const void* sbuf, void* rbuf,
+                                     const size_t* peers, const size_t* soffs,
+                                     const size_t* scnts, const size_t* roffs,
+                                     const size_t* rcnts, size_t npairs,
+                                     mlsl_data_type dt, mlsl_group g, mlsl_request* out);
+
 /* persistent requests: describe once, Start/Wait/Test every iteration
  * (the reference's Session-level persistent-request contract exposed for
  * generic collectives; hot loops skip per-call planning/allocation). */

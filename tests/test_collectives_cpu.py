@@ -96,3 +96,8 @@ def test_pointer_checker():
 @pytest.mark.parametrize("world", [1, 2, 3])
 def test_edge_cases(world):
     run_ranks("edge_cases", world)
+
+
+@pytest.mark.parametrize("world", [2, 4])
+def test_srlist_ring(world):
+    run_ranks("srlist_ring", world)

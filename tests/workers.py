@@ -190,12 +190,21 @@ def hybrid_grid():
 
 
 def srlist_ring():
-    """SendRecvList neighbor exchange via the C++ API's SRLIST (exercised
-    through the planner later; here via raw distribution barrier + manual
-    check is skipped on the Python surface for now)."""
+    """SendRecvList ring: send my 8-element block to next, receive from
+    prev (the ring-attention neighbor-exchange primitive)."""
     mx, rank, size = _init()
     d = mx.Distribution(size, 1)
-    d.barrier("global")
+    n = 8
+    a = np.full(n, float(rank * 11), dtype=np.float32) + np.arange(n, dtype=np.float32)
+    out = np.zeros(n, dtype=np.float32)
+    if size == 1:
+        mx.finalize()
+        return
+    pairs = [((rank + 1) % size, 0, n, 0, 0),
+             ((rank - 1 + size) % size, 0, 0, 0, n)]
+    mx.wait(d.send_recv_list(a, out, pairs, group="data"))
+    prev = (rank - 1 + size) % size
+    assert np.all(out == prev * 11 + np.arange(n)), f"srlist got {out[:3]}"
     mx.finalize()
 
 
