@@ -314,11 +314,7 @@ __global__ void QuantizeKernel(const T* __restrict__ in, T* __restrict__ err,
     const int lane = threadIdx.x & 63;
     const size_t wave = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
     const size_t wstride = (gridDim.x * blockDim.x) >> 6;
-    // Register-cached fast path: whole blocks up to 1024 elems (16/lane)
-    // keep v[] in VGPRs between the max pass and the quantize pass, halving
-    // the in+err read traffic.
-    const bool cached = (block_elems & 3) == 0 && block_elems <= 1024;
-    const int per_lane = static_cast<int>(block_elems >> 6);  // when cached
+    const bool vec4 = (block_elems & 3) == 0;
     for (size_t blk = wave; blk < nblocks; blk += wstride) {
         const size_t base = blk * block_elems;
         const size_t n = min(block_elems, count - base);
@@ -326,47 +322,27 @@ __global__ void QuantizeKernel(const T* __restrict__ in, T* __restrict__ err,
         float* hdr = reinterpret_cast<float*>(wblock);
         int8_t* payload = reinterpret_cast<int8_t*>(wblock + 8);
 
-        if (cached && n == block_elems) {
-            float v[16];
-            float m = 0.f;
-            for (int j = 0; j < per_lane; ++j) {
-                const size_t i = base + lane * per_lane + j;
-                float x = LoadAsF32(in, i);
-                if (USE_ERR) x += LoadAsF32(err, i);
-                v[j] = x;
-                m = fmaxf(m, fabsf(x));
-            }
-            m = WaveMax(m);
-            const float scale = m > 0.f ? m / 127.f : 1.f;
-            if (lane == 0) {
-                hdr[0] = scale;
-                hdr[1] = 0.f;
-            }
-            const float inv = 1.f / scale;
-            // per_lane consecutive int8 per lane: pack in 4-byte pieces
-            for (int j0 = 0; j0 < per_lane; j0 += 4) {
-                int32_t packed = 0;
+        float m = 0.f;
+        if (vec4) {
+            for (size_t i = lane * 4; i + 3 < n; i += 256) {
 #pragma unroll
                 for (int j = 0; j < 4; ++j) {
-                    float q = nearbyintf(v[j0 + j] * inv);
-                    q = fminf(127.f, fmaxf(-127.f, q));
-                    packed |= (static_cast<int32_t>(q) & 0xff) << (8 * j);
-                    if (USE_ERR)
-                        StoreFromF32(err, base + lane * per_lane + j0 + j,
-                                     v[j0 + j] - q * scale);
+                    float v = LoadAsF32(in, base + i + j);
+                    if (USE_ERR) v += LoadAsF32(err, base + i + j);
+                    m = fmaxf(m, fabsf(v));
                 }
-                reinterpret_cast<int32_t*>(payload)[(lane * per_lane + j0) >> 2] =
-                    packed;
             }
-            continue;
-        }
-
-        // generic path (tail block or odd block size)
-        float m = 0.f;
-        for (size_t i = lane; i < n; i += 64) {
-            float x = LoadAsF32(in, base + i);
-            if (USE_ERR) x += LoadAsF32(err, base + i);
-            m = fmaxf(m, fabsf(x));
+            for (size_t i = (n & ~size_t(3)) + lane; i < n; i += 64) {
+                float v = LoadAsF32(in, base + i);
+                if (USE_ERR) v += LoadAsF32(err, base + i);
+                m = fmaxf(m, fabsf(v));
+            }
+        } else {
+            for (size_t i = lane; i < n; i += 64) {
+                float v = LoadAsF32(in, base + i);
+                if (USE_ERR) v += LoadAsF32(err, base + i);
+                m = fmaxf(m, fabsf(v));
+            }
         }
         m = WaveMax(m);
         const float scale = m > 0.f ? m / 127.f : 1.f;
@@ -375,13 +351,38 @@ __global__ void QuantizeKernel(const T* __restrict__ in, T* __restrict__ err,
             hdr[1] = 0.f;
         }
         const float inv = 1.f / scale;
-        for (size_t i = lane; i < n; i += 64) {
-            float x = LoadAsF32(in, base + i);
-            if (USE_ERR) x += LoadAsF32(err, base + i);
-            float q = nearbyintf(x * inv);
-            q = fminf(127.f, fmaxf(-127.f, q));
-            payload[i] = static_cast<int8_t>(q);
-            if (USE_ERR) StoreFromF32(err, base + i, x - q * scale);
+        if (vec4) {
+            int32_t* p4 = reinterpret_cast<int32_t*>(payload);
+            for (size_t i = lane * 4; i + 3 < n; i += 256) {
+                int32_t packed = 0;
+#pragma unroll
+                for (int j = 0; j < 4; ++j) {
+                    float v = LoadAsF32(in, base + i + j);
+                    if (USE_ERR) v += LoadAsF32(err, base + i + j);
+                    float q = nearbyintf(v * inv);
+                    q = fminf(127.f, fmaxf(-127.f, q));
+                    packed |= (static_cast<int32_t>(q) & 0xff) << (8 * j);
+                    if (USE_ERR) StoreFromF32(err, base + i + j, v - q * scale);
+                }
+                p4[i >> 2] = packed;
+            }
+            for (size_t i = (n & ~size_t(3)) + lane; i < n; i += 64) {
+                float v = LoadAsF32(in, base + i);
+                if (USE_ERR) v += LoadAsF32(err, base + i);
+                float q = nearbyintf(v * inv);
+                q = fminf(127.f, fmaxf(-127.f, q));
+                payload[i] = static_cast<int8_t>(q);
+                if (USE_ERR) StoreFromF32(err, base + i, v - q * scale);
+            }
+        } else {
+            for (size_t i = lane; i < n; i += 64) {
+                float v = LoadAsF32(in, base + i);
+                if (USE_ERR) v += LoadAsF32(err, base + i);
+                float q = nearbyintf(v * inv);
+                q = fminf(127.f, fmaxf(-127.f, q));
+                payload[i] = static_cast<int8_t>(q);
+                if (USE_ERR) StoreFromF32(err, base + i, v - q * scale);
+            }
         }
         for (size_t i = n + lane; i < block_elems; i += 64) payload[i] = 0;
     }
