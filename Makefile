@@ -81,6 +81,19 @@ $(BUILD)/cmlsl_sample: samples/cmlsl_sample.c $(LIB)
 	gcc -O2 -Imlsl_amd/csrc/include samples/cmlsl_sample.c \
 	    -Lmlsl_amd -lmlsl_amd -Wl,-rpath,'$$ORIGIN/../mlsl_amd' -o $@
 
+# Host-logic sanitizer build: the schedule algebra is pure C++ (no HIP), so
+# it compiles with g++ + ASan/UBSan — catches indexing/overflow bugs in the
+# collective schedules (the reference had no sanitizer integration).
+ASAN := $(BUILD)/schedule_selftest_asan
+
+asan: $(ASAN)
+	$(ASAN)
+
+$(ASAN): mlsl_amd/csrc/tests/schedule_selftest.cpp mlsl_amd/csrc/comm/schedule.cpp mlsl_amd/csrc/core/log.cpp
+	@mkdir -p $(dir $@)
+	g++ -O1 -g -std=c++17 -fsanitize=address,undefined -fno-omit-frame-pointer \
+	    $^ -pthread -o $@
+
 test: all samples
 	$(SELFTEST)
 	python -m pytest tests/ -x -q -m "not gpu"

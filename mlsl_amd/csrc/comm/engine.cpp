@@ -144,8 +144,13 @@ void Engine::Loop() {
             NotifyDone();
         }
         if (active_.empty() && before == 0) {
+            // exponential idle backoff: hot for the first ~256 polls, then
+            // 50 us sleeps, growing to 1 ms after sustained idleness so an
+            // idle rank does not burn a core (device mode rarely uses the
+            // engine at all).
             if (++idle_spins > 256) {
-                std::this_thread::sleep_for(std::chrono::microseconds(50));
+                const int us = idle_spins > 4096 ? 1000 : 50;
+                std::this_thread::sleep_for(std::chrono::microseconds(us));
             }
         } else {
             idle_spins = 0;

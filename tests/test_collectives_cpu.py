@@ -101,3 +101,8 @@ def test_edge_cases(world):
 @pytest.mark.parametrize("world", [2, 4])
 def test_srlist_ring(world):
     run_ranks("srlist_ring", world)
+
+
+@pytest.mark.parametrize("world", [2, 4])
+def test_stress_random(world):
+    run_ranks("stress_random", world, timeout=300)

@@ -487,6 +487,70 @@ def edge_cases():
 WORKERS["edge_cases"] = edge_cases
 
 
+def stress_random():
+    """Randomized cross-group stress: a seeded, rank-consistent sequence of
+    mixed collectives over world/data/model groups with varying sizes,
+    each verified against a numpy oracle. Catches tag/flow-ordering bugs."""
+    import mlsl_amd as mx
+    mx.init()
+    rank, size = mx.rank(), mx.world_size()
+    mp = 2 if size % 2 == 0 and size > 1 else 1
+    d = mx.Distribution(size // mp, mp)
+    rng = np.random.RandomState(7)   # same sequence on every rank
+
+    def group_info(gname):
+        return d.process_idx(gname), d.process_count(gname)
+
+    iters = int(os.environ.get("STRESS_ITERS", "120"))
+    for it in range(iters):
+        op = rng.choice(["allreduce", "bcast", "reduce_scatter", "allgather",
+                         "alltoall", "reduce"])
+        gname = rng.choice(["data", "model", "global"])
+        n = int(rng.randint(1, 5000))
+        gidx, gsz = group_info(gname)
+        base = np.arange(n, dtype=np.float32) + it
+        if op == "allreduce":
+            a = base + gidx
+            out = np.zeros_like(a)
+            mx.wait(d.all_reduce(a, out, n, op="sum", group=gname))
+            assert np.allclose(out, gsz * base + gsz * (gsz - 1) / 2.0), (it, op)
+        elif op == "bcast":
+            root = int(rng.randint(0, gsz))
+            b = base.copy() if gidx == root else np.zeros_like(base)
+            mx.wait(d.bcast(b, n, root=root, group=gname))
+            assert np.array_equal(b, base), (it, op)
+        elif op == "reduce":
+            root = int(rng.randint(0, gsz))
+            a = base + gidx
+            out = np.zeros_like(a)
+            mx.wait(d.reduce(a, out, n, op="sum", root=root, group=gname))
+            if gidx == root:
+                assert np.allclose(out, gsz * base + gsz * (gsz - 1) / 2.0), (it, op)
+        elif op == "reduce_scatter":
+            a = np.tile(base, gsz) + gidx
+            out = np.zeros(n, dtype=np.float32)
+            mx.wait(d.reduce_scatter(a, out, n, op="sum", group=gname))
+            assert np.allclose(out, gsz * base + gsz * (gsz - 1) / 2.0), (it, op)
+        elif op == "allgather":
+            a = base + 1000 * gidx
+            out = np.zeros(n * gsz, dtype=np.float32)
+            mx.wait(d.all_gather(a, n, out, group=gname))
+            for j in range(gsz):
+                assert np.array_equal(out[j * n:(j + 1) * n], base + 1000 * j), (it, op)
+        elif op == "alltoall":
+            a = np.zeros(n * gsz, dtype=np.float32)
+            for j in range(gsz):
+                a[j * n:(j + 1) * n] = base + gidx * 100 + j
+            out = np.zeros(n * gsz, dtype=np.float32)
+            mx.wait(d.all_to_all(a, n, out, group=gname))
+            for j in range(gsz):
+                assert np.array_equal(out[j * n:(j + 1) * n], base + j * 100 + gidx), (it, op)
+    mx.finalize()
+
+
+WORKERS["stress_random"] = stress_random
+
+
 def main():
     name = sys.argv[1]
     fn = WORKERS.get(name)
