@@ -59,6 +59,13 @@ int mlsl_hip_reduce_nt(void* dst, const void* src, size_t count) {
     OPS_CATCH
 }
 
+int mlsl_hip_copy(void* dst, const void* src, size_t bytes) {
+    OPS_TRY LaunchCopy(dst, src, bytes, nullptr);
+    if (hipStreamSynchronize(nullptr) != hipSuccess)
+        throw Error("hipStreamSynchronize failed");
+    OPS_CATCH
+}
+
 int mlsl_hip_quantize(const void* in, void* err, void* wire, size_t count,
                       size_t block, int dt, int use_err) {
     OPS_TRY LaunchQuantize(in, err, wire, count, block, static_cast<DataType>(dt),

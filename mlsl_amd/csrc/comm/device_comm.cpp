@@ -348,9 +348,8 @@ void IssueFused(CommRequest* req, ChunkExec& ce, ncclComm_t comm, hipStream_t s,
                 NCCL_CHECK(ncclRecv(rbase + offs[i], spec.recv_counts[i], ndt, i, comm, s));
             }
             NCCL_CHECK(ncclGroupEnd());
-            HIP_CHECKD(hipMemcpyAsync(rbase + offs[g->MyIdx()], sbase,
-                                      spec.recv_counts[g->MyIdx()] * es,
-                                      hipMemcpyDeviceToDevice, s));
+            LaunchCopy(rbase + offs[g->MyIdx()], sbase,
+                       spec.recv_counts[g->MyIdx()] * es, s);
             break;
         }
         case CollOp::GATHER: {
@@ -365,8 +364,8 @@ void IssueFused(CommRequest* req, ChunkExec& ce, ncclComm_t comm, hipStream_t s,
             }
             NCCL_CHECK(ncclGroupEnd());
             if (g->MyIdx() == spec.root)
-                HIP_CHECKD(hipMemcpyAsync(rbase + spec.root * spec.count * es, sbase,
-                                          spec.count * es, hipMemcpyDeviceToDevice, s));
+                LaunchCopy(rbase + spec.root * spec.count * es, sbase,
+                           spec.count * es, s);
             break;
         }
         case CollOp::SCATTER: {
@@ -381,8 +380,8 @@ void IssueFused(CommRequest* req, ChunkExec& ce, ncclComm_t comm, hipStream_t s,
             }
             NCCL_CHECK(ncclGroupEnd());
             if (g->MyIdx() == spec.root)
-                HIP_CHECKD(hipMemcpyAsync(rbase, sbase + spec.root * spec.count * es,
-                                          spec.count * es, hipMemcpyDeviceToDevice, s));
+                LaunchCopy(rbase, sbase + spec.root * spec.count * es,
+                           spec.count * es, s);
             break;
         }
         case CollOp::ALLTOALL: {
@@ -393,9 +392,8 @@ void IssueFused(CommRequest* req, ChunkExec& ce, ncclComm_t comm, hipStream_t s,
                 NCCL_CHECK(ncclRecv(rbase + i * spec.count * es, spec.count, ndt, i, comm, s));
             }
             NCCL_CHECK(ncclGroupEnd());
-            HIP_CHECKD(hipMemcpyAsync(rbase + g->MyIdx() * spec.count * es,
-                                      sbase + g->MyIdx() * spec.count * es, spec.count * es,
-                                      hipMemcpyDeviceToDevice, s));
+            LaunchCopy(rbase + g->MyIdx() * spec.count * es,
+                       sbase + g->MyIdx() * spec.count * es, spec.count * es, s);
             break;
         }
         case CollOp::ALLTOALLV: {
@@ -406,10 +404,9 @@ void IssueFused(CommRequest* req, ChunkExec& ce, ncclComm_t comm, hipStream_t s,
                 NCCL_CHECK(ncclRecv(rbase + spec.recv_offs[i] * es, spec.recv_counts[i], ndt, i, comm, s));
             }
             NCCL_CHECK(ncclGroupEnd());
-            HIP_CHECKD(hipMemcpyAsync(rbase + spec.recv_offs[g->MyIdx()] * es,
-                                      sbase + spec.send_offs[g->MyIdx()] * es,
-                                      spec.send_counts[g->MyIdx()] * es,
-                                      hipMemcpyDeviceToDevice, s));
+            LaunchCopy(rbase + spec.recv_offs[g->MyIdx()] * es,
+                       sbase + spec.send_offs[g->MyIdx()] * es,
+                       spec.send_counts[g->MyIdx()] * es, s);
             break;
         }
         case CollOp::BARRIER:
@@ -470,9 +467,7 @@ void IssueSchedule(CommRequest* req, ChunkExec& ce, ncclComm_t comm, hipStream_t
             uint8_t* d = ptr(st.local_dst);
             uint8_t* src = ptr(st.local_src);
             if (st.local == Step::LocalOp::COPY) {
-                if (d != src)
-                    HIP_CHECKD(hipMemcpyAsync(d, src, st.local_src.bytes,
-                                              hipMemcpyDeviceToDevice, s));
+                if (d != src) LaunchCopy(d, src, st.local_src.bytes, s);
             } else if (ce.sch.quant_block > 0) {
                 const size_t blk = ce.sch.quant_block;
                 const size_t units = st.local_dst.bytes / (blk + 8);
@@ -614,9 +609,11 @@ bool DeviceAdvanceRequest(CommRequest* req, DeviceReqState& st) {
                     LaunchDequantize(wire, rbase, req->Spec().count, blk,
                                      req->Dtype(), s0);
                 } else if (ce.sch.result.bytes && sbase != rbase) {
-                    HIP_CHECKD(hipMemcpyAsync(rbase + ce.sch.result.off, sbase,
-                                              ce.sch.result.bytes, hipMemcpyDefault,
-                                              s0));
+                    // NT copy kernel: measured faster than the blit path for
+                    // large streams (docs/BENCHMARKS.md); buffers are device
+                    // memory here (host users are staged above).
+                    LaunchCopy(rbase + ce.sch.result.off, sbase,
+                               ce.sch.result.bytes, s0);
                 }
             }
             if (st.recv_staged)

@@ -21,6 +21,12 @@ void LaunchReduceNT(void* dst, const void* src, size_t count, hipStream_t stream
 void LaunchReduceOut(void* dst, const void* a, const void* b, size_t count,
                      DataType dt, ReduceOp op, hipStream_t stream);
 
+// Streaming device-to-device copy. For large 16-byte-aligned transfers a
+// nontemporal grid-stride kernel beats hipMemcpyAsync's blit path on HBM3E
+// (same reason the NT reduce wins: no L2 retention for stream-once data);
+// misaligned or small copies fall back to hipMemcpyAsync on `stream`.
+void LaunchCopy(void* dst, const void* src, size_t bytes, hipStream_t stream);
+
 // --- int8 block quantization with error feedback (quant/quant.c contract,
 //     fused into the allreduce path; see comm/quant.cpp) ---
 // Wire block layout: [float scale][float reserved][int8 x block_elems].

@@ -23,6 +23,7 @@ def _lib():
         L.mlsl_hip_synchronize.argtypes = []
         L.mlsl_hip_reduce.argtypes = [c.c_void_p, c.c_void_p, c.c_size_t, c.c_int, c.c_int]
         L.mlsl_hip_reduce_nt.argtypes = [c.c_void_p, c.c_void_p, c.c_size_t]
+        L.mlsl_hip_copy.argtypes = [c.c_void_p, c.c_void_p, c.c_size_t]
         L.mlsl_hip_quantize.argtypes = [c.c_void_p, c.c_void_p, c.c_void_p, c.c_size_t,
                                         c.c_size_t, c.c_int, c.c_int]
         L.mlsl_hip_dequantize.argtypes = [c.c_void_p, c.c_void_p, c.c_size_t,
@@ -32,7 +33,7 @@ def _lib():
         L.mlsl_hip_pack.argtypes = ptypes
         L.mlsl_hip_unpack.argtypes = ptypes
         for n in ("mlsl_hip_device_count", "mlsl_hip_synchronize", "mlsl_hip_reduce",
-                  "mlsl_hip_reduce_nt",
+                  "mlsl_hip_reduce_nt", "mlsl_hip_copy",
                   "mlsl_hip_quantize", "mlsl_hip_dequantize", "mlsl_hip_quant_accum",
                   "mlsl_hip_pack", "mlsl_hip_unpack"):
             getattr(L, n).restype = c.c_int
@@ -63,6 +64,13 @@ def reduce_nt(dst, src, count):
     dp, _ = _as_ptr_dtype(dst)
     sp, _ = _as_ptr_dtype(src)
     check(_lib().mlsl_hip_reduce_nt(dp, sp, count))
+
+
+def copy(dst, src, bytes_):
+    """Streaming D2D copy (NT kernel for large aligned transfers)."""
+    dp, _ = _as_ptr_dtype(dst)
+    sp, _ = _as_ptr_dtype(src)
+    check(_lib().mlsl_hip_copy(dp, sp, bytes_))
 
 
 def wire_bytes(count, block=256):
