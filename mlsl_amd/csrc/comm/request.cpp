@@ -196,7 +196,12 @@ void CommRequest::BuildChunks() {
     size_t n_chunks = 1;
     const bool splittable = spec_.op == CollOp::ALLREDUCE || spec_.op == CollOp::BCAST ||
                             spec_.op == CollOp::REDUCE;
-    if (splittable && gs > 1 && !Compressed()) {
+    // Splitting only pays when chunks land on DIFFERENT channel streams
+    // (concurrent RCCL comms over distinct link schedules). On a single
+    // channel the chunks serialize on one stream, so splitting is pure
+    // launch overhead — unlike the reference, whose endpoint servers gave
+    // every chunk its own progress process (comm_ep.cpp GET_EP_PAYLOAD).
+    if (splittable && gs > 1 && !Compressed() && cfg.num_channels > 1) {
         n_chunks = cfg.num_channels;
         if (MessageBytes() >= cfg.large_msg_mb * (1024 * 1024) && cfg.large_msg_chunks > 1)
             n_chunks *= cfg.large_msg_chunks;
