@@ -41,7 +41,15 @@ def main():
     # streaming copy: NT kernel vs torch/HIP blit path (same 256 MiB)
     dt = timed(lambda: ops.copy(a, b, n * 4))
     res["copy_nt_256MiB_TBps"] = round(2 * 4 * n / dt / 1e12, 3)
-    dt = timed(lambda: a.copy_(b))
+    dt = timed(lambda: ops.copy_variant(a, b, n * 4, False))
+    res["copy_plain_256MiB_TBps"] = round(2 * 4 * n / dt / 1e12, 3)
+
+    # ops.* sync per call; make torch pay the same per-call sync for a
+    # fair A/B.
+    def _torch_copy():
+        a.copy_(b)
+        torch.cuda.synchronize()
+    dt = timed(_torch_copy)
     res["copy_torch_256MiB_TBps"] = round(2 * 4 * n / dt / 1e12, 3)
 
     abf = torch.randn(n, device="cuda", dtype=torch.bfloat16)
@@ -59,6 +67,9 @@ def main():
     out = torch.empty_like(a)
     dt = timed(lambda: ops.dequantize(wire, out, n))
     res["dequantize_f32_TBps"] = round((1 + 4) * n / dt / 1e12, 3)
+
+    dt = timed(lambda: ops.dequantize_nt(wire, out, n))
+    res["dequantize_nt_f32_TBps"] = round((1 + 4) * n / dt / 1e12, 3)
 
     wb = torch.empty_like(wire)
     ops.quantize(b, wb, n)

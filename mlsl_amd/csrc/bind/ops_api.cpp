@@ -66,6 +66,13 @@ int mlsl_hip_copy(void* dst, const void* src, size_t bytes) {
     OPS_CATCH
 }
 
+int mlsl_hip_copy_variant(void* dst, const void* src, size_t bytes, int nt) {
+    OPS_TRY LaunchCopyVariant(dst, src, bytes, nt != 0, nullptr);
+    if (hipStreamSynchronize(nullptr) != hipSuccess)
+        throw Error("hipStreamSynchronize failed");
+    OPS_CATCH
+}
+
 int mlsl_hip_quantize(const void* in, void* err, void* wire, size_t count,
                       size_t block, int dt, int use_err) {
     OPS_TRY LaunchQuantize(in, err, wire, count, block, static_cast<DataType>(dt),
@@ -77,6 +84,13 @@ int mlsl_hip_quantize(const void* in, void* err, void* wire, size_t count,
 
 int mlsl_hip_dequantize(const void* wire, void* out, size_t count, size_t block, int dt) {
     OPS_TRY LaunchDequantize(wire, out, count, block, static_cast<DataType>(dt), nullptr);
+    if (hipStreamSynchronize(nullptr) != hipSuccess)
+        throw Error("hipStreamSynchronize failed");
+    OPS_CATCH
+}
+
+int mlsl_hip_dequantize_nt(const void* wire, void* out, size_t count, size_t block, int dt) {
+    OPS_TRY LaunchDequantizeNT(wire, out, count, block, static_cast<DataType>(dt), nullptr);
     if (hipStreamSynchronize(nullptr) != hipSuccess)
         throw Error("hipStreamSynchronize failed");
     OPS_CATCH

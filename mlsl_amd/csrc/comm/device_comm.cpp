@@ -551,8 +551,16 @@ bool DeviceAdvanceRequest(CommRequest* req, DeviceReqState& st) {
     // buffers through persistent HBM so RCCL/kernels see device memory.
     size_t send_b = 0, recv_b = 0;
     IoBytes(req, &send_b, &recv_b);
-    const bool s_host = send_b > 0 && !IsDevicePtr(req->UserSendBuf());
-    const bool r_host = recv_b > 0 && !IsDevicePtr(req->UserRecvBuf());
+    if (st.cls_sptr != req->UserSendBuf()) {
+        st.cls_sptr = req->UserSendBuf();
+        st.cls_s_host = !IsDevicePtr(st.cls_sptr);
+    }
+    if (st.cls_rptr != req->UserRecvBuf()) {
+        st.cls_rptr = req->UserRecvBuf();
+        st.cls_r_host = !IsDevicePtr(st.cls_rptr);
+    }
+    const bool s_host = send_b > 0 && st.cls_s_host;
+    const bool r_host = recv_b > 0 && st.cls_r_host;
     st.recv_staged = r_host;
     if (s_host) {
         if (!st.stage_send || st.stage_send_bytes < send_b) {

@@ -275,39 +275,61 @@ namespace {
 
 using uint4_ev = __attribute__((ext_vector_type(4))) unsigned int;
 
-// Nontemporal streaming copy: 16-B lanes, 2-deep unroll. Stream-once data
-// must not displace L2 (each XCD's L2 is private; retention buys nothing
-// for a pure copy), so both sides use nontemporal accesses.
+// Streaming copy: 16-B lanes, 2-deep unroll. NT template arm bypasses L2
+// retention (each XCD's L2 is private; retention buys nothing for a pure
+// copy) — A/B-measured against plain accesses.
+template <bool NT>
 __global__ void CopyNTKernel(uint4_ev* __restrict__ dst,
                              const uint4_ev* __restrict__ src, size_t n16) {
     const size_t tid = blockIdx.x * blockDim.x + threadIdx.x;
     const size_t stride = gridDim.x * blockDim.x;
     size_t i = tid;
     for (; i + stride < n16; i += 2 * stride) {
-        uint4_ev a = __builtin_nontemporal_load(src + i);
-        uint4_ev b = __builtin_nontemporal_load(src + i + stride);
-        __builtin_nontemporal_store(a, dst + i);
-        __builtin_nontemporal_store(b, dst + i + stride);
+        uint4_ev a, b;
+        if constexpr (NT) {
+            a = __builtin_nontemporal_load(src + i);
+            b = __builtin_nontemporal_load(src + i + stride);
+            __builtin_nontemporal_store(a, dst + i);
+            __builtin_nontemporal_store(b, dst + i + stride);
+        } else {
+            a = src[i];
+            b = src[i + stride];
+            dst[i] = a;
+            dst[i + stride] = b;
+        }
     }
-    for (; i < n16; i += stride)
-        __builtin_nontemporal_store(__builtin_nontemporal_load(src + i), dst + i);
+    for (; i < n16; i += stride) {
+        if constexpr (NT)
+            __builtin_nontemporal_store(__builtin_nontemporal_load(src + i), dst + i);
+        else
+            dst[i] = src[i];
+    }
 }
 
 }  // namespace
 
-void LaunchCopy(void* dst, const void* src, size_t bytes, hipStream_t stream) {
+void LaunchCopyVariant(void* dst, const void* src, size_t bytes, bool nt,
+                       hipStream_t stream) {
     const uintptr_t d = reinterpret_cast<uintptr_t>(dst);
     const uintptr_t s = reinterpret_cast<uintptr_t>(src);
     // Kernel path: 16-B aligned and big enough that launch cost (~5 us)
     // amortizes against the bandwidth win over the blit path.
     if (((d | s | bytes) & 15) == 0 && bytes >= (1u << 20)) {
         const size_t n16 = bytes / 16;
-        CopyNTKernel<<<dim3(GridFor(n16)), dim3(kBlock), 0, stream>>>(
-            static_cast<uint4_ev*>(dst), static_cast<const uint4_ev*>(src), n16);
+        if (nt)
+            CopyNTKernel<true><<<dim3(GridFor(n16)), dim3(kBlock), 0, stream>>>(
+                static_cast<uint4_ev*>(dst), static_cast<const uint4_ev*>(src), n16);
+        else
+            CopyNTKernel<false><<<dim3(GridFor(n16)), dim3(kBlock), 0, stream>>>(
+                static_cast<uint4_ev*>(dst), static_cast<const uint4_ev*>(src), n16);
         HIP_CHECK(hipGetLastError());
         return;
     }
     HIP_CHECK(hipMemcpyAsync(dst, src, bytes, hipMemcpyDeviceToDevice, stream));
+}
+
+void LaunchCopy(void* dst, const void* src, size_t bytes, hipStream_t stream) {
+    LaunchCopyVariant(dst, src, bytes, /*nt=*/true, stream);
 }
 
 void LaunchReduceOut(void* dst, const void* a, const void* b, size_t count,
@@ -427,7 +449,7 @@ __global__ void QuantizeKernel(const T* __restrict__ in, T* __restrict__ err,
     }
 }
 
-template <typename T>
+template <typename T, bool NT = false>
 __global__ void DequantizeKernel(const uint8_t* __restrict__ wire, T* __restrict__ out,
                                  size_t count, size_t block_elems) {
     const size_t nblocks = (count + block_elems - 1) / block_elems;
@@ -446,15 +468,18 @@ __global__ void DequantizeKernel(const uint8_t* __restrict__ wire, T* __restrict
         if (vec4) {
             const int32_t* p4 = reinterpret_cast<const int32_t*>(payload);
             for (size_t i = lane * 4; i + 3 < n; i += 256) {
-                const int32_t packed = p4[i >> 2];
+                const int32_t packed =
+                    NT ? __builtin_nontemporal_load(p4 + (i >> 2)) : p4[i >> 2];
                 if constexpr (sizeof(T) == 4) {
                     // f32: one 16-B store per lane pass
                     float4_ev v;
 #pragma unroll
                     for (int j = 0; j < 4; ++j)
                         v[j] = static_cast<int8_t>((packed >> (8 * j)) & 0xff) * scale;
-                    *reinterpret_cast<float4_ev*>(
-                        reinterpret_cast<float*>(out) + base + i) = v;
+                    float4_ev* dst4 = reinterpret_cast<float4_ev*>(
+                        reinterpret_cast<float*>(out) + base + i);
+                    if constexpr (NT) __builtin_nontemporal_store(v, dst4);
+                    else *dst4 = v;
                 } else {
 #pragma unroll
                     for (int j = 0; j < 4; ++j) {
@@ -595,6 +620,22 @@ void LaunchDequantize(const void* wire, void* out, size_t count,
     } else {
         MLSL_THROW("dequantization supports f32/bf16 only");
     }
+    HIP_CHECK(hipGetLastError());
+}
+
+void LaunchDequantizeNT(const void* wire, void* out, size_t count,
+                        size_t block_elems, DataType dt, hipStream_t stream) {
+    // f32-only NT variant (A/B benchmark hook): stream-once wire reads and
+    // output stores skip L2 retention.
+    if (dt != DataType::F32) {
+        LaunchDequantize(wire, out, count, block_elems, dt, stream);
+        return;
+    }
+    const size_t nblocks = (count + block_elems - 1) / block_elems;
+    dim3 grid(static_cast<uint32_t>(std::min<size_t>((nblocks + 3) / 4, kMaxGrid)));
+    hipLaunchKernelGGL((DequantizeKernel<float, true>), grid, dim3(kBlock), 0, stream,
+                       static_cast<const uint8_t*>(wire), static_cast<float*>(out),
+                       count, block_elems);
     HIP_CHECK(hipGetLastError());
 }
 

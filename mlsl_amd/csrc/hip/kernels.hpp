@@ -26,6 +26,8 @@ void LaunchReduceOut(void* dst, const void* a, const void* b, size_t count,
 // (same reason the NT reduce wins: no L2 retention for stream-once data);
 // misaligned or small copies fall back to hipMemcpyAsync on `stream`.
 void LaunchCopy(void* dst, const void* src, size_t bytes, hipStream_t stream);
+void LaunchCopyVariant(void* dst, const void* src, size_t bytes, bool nt,
+                       hipStream_t stream);
 
 // --- int8 block quantization with error feedback (quant/quant.c contract,
 //     fused into the allreduce path; see comm/quant.cpp) ---
@@ -38,6 +40,8 @@ void LaunchQuantize(const void* in, void* err, void* wire, size_t count,
 // Dequantize wire blocks into out (fp32/bf16).
 void LaunchDequantize(const void* wire, void* out, size_t count,
                       size_t block_elems, DataType dt, hipStream_t stream);
+void LaunchDequantizeNT(const void* wire, void* out, size_t count,
+                        size_t block_elems, DataType dt, hipStream_t stream);
 // Compressed-domain accumulate: acc_wire += wire (dequant-sum-requant per
 // block, matching the reference's reduce_sum plugin hook quant/quant.c:89-94).
 void LaunchQuantAccum(void* acc_wire, const void* wire, size_t count,

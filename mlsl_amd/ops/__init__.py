@@ -24,17 +24,19 @@ def _lib():
         L.mlsl_hip_reduce.argtypes = [c.c_void_p, c.c_void_p, c.c_size_t, c.c_int, c.c_int]
         L.mlsl_hip_reduce_nt.argtypes = [c.c_void_p, c.c_void_p, c.c_size_t]
         L.mlsl_hip_copy.argtypes = [c.c_void_p, c.c_void_p, c.c_size_t]
+        L.mlsl_hip_copy_variant.argtypes = [c.c_void_p, c.c_void_p, c.c_size_t, c.c_int]
         L.mlsl_hip_quantize.argtypes = [c.c_void_p, c.c_void_p, c.c_void_p, c.c_size_t,
                                         c.c_size_t, c.c_int, c.c_int]
         L.mlsl_hip_dequantize.argtypes = [c.c_void_p, c.c_void_p, c.c_size_t,
                                           c.c_size_t, c.c_int]
+        L.mlsl_hip_dequantize_nt.argtypes = L.mlsl_hip_dequantize.argtypes
         L.mlsl_hip_quant_accum.argtypes = [c.c_void_p, c.c_void_p, c.c_size_t, c.c_size_t]
         ptypes = [c.c_void_p, c.c_void_p] + [c.c_size_t] * 8 + [c.c_int]
         L.mlsl_hip_pack.argtypes = ptypes
         L.mlsl_hip_unpack.argtypes = ptypes
         for n in ("mlsl_hip_device_count", "mlsl_hip_synchronize", "mlsl_hip_reduce",
-                  "mlsl_hip_reduce_nt", "mlsl_hip_copy",
-                  "mlsl_hip_quantize", "mlsl_hip_dequantize", "mlsl_hip_quant_accum",
+                  "mlsl_hip_reduce_nt", "mlsl_hip_copy", "mlsl_hip_copy_variant",
+                  "mlsl_hip_quantize", "mlsl_hip_dequantize", "mlsl_hip_dequantize_nt", "mlsl_hip_quant_accum",
                   "mlsl_hip_pack", "mlsl_hip_unpack"):
             getattr(L, n).restype = c.c_int
         _declared = True
@@ -73,6 +75,12 @@ def copy(dst, src, bytes_):
     check(_lib().mlsl_hip_copy(dp, sp, bytes_))
 
 
+def copy_variant(dst, src, bytes_, nt):
+    dp, _ = _as_ptr_dtype(dst)
+    sp, _ = _as_ptr_dtype(src)
+    check(_lib().mlsl_hip_copy_variant(dp, sp, bytes_, 1 if nt else 0))
+
+
 def wire_bytes(count, block=256):
     """Bytes of the int8 wire format for `count` elements."""
     nblocks = (count + block - 1) // block
@@ -93,6 +101,13 @@ def dequantize(wire, out, count, block=256, dtype=None):
     op_, d1 = _as_ptr_dtype(out)
     dt = dtype or d1
     check(_lib().mlsl_hip_dequantize(wp, op_, count, block, DTYPE[dt]))
+
+
+def dequantize_nt(wire, out, count, block=256, dtype=None):
+    wp, _ = _as_ptr_dtype(wire)
+    op_, d1 = _as_ptr_dtype(out)
+    dt = dtype or d1
+    check(_lib().mlsl_hip_dequantize_nt(wp, op_, count, block, DTYPE[dt]))
 
 
 def quant_accum(acc_wire, wire, count, block=256):
