@@ -499,6 +499,51 @@ __global__ void DequantizeKernel(const uint8_t* __restrict__ wire, T* __restrict
 
 // acc_wire += wire in the compressed domain: dequant both, sum, requant with
 // a fresh scale (the reference's external reduce_sum hook, quant/quant.c:89).
+//
+// Fast path (block_elems == 256, full block): one wave owns one block and
+// each lane owns exactly 4 elements, so the dequantized sums live in
+// registers across the max-reduce and the requant — one load pair, one
+// store, half the VALU of the generic two-pass version below.
+__global__ void QuantAccum256Kernel(uint8_t* __restrict__ acc,
+                                    const uint8_t* __restrict__ in,
+                                    size_t nblocks) {
+    constexpr size_t kBlockElems = 256;
+    constexpr size_t kWireBlock = kBlockElems + 8;
+    const int lane = threadIdx.x & 63;
+    const size_t wave = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+    const size_t wstride = (gridDim.x * blockDim.x) >> 6;
+    for (size_t blk = wave; blk < nblocks; blk += wstride) {
+        uint8_t* ablock = acc + blk * kWireBlock;
+        const uint8_t* iblock = in + blk * kWireBlock;
+        float* ahdr = reinterpret_cast<float*>(ablock);
+        const float as = ahdr[0];
+        const float is = reinterpret_cast<const float*>(iblock)[0];
+        int32_t* a4 = reinterpret_cast<int32_t*>(ablock + 8);
+        const int32_t* i4 = reinterpret_cast<const int32_t*>(iblock + 8);
+        const int32_t pa = a4[lane], pb = i4[lane];
+        float v[4];
+        float m = 0.f;
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+            v[j] = static_cast<int8_t>((pa >> (8 * j)) & 0xff) * as +
+                   static_cast<int8_t>((pb >> (8 * j)) & 0xff) * is;
+            m = fmaxf(m, fabsf(v[j]));
+        }
+        m = WaveMax(m);
+        const float ns = m > 0.f ? m / 127.f : 1.f;
+        const float inv = 1.f / ns;
+        int32_t packed = 0;
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+            float q = nearbyintf(v[j] * inv);
+            q = fminf(127.f, fmaxf(-127.f, q));
+            packed |= (static_cast<int32_t>(q) & 0xff) << (8 * j);
+        }
+        a4[lane] = packed;
+        if (lane == 0) ahdr[0] = ns;
+    }
+}
+
 __global__ void QuantAccumKernel(uint8_t* __restrict__ acc, const uint8_t* __restrict__ in,
                                  size_t count, size_t block_elems) {
     const size_t nblocks = (count + block_elems - 1) / block_elems;
@@ -653,6 +698,13 @@ void LaunchQuantAccum(void* acc_wire, const void* wire, size_t count,
                       size_t block_elems, hipStream_t stream) {
     const size_t nblocks = (count + block_elems - 1) / block_elems;
     dim3 grid(static_cast<uint32_t>(std::min<size_t>((nblocks + 3) / 4, kMaxGrid)));
+    if (block_elems == 256 && count % 256 == 0) {
+        hipLaunchKernelGGL(QuantAccum256Kernel, grid, dim3(kBlock), 0, stream,
+                           static_cast<uint8_t*>(acc_wire),
+                           static_cast<const uint8_t*>(wire), nblocks);
+        HIP_CHECK(hipGetLastError());
+        return;
+    }
     hipLaunchKernelGGL(QuantAccumKernel, grid,
                        dim3(kBlock), 0, stream, static_cast<uint8_t*>(acc_wire),
                        static_cast<const uint8_t*>(wire), count, block_elems);

@@ -93,9 +93,12 @@ class TestKernels:
         drift = (total_in - total_out - err).abs().max().item()
         assert drift < 3 * scale, f"error feedback drifting: {drift} vs step {scale}"
 
-    def test_quant_accum_compressed_domain(self):
+    @pytest.mark.parametrize("block", [256, 128])
+    def test_quant_accum_compressed_domain(self, block):
+        # block=256 exercises the register-resident wave-per-block fast
+        # path; 128 the generic two-pass kernel.
         from mlsl_amd import ops
-        n, block = 1 << 18, 256
+        n = 1 << 18
         a = torch.randn(n, device="cuda")
         b = torch.randn(n, device="cuda")
         wa = torch.empty(ops.wire_bytes(n, block), device="cuda", dtype=torch.uint8)
