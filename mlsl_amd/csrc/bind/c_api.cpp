@@ -8,6 +8,7 @@
 #include <string>
 
 #include "../comm/group.hpp"
+#include "../core/log.hpp"
 #include "../dl/session.hpp"
 #include "../include/mlsl/mlsl.hpp"
 
@@ -678,6 +679,163 @@ int mlsl_statistics_get_total_comm_cycles(mlsl_statistics st, unsigned long long
 int mlsl_statistics_get_total_compute_cycles(mlsl_statistics st, unsigned long long* out) {
     C_TRY* out = STATS(st)->GetTotalComputeCycles();
     C_CATCH
+}
+
+
+/* ---- native getters added for reference parity ---- */
+
+int mlsl_activation_get_comm_buf(mlsl_activation a, void** out) {
+    C_TRY* out = ACT(a)->GetCommBuf();
+    C_CATCH
+}
+int mlsl_statistics_is_started(mlsl_statistics st, int* out) {
+    C_TRY* out = STATS(st)->IsStarted() ? 1 : 0;
+    C_CATCH
+}
+int mlsl_session_get_phase_type(mlsl_session s, mlsl_phase* out) {
+    C_TRY* out = static_cast<mlsl_phase>(SES(s)->Phase());
+    C_CATCH
+}
+int mlsl_operation_get_op_type(mlsl_operation o, mlsl_op_type* out) {
+    C_TRY* out = static_cast<mlsl_op_type>(OP(o)->Kind());
+    C_CATCH
+}
+int mlsl_operation_get_session(mlsl_operation o, mlsl_session* out) {
+    C_TRY* out = OP(o)->GetSession();
+    C_CATCH
+}
+int mlsl_operation_has_parameter_sets(mlsl_operation o, int* out) {
+    C_TRY* out = OP(o)->GetParameterSetCount() > 0 ? 1 : 0;
+    C_CATCH
+}
+int mlsl_get_quant_params(size_t* block_elems) {
+    C_TRY* block_elems = Environment::GetEnv().GetQuantizationParams().block_elems;
+    C_CATCH
+}
+
+/* ---- reference-name compatibility layer (exact include/mlsl.h
+ * signatures; forwards to the native functions above). The environment
+ * handle is a token for the singleton. ---- */
+
+static int g_env_token;
+
+int mlsl_environment_get_env(mlsl_environment* env) {
+    C_TRY* env = &g_env_token;
+    C_CATCH
+}
+int mlsl_environment_get_version(int* version) { return mlsl_get_version(version); }
+int mlsl_environment_configure(mlsl_environment, const char* config) {
+    return mlsl_configure(config);
+}
+int mlsl_environment_init(mlsl_environment, int*, char***) { return mlsl_init(-1, -1); }
+int mlsl_environment_finalize(mlsl_environment) { return mlsl_finalize(); }
+int mlsl_environment_is_initialized(mlsl_environment, int* is_initialized) {
+    return mlsl_initialized(is_initialized);
+}
+int mlsl_environment_get_process_idx(mlsl_environment, size_t* process_idx) {
+    return mlsl_rank(process_idx);
+}
+int mlsl_environment_get_process_count(mlsl_environment, size_t* process_count) {
+    return mlsl_world_size(process_count);
+}
+int mlsl_environment_create_session(mlsl_environment, mlsl_phase phase_type,
+                                    mlsl_session* session) {
+    return mlsl_session_create(phase_type, session);
+}
+int mlsl_environment_delete_session(mlsl_environment, mlsl_session session) {
+    return mlsl_session_free(session);
+}
+int mlsl_environment_create_distribution(mlsl_environment, size_t data_partitions,
+                                         size_t model_partitions, mlsl_distribution* dist) {
+    return mlsl_distribution_create(data_partitions, model_partitions, dist);
+}
+int mlsl_environment_delete_distribution(mlsl_environment, mlsl_distribution dist) {
+    return mlsl_distribution_free(dist);
+}
+int mlsl_environment_wait(mlsl_environment, mlsl_comm_req req) {
+    void* result = nullptr;
+    return mlsl_wait(req, &result);
+}
+int mlsl_environment_test(mlsl_environment, mlsl_comm_req req, int* is_completed) {
+    void* result = nullptr;
+    return mlsl_test(req, is_completed, &result);
+}
+int mlsl_environment_alloc(mlsl_environment, size_t size, size_t alignment, void** ptr) {
+    return mlsl_alloc(size, alignment, ptr);
+}
+int mlsl_environment_free(mlsl_environment, void* ptr) { return mlsl_dealloc(ptr); }
+int mlsl_environment_set_quantization_params(mlsl_environment, mlsl_quant_params_t* params) {
+    C_TRY
+    if (!params || params->elem_in_block == 0)
+        throw Error("quant params: elem_in_block must be > 0");
+    if (params->lib_path && params->lib_path[0])
+        MLSL_LOG(INFO, "quant plugin path '%s' ignored: built-in kernels",
+                 params->lib_path);
+    QuantParams qp;
+    qp.block_elems = params->elem_in_block;
+    Environment::GetEnv().SetQuantizationParams(qp);
+    C_CATCH
+}
+int mlsl_environment_get_quantization_params(mlsl_environment, mlsl_quant_params_t* params) {
+    C_TRY
+    const QuantParams& qp = Environment::GetEnv().GetQuantizationParams();
+    params->lib_path = nullptr;
+    params->quant_buffer_func_name = nullptr;
+    params->dequant_buffer_func_name = nullptr;
+    params->reduce_sum_func_name = nullptr;
+    params->elem_in_block = qp.block_elems;
+    params->block_size = qp.WireBlockBytes();
+    C_CATCH
+}
+int mlsl_distribution_get_process_count(mlsl_distribution dist, mlsl_group group_type,
+                                        size_t* process_count) {
+    return mlsl_distribution_process_count(dist, group_type, process_count);
+}
+int mlsl_distribution_get_process_idx(mlsl_distribution dist, mlsl_group group_type,
+                                      size_t* process_idx) {
+    return mlsl_distribution_process_idx(dist, group_type, process_idx);
+}
+int mlsl_session_create_operation_reg_info(mlsl_session session, mlsl_op_type op_type,
+                                           mlsl_op_reg_info* reg_info) {
+    return mlsl_session_create_op_reg_info(session, op_type, reg_info);
+}
+int mlsl_session_delete_operation_reg_info(mlsl_session session, mlsl_op_reg_info reg_info) {
+    return mlsl_session_delete_op_reg_info(session, reg_info);
+}
+int mlsl_session_add_operation_with_distribution(mlsl_session session,
+                                                 mlsl_op_reg_info reg_info,
+                                                 mlsl_distribution dist, size_t* op_idx) {
+    return mlsl_session_add_operation(session, reg_info, dist, op_idx);
+}
+int mlsl_operation_reg_info_set_name(mlsl_op_reg_info reg_info, const char* name) {
+    return mlsl_op_reg_info_set_name(reg_info, name);
+}
+int mlsl_operation_reg_info_add_input(mlsl_op_reg_info reg_info, size_t fm_count,
+                                      size_t fm_size, mlsl_data_type dtype) {
+    size_t idx = 0;
+    return mlsl_op_reg_info_add_input(reg_info, fm_count, fm_size, dtype, &idx);
+}
+int mlsl_operation_reg_info_add_output(mlsl_op_reg_info reg_info, size_t fm_count,
+                                       size_t fm_size, mlsl_data_type dtype) {
+    size_t idx = 0;
+    return mlsl_op_reg_info_add_output(reg_info, fm_count, fm_size, dtype, &idx);
+}
+int mlsl_operation_reg_info_add_parameter_set(mlsl_op_reg_info reg_info, size_t kernel_count,
+                                              size_t kernel_size, mlsl_data_type dtype,
+                                              int dist_update) {
+    size_t idx = 0;
+    return mlsl_op_reg_info_add_parameter_set(reg_info, kernel_count, kernel_size, dtype,
+                                              dist_update, MLSL_CT_NONE, &idx);
+}
+int mlsl_operation_reg_info_add_parameter_set_with_compress(
+    mlsl_op_reg_info reg_info, size_t kernel_count, size_t kernel_size,
+    mlsl_data_type dtype, int dist_update, mlsl_compression compress_type) {
+    size_t idx = 0;
+    return mlsl_op_reg_info_add_parameter_set(reg_info, kernel_count, kernel_size, dtype,
+                                              dist_update, compress_type, &idx);
+}
+int mlsl_operation_reg_info_validate(mlsl_op_reg_info reg_info, mlsl_distribution dist) {
+    return mlsl_op_reg_info_validate(reg_info, dist);
 }
 
 }  // extern "C"

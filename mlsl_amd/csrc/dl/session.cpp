@@ -69,7 +69,16 @@ Activation::Activation(Operation* op, const RegEntry& re, bool is_input, size_t 
     }
 }
 
-Activation::~Activation() = default;
+Activation::~Activation() {
+    if (owned_comm_buf_ && Context::Initialized())
+        Context::Get().Free(owned_comm_buf_);
+}
+
+void* Activation::GetCommBuf() {
+    if (!comm_buf_bytes_) return nullptr;
+    if (!owned_comm_buf_) owned_comm_buf_ = Context::Get().Alloc(comm_buf_bytes_, 64);
+    return owned_comm_buf_;
+}
 
 void Activation::SetPeer(Activation* act) {
     if (act == nullptr) {
@@ -491,7 +500,8 @@ void Operation::Commit() {
 static inline unsigned long long Now() { return __rdtsc(); }
 
 Statistics::Statistics(Session* s) : session_(s) {
-    enabled_ = GlobalConfig().stats;
+    env_enabled_ = GlobalConfig().stats;
+    started_ = env_enabled_;
     last_ts_ = Now();
 }
 
@@ -500,7 +510,7 @@ void Statistics::EnsureSize(size_t n) {
 }
 
 void Statistics::Update(const Event& ev, size_t bytes) {
-    if (!enabled_) return;
+    if (!env_enabled_ || !started_) return;
     EnsureSize(ev.op_idx + 1);
     const unsigned long long now = Now();
     const unsigned long long delta = now - last_ts_;
@@ -519,7 +529,7 @@ void Statistics::Update(const Event& ev, size_t bytes) {
 }
 
 void Statistics::CollectIsolation() {
-    if (!enabled_) return;
+    if (!env_enabled_ || !started_) return;
     // Reference protocol: 10 iterations, skip the first 4
     // (src/mlsl_impl_stats.cpp:48-49). Each entity's request runs standalone
     // against scratch buffers; ranks iterate in identical order so the

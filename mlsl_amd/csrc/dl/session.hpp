@@ -96,6 +96,11 @@ class Activation {
     const CommBlockInfo* GetPackBlock(size_t i) const { return &pack_blocks_[i]; }
     const CommBlockInfo* GetUnpackBlock(size_t i) const { return &unpack_blocks_[i]; }
     size_t GetCommBufSize() const { return comm_buf_bytes_; }
+    // Library-owned comm buffer (reference Activation::GetCommBuf,
+    // include/mlsl.hpp:210-275): lazily allocated at GetCommBufSize() bytes
+    // from Environment::Alloc (HBM in device mode), freed with the
+    // activation. Users may still pass their own buffer to StartComm.
+    void* GetCommBuf();
 
     void StartComm(void* buf);
     void* WaitComm();
@@ -127,6 +132,7 @@ class Activation {
     size_t send_off_bytes_ = 0;   // added to the user buffer at Start
     size_t recv_off_bytes_ = 0;   // recv region offset within the user buffer
     size_t comm_buf_bytes_ = 0;
+    void* owned_comm_buf_ = nullptr;
     std::vector<CommBlockInfo> pack_blocks_, unpack_blocks_;
 };
 
@@ -224,9 +230,12 @@ class Statistics {
     void Update(const Event& ev, size_t bytes = 0);
     void CollectIsolation();   // 10 iterations, skip 4 (ref :48-49)
 
-    void Start() { enabled_ = true; }
-    void Stop() { enabled_ = false; }
-    bool IsEnabled() const { return enabled_; }
+    // IsEnabled = MLSL_STATS env gate; IsStarted = currently collecting
+    // (reference Statistics::IsEnabled/IsStarted, include/mlsl.hpp:655-660).
+    void Start() { started_ = true; }
+    void Stop() { started_ = false; }
+    bool IsEnabled() const { return env_enabled_; }
+    bool IsStarted() const { return started_; }
     void Reset();
     void Print();
 
@@ -252,7 +261,8 @@ class Statistics {
     void EnsureSize(size_t n);
 
     Session* session_;
-    bool enabled_ = false;
+    bool env_enabled_ = false;
+    bool started_ = false;
     unsigned long long last_ts_ = 0;
     std::vector<OpStats> per_op_;
 };

@@ -209,6 +209,63 @@ int mlsl_statistics_get_total_comm_size(mlsl_statistics st, size_t* out);
 int mlsl_statistics_get_total_comm_cycles(mlsl_statistics st, unsigned long long* out);
 int mlsl_statistics_get_total_compute_cycles(mlsl_statistics st, unsigned long long* out);
 
+/* ---- native getters added for reference parity ---- */
+int mlsl_activation_get_comm_buf(mlsl_activation a, void** out);  /* library-owned, lazy */
+int mlsl_statistics_is_started(mlsl_statistics st, int* out);
+int mlsl_session_get_phase_type(mlsl_session s, mlsl_phase* out);
+int mlsl_operation_get_op_type(mlsl_operation o, mlsl_op_type* out);
+int mlsl_operation_get_session(mlsl_operation o, mlsl_session* out);
+int mlsl_operation_has_parameter_sets(mlsl_operation o, int* out);
+int mlsl_get_quant_params(size_t* block_elems);
+
+/* ---- reference-name compatibility layer (drop-in for include/mlsl.h) ----
+ * Exact reference signatures so migrating C code links unchanged.
+ * mlsl_environment is a token handle for the singleton environment;
+ * mlsl_quant_params_t mirrors the reference struct (the dlopen'd plugin
+ * fields are accepted and ignored: quantization kernels are built in). */
+typedef void* mlsl_environment;
+typedef void* mlsl_comm_req;
+typedef struct {
+    char* lib_path;                  /* ignored: built-in CDNA4/CPU kernels */
+    char* quant_buffer_func_name;    /* ignored */
+    char* dequant_buffer_func_name;  /* ignored */
+    char* reduce_sum_func_name;      /* ignored */
+    size_t block_size;               /* wire block bytes (elem_in_block + 8) */
+    size_t elem_in_block;            /* elements per quantization block */
+} mlsl_quant_params_t;
+
+int mlsl_environment_get_env(mlsl_environment* env);
+int mlsl_environment_get_version(int* version);
+int mlsl_environment_configure(mlsl_environment env, const char* config);
+int mlsl_environment_init(mlsl_environment env, int* argc, char** argv[]);
+int mlsl_environment_finalize(mlsl_environment env);
+int mlsl_environment_is_initialized(mlsl_environment env, int* is_initialized);
+int mlsl_environment_get_process_idx(mlsl_environment env, size_t* process_idx);
+int mlsl_environment_get_process_count(mlsl_environment env, size_t* process_count);
+int mlsl_environment_create_session(mlsl_environment env, mlsl_phase phase_type, mlsl_session* session);
+int mlsl_environment_delete_session(mlsl_environment env, mlsl_session session);
+int mlsl_environment_create_distribution(mlsl_environment env, size_t data_partitions, size_t model_partitions, mlsl_distribution* dist);
+int mlsl_environment_delete_distribution(mlsl_environment env, mlsl_distribution dist);
+int mlsl_environment_wait(mlsl_environment env, mlsl_comm_req req);
+int mlsl_environment_test(mlsl_environment env, mlsl_comm_req req, int* is_completed);
+int mlsl_environment_alloc(mlsl_environment env, size_t size, size_t alignment, void** ptr);
+int mlsl_environment_free(mlsl_environment env, void* ptr);
+int mlsl_environment_set_quantization_params(mlsl_environment env, mlsl_quant_params_t* params);
+int mlsl_environment_get_quantization_params(mlsl_environment env, mlsl_quant_params_t* params);
+int mlsl_distribution_get_process_count(mlsl_distribution dist, mlsl_group group_type, size_t* process_count);
+int mlsl_distribution_get_process_idx(mlsl_distribution dist, mlsl_group group_type, size_t* process_idx);
+int mlsl_session_create_operation_reg_info(mlsl_session session, mlsl_op_type op_type, mlsl_op_reg_info* reg_info);
+int mlsl_session_delete_operation_reg_info(mlsl_session session, mlsl_op_reg_info reg_info);
+int mlsl_session_add_operation_with_distribution(mlsl_session session, mlsl_op_reg_info reg_info, mlsl_distribution dist, size_t* op_idx);
+int mlsl_operation_reg_info_set_name(mlsl_op_reg_info reg_info, const char* name);
+int mlsl_operation_reg_info_add_input(mlsl_op_reg_info reg_info, size_t fm_count, size_t fm_size, mlsl_data_type dtype);
+int mlsl_operation_reg_info_add_output(mlsl_op_reg_info reg_info, size_t fm_count, size_t fm_size, mlsl_data_type dtype);
+int mlsl_operation_reg_info_add_parameter_set(mlsl_op_reg_info reg_info, size_t kernel_count, size_t kernel_size, mlsl_data_type dtype, int dist_update);
+int mlsl_operation_reg_info_add_parameter_set_with_compress(mlsl_op_reg_info reg_info, size_t kernel_count, size_t kernel_size, mlsl_data_type dtype, int dist_update, mlsl_compression compress_type);
+int mlsl_operation_reg_info_validate(mlsl_op_reg_info reg_info, mlsl_distribution dist);
+
+
+
 #ifdef __cplusplus
 }
 #endif

@@ -551,6 +551,128 @@ def stress_random():
 WORKERS["stress_random"] = stress_random
 
 
+
+
+def c_env_compat():
+    """Reference-named C compatibility surface drives a full lifecycle:
+    environment init -> alloc -> distribution -> session/planner commit ->
+    gradient comm -> finalize, all through mlsl_environment_* /
+    mlsl_operation_reg_info_* / mlsl_session_*_operation_reg_info names
+    (reference include/mlsl.h signatures)."""
+    import ctypes
+    from ctypes import byref, c_void_p, c_size_t, c_int, c_char_p
+    from mlsl_amd._lib import lib
+    L = lib()
+
+    def ck(rc):
+        assert rc == 0, ctypes.string_at(L.mlsl_last_error()).decode()
+
+    env = c_void_p()
+    ck(L.mlsl_environment_get_env(byref(env)))
+    ck(L.mlsl_environment_init(env, None, None))
+    try:
+        ini = c_int(0)
+        ck(L.mlsl_environment_is_initialized(env, byref(ini)))
+        assert ini.value == 1
+        n = c_size_t(0)
+        ck(L.mlsl_environment_get_process_count(env, byref(n)))
+        size = n.value
+        ck(L.mlsl_environment_get_process_idx(env, byref(n)))
+        rank = n.value
+
+        ptr = c_void_p()
+        ck(L.mlsl_environment_alloc(env, 4096, 64, byref(ptr)))
+
+        dist = c_void_p()
+        ck(L.mlsl_environment_create_distribution(env, size, 1, byref(dist)))
+        cnt = c_size_t(0)
+        ck(L.mlsl_distribution_get_process_count(dist, 0, byref(cnt)))
+        assert cnt.value == size
+
+        # generic collective through the env-named wait
+        buf = (ctypes.c_float * 128)(*([float(rank)] * 128))
+        req = c_void_p()
+        ck(L.mlsl_distribution_all_reduce(dist, buf, buf, 128, 0, 0, 0, byref(req)))
+        ck(L.mlsl_environment_wait(env, req))
+        want = (size - 1) * size / 2.0
+        assert abs(buf[0] - want) < 1e-6 and abs(buf[127] - want) < 1e-6
+
+        sess = c_void_p()
+        ck(L.mlsl_environment_create_session(env, 0, byref(sess)))
+        ck(L.mlsl_session_set_global_minibatch_size(sess, 8 * size))
+        ri = c_void_p()
+        ck(L.mlsl_session_create_operation_reg_info(sess, 0, byref(ri)))
+        ck(L.mlsl_operation_reg_info_set_name(ri, b"fc1"))
+        ck(L.mlsl_operation_reg_info_add_input(ri, 16, 4, 0))
+        ck(L.mlsl_operation_reg_info_add_output(ri, 16, 4, 0))
+        ck(L.mlsl_operation_reg_info_add_parameter_set(ri, 16, 9, 0, 0))
+        idx = c_size_t(0)
+        ck(L.mlsl_session_add_operation_with_distribution(sess, ri, dist, byref(idx)))
+        ck(L.mlsl_session_delete_operation_reg_info(sess, ri))
+        ck(L.mlsl_session_commit(sess))
+
+        op = c_void_p()
+        ck(L.mlsl_session_get_operation(sess, idx.value, byref(op)))
+        hp = c_int(0)
+        ck(L.mlsl_operation_has_parameter_sets(op, byref(hp)))
+        assert hp.value == 1
+        ot = c_int(-1)
+        ck(L.mlsl_operation_get_op_type(op, byref(ot)))
+        assert ot.value == 0
+        ph = c_int(-1)
+        ck(L.mlsl_session_get_phase_type(sess, byref(ph)))
+        assert ph.value == 0
+
+        ps = c_void_p()
+        ck(L.mlsl_operation_get_parameter_set(op, 0, byref(ps)))
+        import numpy as np
+        g = np.full(16 * 9, float(rank + 1), dtype=np.float32)
+        ck(L.mlsl_parameter_set_start_gradient_comm(
+            ps, g.ctypes.data_as(c_void_p)))
+        res = c_void_p()
+        ck(L.mlsl_parameter_set_wait_gradient_comm(ps, byref(res)))
+        got = np.ctypeslib.as_array(
+            ctypes.cast(res, ctypes.POINTER(ctypes.c_float)), shape=(16 * 9,))
+        assert np.allclose(got, size * (size + 1) / 2.0), got[:4]
+
+        # activation comm buf: library-owned lazy buffer
+        act = c_void_p()
+        ck(L.mlsl_operation_get_input(op, 0, byref(act)))
+        cb = c_void_p()
+        ck(L.mlsl_activation_get_comm_buf(act, byref(cb)))
+        cbs = c_size_t(0)
+        ck(L.mlsl_activation_get_comm_buf_size(act, byref(cbs)))
+        assert (cb.value is not None) == (cbs.value > 0)
+
+        # quant params struct roundtrip
+        class QP(ctypes.Structure):
+            _fields_ = [("lib_path", c_char_p), ("quant", c_char_p),
+                        ("dequant", c_char_p), ("reduce_sum", c_char_p),
+                        ("block_size", c_size_t), ("elem_in_block", c_size_t)]
+        qp = QP(None, None, None, None, 0, 128)
+        ck(L.mlsl_environment_set_quantization_params(env, byref(qp)))
+        qp2 = QP()
+        ck(L.mlsl_environment_get_quantization_params(env, byref(qp2)))
+        assert qp2.elem_in_block == 128 and qp2.block_size == 128 + 8
+        qp.elem_in_block = 256
+        ck(L.mlsl_environment_set_quantization_params(env, byref(qp)))
+
+        st = c_void_p()
+        ck(L.mlsl_session_get_stats(sess, byref(st)))
+        started = c_int(-1)
+        ck(L.mlsl_statistics_is_started(st, byref(started)))
+        assert started.value in (0, 1)
+
+        ck(L.mlsl_environment_delete_session(env, sess))
+        ck(L.mlsl_environment_delete_distribution(env, dist))
+        ck(L.mlsl_environment_free(env, ptr))
+    finally:
+        ck(L.mlsl_environment_finalize(env))
+
+
+WORKERS["c_env_compat"] = c_env_compat
+
+
 def main():
     name = sys.argv[1]
     fn = WORKERS.get(name)
