@@ -103,3 +103,4 @@ clean:
 
 -include $(OBJS:.o=.d)
 -include $(BUILD)/mlsl_amd/csrc/tests/schedule_selftest.d
+-include $(BUILD)/mlsl_amd/csrc/tests/api_selftest.d
