@@ -71,6 +71,20 @@ def main():
     dt = timed(lambda: ops.dequantize_nt(wire, out, n))
     res["dequantize_nt_f32_TBps"] = round((1 + 4) * n / dt / 1e12, 3)
 
+    # bf16 quantize/dequantize (driver config 5 dtype; two-blocks-per-wave
+    # fast path)
+    nb2 = n  # 64M bf16 = 128 MiB
+    abf2 = torch.randn(nb2, device="cuda", dtype=torch.bfloat16)
+    errb = torch.zeros(nb2, device="cuda", dtype=torch.bfloat16)
+    wireb = torch.empty(ops.wire_bytes(nb2), device="cuda", dtype=torch.uint8)
+    dt = timed(lambda: ops.quantize(abf2, wireb, nb2, err=errb, dtype="bf16"))
+    # reads in+err (4B/elem at HBM), writes wire(~1B)+err(2B)
+    res["quantize_bf16_TBps"] = round((4 + 1 + 2) * nb2 / dt / 1e12, 3)
+    res["quantize_bf16_Gelems_s"] = round(nb2 / dt / 1e9, 2)
+    outb = torch.empty_like(abf2)
+    dt = timed(lambda: ops.dequantize(wireb, outb, nb2, dtype="bf16"))
+    res["dequantize_bf16_TBps"] = round((1 + 2) * nb2 / dt / 1e12, 3)
+
     wb = torch.empty_like(wire)
     ops.quantize(b, wb, n)
     dt = timed(lambda: ops.quant_accum(wire, wb, n))

@@ -207,6 +207,14 @@ __global__ void ReduceF16Kernel(__half* __restrict__ dst,
         }                                                                             \
     } while (0)
 
+// Half-wave (32-lane) max: __shfl_xor with width 32 stays inside each half.
+__device__ __forceinline__ float HalfWaveMax(float m) {
+#pragma unroll
+    for (int off = 16; off > 0; off >>= 1)
+        m = fmaxf(m, __shfl_xor(m, off, 32));
+    return m;
+}
+
 template <typename T>
 void LaunchScalarByOp(T* dst, const T* src, size_t n, ReduceOp op, hipStream_t stream) {
     const int grid = GridFor(n);
@@ -497,6 +505,100 @@ __global__ void DequantizeKernel(const uint8_t* __restrict__ wire, T* __restrict
     }
 }
 
+// bf16 fast path (block_elems == 256, whole blocks): one 64-lane wave owns
+// TWO wire blocks — each 32-lane half owns one block at 8 elements (16 B)
+// per lane, doubling the per-lane access width over the generic path
+// (bf16 at 4 elems/lane is only 8-B loads). Scale reduction is a half-wave
+// shuffle; measured A/B against the generic kernel in docs/BENCHMARKS.md.
+using ushort8_t = __attribute__((ext_vector_type(8))) unsigned short;
+using uint2_ev = __attribute__((ext_vector_type(2))) unsigned int;
+
+template <bool USE_ERR>
+__global__ void QuantizeBf16x2Kernel(const unsigned short* __restrict__ in,
+                                     unsigned short* __restrict__ err,
+                                     uint8_t* __restrict__ wire, size_t nblocks) {
+    constexpr size_t kBE = 256, kWB = kBE + 8;
+    const int lane = threadIdx.x & 63;
+    const int half = lane >> 5, sub = lane & 31;
+    const size_t wave = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+    const size_t wstride = (gridDim.x * blockDim.x) >> 6;
+    const size_t npairs = (nblocks + 1) / 2;
+    for (size_t pair = wave; pair < npairs; pair += wstride) {
+        const size_t blk = pair * 2 + half;
+        if (blk >= nblocks) continue;
+        const size_t base = blk * kBE + sub * 8;
+        float v[8];
+        {
+            ushort8_t a = *reinterpret_cast<const ushort8_t*>(in + base);
+            if (USE_ERR) {
+                ushort8_t e = *reinterpret_cast<const ushort8_t*>(err + base);
+#pragma unroll
+                for (int j = 0; j < 8; ++j)
+                    v[j] = __bfloat162float(__hip_bfloat16_raw{a[j]}) +
+                           __bfloat162float(__hip_bfloat16_raw{e[j]});
+            } else {
+#pragma unroll
+                for (int j = 0; j < 8; ++j)
+                    v[j] = __bfloat162float(__hip_bfloat16_raw{a[j]});
+            }
+        }
+        float m = 0.f;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) m = fmaxf(m, fabsf(v[j]));
+        m = HalfWaveMax(m);
+        const float scale = m > 0.f ? m / 127.f : 1.f;
+        const float inv = 1.f / scale;
+        uint8_t* wblock = wire + blk * kWB;
+        if (sub == 0) {
+            float* hdr = reinterpret_cast<float*>(wblock);
+            hdr[0] = scale;
+            hdr[1] = 0.f;
+        }
+        uint2_ev packed{0, 0};
+        ushort8_t res;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+            float q = nearbyintf(v[j] * inv);
+            q = fminf(127.f, fmaxf(-127.f, q));
+            packed[j >> 2] |= (static_cast<unsigned int>(static_cast<int32_t>(q)) & 0xffu)
+                              << (8 * (j & 3));
+            if (USE_ERR) {
+                __hip_bfloat16 hr = __float2bfloat16(v[j] - q * scale);
+                res[j] = reinterpret_cast<unsigned short&>(hr);
+            }
+        }
+        *reinterpret_cast<uint2_ev*>(wblock + 8 + sub * 8) = packed;
+        if (USE_ERR) *reinterpret_cast<ushort8_t*>(err + base) = res;
+    }
+}
+
+__global__ void DequantizeBf16x2Kernel(const uint8_t* __restrict__ wire,
+                                       unsigned short* __restrict__ out,
+                                       size_t nblocks) {
+    constexpr size_t kBE = 256, kWB = kBE + 8;
+    const int lane = threadIdx.x & 63;
+    const int half = lane >> 5, sub = lane & 31;
+    const size_t wave = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+    const size_t wstride = (gridDim.x * blockDim.x) >> 6;
+    const size_t npairs = (nblocks + 1) / 2;
+    for (size_t pair = wave; pair < npairs; pair += wstride) {
+        const size_t blk = pair * 2 + half;
+        if (blk >= nblocks) continue;
+        const uint8_t* wblock = wire + blk * kWB;
+        const float scale = reinterpret_cast<const float*>(wblock)[0];
+        const uint2_ev packed =
+            *reinterpret_cast<const uint2_ev*>(wblock + 8 + sub * 8);
+        ushort8_t o;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+            const int8_t q = static_cast<int8_t>((packed[j >> 2] >> (8 * (j & 3))) & 0xff);
+            __hip_bfloat16 h = __float2bfloat16(static_cast<float>(q) * scale);
+            o[j] = reinterpret_cast<unsigned short&>(h);
+        }
+        *reinterpret_cast<ushort8_t*>(out + blk * kBE + sub * 8) = o;
+    }
+}
+
 // acc_wire += wire in the compressed domain: dequant both, sum, requant with
 // a fresh scale (the reference's external reduce_sum hook, quant/quant.c:89).
 //
@@ -634,7 +736,26 @@ void LaunchQuantize(const void* in, void* err, void* wire, size_t count,
                                static_cast<const float*>(in), static_cast<float*>(err),
                                static_cast<uint8_t*>(wire), count, block_elems);
     } else if (dt == DataType::BF16) {
-        if (use_err)
+        // two-blocks-per-wave fast path: whole 256-elem blocks, 16-B
+        // aligned bf16 buffers (see QuantizeBf16x2Kernel).
+        const bool fast = block_elems == 256 && count % 256 == 0 &&
+                          (reinterpret_cast<uintptr_t>(in) & 15) == 0 &&
+                          (reinterpret_cast<uintptr_t>(err) & 15) == 0 &&
+                          (reinterpret_cast<uintptr_t>(wire) & 7) == 0;
+        if (fast) {
+            const size_t npairs = (nblocks + 1) / 2;
+            dim3 g2(static_cast<uint32_t>(std::min<size_t>((npairs + 3) / 4, kMaxGrid)));
+            if (use_err)
+                hipLaunchKernelGGL((QuantizeBf16x2Kernel<true>), g2, dim3(kBlock), 0, stream,
+                                   static_cast<const unsigned short*>(in),
+                                   static_cast<unsigned short*>(err),
+                                   static_cast<uint8_t*>(wire), nblocks);
+            else
+                hipLaunchKernelGGL((QuantizeBf16x2Kernel<false>), g2, dim3(kBlock), 0, stream,
+                                   static_cast<const unsigned short*>(in),
+                                   static_cast<unsigned short*>(err),
+                                   static_cast<uint8_t*>(wire), nblocks);
+        } else if (use_err)
             hipLaunchKernelGGL((QuantizeKernel<unsigned short, true>), grid, dim3(kBlock), 0, stream,
                                static_cast<const unsigned short*>(in),
                                static_cast<unsigned short*>(err),
@@ -669,6 +790,18 @@ void LaunchDequantize(const void* wire, void* out, size_t count,
                            static_cast<const uint8_t*>(wire), static_cast<float*>(out),
                            count, block_elems);
     } else if (dt == DataType::BF16) {
+        const bool fast = block_elems == 256 && count % 256 == 0 &&
+                          (reinterpret_cast<uintptr_t>(out) & 15) == 0 &&
+                          (reinterpret_cast<uintptr_t>(wire) & 7) == 0;
+        if (fast) {
+            const size_t npairs = (nblocks + 1) / 2;
+            dim3 g2(static_cast<uint32_t>(std::min<size_t>((npairs + 3) / 4, kMaxGrid)));
+            hipLaunchKernelGGL(DequantizeBf16x2Kernel, g2, dim3(kBlock), 0, stream,
+                               static_cast<const uint8_t*>(wire),
+                               static_cast<unsigned short*>(out), nblocks);
+            HIP_CHECK(hipGetLastError());
+            return;
+        }
         hipLaunchKernelGGL((DequantizeKernel<unsigned short>), grid, dim3(kBlock), 0, stream,
                            static_cast<const uint8_t*>(wire),
                            static_cast<unsigned short*>(out), count, block_elems);

@@ -72,6 +72,35 @@ class TestKernels:
         # error feedback: residual = input - quantized
         assert torch.allclose(g, out + err, atol=1e-5)
 
+    @pytest.mark.parametrize("block,n", [(256, 1 << 18), (256, (1 << 18) - 128),
+                                         (128, 1 << 16)])
+    def test_quant_roundtrip_bf16(self, block, n):
+        # (256, 256-multiple) exercises the two-blocks-per-wave bf16 fast
+        # path; the others the generic kernel.
+        from mlsl_amd import ops
+        a = torch.randn(n, device="cuda", dtype=torch.bfloat16)
+        err = torch.zeros(n, device="cuda", dtype=torch.bfloat16)
+        wire = torch.empty(ops.wire_bytes(n, block), device="cuda",
+                           dtype=torch.uint8)
+        ops.quantize(a, wire, n, err=err, block=block, dtype="bf16")
+        out = torch.empty_like(a)
+        ops.dequantize(wire, out, n, block=block, dtype="bf16")
+        torch.cuda.synchronize()
+        af = a.float()
+        # per-block scale: max abs / 127 -> error bounded by ~scale/2 (+
+        # bf16 rounding); residual holds the rest
+        nb = (n + block - 1) // block
+        pad = nb * block - n
+        blocks = torch.nn.functional.pad(af, (0, pad)).view(nb, block)
+        step = blocks.abs().max(dim=1).values / 127.0
+        err_bound = step.repeat_interleave(block)[:n] * 0.6 + 0.02 * af.abs()
+        diff = (out.float() - af).abs()
+        assert (diff <= err_bound + 1e-6).float().mean().item() > 0.999, \
+            f"bf16 roundtrip error too large: {diff.max()}"
+        # error feedback: residual + dequant ~= original
+        rec = out.float() + err.float()
+        assert (rec - af).abs().max().item() < 0.05 * af.abs().max().item() + 0.05
+
     def test_quant_error_feedback_accumulates(self):
         from mlsl_amd import ops
         n, block = 1 << 16, 256
