@@ -218,11 +218,14 @@ void DeviceSetupRequest(CommRequest* req, DeviceReqState& st) {
     // staging), and the barrier token. The fused RCCL path allocates
     // nothing — one-shot hot-loop requests must not pay a hipMalloc+memset
     // per iteration.
+    // 16-B-align each region so the vectorized quant kernels' fast paths
+    // apply (wire blocks are 264 B, so raw offsets are only 8-B aligned).
+    auto al16 = [](size_t x) { return (x + 15) & ~size_t(15); };
     size_t tmp = 0;
     if (req->Compressed()) {
         for (auto& ce : req->Chunks())
-            tmp += ce.sch.result.bytes + ce.sch.tmp_bytes +
-                   req->Spec().count * DtypeSize(req->Dtype());
+            tmp += al16(ce.sch.result.bytes) + al16(ce.sch.tmp_bytes) +
+                   al16(req->Spec().count * DtypeSize(req->Dtype()));
     } else if (req->UsesDeviceSchedule()) {
         for (auto& ce : req->Chunks()) tmp += ce.sch.tmp_bytes;
     }
@@ -610,7 +613,8 @@ bool DeviceAdvanceRequest(CommRequest* req, DeviceReqState& st) {
                     // quantize -> dequantize keeps quantized-allreduce
                     // semantics (and error feedback) at n=1
                     uint8_t* wire = static_cast<uint8_t*>(st.tmp_dev);
-                    uint8_t* err = wire + ce.sch.result.bytes + ce.sch.tmp_bytes;
+                    uint8_t* err = wire + ((ce.sch.result.bytes + 15) & ~size_t(15)) +
+                                   ((ce.sch.tmp_bytes + 15) & ~size_t(15));
                     const size_t blk = req->QParams().block_elems;
                     LaunchQuantize(sbase, err, wire, req->Spec().count, blk,
                                    req->Dtype(), true, s0);
@@ -638,17 +642,18 @@ bool DeviceAdvanceRequest(CommRequest* req, DeviceReqState& st) {
             if (compressed) {
                 // quantize -> compressed-domain ring -> dequantize (driver
                 // config 5: int8 allreduce of bf16/f32 grads)
-                const size_t wire_b = ce.sch.result.bytes;
+                const size_t wire_b = (ce.sch.result.bytes + 15) & ~size_t(15);
                 uint8_t* wire = tbase;
                 uint8_t* scratch = tbase + wire_b;
-                uint8_t* err = scratch + ce.sch.tmp_bytes;
+                uint8_t* err = scratch + ((ce.sch.tmp_bytes + 15) & ~size_t(15));
                 const size_t blk = req->QParams().block_elems;
                 LaunchQuantize(req->SendBuf(), err, wire, req->Spec().count, blk,
                                req->Dtype(), true, strm_);
                 IssueSchedule(req, ce, comm_, strm_, wire, wire, scratch);
                 LaunchDequantize(wire, req->RecvBuf(), req->Spec().count, blk,
                                  req->Dtype(), strm_);
-                tmp_off += wire_b + ce.sch.tmp_bytes + req->Spec().count * es;
+                tmp_off += wire_b + ((ce.sch.tmp_bytes + 15) & ~size_t(15)) +
+                           ((req->Spec().count * es + 15) & ~size_t(15));
             } else if (use_schedule) {
                 IssueSchedule(req, ce, comm_, strm_,
                               req->SendBuf() + ce.elem_off * es,
