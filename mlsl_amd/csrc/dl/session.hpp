@@ -184,6 +184,8 @@ class Operation {
     Distribution* GetDistribution() const { return dist_; }
     const char* GetName() const { return name_.c_str(); }
     OpKind Kind() const { return kind_; }
+    OpKind GetOpType() const { return kind_; }            // reference name
+    bool HasParameterSets() const { return !params_.empty(); }
     size_t OpIndex() const { return op_idx_; }
     Session* GetSession() const { return session_; }
 
@@ -275,6 +277,7 @@ class Session {
     void SetGlobalMinibatchSize(size_t mb);
     size_t GetGlobalMinibatchSize() const { return global_mb_; }
     PhaseKind Phase() const { return phase_; }
+    PhaseKind GetPhaseType() const { return phase_; }     // reference name
 
     OperationRegInfo* CreateOperationRegInfo(OpKind kind);
     void DeleteOperationRegInfo(OperationRegInfo* i);
