@@ -64,3 +64,7 @@ def test_cpp_api_selftest(mp, du):
         out, _ = p.communicate(timeout=120)
         assert p.returncode == 0, f"rank {r}: {out}"
         assert "PASSED" in out
+
+
+def test_train_ddp_sample():
+    run_ranks("train_ddp_sample", 2, timeout=240)

@@ -673,6 +673,23 @@ def c_env_compat():
 WORKERS["c_env_compat"] = c_env_compat
 
 
+
+
+def train_ddp_sample():
+    """samples/train_ddp.py as a worker: full DDP training loop, asserts
+    cross-rank parameter identity (exit code is the verdict)."""
+    import runpy
+    import os
+    os.environ["STEPS"] = "5"
+    import sys
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    runpy.run_path(os.path.join(repo, "samples", "train_ddp.py"),
+                   run_name="__main__")
+
+
+WORKERS["train_ddp_sample"] = train_ddp_sample
+
+
 def main():
     name = sys.argv[1]
     fn = WORKERS.get(name)
