@@ -54,7 +54,8 @@ class TestKernels:
         assert torch.allclose(a, want)
         assert torch.equal(a32, want32)
 
-    @pytest.mark.parametrize("n,block", [(1 << 20, 256), (100_000, 256), (5000, 512)])
+    @pytest.mark.parametrize("n,block", [(1 << 20, 256), (100_000, 256), (5000, 512),
+                                         (1 << 21, 256)])  # >=4MiB: f32 x2 fast path
     def test_quant_roundtrip_f32(self, n, block):
         from mlsl_amd import ops
         g = torch.randn(n, device="cuda") * 3.0
