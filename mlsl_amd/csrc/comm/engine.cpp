@@ -58,6 +58,13 @@ void Engine::Submit(CommRequest* req) {
         std::lock_guard<std::mutex> lk(inbox_mu_);
         inbox_overflow_.push_back(req);
     }
+    if (deep_idle_.load(std::memory_order_acquire)) {
+        // Pairs with the Loop's check-then-wait under idle_mu_: taking the
+        // lock here means the loop either saw the push or is parked and
+        // gets the notify.
+        std::lock_guard<std::mutex> lk(idle_mu_);
+        idle_cv_.notify_one();
+    }
 }
 
 void Engine::DrainInbox() {
@@ -144,13 +151,17 @@ void Engine::Loop() {
             NotifyDone();
         }
         if (active_.empty() && before == 0) {
-            // exponential idle backoff: hot for the first ~256 polls, then
-            // 50 us sleeps, growing to 1 ms after sustained idleness so an
-            // idle rank does not burn a core (device mode rarely uses the
-            // engine at all).
-            if (++idle_spins > 256) {
-                const int us = idle_spins > 4096 ? 1000 : 50;
-                std::this_thread::sleep_for(std::chrono::microseconds(us));
+            // Hot for the first ~4096 polls (sub-ms window covering the
+            // gaps of a busy training loop), then park on the condvar;
+            // Submit nudges it, and the bounded wait re-checks the ring so
+            // a missed nudge costs at most one timeout.
+            if (++idle_spins > 4096) {
+                std::unique_lock<std::mutex> lk(idle_mu_);
+                deep_idle_.store(true, std::memory_order_release);
+                if (ring_.Empty() && inbox_overflow_.empty() &&
+                    !stop_.load(std::memory_order_acquire))
+                    idle_cv_.wait_for(lk, std::chrono::milliseconds(1));
+                deep_idle_.store(false, std::memory_order_release);
             }
         } else {
             idle_spins = 0;

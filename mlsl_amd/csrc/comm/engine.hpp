@@ -39,6 +39,10 @@ class SpscRing {
         tail_.store(t + 1, std::memory_order_release);
         return true;
     }
+    bool Empty() const {
+        return head_.load(std::memory_order_acquire) ==
+               tail_.load(std::memory_order_acquire);
+    }
     CommRequest* Pop() {
         const uint64_t h = head_.load(std::memory_order_relaxed);
         if (h == tail_.load(std::memory_order_acquire)) return nullptr;
@@ -88,6 +92,14 @@ class Engine {
     std::vector<CommRequest*> active_;
     std::mutex done_mu_;
     std::condition_variable done_cv_;
+
+    // Deep-idle parking: after the hot spin window the loop waits on this
+    // condvar (bounded) instead of sleeping blind; Submit nudges it so
+    // wake-up latency is a notify, not a sleep quantum (the reference's
+    // ep_server spins 100% forever — we spin hot briefly, then park).
+    std::mutex idle_mu_;
+    std::condition_variable idle_cv_;
+    std::atomic<bool> deep_idle_{false};
 
     friend class CommRequest;
     void NotifyDone();
