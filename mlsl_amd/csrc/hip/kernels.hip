@@ -989,6 +989,52 @@ __global__ void PackKernel(const T* __restrict__ src, T* __restrict__ dst,
     }
 }
 
+// Magic-number unsigned division (libdivide-style): q = hi32(n * M) >> s,
+// valid for 32-bit n and d. Replaces the per-element 64-bit div/mod pair
+// that dominated the naive pack kernel (integer division has no hardware
+// unit on CDNA4 — it expands to a long instruction sequence).
+struct FastDiv {
+    uint32_t M;
+    uint32_t s;
+};
+
+inline FastDiv MakeFastDiv(uint32_t d) {
+    FastDiv r;
+    if (d == 1) {
+        r.M = 0xFFFFFFFFu;
+        r.s = 0;
+        return r;
+    }
+    uint32_t lg = 0;
+    while ((1ull << lg) < d) ++lg;
+    r.M = static_cast<uint32_t>(((1ull << (32 + lg)) + d - 1) / d);
+    r.s = lg;
+    return r;
+}
+
+__device__ __forceinline__ uint32_t FDiv(uint32_t n, FastDiv f) {
+    return static_cast<uint32_t>((static_cast<uint64_t>(n) * f.M) >> 32) >> f.s;
+}
+
+template <typename T, bool PACK>
+__global__ void PackFastKernel(const T* __restrict__ src, T* __restrict__ dst,
+                               PackBlockDesc d, FastDiv dfs, FastDiv dfc,
+                               uint32_t total) {
+    const uint32_t tid = blockIdx.x * blockDim.x + threadIdx.x;
+    const uint32_t stride = gridDim.x * blockDim.x;
+    for (uint32_t i = tid; i < total; i += stride) {
+        const uint32_t row = FDiv(i, dfs);                    // i / fm_size
+        const uint32_t k = i - row * static_cast<uint32_t>(d.fm_size);
+        const uint32_t mb = FDiv(row, dfc);                   // row / fm_count
+        const uint32_t fm = row - mb * static_cast<uint32_t>(d.fm_count);
+        const size_t layer_idx =
+            ((d.mb_offset + mb) * d.local_fm_count + d.fm_offset + fm) * d.fm_size + k;
+        const size_t buf_idx = d.buf_offset + i;
+        if (PACK) dst[buf_idx] = src[layer_idx];
+        else dst[layer_idx] = src[buf_idx];
+    }
+}
+
 }  // namespace
 
 namespace {
@@ -1005,35 +1051,54 @@ bool Pack16(const PackBlockDesc& d, size_t es, PackBlockDesc* out) {
 
 }  // namespace
 
+namespace {
+
+// Launch the magic-div kernel when totals fit 32 bits (any realistic
+// activation block), else the size_t-div fallback.
+template <typename T, bool PACK>
+void LaunchPackTyped(const T* src, T* dst, const PackBlockDesc& d,
+                     hipStream_t stream) {
+    const size_t total = d.mb_count * d.fm_count * d.fm_size;
+    const int grid = GridFor(total);
+    if (total < (1ull << 32) && d.fm_size < (1ull << 32) &&
+        d.fm_count < (1ull << 32)) {
+        PackFastKernel<T, PACK><<<dim3(grid), dim3(kBlock), 0, stream>>>(
+            src, dst, d, MakeFastDiv(static_cast<uint32_t>(d.fm_size)),
+            MakeFastDiv(static_cast<uint32_t>(d.fm_count)),
+            static_cast<uint32_t>(total));
+    } else {
+        PackKernel<T, PACK><<<dim3(grid), dim3(kBlock), 0, stream>>>(src, dst, d);
+    }
+    HIP_CHECK(hipGetLastError());
+}
+
+}  // namespace
+
 void LaunchPack(const void* src, void* dst, const PackBlockDesc& d, DataType dt,
                 hipStream_t stream) {
     const size_t es = DtypeSize(dt);
     PackBlockDesc d16;
     if (Pack16(d, es, &d16)) {
-        const size_t total16 = d16.mb_count * d16.fm_count * d16.fm_size;
-        PackKernel<uint4, true><<<dim3(GridFor(total16)), dim3(kBlock), 0, stream>>>(
-            static_cast<const uint4*>(src), static_cast<uint4*>(dst), d16);
-        HIP_CHECK(hipGetLastError());
+        LaunchPackTyped<uint4, true>(static_cast<const uint4*>(src),
+                                     static_cast<uint4*>(dst), d16, stream);
         return;
     }
-    const size_t total = d.mb_count * d.fm_count * d.fm_size;
-    const int grid = GridFor(total);
     switch (DtypeSize(dt)) {
         case 4:
-            hipLaunchKernelGGL((PackKernel<uint32_t, true>), dim3(grid), dim3(kBlock), 0, stream,
-                               static_cast<const uint32_t*>(src), static_cast<uint32_t*>(dst), d);
+            LaunchPackTyped<uint32_t, true>(static_cast<const uint32_t*>(src),
+                               static_cast<uint32_t*>(dst), d, stream);
             break;
         case 8:
-            hipLaunchKernelGGL((PackKernel<uint64_t, true>), dim3(grid), dim3(kBlock), 0, stream,
-                               static_cast<const uint64_t*>(src), static_cast<uint64_t*>(dst), d);
+            LaunchPackTyped<uint64_t, true>(static_cast<const uint64_t*>(src),
+                               static_cast<uint64_t*>(dst), d, stream);
             break;
         case 2:
-            hipLaunchKernelGGL((PackKernel<uint16_t, true>), dim3(grid), dim3(kBlock), 0, stream,
-                               static_cast<const uint16_t*>(src), static_cast<uint16_t*>(dst), d);
+            LaunchPackTyped<uint16_t, true>(static_cast<const uint16_t*>(src),
+                               static_cast<uint16_t*>(dst), d, stream);
             break;
         default:
-            hipLaunchKernelGGL((PackKernel<uint8_t, true>), dim3(grid), dim3(kBlock), 0, stream,
-                               static_cast<const uint8_t*>(src), static_cast<uint8_t*>(dst), d);
+            LaunchPackTyped<uint8_t, true>(static_cast<const uint8_t*>(src),
+                               static_cast<uint8_t*>(dst), d, stream);
     }
     HIP_CHECK(hipGetLastError());
 }
@@ -1043,30 +1108,26 @@ void LaunchUnpack(const void* src, void* dst, const PackBlockDesc& d, DataType d
     const size_t es = DtypeSize(dt);
     PackBlockDesc d16;
     if (Pack16(d, es, &d16)) {
-        const size_t total16 = d16.mb_count * d16.fm_count * d16.fm_size;
-        PackKernel<uint4, false><<<dim3(GridFor(total16)), dim3(kBlock), 0, stream>>>(
-            static_cast<const uint4*>(src), static_cast<uint4*>(dst), d16);
-        HIP_CHECK(hipGetLastError());
+        LaunchPackTyped<uint4, false>(static_cast<const uint4*>(src),
+                                      static_cast<uint4*>(dst), d16, stream);
         return;
     }
-    const size_t total = d.mb_count * d.fm_count * d.fm_size;
-    const int grid = GridFor(total);
     switch (DtypeSize(dt)) {
         case 4:
-            hipLaunchKernelGGL((PackKernel<uint32_t, false>), dim3(grid), dim3(kBlock), 0, stream,
-                               static_cast<const uint32_t*>(src), static_cast<uint32_t*>(dst), d);
+            LaunchPackTyped<uint32_t, false>(static_cast<const uint32_t*>(src),
+                               static_cast<uint32_t*>(dst), d, stream);
             break;
         case 8:
-            hipLaunchKernelGGL((PackKernel<uint64_t, false>), dim3(grid), dim3(kBlock), 0, stream,
-                               static_cast<const uint64_t*>(src), static_cast<uint64_t*>(dst), d);
+            LaunchPackTyped<uint64_t, false>(static_cast<const uint64_t*>(src),
+                               static_cast<uint64_t*>(dst), d, stream);
             break;
         case 2:
-            hipLaunchKernelGGL((PackKernel<uint16_t, false>), dim3(grid), dim3(kBlock), 0, stream,
-                               static_cast<const uint16_t*>(src), static_cast<uint16_t*>(dst), d);
+            LaunchPackTyped<uint16_t, false>(static_cast<const uint16_t*>(src),
+                               static_cast<uint16_t*>(dst), d, stream);
             break;
         default:
-            hipLaunchKernelGGL((PackKernel<uint8_t, false>), dim3(grid), dim3(kBlock), 0, stream,
-                               static_cast<const uint8_t*>(src), static_cast<uint8_t*>(dst), d);
+            LaunchPackTyped<uint8_t, false>(static_cast<const uint8_t*>(src),
+                               static_cast<uint8_t*>(dst), d, stream);
     }
     HIP_CHECK(hipGetLastError());
 }
