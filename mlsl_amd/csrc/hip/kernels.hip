@@ -989,31 +989,24 @@ __global__ void PackKernel(const T* __restrict__ src, T* __restrict__ dst,
     }
 }
 
-// Magic-number unsigned division (libdivide-style): q = hi32(n * M) >> s,
-// valid for 32-bit n and d. Replaces the per-element 64-bit div/mod pair
-// that dominated the naive pack kernel (integer division has no hardware
-// unit on CDNA4 — it expands to a long instruction sequence).
+// Magic-number unsigned division: M = ceil(2^64 / d); for n < 2^32 the
+// high half of n*M is exactly floor(n/d) (error term n*e/(d*2^64) < 2^-32).
+// Replaces the per-element 64-bit div/mod pair that dominated the naive
+// pack kernel (integer division has no hardware unit on CDNA4 — it
+// expands to a long instruction sequence).
 struct FastDiv {
-    uint32_t M;
-    uint32_t s;
+    uint64_t M;     // ceil(2^64 / d); 0 means d == 1 (identity)
 };
 
 inline FastDiv MakeFastDiv(uint32_t d) {
     FastDiv r;
-    if (d == 1) {
-        r.M = 0xFFFFFFFFu;
-        r.s = 0;
-        return r;
-    }
-    uint32_t lg = 0;
-    while ((1ull << lg) < d) ++lg;
-    r.M = static_cast<uint32_t>(((1ull << (32 + lg)) + d - 1) / d);
-    r.s = lg;
+    r.M = d <= 1 ? 0 : (~0ull / d) + 1;
     return r;
 }
 
 __device__ __forceinline__ uint32_t FDiv(uint32_t n, FastDiv f) {
-    return static_cast<uint32_t>((static_cast<uint64_t>(n) * f.M) >> 32) >> f.s;
+    if (f.M == 0) return n;
+    return static_cast<uint32_t>(__umul64hi(static_cast<uint64_t>(n), f.M));
 }
 
 template <typename T, bool PACK>
