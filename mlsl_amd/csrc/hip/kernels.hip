@@ -1009,7 +1009,7 @@ __device__ __forceinline__ uint32_t FDiv(uint32_t n, FastDiv f) {
     return static_cast<uint32_t>(__umul64hi(static_cast<uint64_t>(n), f.M));
 }
 
-template <typename T, bool PACK>
+template <typename T, bool PACK, bool NT = false>
 __global__ void PackFastKernel(const T* __restrict__ src, T* __restrict__ dst,
                                PackBlockDesc d, FastDiv dfs, FastDiv dfc,
                                uint32_t total) {
@@ -1023,8 +1023,21 @@ __global__ void PackFastKernel(const T* __restrict__ src, T* __restrict__ dst,
         const size_t layer_idx =
             ((d.mb_offset + mb) * d.local_fm_count + d.fm_offset + fm) * d.fm_size + k;
         const size_t buf_idx = d.buf_offset + i;
-        if (PACK) dst[buf_idx] = src[layer_idx];
-        else dst[layer_idx] = src[buf_idx];
+        const T* s_ = PACK ? src + layer_idx : src + buf_idx;
+        T* d_ = PACK ? dst + buf_idx : dst + layer_idx;
+        if constexpr (NT) {
+            if constexpr (sizeof(T) == 16) {
+                // uint4 is a struct type; the NT builtins need an
+                // ext-vector-compatible pointee.
+                __builtin_nontemporal_store(
+                    __builtin_nontemporal_load(reinterpret_cast<const uint4_ev*>(s_)),
+                    reinterpret_cast<uint4_ev*>(d_));
+            } else {
+                __builtin_nontemporal_store(__builtin_nontemporal_load(s_), d_);
+            }
+        } else {
+            *d_ = *s_;
+        }
     }
 }
 
@@ -1055,10 +1068,14 @@ void LaunchPackTyped(const T* src, T* dst, const PackBlockDesc& d,
     const int grid = GridFor(total);
     if (total < (1ull << 32) && d.fm_size < (1ull << 32) &&
         d.fm_count < (1ull << 32)) {
-        PackFastKernel<T, PACK><<<dim3(grid), dim3(kBlock), 0, stream>>>(
-            src, dst, d, MakeFastDiv(static_cast<uint32_t>(d.fm_size)),
-            MakeFastDiv(static_cast<uint32_t>(d.fm_count)),
-            static_cast<uint32_t>(total));
+        const FastDiv dfs = MakeFastDiv(static_cast<uint32_t>(d.fm_size));
+        const FastDiv dfc = MakeFastDiv(static_cast<uint32_t>(d.fm_count));
+        if (total * sizeof(T) >= (16u << 20))
+            PackFastKernel<T, PACK, true><<<dim3(grid), dim3(kBlock), 0, stream>>>(
+                src, dst, d, dfs, dfc, static_cast<uint32_t>(total));
+        else
+            PackFastKernel<T, PACK, false><<<dim3(grid), dim3(kBlock), 0, stream>>>(
+                src, dst, d, dfs, dfc, static_cast<uint32_t>(total));
     } else {
         PackKernel<T, PACK><<<dim3(grid), dim3(kBlock), 0, stream>>>(src, dst, d);
     }
