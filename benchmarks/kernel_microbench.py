@@ -90,8 +90,10 @@ def main():
     dt = timed(lambda: ops.quant_accum(wire, wb, n))
     res["quant_accum_TBps"] = round(3 * 1 * n / dt / 1e12, 3)
 
-    # pack: half the fms of a [mb][fm][s] block
-    mb, fm, s = 64, 512, 256
+    # pack: half the fms of a [mb][fm][s] block. Sized so the move is
+    # ~128 MiB: at 16 MiB the old size, per-call sync+launch (~12 us)
+    # dominated the ~7 us copy and under-reported bandwidth 2.5x.
+    mb, fm, s = 64, 2048, 512
     src = torch.randn(mb * fm * s, device="cuda")
     dst = torch.empty(mb * (fm // 2) * s, device="cuda")
     kw = dict(mb_offset=0, mb_count=mb, fm_offset=fm // 4, fm_count=fm // 2,
