@@ -67,6 +67,14 @@ $(APITEST): $(BUILD)/mlsl_amd/csrc/tests/api_selftest.o $(LIB)
 	@mkdir -p $(dir $@)
 	$(HIPCC) $< -Lmlsl_amd -lmlsl_amd -Wl,-rpath,'$$ORIGIN/../mlsl_amd' -pthread -o $@
 
+BENCHLAT := $(BUILD)/bench_latency
+
+benchlat: $(BENCHLAT)
+
+$(BENCHLAT): $(BUILD)/mlsl_amd/csrc/tests/bench_latency.o $(LIB)
+	@mkdir -p $(dir $@)
+	$(HIPCC) $< -Lmlsl_amd -lmlsl_amd -Wl,-rpath,'$$ORIGIN/../mlsl_amd' -pthread -o $@
+
 SAMPLES := $(BUILD)/mlsl_sample $(BUILD)/cmlsl_sample
 
 samples: $(SAMPLES)
@@ -104,3 +112,4 @@ clean:
 -include $(OBJS:.o=.d)
 -include $(BUILD)/mlsl_amd/csrc/tests/schedule_selftest.d
 -include $(BUILD)/mlsl_amd/csrc/tests/api_selftest.d
+-include $(BUILD)/mlsl_amd/csrc/tests/bench_latency.d

@@ -121,8 +121,10 @@ bool Engine::AdvanceOne(CommRequest* req) {
     }
 }
 
-void Engine::ProgressAll() {
+bool Engine::ProgressAll() {
+    const size_t before_drain = active_.size();
     DrainInbox();
+    bool did_work = active_.size() != before_drain;
     if (mesh_) mesh_->Progress();
     bool any_done = false;
     for (size_t i = 0; i < active_.size();) {
@@ -134,14 +136,15 @@ void Engine::ProgressAll() {
         }
     }
     if (any_done) NotifyDone();
+    return did_work || any_done || !active_.empty();
 }
 
 void Engine::Loop() {
     int idle_spins = 0;
     while (!stop_.load(std::memory_order_acquire)) {
-        const size_t before = active_.size();
+        bool did_work = false;
         try {
-            ProgressAll();
+            did_work = ProgressAll();
         } catch (const std::exception& e) {
             // Transport-level failure (not attributable to one request):
             // fail everything in flight rather than terminating the process.
@@ -150,7 +153,7 @@ void Engine::Loop() {
             active_.clear();
             NotifyDone();
         }
-        if (active_.empty() && before == 0) {
+        if (!did_work) {
             // Hot for the first ~4096 polls (sub-ms window covering the
             // gaps of a busy training loop), then park on the condvar;
             // Submit nudges it, and the bounded wait re-checks the ring so

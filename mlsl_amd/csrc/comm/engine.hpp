@@ -76,7 +76,7 @@ class Engine {
     void Loop();
     bool AdvanceOne(CommRequest* req);
     void DrainInbox();
-    void ProgressAll();
+    bool ProgressAll();  // true if it drained, advanced or completed work
 
     Mesh* mesh_;
     ProgressMode mode_;
