@@ -128,6 +128,11 @@ void TcpSetNonBlocking(int fd, bool nb) {
 void TcpSetNoDelay(int fd) {
     int one = 1;
     ::setsockopt(fd, IPPROTO_TCP, TCP_NODELAY, &one, sizeof(one));
+    // Large socket buffers: the mesh streams multi-MB collective segments;
+    // default 200KB-ish buffers stall the nonblocking sender.
+    int buf = 4 << 20;
+    ::setsockopt(fd, SOL_SOCKET, SO_SNDBUF, &buf, sizeof(buf));
+    ::setsockopt(fd, SOL_SOCKET, SO_RCVBUF, &buf, sizeof(buf));
 }
 
 Bootstrap::Bootstrap(int rank, int size) {
