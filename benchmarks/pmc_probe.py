@@ -46,7 +46,24 @@ def main():
     for _ in range(ITERS):
         ops.quantize(a, wire, n, err=err)   # QuantizeKernel
     for _ in range(ITERS):
-        ops.dequantize(wire, out, n)  # DequantizeKernel
+        ops.dequantize(wire, out, n)  # DequantizeF32x2Kernel
+    # bf16 two-blocks-per-wave pair
+    abf = torch.randn(n, device="cuda", dtype=torch.bfloat16)
+    errb = torch.zeros(n, device="cuda", dtype=torch.bfloat16)
+    wireb = torch.empty(ops.wire_bytes(n), device="cuda", dtype=torch.uint8)
+    outb = torch.empty_like(abf)
+    for _ in range(ITERS):
+        ops.quantize(abf, wireb, n, err=errb, dtype="bf16")  # QuantizeBf16x2Kernel
+    for _ in range(ITERS):
+        ops.dequantize(wireb, outb, n, dtype="bf16")  # DequantizeBf16x2Kernel
+    # pack: half the fms of a 256 MiB block (PackFastKernel<uint4,NT>)
+    mb, fm, sdim = 64, 2048, 512
+    src = torch.randn(mb * fm * sdim, device="cuda")
+    dst = torch.empty(mb * (fm // 2) * sdim, device="cuda")
+    for _ in range(ITERS):
+        ops.pack(src, dst, mb_offset=0, mb_count=mb, fm_offset=fm // 4,
+                 fm_count=fm // 2, fm_size=sdim, buf_offset=0,
+                 local_fm_count=fm, local_mb_count=mb, dtype="f32")
     torch.cuda.synchronize()
     print("pmc probe done")
 
