@@ -255,3 +255,39 @@ class TestGraphReplay:
             mx.finalize()
         finally:
             os.environ.pop("MLSL_USE_GRAPHS", None)
+
+
+@pytest.mark.gpu
+class TestSoak:
+    def test_no_device_memory_growth(self):
+        """200 one-shot + 200 persistent request iterations must not grow
+        device memory: scratch comes from the HBM pool, one-shot requests
+        are freed at wait (RequestStorage removal), events are reused."""
+        import mlsl_amd as mx
+        import torch
+        mx.init()
+        d = mx.Distribution(1, 1)
+        n = 1 << 20
+        a = torch.randn(n, device="cuda")
+        b = torch.empty_like(a)
+        # warm every path once (pool high-water, events, staging)
+        preq = mx.PersistentRequest(d, "all_reduce", n, dtype="f32", op="sum",
+                                    group="data")
+        qreq = mx.PersistentRequest(d, "all_reduce", n, dtype="f32", op="sum",
+                                    group="data", quantized=True)
+        for _ in range(3):
+            mx.wait(d.all_reduce(a, b, n, op="sum", group="data"))
+            preq.start(a, b); preq.wait()
+            qreq.start(a, b); qreq.wait()
+        torch.cuda.synchronize()
+        free0, _ = torch.cuda.mem_get_info()
+        for _ in range(200):
+            mx.wait(d.all_reduce(a, b, n, op="sum", group="data"))
+            preq.start(a, b); preq.wait()
+            qreq.start(a, b); qreq.wait()
+        torch.cuda.synchronize()
+        free1, _ = torch.cuda.mem_get_info()
+        grew = free0 - free1
+        assert grew < (64 << 20), f"device memory grew {grew / 1e6:.1f} MB"
+        preq.destroy(); qreq.destroy()
+        mx.finalize()
