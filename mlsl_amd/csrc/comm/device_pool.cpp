@@ -12,8 +12,11 @@ DevicePool::DevicePool(size_t max_cached_bytes) : max_cached_(max_cached_bytes) 
 DevicePool::~DevicePool() { Trim(); }
 
 size_t DevicePool::Bucket(size_t bytes) const {
-    // round up to the next power of two >= 512 B (wasted tail is cheap
-    // against 288 GB; exact-size reuse hits the common persistent case)
+    // Small: next power of two >= 512 B (wasted tail is cheap against
+    // 288 GB; exact-size reuse hits the common persistent case).
+    // Large (>=64 MB): 2 MB granularity — pow2 would waste up to 2x on
+    // the multi-hundred-MB quantized-wire and staging buffers.
+    if (bytes >= (64u << 20)) return (bytes + (2u << 20) - 1) & ~size_t((2u << 20) - 1);
     size_t b = 512;
     while (b < bytes) b <<= 1;
     return b;
