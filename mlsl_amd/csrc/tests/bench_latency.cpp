@@ -11,6 +11,8 @@
 #include <cstdlib>
 #include <cstring>
 
+#include <hip/hip_runtime.h>
+
 #include "../include/mlsl/c_api.h"
 
 #define CK(call)                                                              \
@@ -38,7 +40,18 @@ int main(int argc, char** argv) {
         void *sbuf = nullptr, *rbuf = nullptr;
         CK(mlsl_alloc(bytes, 64, &sbuf));
         CK(mlsl_alloc(bytes, 64, &rbuf));
-
+        // mlsl_alloc returns HBM in device mode: touch via hipMemset (it
+        // falls back to plain memset semantics for host pointers too).
+        hipPointerAttribute_t attr;
+        if (hipPointerGetAttributes(&attr, rbuf) == hipSuccess &&
+            attr.type == hipMemoryTypeDevice) {
+            (void)hipMemset(sbuf, 0, bytes);
+            (void)hipMemset(rbuf, 0, bytes);
+            (void)hipDeviceSynchronize();
+        } else {
+            std::memset(sbuf, 0, bytes);
+            std::memset(rbuf, 0, bytes);
+        }
         mlsl_request req = nullptr;
         CK(mlsl_persistent_all_reduce(dist, count, MLSL_DT_F32, MLSL_RT_SUM,
                                       MLSL_GT_DATA, /*quantized=*/0, &req));
