@@ -529,19 +529,30 @@ bool DeviceAdvanceRequest(CommRequest* req, DeviceReqState& st) {
         HIP_CHECKD(hipStreamCreateWithFlags(&s0, hipStreamNonBlocking));
         gc.streams.push_back(s0);
     }
-    // Order after the caller's compute stream: the producer kernels
-    // (e.g. torch backward on the default stream) must land before the
-    // collective reads the buffers.
-    if (!st.dep_event)
-        HIP_CHECKD(hipEventCreateWithFlags(&st.dep_event, hipEventDisableTiming));
-    HIP_CHECKD(hipEventRecord(st.dep_event,
-                              static_cast<hipStream_t>(rt->ComputeStream())));
-    for (hipStream_t cs : gc.streams)
-        HIP_CHECKD(hipStreamWaitEvent(cs, st.dep_event, 0));
+    const Config& cfg = GlobalConfig();
+    // Reference MLSL_MAX_SHORT_MSG_SIZE analog: small world-1 messages
+    // issue directly on the compute stream — same-stream ordering makes
+    // the dep-event handshake (two extra HIP calls of the small-message
+    // floor) unnecessary. Large messages keep the side stream for overlap.
+    const bool short_local = gc.comms.empty() &&
+                             req->MessageBytes() <= cfg.max_short_msg;
+    hipStream_t base_s = short_local ? static_cast<hipStream_t>(rt->ComputeStream())
+                                     : gc.streams[0];
+    if (!short_local) {
+        // Order after the caller's compute stream: the producer kernels
+        // (e.g. torch backward on the default stream) must land before the
+        // collective reads the buffers.
+        if (!st.dep_event)
+            HIP_CHECKD(hipEventCreateWithFlags(&st.dep_event, hipEventDisableTiming));
+        HIP_CHECKD(hipEventRecord(st.dep_event,
+                                  static_cast<hipStream_t>(rt->ComputeStream())));
+        for (hipStream_t cs : gc.streams)
+            HIP_CHECKD(hipStreamWaitEvent(cs, st.dep_event, 0));
+    }
 
     // hipGraph replay (MLSL_USE_GRAPHS): same request + same buffers ->
     // launch the captured graph instead of re-enqueueing.
-    if (st.graph_exec && req->SendBuf() == st.captured_sbuf &&
+    if (!short_local && st.graph_exec && req->SendBuf() == st.captured_sbuf &&
         req->RecvBuf() == st.captured_rbuf) {
         HIP_CHECKD(hipGraphLaunch(st.graph_exec, gc.streams[0]));
         EnsureEvents(st, 1);
@@ -572,7 +583,7 @@ bool DeviceAdvanceRequest(CommRequest* req, DeviceReqState& st) {
             st.stage_send_bytes = send_b;
         }
         HIP_CHECKD(hipMemcpyAsync(st.stage_send, req->UserSendBuf(), send_b,
-                                  hipMemcpyHostToDevice, gc.streams[0]));
+                                  hipMemcpyHostToDevice, base_s));
         if (gc.streams.size() > 1) {
             HIP_CHECKD(hipEventRecord(st.dep_event, gc.streams[0]));
             for (size_t i = 1; i < gc.streams.size(); ++i)
@@ -589,7 +600,6 @@ bool DeviceAdvanceRequest(CommRequest* req, DeviceReqState& st) {
     auto& chunks = req->Chunks();
     const bool use_schedule = req->UsesDeviceSchedule();
     const bool compressed = req->Compressed();
-    const Config& cfg = GlobalConfig();
     const bool prio = cfg.msg_priority && gc.prio_comm &&
                       req->MessageBytes() >= cfg.msg_priority_threshold;
     if (prio) HIP_CHECKD(hipStreamWaitEvent(gc.prio_stream, st.dep_event, 0));
@@ -597,7 +607,8 @@ bool DeviceAdvanceRequest(CommRequest* req, DeviceReqState& st) {
     // Graph capture only when everything lands on ONE capturable stream:
     // no pageable staging, no priority lane, single channel.
     const bool try_capture = cfg.use_graphs && !st.graph_failed && !s_host &&
-                             !r_host && !prio && GraphEligible(req, gc);
+                             !r_host && !prio && !short_local &&
+                             GraphEligible(req, gc);
 
     const size_t es = DtypeSize(req->Dtype());
     const size_t nch = gc.comms.empty() ? 1 : gc.comms.size();
@@ -605,7 +616,7 @@ bool DeviceAdvanceRequest(CommRequest* req, DeviceReqState& st) {
 
     auto issue_all = [&]() {
         if (gc.comms.empty()) {
-            hipStream_t s0 = gc.streams[0];
+            hipStream_t s0 = base_s;
             for (auto& ce : chunks) {
                 const uint8_t* sbase = req->SendBuf() + ce.elem_off * es;
                 uint8_t* rbase = req->RecvBuf() + ce.elem_off * es;
@@ -706,7 +717,8 @@ bool DeviceAdvanceRequest(CommRequest* req, DeviceReqState& st) {
     // path complete on stream 0; multi-channel records one per channel.
     if (st.graph_exec || gc.comms.empty()) {
         EnsureEvents(st, 1);
-        HIP_CHECKD(hipEventRecord(st.events[0], gc.streams[0]));
+        HIP_CHECKD(hipEventRecord(st.events[0],
+                                  gc.comms.empty() ? base_s : gc.streams[0]));
     } else {
         EnsureEvents(st, used);
         for (size_t ch = 0; ch < used; ++ch)
