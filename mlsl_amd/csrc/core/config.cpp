@@ -34,7 +34,7 @@ Config Config::FromEnv() {
     c.num_channels = EnvSize("MLSL_NUM_CHANNELS", EnvSize("MLSL_NUM_SERVERS", 1));
     if (c.num_channels < 1) c.num_channels = 1;
     if (c.num_channels > 16) c.num_channels = 16;
-    c.max_short_msg = EnvSize("MLSL_MAX_SHORT_MSG_SIZE", 8192);
+    c.max_short_msg = EnvSize("MLSL_MAX_SHORT_MSG_SIZE", 65536);
     c.large_msg_mb = EnvSize("MLSL_LARGE_MSG_SIZE_MB", 128);
     c.large_msg_chunks = EnvSize("MLSL_LARGE_MSG_CHUNKS", 4);
     if (const char* e = std::getenv("MLSL_ALLREDUCE_ALGO")) {

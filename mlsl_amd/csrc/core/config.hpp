@@ -29,7 +29,7 @@ struct Config {
     // a large message is chunked over. The xGMI analog of the reference's
     // endpoint servers (MLSL_NUM_SERVERS, default 4, src/comm_ep.cpp:123).
     size_t num_channels = 1;       // MLSL_NUM_CHANNELS (alias MLSL_NUM_SERVERS)
-    size_t max_short_msg = 8192;   // MLSL_MAX_SHORT_MSG_SIZE (bytes): world-1
+    size_t max_short_msg = 65536;  // MLSL_MAX_SHORT_MSG_SIZE (bytes): world-1
                                    // messages at or below issue on the
                                    // compute stream (no dep-event handshake)
     size_t large_msg_mb = 128;     // MLSL_LARGE_MSG_SIZE_MB: chunk-harder threshold
