@@ -40,3 +40,8 @@ def test_case2_allreduce():
 
 def test_case3_repartition():
     run_ranks("case3_repartition", 4)
+
+
+@pytest.mark.parametrize("world", [2, 4])
+def test_quantized_paramset(world):
+    run_ranks("quantized_paramset", world)

@@ -399,6 +399,49 @@ def case3_repartition():
     mx.finalize()
 
 
+
+
+def quantized_paramset():
+    """ParameterSet with compression=int8 (reference
+    add_parameter_set_with_compress + quant-path relative-error check,
+    mlsl_test.cpp:407-428): the gradient allreduce runs through the
+    quantized engine; the reduced result must match the exact sum within
+    the per-block quantization step."""
+    mx, rank, size = _init()
+    d = mx.Distribution(size, 1)
+    s = mx.Session()
+    s.set_global_minibatch_size(4 * size)
+    info = s.create_op_reg_info("cc")
+    info.set_name("qfc")
+    info.add_input(8, 6, "f32")
+    info.add_output(8, 6, "f32")
+    info.add_parameter_set(64 * 64, 4, "f32", compression="int8")
+    op = s.operation(s.add_operation(info, d))
+    s.commit()
+    ps = op.parameter_set(0)
+
+    n = ps.local_kernel_count * ps.kernel_size
+    rng = np.random.RandomState(11 + rank)
+    g = (rng.randn(n) * 2).astype(np.float32)
+    all_g = [(np.random.RandomState(11 + r).randn(n) * 2).astype(np.float32)
+             for r in range(size)]
+    want = np.sum(all_g, axis=0)
+
+    ps.start_gradient_comm(g)
+    res = ps.wait_gradient_comm()
+    got = _as_np(res, n)
+    # per-block int8 bound: |err| <= ~1 step per rank-hop; generous 4 steps
+    block = 256
+    nb = (n + block - 1) // block
+    pad = nb * block - n
+    wb = np.pad(want, (0, pad)).reshape(nb, block)
+    step = np.abs(wb).max(axis=1) / 127.0
+    bound = np.repeat(step, block)[:n] * (2.0 * max(size, 2)) + 1e-5
+    err = np.abs(got - want)
+    assert (err <= bound).mean() > 0.999, (err.max(), bound.max())
+    mx.finalize()
+
+
 WORKERS = {
     "grid_shapes": grid_shapes,
     "mlsl_net": mlsl_net,
@@ -406,3 +449,5 @@ WORKERS = {
     "case2_allreduce": case2_allreduce,
     "case3_repartition": case3_repartition,
 }
+
+WORKERS["quantized_paramset"] = quantized_paramset
