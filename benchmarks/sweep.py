@@ -59,7 +59,11 @@ def main():
         backend = "nccl" if use_cuda else "gloo"
         if size > 1:
             td.init_process_group(backend=backend)
-            tdist = td
+        else:
+            # world-1 comparison point (RCCL's own degenerate path)
+            td.init_process_group(backend=backend, init_method="tcp://127.0.0.1:29712",
+                                  rank=0, world_size=1)
+        tdist = td
 
     sizes = []
     b = args.min_kb * 1024
