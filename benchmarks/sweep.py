@@ -117,14 +117,24 @@ def main():
         preq.destroy()
 
         if tdist is not None:
+            # torch's all_reduce is in-place only; compose copy+allreduce so
+            # both impls do the same out-of-place semantics (at world 1 the
+            # in-place op is a no-op and would not be a comparison at all).
             tb = buf.clone() if use_cuda else buf.copy()
-            for _ in range(args.warmup):
+
+            def trun():
+                if use_cuda:
+                    tb.copy_(buf)
+                else:
+                    tb[:] = buf
                 tdist.all_reduce(tb)
+            for _ in range(args.warmup):
+                trun()
             sync()
             tdist.barrier()
             t0 = time.perf_counter()
             for _ in range(args.iters):
-                tdist.all_reduce(tb)
+                trun()
             sync()
             dt2 = (time.perf_counter() - t0) / args.iters
             rec2 = dict(rec, impl="torch.distributed", lat_us=round(dt2 * 1e6, 2),
