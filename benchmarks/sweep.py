@@ -123,11 +123,16 @@ def main():
             tb = buf.clone() if use_cuda else buf.copy()
 
             def trun():
+                # same round-trip semantics as preq.start+wait: out-of-place
+                # result, host observes completion every iteration (torch's
+                # all_reduce only enqueues; without the sync it would be
+                # measuring enqueue pipelining against our blocking wait)
                 if use_cuda:
                     tb.copy_(buf)
                 else:
                     tb[:] = buf
                 tdist.all_reduce(tb)
+                sync()
             for _ in range(args.warmup):
                 trun()
             sync()
