@@ -591,6 +591,9 @@ bool DeviceAdvanceRequest(CommRequest* req, DeviceReqState& st) {
         HIP_CHECKD(hipMemcpyAsync(st.stage_send, req->UserSendBuf(), send_b,
                                   hipMemcpyHostToDevice, base_s));
         if (gc.streams.size() > 1) {
+            if (!st.dep_event)
+                HIP_CHECKD(hipEventCreateWithFlags(&st.dep_event,
+                                                   hipEventDisableTiming));
             HIP_CHECKD(hipEventRecord(st.dep_event, gc.streams[0]));
             for (size_t i = 1; i < gc.streams.size(); ++i)
                 HIP_CHECKD(hipStreamWaitEvent(gc.streams[i], st.dep_event, 0));
