@@ -79,6 +79,12 @@ SAMPLES := $(BUILD)/mlsl_sample $(BUILD)/cmlsl_sample
 
 samples: $(SAMPLES)
 
+quantplugin: $(BUILD)/libquant_plugin.so
+
+$(BUILD)/libquant_plugin.so: samples/quant_plugin.c
+	@mkdir -p $(dir $@)
+	gcc -O2 -shared -fPIC -o $@ $< -lm
+
 $(BUILD)/mlsl_sample: samples/mlsl_sample.cpp $(LIB)
 	@mkdir -p $(dir $@)
 	$(HIPCC) -O2 -std=c++17 -Imlsl_amd/csrc/include samples/mlsl_sample.cpp \

@@ -768,11 +768,21 @@ int mlsl_environment_set_quantization_params(mlsl_environment, mlsl_quant_params
     C_TRY
     if (!params || params->elem_in_block == 0)
         throw Error("quant params: elem_in_block must be > 0");
-    if (params->lib_path && params->lib_path[0])
-        MLSL_LOG(INFO, "quant plugin path '%s' ignored: built-in kernels",
-                 params->lib_path);
     QuantParams qp;
     qp.block_elems = params->elem_in_block;
+    if (params->lib_path && params->lib_path[0]) {
+        // dlopen'd compression plugin (reference quant/quant.c): the host
+        // compressed path calls these three functions; block_size declares
+        // the plugin's wire-block bytes.
+        qp.lib_path = params->lib_path;
+        if (params->quant_buffer_func_name)
+            qp.quant_fn = params->quant_buffer_func_name;
+        if (params->dequant_buffer_func_name)
+            qp.dequant_fn = params->dequant_buffer_func_name;
+        if (params->reduce_sum_func_name)
+            qp.reduce_fn = params->reduce_sum_func_name;
+        if (params->block_size) qp.block_bytes = params->block_size;
+    }
     Environment::GetEnv().SetQuantizationParams(qp);
     C_CATCH
 }

@@ -223,6 +223,13 @@ void DeviceSetupRequest(CommRequest* req, DeviceReqState& st) {
     auto al16 = [](size_t x) { return (x + 15) & ~size_t(15); };
     size_t tmp = 0;
     if (req->Compressed()) {
+        // The dlopen'd compression plugin is a host-CPU contract
+        // (reference ran it inside ep_server processes); silently running
+        // the built-in kernels instead would not honor the user's library.
+        MLSL_CHECK(req->Plugin() == nullptr,
+                   "quantization plugin (lib_path) is supported on the host "
+                   "transport only; device mode uses the built-in CDNA4 "
+                   "kernels (unset lib_path)");
         for (auto& ce : req->Chunks())
             tmp += al16(ce.sch.result.bytes) + al16(ce.sch.tmp_bytes) +
                    al16(req->Spec().count * DtypeSize(req->Dtype()));

@@ -1,8 +1,13 @@
 #include "quant.hpp"
 
+#include <dlfcn.h>
+
 #include <algorithm>
 #include <cmath>
 #include <cstring>
+#include <mutex>
+#include <string>
+#include <unordered_map>
 
 #include "../core/log.hpp"
 
@@ -121,6 +126,31 @@ void HostQuantAccum(void* acc_wire, const void* wire, size_t count, size_t block
         }
         ahdr[0] = ns;
     }
+}
+
+const QuantPluginApi* LoadQuantPlugin(const QuantParams& qp) {
+    if (qp.lib_path.empty()) return nullptr;
+    static std::mutex mu;
+    static std::unordered_map<std::string, QuantPluginApi> cache;
+    std::lock_guard<std::mutex> lk(mu);
+    const std::string key = qp.lib_path + "|" + qp.quant_fn;
+    auto it = cache.find(key);
+    if (it != cache.end()) return &it->second;
+
+    void* h = dlopen(qp.lib_path.c_str(), RTLD_NOW);
+    if (!h) MLSL_THROW(std::string("quant plugin dlopen failed: ") + dlerror());
+    QuantPluginApi api{};
+    api.quant = reinterpret_cast<decltype(api.quant)>(dlsym(h, qp.quant_fn.c_str()));
+    api.dequant =
+        reinterpret_cast<decltype(api.dequant)>(dlsym(h, qp.dequant_fn.c_str()));
+    api.reduce_sum = reinterpret_cast<decltype(api.reduce_sum)>(
+        dlsym(h, qp.reduce_fn.c_str()));
+    if (!api.quant || !api.dequant || !api.reduce_sum)
+        MLSL_THROW("quant plugin missing symbol(s): " + qp.quant_fn + "/" +
+                   qp.dequant_fn + "/" + qp.reduce_fn);
+    MLSL_LOG(INFO, "quant plugin loaded: %s (host path; device path keeps "
+             "built-in CDNA4 kernels)", qp.lib_path.c_str());
+    return &cache.emplace(key, api).first->second;
 }
 
 }  // namespace mlsl

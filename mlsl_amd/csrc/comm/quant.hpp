@@ -26,4 +26,17 @@ void HostDequantize(const void* wire, void* out, size_t count, size_t block,
 // acc_wire += wire in the compressed domain (dequant-sum-requant per block).
 void HostQuantAccum(void* acc_wire, const void* wire, size_t count, size_t block);
 
+// dlopen'd compression plugin (reference quant/quant.c ABI — Intel
+// DL-comp style). Loaded once per lib_path; used by the HOST compressed
+// path when QuantParams.lib_path is set. Signatures quant/quant.c:57-65.
+struct QuantPluginApi {
+    int (*quant)(void* src, void* dst, size_t count, void* diff,
+                 int src_data_type, size_t comp_ratio, int method);
+    int (*dequant)(void* src, void* dst, size_t count);
+    int (*reduce_sum)(const void* in, void* inout, size_t block_count);
+};
+// nullptr when qp.lib_path is empty; throws if the library or a symbol
+// cannot be loaded (reference ASSERTs the same way, quant.c:112-126).
+const QuantPluginApi* LoadQuantPlugin(const QuantParams& qp);
+
 }  // namespace mlsl

@@ -138,6 +138,7 @@ void CommRequest::SetCompression(Compression c, const QuantParams& qp) {
     MLSL_CHECK(!setup_done_, "SetCompression after Setup");
     comp_ = c;
     qparams_ = qp;
+    plugin_ = LoadQuantPlugin(qparams_);
 }
 
 bool CommRequest::Compressed() const {
@@ -239,7 +240,8 @@ void CommRequest::BuildChunks() {
                     // 2283). Single chunk; the wire lives in TMP.
                     const size_t blk = qparams_.block_elems;
                     const size_t nblocks = (cnt + blk - 1) / blk;
-                    ce.sch = BuildAllReduceRingUnits(gr, gs, nblocks, blk + 8, blk);
+                    ce.sch = BuildAllReduceRingUnits(gr, gs, nblocks,
+                                                     qparams_.WireBlockBytes(), blk);
                 } else {
                     ce.sch = (algo == AllReduceAlgo::RHD && (gs & (gs - 1)) == 0)
                                  ? BuildAllReduceRHD(gr, gs, cnt, dtype_, spec_.rop)
@@ -444,8 +446,19 @@ bool CommRequest::AdvanceHost(Mesh* mesh) {
         if (compressed && !ce.prologue_done) {
             // quantize user send (+ residual) into the wire
             uint8_t* err = ce.tmp.data() + wire_bytes + ce.sch.tmp_bytes;
-            HostQuantize(sbase, err, wire, spec_.count, qparams_.block_elems,
-                         dtype_, true);
+            if (plugin_) {
+                // reference ABI: dtype enum {INT8=0,F16=1,F32=2,F64=3},
+                // comp_ratio = input-elem-bytes (int8 wire), method DFP=1
+                const int sdt = dtype_ == DataType::F32 ? 2
+                                : dtype_ == DataType::F64 ? 3 : 1;
+                int rc = plugin_->quant(const_cast<uint8_t*>(sbase), wire,
+                                        spec_.count, err, sdt,
+                                        DtypeSize(dtype_), 1);
+                MLSL_CHECK(rc == 0, "quant plugin compress failed");
+            } else {
+                HostQuantize(sbase, err, wire, spec_.count, qparams_.block_elems,
+                             dtype_, true);
+            }
             ce.prologue_done = true;
         }
 
@@ -479,8 +492,14 @@ bool CommRequest::AdvanceHost(Mesh* mesh) {
                             if (d != s) std::memmove(d, s, st.local_src.bytes);
                         } else if (ce.sch.quant_block > 0) {
                             const size_t blk = ce.sch.quant_block;
-                            const size_t units = st.local_dst.bytes / (blk + 8);
-                            HostQuantAccum(d, s, units * blk, blk);
+                            const size_t units =
+                                st.local_dst.bytes / qparams_.WireBlockBytes();
+                            if (plugin_) {
+                                int rc = plugin_->reduce_sum(s, d, units);
+                                MLSL_CHECK(rc == 0, "quant plugin reduce failed");
+                            } else {
+                                HostQuantAccum(d, s, units * blk, blk);
+                            }
                         } else {
                             HostReduce(d, s, st.local_dst.bytes / es, dtype_, ce.sch.rop);
                         }
@@ -495,8 +514,13 @@ bool CommRequest::AdvanceHost(Mesh* mesh) {
             if (ce.cur_phase >= ce.sch.num_phases) {
                 ce.finished = true;
                 if (compressed) {
-                    HostDequantize(wire, rbase, spec_.count, qparams_.block_elems,
-                                   dtype_);
+                    if (plugin_) {
+                        int rc = plugin_->dequant(wire, rbase, spec_.count);
+                        MLSL_CHECK(rc == 0, "quant plugin decompress failed");
+                    } else {
+                        HostDequantize(wire, rbase, spec_.count,
+                                       qparams_.block_elems, dtype_);
+                    }
                 }
             }
         }

@@ -108,6 +108,7 @@ class CommRequest {
     void SetCompression(Compression c, const QuantParams& qp);
     Compression GetCompression() const { return comp_; }
     const QuantParams& QParams() const { return qparams_; }
+    const struct QuantPluginApi* Plugin() const { return plugin_; }
     bool Compressed() const;
     // True when the device executor will walk the schedule (custom ring/RHD
     // over ncclSend/Recv) instead of the fused RCCL op.
@@ -166,6 +167,7 @@ class CommRequest {
     bool setup_done_ = false;
     Compression comp_ = Compression::NONE;
     QuantParams qparams_;
+    const struct QuantPluginApi* plugin_ = nullptr;  // dlopen'd compression
 
     std::vector<ChunkExec> chunks_;
     size_t total_tmp_bytes_ = 0;

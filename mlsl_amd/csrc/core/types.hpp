@@ -93,13 +93,23 @@ enum class Compression : int {
     QUANT_INT8 = 1,   // block int8 quantization with error feedback
 };
 
-// Block-quantization parameters (reference QuantParams include/mlsl.hpp:162-171;
-// the dlopen'd library interface of quant/quant.c is replaced by built-in
-// HIP/CPU kernels — see comm/quant.*).
+// Block-quantization parameters (reference QuantParams
+// include/mlsl.hpp:162-171). Built-in int8 HIP/CPU kernels by default; a
+// dlopen'd user library (Intel DL-comp-style quantize/dequantize/
+// reduce_sum, quant/quant.c ABI) takes over the HOST path when lib_path
+// is set — the device path keeps the fused CDNA4 kernels.
 struct QuantParams {
     size_t block_elems = 256;   // elements per quantization block
-    // Derived: on-wire block = block_elems int8 payload + fp32 scale + fp32 zero.
-    size_t WireBlockBytes() const { return block_elems + 2 * sizeof(float); }
+    std::string lib_path;       // dlopen plugin ("" = built-in kernels)
+    std::string quant_fn = "dl_comp_compress_buffer";
+    std::string dequant_fn = "dl_comp_decompress_buffer";
+    std::string reduce_fn = "dl_comp_compressed_buffer_reduce_sum";
+    size_t block_bytes = 0;     // plugin wire-block bytes (0 = derived)
+    // On-wire block: plugin-declared size, else int8 payload + fp32 scale
+    // + fp32 reserved.
+    size_t WireBlockBytes() const {
+        return block_bytes ? block_bytes : block_elems + 2 * sizeof(float);
+    }
 };
 
 class Error : public std::runtime_error {

@@ -106,3 +106,8 @@ def test_srlist_ring(world):
 @pytest.mark.parametrize("world", [2, 4])
 def test_stress_random(world):
     run_ranks("stress_random", world, timeout=300)
+
+
+@pytest.mark.parametrize("world", [2, 3])
+def test_quant_plugin_dlopen(world):
+    run_ranks("quant_plugin", world)

@@ -690,6 +690,68 @@ def train_ddp_sample():
 WORKERS["train_ddp_sample"] = train_ddp_sample
 
 
+
+
+def quant_plugin():
+    """dlopen'd compression plugin (reference quant/quant.c ABI): load the
+    sample int8 plugin via the reference-named C API, run a quantized
+    allreduce over the host transport, verify against the exact sum within
+    the per-block quantization step."""
+    import ctypes
+    import subprocess
+    from ctypes import byref, c_void_p, c_size_t, c_char_p
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    so = os.path.join(repo, "build", "libquant_plugin.so")
+    if not os.path.exists(so):
+        subprocess.run(["make", "quantplugin"], cwd=repo, check=True,
+                       capture_output=True, timeout=300)
+    from mlsl_amd._lib import lib
+    L = lib()
+
+    class QP(ctypes.Structure):
+        _fields_ = [("lib_path", c_char_p), ("quant", c_char_p),
+                    ("dequant", c_char_p), ("reduce_sum", c_char_p),
+                    ("block_size", c_size_t), ("elem_in_block", c_size_t)]
+
+    import mlsl_amd as mx
+    mx.init()
+    rank, size = mx.rank(), mx.world_size()
+    qp = QP(so.encode(), None, None, None, 264, 256)
+    env = c_void_p()
+    assert L.mlsl_environment_get_env(byref(env)) == 0
+    assert L.mlsl_environment_set_quantization_params(env, byref(qp)) == 0, \
+        ctypes.string_at(L.mlsl_last_error()).decode()
+
+    d = mx.Distribution(size, 1)
+    n = 70000   # includes a partial tail block
+    rng = np.random.RandomState(3 + rank)
+    g = (rng.randn(n) * 2).astype(np.float32)
+    all_g = [(np.random.RandomState(3 + r).randn(n) * 2).astype(np.float32)
+             for r in range(size)]
+    want = np.sum(all_g, axis=0)
+    out = np.zeros_like(g)
+    preq = mx.PersistentRequest(d, "all_reduce", n, dtype="f32", op="sum",
+                                group="data", quantized=True)
+    preq.start(g, out)
+    preq.wait()
+    preq.destroy()
+    block = 256
+    nb = (n + block - 1) // block
+    pad = nb * block - n
+    wb = np.pad(want, (0, pad)).reshape(nb, block)
+    step = np.abs(wb).max(axis=1) / 127.0
+    bound = np.repeat(step, block)[:n] * (2.0 * max(size, 2)) + 1e-5
+    err = np.abs(out - want)
+    assert (err <= bound).mean() > 0.999, (err.max(), bound.max())
+    # restore defaults for any later use in this process
+    qp2 = QP(None, None, None, None, 0, 256)
+    L.mlsl_environment_set_quantization_params(env, byref(qp2))
+    mx.finalize()
+
+
+WORKERS["quant_plugin"] = quant_plugin
+
+
 def main():
     name = sys.argv[1]
     fn = WORKERS.get(name)
