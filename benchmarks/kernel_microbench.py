@@ -38,6 +38,9 @@ def main():
     dt = timed(lambda: ops.reduce_nt(a, b, n))
     res["reduce_f32_nt_TBps"] = round(3 * 4 * n / dt / 1e12, 3)
 
+    dt = timed(lambda: ops.reduce_nt2(a, b, n))
+    res["reduce_f32_nt2_TBps"] = round(3 * 4 * n / dt / 1e12, 3)
+
     # streaming copy: NT kernel vs torch/HIP blit path (same 256 MiB)
     dt = timed(lambda: ops.copy(a, b, n * 4))
     res["copy_nt_256MiB_TBps"] = round(2 * 4 * n / dt / 1e12, 3)

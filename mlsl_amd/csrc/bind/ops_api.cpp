@@ -59,6 +59,13 @@ int mlsl_hip_reduce_nt(void* dst, const void* src, size_t count) {
     OPS_CATCH
 }
 
+int mlsl_hip_reduce_nt2(void* dst, const void* src, size_t count) {
+    OPS_TRY LaunchReduceNT2(dst, src, count, nullptr);
+    if (hipStreamSynchronize(nullptr) != hipSuccess)
+        throw Error("hipStreamSynchronize failed");
+    OPS_CATCH
+}
+
 int mlsl_hip_copy(void* dst, const void* src, size_t bytes) {
     OPS_TRY LaunchCopy(dst, src, bytes, nullptr);
     if (hipStreamSynchronize(nullptr) != hipSuccess)

@@ -104,6 +104,32 @@ __global__ void ReduceF32NTKernel(float* __restrict__ dst,
         dst[j] = Apply<float, OP>(dst[j], src[j]);
 }
 
+// 32 B/lane variant (two float4 per iteration, consecutive): A/B hook.
+template <ReduceOp OP>
+__global__ void ReduceF32NT2Kernel(float* __restrict__ dst,
+                                   const float* __restrict__ src, size_t n) {
+    const size_t tid = blockIdx.x * blockDim.x + threadIdx.x;
+    const size_t stride = gridDim.x * blockDim.x;
+    const size_t n8 = n / 8;
+    const float4_ev* s4 = reinterpret_cast<const float4_ev*>(src);
+    float4_ev* d4 = reinterpret_cast<float4_ev*>(dst);
+    for (size_t i = tid; i < n8; i += stride) {
+        float4_ev a0 = __builtin_nontemporal_load(d4 + 2 * i);
+        float4_ev a1 = __builtin_nontemporal_load(d4 + 2 * i + 1);
+        float4_ev b0 = __builtin_nontemporal_load(s4 + 2 * i);
+        float4_ev b1 = __builtin_nontemporal_load(s4 + 2 * i + 1);
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+            a0[j] = Apply<float, OP>(a0[j], b0[j]);
+            a1[j] = Apply<float, OP>(a1[j], b1[j]);
+        }
+        __builtin_nontemporal_store(a0, d4 + 2 * i);
+        __builtin_nontemporal_store(a1, d4 + 2 * i + 1);
+    }
+    for (size_t j = n8 * 8 + tid; j < n; j += stride)
+        dst[j] = Apply<float, OP>(dst[j], src[j]);
+}
+
 // ---- generic scalar fallback (f64/i32/i64/u8) ----
 template <typename T, ReduceOp OP>
 __global__ void ReduceScalarKernel(T* __restrict__ dst, const T* __restrict__ src,
@@ -275,6 +301,12 @@ void LaunchReduce(void* dst, const void* src, size_t count, DataType dt,
 
 void LaunchReduceNT(void* dst, const void* src, size_t count, hipStream_t stream) {
     MLSL_LAUNCH_BY_OP(ReduceF32NTKernel, static_cast<float*>(dst),
+                      static_cast<const float*>(src), count, ReduceOp::SUM, stream);
+    HIP_CHECK(hipGetLastError());
+}
+
+void LaunchReduceNT2(void* dst, const void* src, size_t count, hipStream_t stream) {
+    MLSL_LAUNCH_BY_OP(ReduceF32NT2Kernel, static_cast<float*>(dst),
                       static_cast<const float*>(src), count, ReduceOp::SUM, stream);
     HIP_CHECK(hipGetLastError());
 }

@@ -17,6 +17,7 @@ void LaunchReduce(void* dst, const void* src, size_t count, DataType dt,
 // fused pipelines).
 // f32 SUM with nontemporal loads/stores (A/B benchmarking).
 void LaunchReduceNT(void* dst, const void* src, size_t count, hipStream_t stream);
+void LaunchReduceNT2(void* dst, const void* src, size_t count, hipStream_t stream);
 
 void LaunchReduceOut(void* dst, const void* a, const void* b, size_t count,
                      DataType dt, ReduceOp op, hipStream_t stream);

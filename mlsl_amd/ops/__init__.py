@@ -24,6 +24,7 @@ def _lib():
         L.mlsl_hip_reduce.argtypes = [c.c_void_p, c.c_void_p, c.c_size_t, c.c_int, c.c_int]
         L.mlsl_hip_reduce_nt.argtypes = [c.c_void_p, c.c_void_p, c.c_size_t]
         L.mlsl_hip_copy.argtypes = [c.c_void_p, c.c_void_p, c.c_size_t]
+        L.mlsl_hip_reduce_nt2.argtypes = [c.c_void_p, c.c_void_p, c.c_size_t]
         L.mlsl_hip_copy_variant.argtypes = [c.c_void_p, c.c_void_p, c.c_size_t, c.c_int]
         L.mlsl_hip_quantize.argtypes = [c.c_void_p, c.c_void_p, c.c_void_p, c.c_size_t,
                                         c.c_size_t, c.c_int, c.c_int]
@@ -35,7 +36,7 @@ def _lib():
         L.mlsl_hip_pack.argtypes = ptypes
         L.mlsl_hip_unpack.argtypes = ptypes
         for n in ("mlsl_hip_device_count", "mlsl_hip_synchronize", "mlsl_hip_reduce",
-                  "mlsl_hip_reduce_nt", "mlsl_hip_copy", "mlsl_hip_copy_variant",
+                  "mlsl_hip_reduce_nt", "mlsl_hip_reduce_nt2", "mlsl_hip_copy", "mlsl_hip_copy_variant",
                   "mlsl_hip_quantize", "mlsl_hip_dequantize", "mlsl_hip_dequantize_nt", "mlsl_hip_quant_accum",
                   "mlsl_hip_pack", "mlsl_hip_unpack"):
             getattr(L, n).restype = c.c_int
@@ -66,6 +67,12 @@ def reduce_nt(dst, src, count):
     dp, _ = _as_ptr_dtype(dst)
     sp, _ = _as_ptr_dtype(src)
     check(_lib().mlsl_hip_reduce_nt(dp, sp, count))
+
+
+def reduce_nt2(dst, src, count):
+    dp, _ = _as_ptr_dtype(dst)
+    sp, _ = _as_ptr_dtype(src)
+    check(_lib().mlsl_hip_reduce_nt2(dp, sp, count))
 
 
 def copy(dst, src, bytes_):
