@@ -12,5 +12,6 @@ from . import seqpar  # noqa: F401
 
 try:  # torch is optional at import time
     from .ddp import DistributedData  # noqa: F401
+    from .zero1 import ShardedOptimizer  # noqa: F401
 except ImportError:  # pragma: no cover
     pass

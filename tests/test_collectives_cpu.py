@@ -111,3 +111,8 @@ def test_stress_random(world):
 @pytest.mark.parametrize("world", [2, 3])
 def test_quant_plugin_dlopen(world):
     run_ranks("quant_plugin", world)
+
+
+@pytest.mark.parametrize("world", [1, 2, 4])
+def test_zero1_sharded_optimizer(world):
+    run_ranks("zero1_sharded_opt", world)

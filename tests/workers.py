@@ -752,6 +752,55 @@ def quant_plugin():
 WORKERS["quant_plugin"] = quant_plugin
 
 
+
+
+def zero1_sharded_opt():
+    """ShardedOptimizer (ZeRO-1 / distributedUpdate analog): RS -> sharded
+    AdamW step -> AG must match a full-size AdamW on the averaged grads."""
+    import torch
+    from mlsl_amd.parallel.zero1 import ShardedOptimizer
+    import mlsl_amd as mx
+    mx.init()
+    rank, size = mx.rank(), mx.world_size()
+    torch.manual_seed(99)
+    model = torch.nn.Sequential(torch.nn.Linear(37, 53), torch.nn.Linear(53, 11))
+    ref = torch.nn.Sequential(torch.nn.Linear(37, 53), torch.nn.Linear(53, 11))
+    ref.load_state_dict(model.state_dict())
+
+    d = mx.Distribution(size, 1)
+    sopt = ShardedOptimizer(model.parameters(), torch.optim.AdamW, d,
+                            reduce="rs", lr=1e-2)
+    ref_opt = torch.optim.AdamW(ref.parameters(), lr=1e-2)
+
+    for it in range(4):
+        grads = []
+        for i, p in enumerate(model.parameters()):
+            g = torch.full_like(p, 0.01 * (it + 1) * (i + 1) * (rank + 1))
+            p.grad = g
+            # mean over ranks: 0.01*(it+1)*(i+1) * (1+..+size)/size
+            grads.append(torch.full_like(p, 0.01 * (it + 1) * (i + 1) *
+                                         (size + 1) / 2.0))
+        for p, g in zip(ref.parameters(), grads):
+            p.grad = g
+        sopt.step()
+        ref_opt.step()
+        sopt.zero_grad()
+        ref_opt.zero_grad()
+
+    for p, q in zip(model.parameters(), ref.parameters()):
+        assert torch.allclose(p.data, q.data, atol=1e-6), \
+            (p.data - q.data).abs().max()
+    # shard-local optimizer state only
+    n_state = sum(v.numel() for st in sopt.opt.state.values()
+                  for v in st.values() if torch.is_tensor(v))
+    total = sum(p.numel() for p in model.parameters())
+    assert n_state <= 2 * ((total + size - 1) // size) + 4, n_state
+    mx.finalize()
+
+
+WORKERS["zero1_sharded_opt"] = zero1_sharded_opt
+
+
 def main():
     name = sys.argv[1]
     fn = WORKERS.get(name)
