@@ -258,6 +258,27 @@ class TestGraphReplay:
 
 
 @pytest.mark.gpu
+class TestTorchWrappers:
+    def test_zero1_on_device(self):
+        import mlsl_amd as mx
+        from mlsl_amd.parallel.zero1 import ShardedOptimizer
+        mx.init()
+        torch.manual_seed(5)
+        model = torch.nn.Linear(64, 64).cuda()
+        d = mx.Distribution(1, 1)
+        sopt = ShardedOptimizer(model.parameters(), torch.optim.SGD, d,
+                                reduce="rs", lr=0.1)
+        x = torch.randn(8, 64, device="cuda")
+        loss = model(x).pow(2).mean()
+        loss.backward()
+        before = model.weight.detach().clone()
+        sopt.step()
+        torch.cuda.synchronize()
+        assert not torch.equal(before, model.weight.detach())
+        mx.finalize()
+
+
+@pytest.mark.gpu
 class TestSoak:
     def test_no_device_memory_growth(self):
         """200 one-shot + 200 persistent request iterations must not grow
