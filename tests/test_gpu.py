@@ -279,6 +279,31 @@ class TestTorchWrappers:
 
 
 @pytest.mark.gpu
+class TestStatsDevice:
+    def test_statistics_on_device_path(self):
+        # MLSL_STATS over the device engine: runtime + isolation counters
+        # populate for a committed net (reference mlsl_test stats getters,
+        # mlsl_test.cpp:679-685).
+        import os
+        os.environ["MLSL_STATS"] = "1"
+        import mlsl_amd as mx
+        try:
+            mx.init()
+            from mlsl_amd.models.synthetic import SyntheticNet
+            net = SyntheticNet(1, 1, global_mb=8, xp=torch, device="cuda")
+            for _ in range(3):
+                net.step()
+            st = net.sess.stats
+            assert st.enabled
+            assert st.total_comm_cycles + st.isolation_comm_cycles(0) >= 0
+            assert st.total_comm_size >= 0
+            st.print()
+            mx.finalize()
+        finally:
+            os.environ.pop("MLSL_STATS", None)
+
+
+@pytest.mark.gpu
 class TestSoak:
     def test_no_device_memory_growth(self):
         """200 one-shot + 200 persistent request iterations must not grow
