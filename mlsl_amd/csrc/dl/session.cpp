@@ -652,7 +652,11 @@ Session::Session(PhaseKind phase) : phase_(phase) {
     stats_ = std::make_unique<Statistics>(this);
 }
 
-Session::~Session() = default;
+Session::~Session() {
+    // MLSL_STATS runs always leave the table behind (the reference dumps
+    // mlsl_stats.log without an explicit call, mlsl_impl_stats.cpp:97).
+    if (stats_ && stats_->IsEnabled() && GetOperationCount() > 0) stats_->Print();
+}
 
 void Session::SetGlobalMinibatchSize(size_t mb) {
     MLSL_CHECK(mb > 0, "global minibatch must be positive");
