@@ -67,6 +67,9 @@ def main():
     res["quantize_f32_TBps"] = round((8 + 1 + 4) * n / dt / 1e12, 3)
     res["quantize_f32_Gelems_s"] = round(n / dt / 1e9, 2)
 
+    dt = timed(lambda: ops.quantize_f32_nt(a, wire, n, err))
+    res["quantize_f32_ntst_Gelems_s"] = round(n / dt / 1e9, 2)
+
     out = torch.empty_like(a)
     dt = timed(lambda: ops.dequantize(wire, out, n))
     res["dequantize_f32_TBps"] = round((1 + 4) * n / dt / 1e12, 3)

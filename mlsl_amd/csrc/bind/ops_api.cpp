@@ -96,6 +96,14 @@ int mlsl_hip_dequantize(const void* wire, void* out, size_t count, size_t block,
     OPS_CATCH
 }
 
+int mlsl_hip_quantize_f32_nt(const void* in, void* err, void* wire, size_t count,
+                             size_t block) {
+    OPS_TRY LaunchQuantizeF32NT(in, err, wire, count, block, nullptr);
+    if (hipStreamSynchronize(nullptr) != hipSuccess)
+        throw Error("hipStreamSynchronize failed");
+    OPS_CATCH
+}
+
 int mlsl_hip_dequantize_nt(const void* wire, void* out, size_t count, size_t block, int dt) {
     OPS_TRY LaunchDequantizeNT(wire, out, count, block, static_cast<DataType>(dt), nullptr);
     if (hipStreamSynchronize(nullptr) != hipSuccess)

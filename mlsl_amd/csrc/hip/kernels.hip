@@ -634,7 +634,7 @@ __global__ void DequantizeBf16x2Kernel(const uint8_t* __restrict__ wire,
 // f32 variants of the same two-blocks-per-wave layout: 8 f32 (2x16 B) per
 // lane. Doubles per-lane access width and ILP over the 4-elems/lane
 // generic kernel; dequant uses NT stores (launcher gates on message size).
-template <bool USE_ERR>
+template <bool USE_ERR, bool NT_ST = false>
 __global__ void QuantizeF32x2Kernel(const float* __restrict__ in,
                                     float* __restrict__ err,
                                     uint8_t* __restrict__ wire, size_t nblocks) {
@@ -687,10 +687,19 @@ __global__ void QuantizeF32x2Kernel(const float* __restrict__ in,
                 else r1[j - 4] = r;
             }
         }
-        *reinterpret_cast<uint2_ev*>(wblock + 8 + sub * 8) = packed;
-        if (USE_ERR) {
-            *reinterpret_cast<float4_ev*>(err + base) = r0;
-            *reinterpret_cast<float4_ev*>(err + base + 4) = r1;
+        if constexpr (NT_ST) {
+            __builtin_nontemporal_store(packed,
+                reinterpret_cast<uint2_ev*>(wblock + 8 + sub * 8));
+            if (USE_ERR) {
+                __builtin_nontemporal_store(r0, reinterpret_cast<float4_ev*>(err + base));
+                __builtin_nontemporal_store(r1, reinterpret_cast<float4_ev*>(err + base + 4));
+            }
+        } else {
+            *reinterpret_cast<uint2_ev*>(wblock + 8 + sub * 8) = packed;
+            if (USE_ERR) {
+                *reinterpret_cast<float4_ev*>(err + base) = r0;
+                *reinterpret_cast<float4_ev*>(err + base + 4) = r1;
+            }
         }
     }
 }
@@ -965,6 +974,19 @@ void LaunchDequantize(const void* wire, void* out, size_t count,
     } else {
         MLSL_THROW("dequantization supports f32/bf16 only");
     }
+    HIP_CHECK(hipGetLastError());
+}
+
+void LaunchQuantizeF32NT(const void* in, void* err, void* wire, size_t count,
+                         size_t block_elems, hipStream_t stream) {
+    // A/B hook: f32 two-blocks-per-wave with NT stores for wire+err.
+    const size_t nblocks = (count + block_elems - 1) / block_elems;
+    MLSL_CHECK(block_elems == 256 && count % 256 == 0, "NT A/B needs whole 256-blocks");
+    const size_t npairs = (nblocks + 1) / 2;
+    dim3 g2(static_cast<uint32_t>(std::min<size_t>((npairs + 3) / 4, kMaxGrid)));
+    hipLaunchKernelGGL((QuantizeF32x2Kernel<true, true>), g2, dim3(kBlock), 0, stream,
+                       static_cast<const float*>(in), static_cast<float*>(err),
+                       static_cast<uint8_t*>(wire), nblocks);
     HIP_CHECK(hipGetLastError());
 }
 

@@ -41,6 +41,8 @@ void LaunchQuantize(const void* in, void* err, void* wire, size_t count,
 // Dequantize wire blocks into out (fp32/bf16).
 void LaunchDequantize(const void* wire, void* out, size_t count,
                       size_t block_elems, DataType dt, hipStream_t stream);
+void LaunchQuantizeF32NT(const void* in, void* err, void* wire, size_t count,
+                         size_t block_elems, hipStream_t stream);
 void LaunchDequantizeNT(const void* wire, void* out, size_t count,
                         size_t block_elems, DataType dt, hipStream_t stream);
 // Compressed-domain accumulate: acc_wire += wire (dequant-sum-requant per

@@ -31,13 +31,15 @@ def _lib():
         L.mlsl_hip_dequantize.argtypes = [c.c_void_p, c.c_void_p, c.c_size_t,
                                           c.c_size_t, c.c_int]
         L.mlsl_hip_dequantize_nt.argtypes = L.mlsl_hip_dequantize.argtypes
+        L.mlsl_hip_quantize_f32_nt.argtypes = [c.c_void_p, c.c_void_p, c.c_void_p,
+                                               c.c_size_t, c.c_size_t]
         L.mlsl_hip_quant_accum.argtypes = [c.c_void_p, c.c_void_p, c.c_size_t, c.c_size_t]
         ptypes = [c.c_void_p, c.c_void_p] + [c.c_size_t] * 8 + [c.c_int]
         L.mlsl_hip_pack.argtypes = ptypes
         L.mlsl_hip_unpack.argtypes = ptypes
         for n in ("mlsl_hip_device_count", "mlsl_hip_synchronize", "mlsl_hip_reduce",
                   "mlsl_hip_reduce_nt", "mlsl_hip_reduce_nt2", "mlsl_hip_copy", "mlsl_hip_copy_variant",
-                  "mlsl_hip_quantize", "mlsl_hip_dequantize", "mlsl_hip_dequantize_nt", "mlsl_hip_quant_accum",
+                  "mlsl_hip_quantize", "mlsl_hip_dequantize", "mlsl_hip_dequantize_nt", "mlsl_hip_quantize_f32_nt", "mlsl_hip_quant_accum",
                   "mlsl_hip_pack", "mlsl_hip_unpack"):
             getattr(L, n).restype = c.c_int
         _declared = True
@@ -108,6 +110,13 @@ def dequantize(wire, out, count, block=256, dtype=None):
     op_, d1 = _as_ptr_dtype(out)
     dt = dtype or d1
     check(_lib().mlsl_hip_dequantize(wp, op_, count, block, DTYPE[dt]))
+
+
+def quantize_f32_nt(inp, wire, count, err, block=256):
+    ip, _ = _as_ptr_dtype(inp)
+    wp, _ = _as_ptr_dtype(wire)
+    ep, _ = _as_ptr_dtype(err)
+    check(_lib().mlsl_hip_quantize_f32_nt(ip, ep, wp, count, block))
 
 
 def dequantize_nt(wire, out, count, block=256, dtype=None):
