@@ -116,3 +116,8 @@ def test_quant_plugin_dlopen(world):
 @pytest.mark.parametrize("world", [1, 2, 4])
 def test_zero1_sharded_optimizer(world):
     run_ranks("zero1_sharded_opt", world)
+
+
+@pytest.mark.parametrize("world", [2, 4])
+def test_inplace_collectives(world):
+    run_ranks("inplace_collectives", world)

@@ -801,6 +801,47 @@ def zero1_sharded_opt():
 WORKERS["zero1_sharded_opt"] = zero1_sharded_opt
 
 
+
+
+def inplace_collectives():
+    """NCCL-convention in-place buffers: allgather with sbuf = rbuf +
+    rank*count, reduce_scatter with rbuf = sbuf + rank*count, bcast
+    in-place. (Reference hard-part: in-place rules per op,
+    src/comm_ep.cpp:623-736.)"""
+    mx, rank, size = _init()
+    d = mx.Distribution(size, 1)
+    cnt = 1000
+
+    # all_gather in-place: contribute from own slot of the full buffer
+    full = np.zeros(size * cnt, dtype=np.float32)
+    full[rank * cnt:(rank + 1) * cnt] = np.arange(cnt) + 100.0 * rank
+    sb = full[rank * cnt:(rank + 1) * cnt]
+    mx.wait(d.all_gather(sb, cnt, full, group="data"))
+    for r in range(size):
+        assert np.allclose(full[r * cnt:(r + 1) * cnt],
+                           np.arange(cnt) + 100.0 * r), "allgather in-place"
+
+    # reduce_scatter in-place: result lands in own slot of the send buffer
+    send = np.tile(np.arange(size * cnt, dtype=np.float32), 1) + rank
+    rb = send[rank * cnt:(rank + 1) * cnt]
+    mx.wait(d.reduce_scatter(send, rb, cnt, op="sum", group="data"))
+    want = size * (np.arange(cnt) + rank * cnt + 0.0) + size * (size - 1) / 2.0
+    # careful: expected = sum over ranks of (idx_global + r) at slot rank
+    idxg = np.arange(rank * cnt, (rank + 1) * cnt, dtype=np.float32)
+    want = size * idxg + size * (size - 1) / 2.0
+    assert np.allclose(rb, want), ("reduce_scatter in-place", rb[:4], want[:4])
+
+    # bcast is naturally in-place
+    b = (np.arange(cnt, dtype=np.float32) if rank == 0
+         else np.zeros(cnt, dtype=np.float32))
+    mx.wait(d.bcast(b, cnt, root=0, group="data"))
+    assert np.allclose(b, np.arange(cnt)), "bcast in-place"
+    mx.finalize()
+
+
+WORKERS["inplace_collectives"] = inplace_collectives
+
+
 def main():
     name = sys.argv[1]
     fn = WORKERS.get(name)
