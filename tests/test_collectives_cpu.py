@@ -121,3 +121,8 @@ def test_zero1_sharded_optimizer(world):
 @pytest.mark.parametrize("world", [2, 4])
 def test_inplace_collectives(world):
     run_ranks("inplace_collectives", world)
+
+
+@pytest.mark.parametrize("world", [2, 4])
+def test_torch_distributed_backend(world):
+    run_ranks("torch_backend", world)

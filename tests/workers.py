@@ -842,6 +842,80 @@ def inplace_collectives():
 WORKERS["inplace_collectives"] = inplace_collectives
 
 
+
+
+def torch_backend():
+    """torch.distributed backend "mlsl": init_process_group + the standard
+    collective calls run over our engine with analytic expectations."""
+    import torch
+    import torch.distributed as td
+    import mlsl_amd.torch_backend  # noqa: F401 (registers "mlsl")
+    rank = int(os.environ["RANK"])
+    size = int(os.environ["WORLD_SIZE"])
+    # torch's env:// store needs its own port; our bootstrap uses MLSL_PORT
+    os.environ.setdefault("MASTER_PORT",
+                          str(int(os.environ.get("MLSL_PORT", "29600")) + 10))
+    td.init_process_group(backend="mlsl", rank=rank, world_size=size)
+    assert td.get_rank() == rank and td.get_world_size() == size
+
+    t = torch.full((1000,), float(rank))
+    td.all_reduce(t)
+    assert torch.all(t == size * (size - 1) / 2.0), "all_reduce"
+
+    t = torch.arange(64, dtype=torch.float32) if rank == 1 % size \
+        else torch.zeros(64)
+    td.broadcast(t, src=1 % size)
+    assert torch.equal(t, torch.arange(64, dtype=torch.float32)), "broadcast"
+
+    inp = torch.full((10,), float(rank))
+    outs = [torch.zeros(10) for _ in range(size)]
+    td.all_gather(outs, inp)
+    for r in range(size):
+        assert torch.all(outs[r] == r), "all_gather"
+
+    big = torch.zeros(10 * size)
+    td.all_gather_into_tensor(big, inp)
+    for r in range(size):
+        assert torch.all(big[r * 10:(r + 1) * 10] == r), "all_gather_into_tensor"
+
+    src = torch.arange(size * 5, dtype=torch.float32) + rank
+    out = torch.zeros(5)
+    td.reduce_scatter_tensor(out, src)
+    want = size * (torch.arange(5, dtype=torch.float32) + rank * 5) \
+        + size * (size - 1) / 2.0
+    # careful: slot for this rank
+    want = size * (torch.arange(rank * 5, (rank + 1) * 5, dtype=torch.float32)) \
+        + size * (size - 1) / 2.0
+    assert torch.allclose(out, want), ("reduce_scatter_tensor", out, want)
+
+    a2a_in = torch.arange(size * 3, dtype=torch.float32) + 100.0 * rank
+    a2a_out = torch.zeros(size * 3)
+    td.all_to_all_single(a2a_out, a2a_in)
+    for r in range(size):
+        want = torch.arange(rank * 3, rank * 3 + 3, dtype=torch.float32) + 100.0 * r
+        assert torch.equal(a2a_out[r * 3:(r + 1) * 3], want), "all_to_all_single"
+
+    td.barrier()
+
+    # torch-native DDP over the mlsl backend (the C++ reducer drives
+    # broadcast + bucketed allreduce through our ProcessGroup)
+    torch.manual_seed(17)
+    m = torch.nn.Linear(16, 4)
+    ddp = torch.nn.parallel.DistributedDataParallel(m)
+    x = torch.randn(8, 16, generator=torch.Generator().manual_seed(23 + rank))
+    ddp(x).sum().backward()
+    g0 = torch.cat([p.grad.flatten() for p in ddp.parameters()])
+    lo, hi = g0.clone(), g0.clone()
+    td.all_reduce(lo, op=td.ReduceOp.MIN)
+    td.all_reduce(hi, op=td.ReduceOp.MAX)
+    assert torch.allclose(lo, hi, atol=1e-6), "DDP grads differ across ranks"
+
+    td.destroy_process_group()
+
+
+WORKERS["torch_backend"] = torch_backend
+
+
 def main():
     name = sys.argv[1]
     fn = WORKERS.get(name)
