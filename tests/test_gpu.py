@@ -259,6 +259,25 @@ class TestGraphReplay:
 
 @pytest.mark.gpu
 class TestTorchWrappers:
+    def test_torch_backend_on_device(self):
+        import os
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29756")
+        os.environ.setdefault("RANK", "0")
+        os.environ.setdefault("WORLD_SIZE", "1")
+        import torch.distributed as td
+        import mlsl_amd.torch_backend  # noqa: F401
+        td.init_process_group(backend="mlsl", rank=0, world_size=1)
+        t = torch.randn(1 << 20, device="cuda")
+        ref = t.clone()
+        td.all_reduce(t)
+        assert torch.allclose(t, ref)
+        big = torch.zeros(1 << 20, device="cuda")
+        td.all_gather_into_tensor(big, t)
+        assert torch.allclose(big, t)
+        td.barrier()
+        td.destroy_process_group()
+
     def test_zero1_on_device(self):
         import mlsl_amd as mx
         from mlsl_amd.parallel.zero1 import ShardedOptimizer
