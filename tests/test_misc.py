@@ -126,3 +126,22 @@ def test_environment_compat_roundtrip():
     session/commit/finalize cycle."""
     from tests.mp import run_ranks
     run_ranks("c_env_compat", 1)
+
+
+def test_reinit_cycle():
+    """Environment Init -> Finalize -> Init again in one process (world 1):
+    the context must rebuild cleanly (fresh groups, engine, allocator)."""
+    import os
+    os.environ.setdefault("MLSL_TRANSPORT", "tcp")
+    import mlsl_amd as mx
+    import numpy as np
+    for cycle in range(3):
+        mx.init()
+        d = mx.Distribution(1, 1)
+        a = np.arange(100, dtype=np.float32) * (cycle + 1)
+        out = np.zeros_like(a)
+        mx.wait(d.all_reduce(a, out, 100, op="sum", group="data"))
+        assert np.allclose(out, a)
+        p = mx.alloc(4096)
+        mx.free(p)
+        mx.finalize()
