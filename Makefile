@@ -36,9 +36,9 @@ HIPSRC := mlsl_amd/csrc/hip/kernels.hip
 
 OBJS := $(CSRC:%.cpp=$(BUILD)/%.o) $(HIPSRC:%.hip=$(BUILD)/%.o)
 
-.PHONY: all lib selftest clean test
+.PHONY: all lib selftest apitest benchlat samples quantplugin clean test
 
-all: lib selftest apitest
+all: lib selftest apitest benchlat samples quantplugin
 
 lib: $(LIB)
 
