@@ -47,6 +47,10 @@ def _sizes(arr):
 
 
 def init(rank=-1, size=-1):
+    """Idempotent: a no-op when the environment is already up (torch-style
+    convenience; the C/C++ Init still errors on double-init)."""
+    if is_initialized():
+        return
     check(lib().mlsl_init(rank, size))
 
 

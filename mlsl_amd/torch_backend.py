@@ -10,6 +10,10 @@ the right default for the TCP host transport and for correctness-first
 device use; overlap-sensitive code should use the native API
 (PersistentRequest / DistributedData) directly.
 
+``destroy_process_group`` does not tear down the mlsl environment (other
+process groups may still use it); call ``mlsl_amd.finalize()`` when done,
+or let process exit clean it up.
+
 Implementation note: a pure-Python ``dist.ProcessGroup`` subclass (the
 pattern torch itself uses for in-process test groups); torch's store/
 rendezvous is left untouched — our bootstrap rendezvous uses

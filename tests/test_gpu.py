@@ -277,6 +277,8 @@ class TestTorchWrappers:
         assert torch.allclose(big, t)
         td.barrier()
         td.destroy_process_group()
+        import mlsl_amd as mx
+        mx.finalize()
 
     def test_zero1_on_device(self):
         import mlsl_amd as mx
