@@ -147,6 +147,26 @@ class MlslProcessGroup(dist.ProcessGroup):
                 dtype=_DT[input_buffer.dtype], group="data"))
         return _done(output_buffer)
 
+    def send(self, tensor_list, dst_rank, tag=0):
+        # matched pairwise with the peer's recv (NCCL-style semantics);
+        # implemented over send_recv_list with a send-only pair
+        for t in tensor_list:
+            c = t.contiguous()
+            mx.wait(self._dist.send_recv_list(
+                c, c, [(dst_rank, 0, c.numel(), 0, 0)],
+                dtype=_DT[t.dtype], group="data"))
+        return _done(tensor_list)
+
+    def recv(self, tensor_list, src_rank, tag=0):
+        for t in tensor_list:
+            c = t if t.is_contiguous() else t.contiguous()
+            mx.wait(self._dist.send_recv_list(
+                c, c, [(src_rank, 0, 0, 0, c.numel())],
+                dtype=_DT[t.dtype], group="data"))
+            if c.data_ptr() != t.data_ptr():
+                t.copy_(c)
+        return _done(tensor_list)
+
     def barrier(self, opts=None):
         self._dist.barrier("data")
         return _done(True)

@@ -897,6 +897,18 @@ def torch_backend():
 
     td.barrier()
 
+    # p2p ring: send to next, recv from prev (pipeline-parallel pattern)
+    nxt, prv = (rank + 1) % size, (rank - 1) % size
+    ps = torch.full((32,), float(rank))
+    pr = torch.zeros(32)
+    if rank % 2 == 0:
+        td.send(ps, nxt)
+        td.recv(pr, prv)
+    else:
+        td.recv(pr, prv)
+        td.send(ps, nxt)
+    assert torch.all(pr == prv), "p2p ring"
+
     # torch-native DDP over the mlsl backend (the C++ reducer drives
     # broadcast + bucketed allreduce through our ProcessGroup)
     torch.manual_seed(17)
