@@ -5,6 +5,7 @@
 #pragma once
 
 #include <cstdint>
+#include <unordered_map>
 #include <vector>
 
 namespace mlsl {
@@ -25,6 +26,13 @@ class ProcessGroup {
 
     uint32_t NextFlow() { return flow_seq_++; }
 
+    // Per-directed-edge message sequence for point-to-point (SRLIST) tags:
+    // sender counts messages it sent TO peer, receiver counts messages it
+    // received FROM peer — the two counters advance in lock-step per edge
+    // (NCCL-style p2p matching), independent of any other group activity.
+    uint32_t NextSendSeq(int peer) { return send_seq_[peer]++; }
+    uint32_t NextRecvSeq(int peer) { return recv_seq_[peer]++; }
+
     // Device-side communicators (RCCL comms + streams per channel), created
     // lazily by the device transport. Owned here so persistent requests can
     // share them.
@@ -36,6 +44,7 @@ class ProcessGroup {
     std::vector<int> ranks_;
     int my_idx_ = -1;
     uint32_t flow_seq_ = 0;
+    std::unordered_map<int, uint32_t> send_seq_, recv_seq_;
     DeviceComm* device_ = nullptr;
 };
 

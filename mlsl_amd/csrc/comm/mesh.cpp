@@ -125,7 +125,7 @@ void Channel::ProgressRecv() {
             } else {
                 UnexpectedMsg um;
                 um.data.resize(recv_len_);
-                recv_unexp_ = &unexpected_.emplace(recv_tag_, std::move(um)).first->second;
+                recv_unexp_ = &unexpected_.emplace(recv_tag_, std::move(um))->second;
             }
         }
         // Payload.

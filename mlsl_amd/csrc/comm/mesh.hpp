@@ -73,7 +73,9 @@ class Channel {
     PendingRecv* recv_cur_ = nullptr;        // posted target, or
     UnexpectedMsg* recv_unexp_ = nullptr;    // unexpected buffer
     std::unordered_map<uint64_t, PendingRecv> posted_;
-    std::unordered_map<uint64_t, UnexpectedMsg> unexpected_;
+    // multimap: several in-flight messages may carry the same tag (e.g. a
+    // buggy or replayed sender) — aliasing them corrupted stream framing
+    std::unordered_multimap<uint64_t, UnexpectedMsg> unexpected_;
 };
 
 class Mesh {

@@ -309,6 +309,13 @@ void CommRequest::Setup() {
     setup_done_ = true;
 }
 
+uint64_t CommRequest::PairTag(uint32_t seq) const {
+    // phase byte 0xFF marks an edge-sequenced p2p tag (schedule phases
+    // never reach 255); seq wraps at 2^24 in-flight-window safety.
+    return (static_cast<uint64_t>(group_->Uid() & 0xFFFFF) << 44) |
+           (static_cast<uint64_t>(seq & 0xFFFFFF) << 8) | 0xFFull;
+}
+
 uint64_t CommRequest::MakeTag(size_t chunk, int phase) const {
     return (static_cast<uint64_t>(group_->Uid() & 0xFFFFF) << 44) |
            (static_cast<uint64_t>(flow_ & 0xFFFFFF) << 20) |
@@ -325,7 +332,7 @@ void CommRequest::Start(const void* sbuf, void* rbuf) {
     rbuf_ = static_cast<uint8_t*>(rbuf);
     dev_sbuf_ = nullptr;
     dev_rbuf_ = nullptr;
-    flow_ = group_->NextFlow();
+    flow_ = spec_.op == CollOp::SRLIST ? 0 : group_->NextFlow();
     for (auto& ce : chunks_) ce.Reset();
     if (dev_) dev_->issued = false;
     error_.clear();
@@ -469,16 +476,27 @@ bool CommRequest::AdvanceHost(Mesh* mesh) {
                 if (st.phase != ce.cur_phase) continue;
                 auto& ss = ce.state[i];
                 // Post receive first (so early sends always land).
+                // SRLIST tags come from per-directed-edge sequences (p2p
+                // matching like NCCL — the sequence is drawn once per step
+                // and survives retries); collective tags from the group
+                // flow, which SRLIST requests deliberately do not consume.
+                const bool p2p = spec_.op == CollOp::SRLIST;
                 if (st.recv_peer >= 0 && st.recv.bytes > 0 && !ss.recv_posted) {
-                    mesh->PostRecv(group_->WorldRank(st.recv_peer),
-                                   MakeTag(ce.chunk_idx, st.phase), ptr(st.recv),
-                                   st.recv.bytes, &ss.recv_done);
+                    const uint64_t tag =
+                        p2p ? PairTag(group_->NextRecvSeq(st.recv_peer))
+                            : MakeTag(ce.chunk_idx, st.phase);
+                    mesh->PostRecv(group_->WorldRank(st.recv_peer), tag,
+                                   ptr(st.recv), st.recv.bytes, &ss.recv_done);
                     ss.recv_posted = true;
                 }
                 if (st.send_peer >= 0 && st.send.bytes > 0 && !ss.send_started) {
-                    if (mesh->StartSend(group_->WorldRank(st.send_peer),
-                                        MakeTag(ce.chunk_idx, st.phase), ptr(st.send),
-                                        st.send.bytes, &ss.send_done))
+                    if (!ss.tag_drawn) {
+                        ss.tag = p2p ? PairTag(group_->NextSendSeq(st.send_peer))
+                                     : MakeTag(ce.chunk_idx, st.phase);
+                        ss.tag_drawn = true;
+                    }
+                    if (mesh->StartSend(group_->WorldRank(st.send_peer), ss.tag,
+                                        ptr(st.send), st.send.bytes, &ss.send_done))
                         ss.send_started = true;
                 }
                 const bool send_ok =

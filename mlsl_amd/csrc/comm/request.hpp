@@ -73,6 +73,8 @@ struct ChunkExec {
         bool send_started = false, send_done = false;
         bool recv_posted = false, recv_done = false;
         bool local_done = false;
+        bool tag_drawn = false;   // p2p: edge sequence drawn exactly once
+        uint64_t tag = 0;
     };
     std::vector<StepState> state;
     std::vector<uint8_t> tmp;  // host scratch
@@ -158,6 +160,7 @@ class CommRequest {
     friend class Engine;
     void BuildChunks();
     uint64_t MakeTag(size_t chunk, int phase) const;
+    uint64_t PairTag(uint32_t seq) const;
 
     ProcessGroup* group_;
     DataType dtype_;

@@ -126,3 +126,8 @@ def test_inplace_collectives(world):
 @pytest.mark.parametrize("world", [2, 4])
 def test_torch_distributed_backend(world):
     run_ranks("torch_backend", world)
+
+
+@pytest.mark.parametrize("world", [2, 4])
+def test_p2p_asymmetric(world):
+    run_ranks("p2p_asymmetric", world)

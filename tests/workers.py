@@ -928,6 +928,45 @@ def torch_backend():
 WORKERS["torch_backend"] = torch_backend
 
 
+
+
+def p2p_asymmetric():
+    """Asymmetric point-to-point (rank0 sends 3 messages to rank1 while the
+    rest of the group is idle) followed by a group collective: per-edge
+    sequence tags must keep both the p2p matching and the later collective
+    tags aligned (a group-flow-based tag scheme breaks exactly here)."""
+    mx, rank, size = _init()
+    d = mx.Distribution(size, 1)
+    n = 512
+    if rank == 0:
+        for i in range(3):
+            buf = np.full(n, 10.0 * (i + 1), dtype=np.float32)
+            mx.wait(d.send_recv_list(buf, buf, [(1, 0, n, 0, 0)], group="data"))
+    elif rank == 1:
+        for i in range(3):
+            out = np.zeros(n, dtype=np.float32)
+            mx.wait(d.send_recv_list(out, out, [(0, 0, 0, 0, n)], group="data"))
+            assert np.all(out == 10.0 * (i + 1)), (i, out[:3])
+    # now a collective over the whole group: flows must still agree
+    a = np.full(64, float(rank), dtype=np.float32)
+    out = np.zeros_like(a)
+    mx.wait(d.all_reduce(a, out, 64, op="sum", group="data"))
+    assert np.all(out == size * (size - 1) / 2.0), "post-p2p allreduce"
+    # interleave: p2p both directions with different counts again
+    if rank == 0:
+        got = np.zeros(n, dtype=np.float32)
+        mx.wait(d.send_recv_list(got, got, [(1, 0, 0, 0, n)], group="data"))
+        assert np.all(got == 77.0)
+    elif rank == 1:
+        msg = np.full(n, 77.0, dtype=np.float32)
+        mx.wait(d.send_recv_list(msg, msg, [(0, 0, n, 0, 0)], group="data"))
+    d.barrier("global")
+    mx.finalize()
+
+
+WORKERS["p2p_asymmetric"] = p2p_asymmetric
+
+
 def main():
     name = sys.argv[1]
     fn = WORKERS.get(name)
