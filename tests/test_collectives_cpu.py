@@ -131,3 +131,8 @@ def test_torch_distributed_backend(world):
 @pytest.mark.parametrize("world", [2, 4])
 def test_p2p_asymmetric(world):
     run_ranks("p2p_asymmetric", world)
+
+
+@pytest.mark.parametrize("world", [2, 4])
+def test_p2p_stress(world):
+    run_ranks("p2p_stress", world, timeout=240)

@@ -967,6 +967,49 @@ def p2p_asymmetric():
 WORKERS["p2p_asymmetric"] = p2p_asymmetric
 
 
+
+
+def p2p_stress():
+    """Randomized asymmetric p2p + collective interleave: a seeded global
+    schedule of directed messages (uneven per-rank counts, bursts that
+    queue unexpected) with interspersed group collectives. Exercises the
+    per-edge sequence tags under load."""
+    mx, rank, size = _init()
+    d = mx.Distribution(size, 1)
+    rng = np.random.RandomState(13)   # same schedule on every rank
+    iters = int(os.environ.get("P2P_STRESS_ITERS", "80"))
+    for it in range(iters):
+        kind = rng.choice(["msg", "burst", "coll"], p=[0.5, 0.2, 0.3])
+        if kind in ("msg", "burst"):
+            src = int(rng.randint(0, size))
+            dst = int((src + 1 + rng.randint(0, size - 1)) % size)
+            n = int(rng.randint(1, 3000))
+            reps = int(rng.randint(2, 5)) if kind == "burst" else 1
+            for k in range(reps):
+                val = it * 1000.0 + k
+                if rank == src:
+                    buf = np.full(n, val, dtype=np.float32)
+                    mx.wait(d.send_recv_list(buf, buf, [(dst, 0, n, 0, 0)],
+                                             group="data"))
+                elif rank == dst:
+                    out = np.zeros(n, dtype=np.float32)
+                    mx.wait(d.send_recv_list(out, out, [(src, 0, 0, 0, n)],
+                                             group="data"))
+                    assert np.all(out == val), (it, k, out[:3])
+        else:
+            n = int(rng.randint(1, 2000))
+            a = np.arange(n, dtype=np.float32) + rank
+            out = np.zeros_like(a)
+            mx.wait(d.all_reduce(a, out, n, op="sum", group="data"))
+            want = size * np.arange(n, dtype=np.float32) + size * (size - 1) / 2.0
+            assert np.allclose(out, want), (it, "coll")
+    d.barrier("global")
+    mx.finalize()
+
+
+WORKERS["p2p_stress"] = p2p_stress
+
+
 def main():
     name = sys.argv[1]
     fn = WORKERS.get(name)
