@@ -147,6 +147,48 @@ class MlslProcessGroup(dist.ProcessGroup):
                 dtype=_DT[input_buffer.dtype], group="data"))
         return _done(output_buffer)
 
+    def reduce(self, tensor_list, opts=None):
+        op = opts.reduceOp if opts is not None else dist.ReduceOp.SUM
+        root = int(opts.rootRank) if opts is not None else 0
+        for t in tensor_list:
+            c = t if t.is_contiguous() else t.contiguous()
+            res = torch.empty_like(c)
+            mx.wait(self._dist.reduce(c, res, c.numel(), op=_op(op), root=root,
+                                      dtype=_DT[t.dtype], group="data"))
+            if self._rank == root:
+                t.copy_(res)
+        return _done(tensor_list)
+
+    def gather(self, output_tensors, input_tensors, opts=None):
+        root = int(opts.rootRank) if opts is not None else 0
+        inp = input_tensors[0].contiguous()
+        if self._rank == root:
+            outs = output_tensors[0]
+            flat = torch.empty(self._world * inp.numel(), dtype=inp.dtype,
+                               device=inp.device)
+        else:
+            flat = torch.empty(1, dtype=inp.dtype, device=inp.device)
+        mx.wait(self._dist.gather(inp, inp.numel(), flat, root=root,
+                                  dtype=_DT[inp.dtype], group="data"))
+        if self._rank == root:
+            for r, o in enumerate(outs):
+                o.copy_(flat[r * inp.numel():(r + 1) * inp.numel()].view_as(o))
+        return _done(output_tensors)
+
+    def scatter(self, output_tensors, input_tensors, opts=None):
+        root = int(opts.rootRank) if opts is not None else 0
+        out = output_tensors[0]
+        o = out if out.is_contiguous() else out.contiguous()
+        if self._rank == root:
+            flat = torch.cat([t.reshape(-1) for t in input_tensors[0]])
+        else:
+            flat = torch.empty(1, dtype=out.dtype, device=out.device)
+        mx.wait(self._dist.scatter(flat, o, o.numel(), root=root,
+                                   dtype=_DT[out.dtype], group="data"))
+        if o.data_ptr() != out.data_ptr():
+            out.copy_(o)
+        return _done(output_tensors)
+
     def send(self, tensor_list, dst_rank, tag=0):
         # matched pairwise with the peer's recv (NCCL-style semantics);
         # implemented over send_recv_list with a send-only pair

@@ -897,6 +897,22 @@ def torch_backend():
 
     td.barrier()
 
+    # reduce / gather / scatter
+    t = torch.full((50,), float(rank + 1))
+    td.reduce(t, dst=0)
+    if rank == 0:
+        assert torch.all(t == size * (size + 1) / 2.0), "reduce"
+    gin = torch.full((6,), float(rank))
+    gouts = [torch.zeros(6) for _ in range(size)] if rank == 1 % size else None
+    td.gather(gin, gouts, dst=1 % size)
+    if rank == 1 % size:
+        for r in range(size):
+            assert torch.all(gouts[r] == r), "gather"
+    sout = torch.zeros(4)
+    sins = [torch.full((4,), float(r * 2)) for r in range(size)] if rank == 0 else None
+    td.scatter(sout, sins, src=0)
+    assert torch.all(sout == rank * 2), "scatter"
+
     # p2p ring: send to next, recv from prev (pipeline-parallel pattern)
     nxt, prv = (rank + 1) % size, (rank - 1) % size
     ps = torch.full((32,), float(rank))
