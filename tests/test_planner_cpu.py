@@ -45,3 +45,7 @@ def test_case3_repartition():
 @pytest.mark.parametrize("world", [2, 4])
 def test_quantized_paramset(world):
     run_ranks("quantized_paramset", world)
+
+
+def test_stats_log_dump():
+    run_ranks("stats_log_dump", 2)

@@ -442,6 +442,43 @@ def quantized_paramset():
     mx.finalize()
 
 
+
+
+def stats_log_dump():
+    """MLSL_STATS=1 run leaves mlsl_stats.log with per-op rows at session
+    teardown (reference auto-dump behavior, mlsl_impl_stats.cpp:97)."""
+    import os as _os
+    import tempfile
+    _os.environ["MLSL_STATS"] = "1"
+    workdir = tempfile.mkdtemp(prefix="mlslstats")
+    _os.chdir(workdir)
+    mx, rank, size = _init()
+    d = mx.Distribution(size, 1)
+    s2 = mx.Session()
+    s2.set_global_minibatch_size(4 * size)
+    info = s2.create_op_reg_info("cc")
+    info.set_name("statop")
+    info.add_input(8, 4, "f32")
+    info.add_output(8, 4, "f32")
+    info.add_parameter_set(16, 3, "f32")
+    op = s2.operation(s2.add_operation(info, d))
+    s2.commit()
+    ps = op.parameter_set(0)
+    g = np.ones(ps.local_kernel_count * ps.kernel_size, dtype=np.float32)
+    for _ in range(3):
+        ps.start_gradient_comm(g)
+        ps.wait_gradient_comm()
+    st = s2.stats
+    assert st.enabled
+    assert st.total_comm_size > 0
+    s2.close()
+    assert _os.path.exists("mlsl_stats.log"), "stats log not dumped"
+    txt = open("mlsl_stats.log").read()
+    assert "statop" in txt and "comm" in txt, txt[:200]
+    mx.finalize()
+
+
+
 WORKERS = {
     "grid_shapes": grid_shapes,
     "mlsl_net": mlsl_net,
@@ -451,3 +488,4 @@ WORKERS = {
 }
 
 WORKERS["quantized_paramset"] = quantized_paramset
+WORKERS["stats_log_dump"] = stats_log_dump
