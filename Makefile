@@ -108,6 +108,17 @@ $(ASAN): mlsl_amd/csrc/tests/schedule_selftest.cpp mlsl_amd/csrc/comm/schedule.c
 	g++ -O1 -g -std=c++17 -fsanitize=address,undefined -fno-omit-frame-pointer \
 	    $^ -pthread -o $@
 
+# Full-stack ASan: the API selftest (planner+engine+mesh, TCP world
+# matrix) host-compiled with g++ + ASan/UBSan; device kernels stubbed.
+ASAN_API := $(BUILD)/api_selftest_asan
+ASAN_SRC := $(CSRC) mlsl_amd/csrc/tests/api_selftest.cpp             mlsl_amd/csrc/tests/asan_kernel_stubs.cpp
+
+asan-api: $(ASAN_API)
+
+$(ASAN_API): $(ASAN_SRC)
+	@mkdir -p $(dir $@)
+	g++ -O1 -g -std=c++17 -D__HIP_PLATFORM_AMD__ -I/opt/rocm/include 	    -fsanitize=address,undefined -fno-omit-frame-pointer 	    $(ASAN_SRC) -L/opt/rocm/lib -lrccl -lamdhip64 -pthread -o $@
+
 test: all samples
 	$(SELFTEST)
 	python -m pytest tests/ -x -q -m "not gpu"

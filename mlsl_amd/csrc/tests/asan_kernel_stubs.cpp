@@ -1,0 +1,22 @@
+// Host-only kernel-launcher stubs for the ASan build of the full API
+// selftest: the CPU/TCP matrix never launches device kernels, but
+// device_comm.cpp references the symbols. Any call is a hard error.
+#include "../core/types.hpp"
+#include "../hip/kernels.hpp"
+
+namespace mlsl {
+#define STUB(...) { MLSL_THROW("kernel launcher called in host-only ASan build"); }
+void LaunchReduce(void*, const void*, size_t, DataType, ReduceOp, hipStream_t) STUB()
+void LaunchReduceNT(void*, const void*, size_t, hipStream_t) STUB()
+void LaunchReduceNT2(void*, const void*, size_t, hipStream_t) STUB()
+void LaunchReduceOut(void*, const void*, const void*, size_t, DataType, ReduceOp, hipStream_t) STUB()
+void LaunchCopy(void*, const void*, size_t, hipStream_t) STUB()
+void LaunchCopyVariant(void*, const void*, size_t, bool, hipStream_t) STUB()
+void LaunchQuantize(const void*, void*, void*, size_t, size_t, DataType, bool, hipStream_t) STUB()
+void LaunchQuantizeF32NT(const void*, void*, void*, size_t, size_t, hipStream_t) STUB()
+void LaunchDequantize(const void*, void*, size_t, size_t, DataType, hipStream_t) STUB()
+void LaunchDequantizeNT(const void*, void*, size_t, size_t, DataType, hipStream_t) STUB()
+void LaunchQuantAccum(void*, const void*, size_t, size_t, hipStream_t) STUB()
+void LaunchPack(const void*, void*, const PackBlockDesc&, DataType, hipStream_t) STUB()
+void LaunchUnpack(const void*, void*, const PackBlockDesc&, DataType, hipStream_t) STUB()
+}  // namespace mlsl

@@ -23,3 +23,30 @@ def test_schedule_selftest_asan():
     out = subprocess.run([exe], capture_output=True, text=True, timeout=600)
     assert out.returncode == 0, out.stdout + out.stderr
     assert "PASSED" in out.stdout
+
+
+def test_api_selftest_asan():
+    """Full-stack ASan/UBSan: planner + engine + TCP mesh, world 4, two
+    grid configs (device kernels stubbed; host paths fully sanitized)."""
+    import os
+    import subprocess
+    from tests.mp import free_port
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    path = os.path.join(repo, "build", "api_selftest_asan")
+    if not os.path.exists(path):
+        subprocess.run(["make", "asan-api"], cwd=repo, check=True,
+                       capture_output=True, timeout=900)
+    for mp, du in ((2, 0), (2, 1)):
+        port = free_port()
+        procs = []
+        for r in range(4):
+            env = dict(os.environ, RANK=str(r), WORLD_SIZE="4",
+                       MASTER_ADDR="127.0.0.1", MLSL_PORT=str(port),
+                       MLSL_TRANSPORT="tcp", MP=str(mp), DIST_UPDATE=str(du),
+                       ASAN_OPTIONS="detect_leaks=0")
+            procs.append(subprocess.Popen([path], env=env, cwd=repo,
+                                          stdout=subprocess.PIPE,
+                                          stderr=subprocess.STDOUT, text=True))
+        for r, p in enumerate(procs):
+            out, _ = p.communicate(timeout=180)
+            assert p.returncode == 0 and "PASSED" in out, f"rank {r}: {out[-2000:]}"
