@@ -119,6 +119,16 @@ $(ASAN_API): $(ASAN_SRC)
 	@mkdir -p $(dir $@)
 	g++ -O1 -g -std=c++17 -D__HIP_PLATFORM_AMD__ -I/opt/rocm/include 	    -fsanitize=address,undefined -fno-omit-frame-pointer 	    $(ASAN_SRC) -L/opt/rocm/lib -lrccl -lamdhip64 -pthread -o $@
 
+TSAN_API := $(BUILD)/api_selftest_tsan
+
+tsan-api: $(TSAN_API)
+
+$(TSAN_API): $(ASAN_SRC)
+	@mkdir -p $(dir $@)
+	g++ -O1 -g -std=c++17 -D__HIP_PLATFORM_AMD__ -I/opt/rocm/include \
+	    -fsanitize=thread -fno-omit-frame-pointer \
+	    $(ASAN_SRC) -L/opt/rocm/lib -lrccl -lamdhip64 -pthread -o $@
+
 test: all samples
 	$(SELFTEST)
 	python -m pytest tests/ -x -q -m "not gpu"
