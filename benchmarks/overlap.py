@@ -64,7 +64,14 @@ def main():
             return a @ b
 
     d = mx.Distribution(size, 1)
-    buck = GradBucketer(d, bufs)
+    # out-of-place buckets: at world 1 an in-place allreduce is an identity
+    # no-op, which would make "comm" vacuous — out-of-place keeps real data
+    # motion at every world size (2x bucket bytes of HBM at n=1).
+    if use_cuda:
+        outs = [torch.empty_like(b) for b in bufs]
+    else:
+        outs = [np.empty_like(b) for b in bufs]
+    buck = GradBucketer(d, bufs, outputs=outs)
 
     def sync():
         if use_cuda:

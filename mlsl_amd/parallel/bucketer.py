@@ -16,9 +16,12 @@ from mlsl_amd.api import _as_ptr_dtype
 
 
 class GradBucketer:
-    def __init__(self, dist, buffers, op="sum", group="data", dtype=None):
+    def __init__(self, dist, buffers, op="sum", group="data", dtype=None,
+                 outputs=None):
         self.dist = dist
         self.buffers = list(buffers)
+        self.outputs = list(outputs) if outputs is not None else self.buffers
+        assert len(self.outputs) == len(self.buffers)
         self.op = op
         self.group = group
         self.dtypes = []
@@ -31,7 +34,8 @@ class GradBucketer:
         b = self.buffers[idx]
         n = b.size if hasattr(b, "size") and isinstance(b.size, int) else (
             b.numel() if hasattr(b, "numel") else len(b))
-        self._reqs[idx] = self.dist.all_reduce(b, b, n, op=self.op,
+        self._reqs[idx] = self.dist.all_reduce(b, self.outputs[idx], n,
+                                               op=self.op,
                                                dtype=self.dtypes[idx],
                                                group=self.group)
 
