@@ -31,6 +31,8 @@ class DistributedData:
         self.average = average
         self.bucket_bytes = bucket_mb * 1024 * 1024
         self._params = [p for p in module.parameters() if p.requires_grad]
+        if torch.cuda.is_available():
+            mx.set_compute_stream(torch.cuda.current_stream().cuda_stream)
 
         # broadcast initial parameters from rank 0 (reference test protocol,
         # mlsl_test.cpp:651-652)

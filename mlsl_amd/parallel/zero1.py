@@ -52,6 +52,8 @@ class ShardedOptimizer:
         assert reduce in ("none", "rs")
         self.reduce = reduce
 
+        if torch.cuda.is_available() and self.params[0].is_cuda:
+            mx.set_compute_stream(torch.cuda.current_stream().cuda_stream)
         dev = self.params[0].device
         dt = self.params[0].dtype
         assert all(p.dtype == dt for p in self.params), \

@@ -55,6 +55,8 @@ class MlslProcessGroup(dist.ProcessGroup):
         self._dist = mx.Distribution(world_size, 1)
         self._rank = rank
         self._world = world_size
+        if torch.cuda.is_available():
+            mx.set_compute_stream(torch.cuda.current_stream().cuda_stream)
 
     def getBackendName(self):
         return "mlsl"
