@@ -129,6 +129,19 @@ $(TSAN_API): $(ASAN_SRC)
 	    -fsanitize=thread -fno-omit-frame-pointer \
 	    $(ASAN_SRC) -L/opt/rocm/lib -lrccl -lamdhip64 -pthread -o $@
 
+# Concurrent-submit race test (multi-producer MPSC ring) under TSan.
+TSAN_SUBMIT := $(BUILD)/submit_race_tsan
+TSAN_SUBMIT_SRC := $(CSRC) mlsl_amd/csrc/tests/submit_race_selftest.cpp \
+    mlsl_amd/csrc/tests/asan_kernel_stubs.cpp
+
+tsan-submit: $(TSAN_SUBMIT)
+
+$(TSAN_SUBMIT): $(TSAN_SUBMIT_SRC)
+	@mkdir -p $(dir $@)
+	g++ -O1 -g -std=c++17 -D__HIP_PLATFORM_AMD__ -I/opt/rocm/include \
+	    -fsanitize=thread -fno-omit-frame-pointer \
+	    $(TSAN_SUBMIT_SRC) -L/opt/rocm/lib -lrccl -lamdhip64 -pthread -o $@
+
 test: all samples
 	$(SELFTEST)
 	python -m pytest tests/ -x -q -m "not gpu"

@@ -10,11 +10,17 @@ or  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
         --master-addr 127.0.0.1 --master-port P bench.py --gpus N ...
 (one rank per GPU; RANK/LOCAL_RANK/WORLD_SIZE/MASTER_* read from the env).
 
-One step = one in-place AllReduce of a fixed fp32 message (default 256 MiB)
-on this rank's GPU. `value` is the whole-job aggregate algorithmic
-bandwidth: N * message_bytes / t_step (GB/s), t_step = max over ranks.
-n_gpus=1 degenerates to the library's local-copy path and is reported
-as such in config.note.
+One step = one out-of-place AllReduce of a fixed fp32 message (default
+256 MiB). Metric definition (pinned; BASELINE.md and docs/BENCHMARKS.md
+use the same formulas):
+
+    algbw (the reported `value`) = message_bytes / t_step      [GB/s]
+    busbw (auxiliary)            = 2*(n-1)/n * algbw           [GB/s]
+
+t_step = max over ranks. algbw is the standard per-collective algorithmic
+bandwidth (nccl-tests convention) — NOT multiplied by N. n_gpus=1
+degenerates to the library's local-copy path (algbw = copy bandwidth) and
+is reported as such in config.note.
 """
 import argparse
 import json
@@ -111,12 +117,14 @@ def main():
 
     ms_per_step = elapsed_max / args.steps * 1e3
     msg_bytes = count * 4
-    value = size * msg_bytes / (elapsed_max / args.steps) / 1e9  # GB/s aggregate
+    # Pinned metric (see module docstring): algbw = S/t, busbw = 2(n-1)/n * algbw.
+    algbw = msg_bytes / (elapsed_max / args.steps) / 1e9
+    busbw = (2.0 * (size - 1) / size) * algbw if size > 1 else 0.0
 
     if rank == 0:
         out = {
-            "metric": "allreduce_algbw_total_GBps",
-            "value": round(value, 3),
+            "metric": "allreduce_algbw_GBps",
+            "value": round(algbw, 3),
             "unit": "GB/s",
             "n_gpus": size,
             "steps": args.steps,
@@ -129,6 +137,7 @@ def main():
             "data": "synthetic",
             "config": {
                 "model": "allreduce-sweep-headline",
+                "busbw_GBps": round(busbw, 3),
                 "message_mib": args.mbytes,
                 "global_batch": None,
                 "seq_len": None,

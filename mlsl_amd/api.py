@@ -24,6 +24,22 @@ _TORCH_DTYPE = {"torch.float32": "f32", "torch.float64": "f64", "torch.uint8": "
                 "torch.int32": "i32", "torch.int64": "i64"}
 
 
+_registered_stream = [None]
+
+
+def _sync_compute_stream(t):
+    """Re-register torch's CURRENT stream as the compute stream whenever a
+    CUDA tensor enters a collective. Registering only once (at wrapper
+    construction) ordered collectives against a stale stream if the
+    application later produced gradients on a different torch stream —
+    silent corruption. The ctypes call is skipped when unchanged."""
+    import torch
+    s = torch.cuda.current_stream(t.device).cuda_stream
+    if s != _registered_stream[0]:
+        set_compute_stream(s)
+        _registered_stream[0] = s
+
+
 def _as_ptr_dtype(buf, dtype=None):
     """Return (void_ptr, inferred_dtype_name or None)."""
     if buf is None:
@@ -38,6 +54,8 @@ def _as_ptr_dtype(buf, dtype=None):
         return c_void_p(ai["data"][0]), dtype or _NUMPY_DTYPE.get(str(buf.dtype))
     # torch
     if hasattr(buf, "data_ptr"):
+        if getattr(buf, "is_cuda", False):
+            _sync_compute_stream(buf)
         return c_void_p(buf.data_ptr()), dtype or _TORCH_DTYPE.get(str(buf.dtype))
     raise TypeError(f"unsupported buffer type {type(buf)}")
 
@@ -118,6 +136,7 @@ def set_compute_stream(stream_ptr):
         L.mlsl_set_compute_stream.restype = _c.c_int
         L._scs_declared = True
     check(L.mlsl_set_compute_stream(_c.c_void_p(stream_ptr)))
+    _registered_stream[0] = stream_ptr
 
 
 def wait(req):

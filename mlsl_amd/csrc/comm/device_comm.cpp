@@ -578,16 +578,12 @@ bool DeviceAdvanceRequest(CommRequest* req, DeviceReqState& st) {
     // buffers through persistent HBM so RCCL/kernels see device memory.
     size_t send_b = 0, recv_b = 0;
     IoBytes(req, &send_b, &recv_b);
-    if (st.cls_sptr != req->UserSendBuf()) {
-        st.cls_sptr = req->UserSendBuf();
-        st.cls_s_host = !IsDevicePtr(st.cls_sptr);
-    }
-    if (st.cls_rptr != req->UserRecvBuf()) {
-        st.cls_rptr = req->UserRecvBuf();
-        st.cls_r_host = !IsDevicePtr(st.cls_rptr);
-    }
-    const bool s_host = send_b > 0 && st.cls_s_host;
-    const bool r_host = recv_b > 0 && st.cls_r_host;
+    // Classified fresh every Start: hipPointerGetAttributes is ~0.3 us and a
+    // pointer-value cache would misclassify a freed-and-reallocated host
+    // pointer landing at an address that used to be device memory (or vice
+    // versa) — silent corruption for a sub-microsecond saving.
+    const bool s_host = send_b > 0 && !IsDevicePtr(req->UserSendBuf());
+    const bool r_host = recv_b > 0 && !IsDevicePtr(req->UserRecvBuf());
     st.recv_staged = r_host;
     if (s_host) {
         if (!st.stage_send || st.stage_send_bytes < send_b) {
@@ -618,7 +614,18 @@ bool DeviceAdvanceRequest(CommRequest* req, DeviceReqState& st) {
     const bool compressed = req->Compressed();
     const bool prio = cfg.msg_priority && gc.prio_comm &&
                       req->MessageBytes() >= cfg.msg_priority_threshold;
-    if (prio) HIP_CHECKD(hipStreamWaitEvent(gc.prio_stream, st.dep_event, 0));
+    if (prio) {
+        // Order the priority lane after the producers and any staging H2D:
+        // both dependencies are carried by channel stream 0 at this point
+        // (it waited on the compute stream above if that was busy, and the
+        // staging copy was enqueued on it), so record a fresh event there
+        // unconditionally — st.dep_event may not exist yet when the compute
+        // stream was idle and no staging ran.
+        if (!st.dep_event)
+            HIP_CHECKD(hipEventCreateWithFlags(&st.dep_event, hipEventDisableTiming));
+        HIP_CHECKD(hipEventRecord(st.dep_event, gc.streams[0]));
+        HIP_CHECKD(hipStreamWaitEvent(gc.prio_stream, st.dep_event, 0));
+    }
 
     // Graph capture only when everything lands on ONE capturable stream:
     // no pageable staging, no priority lane, single channel.

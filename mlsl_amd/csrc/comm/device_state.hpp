@@ -26,12 +26,6 @@ struct DeviceReqState {
     void* stage_recv = nullptr;
     size_t stage_send_bytes = 0, stage_recv_bytes = 0;
     bool recv_staged = false;
-    // Cached host/device classification of the user buffers: persistent
-    // requests re-Start with the same pointers every step, so the
-    // hipPointerGetAttributes pair is paid once, not per iteration.
-    const void* cls_sptr = reinterpret_cast<const void*>(~uintptr_t(0));
-    const void* cls_rptr = reinterpret_cast<const void*>(~uintptr_t(0));
-    bool cls_s_host = false, cls_r_host = false;
     // hipGraph replay (MLSL_USE_GRAPHS): the issue sequence captured once,
     // replayed on subsequent Starts with the same buffers.
     hipGraphExec_t graph_exec = nullptr;

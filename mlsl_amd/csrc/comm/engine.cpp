@@ -52,11 +52,12 @@ void Engine::Submit(CommRequest* req) {
         }
         return;
     }
-    // Progress-thread mode: hand off through the SPSC ring (overflow to the
-    // locked deque if the ring is full or there are competing producers).
+    // Progress-thread mode: hand off through the lock-free MPSC ring
+    // (overflow to the locked deque only when the ring is full).
     if (!ring_.Push(req)) {
         std::lock_guard<std::mutex> lk(inbox_mu_);
         inbox_overflow_.push_back(req);
+        overflow_pending_.store(true, std::memory_order_release);
     }
     if (deep_idle_.load(std::memory_order_acquire)) {
         // Pairs with the Loop's check-then-wait under idle_mu_: taking the
@@ -72,13 +73,14 @@ void Engine::DrainInbox() {
         r->state_.store(ReqState::ACTIVE, std::memory_order_release);
         active_.push_back(r);
     }
-    if (!inbox_overflow_.empty()) {
+    if (overflow_pending_.load(std::memory_order_acquire)) {
         std::lock_guard<std::mutex> lk(inbox_mu_);
         for (CommRequest* r : inbox_overflow_) {
             r->state_.store(ReqState::ACTIVE, std::memory_order_release);
             active_.push_back(r);
         }
         inbox_overflow_.clear();
+        overflow_pending_.store(false, std::memory_order_release);
     }
     // Newest-first priority above the size threshold (reference
     // MLSL_MSG_PRIORITY head-first scan, eplib/allreduce_pr.c:69-81 +
@@ -161,7 +163,8 @@ void Engine::Loop() {
             if (++idle_spins > 4096) {
                 std::unique_lock<std::mutex> lk(idle_mu_);
                 deep_idle_.store(true, std::memory_order_release);
-                if (ring_.Empty() && inbox_overflow_.empty() &&
+                if (ring_.Empty() &&
+                    !overflow_pending_.load(std::memory_order_acquire) &&
                     !stop_.load(std::memory_order_acquire))
                     idle_cv_.wait_for(lk, std::chrono::milliseconds(1));
                 deep_idle_.store(false, std::memory_order_release);

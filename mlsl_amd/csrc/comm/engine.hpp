@@ -2,8 +2,8 @@
 // servers (eplib/server.c, cqueue.c). Instead of proxy MPI processes fed by
 // shared-memory rings, a host progress THREAD owns the TCP mesh / HIP
 // streams and advances resumable request state machines; the API thread
-// talks to it through a lock-free SPSC command ring (cqueue analog) with a
-// mutex fallback for multi-threaded producers.
+// talks to it through a lock-free MPSC command ring (cqueue analog; any
+// application thread may submit) with a mutex-guarded overflow deque.
 //
 // MLSL_PROGRESS=inline gives the reference's "thread mode" analog: work is
 // issued on the calling thread (device mode stays async via streams).
@@ -26,18 +26,41 @@ namespace mlsl {
 class CommRequest;
 class Mesh;
 
-// Fixed-size lock-free single-producer/single-consumer ring (the cqueue_t
+// Fixed-size lock-free multi-producer/single-consumer ring (the cqueue_t
 // analog, eplib/cqueue.h:95-183 — 1000-entry cache-line ring; ours carries
-// pointers, payloads stay in the request).
-class SpscRing {
+// pointers, payloads stay in the request). Multi-producer because any
+// application thread may Start() a request concurrently (torch autograd
+// hooks, user threads); per-slot sequence numbers (Vyukov bounded queue)
+// make Push safe from any thread while Pop stays single-consumer (the
+// progress thread).
+class MpscRing {
   public:
     static constexpr size_t kCap = 1024;
+    MpscRing() {
+        for (size_t i = 0; i < kCap; ++i)
+            slots_[i].seq.store(i, std::memory_order_relaxed);
+    }
     bool Push(CommRequest* r) {
-        const uint64_t t = tail_.load(std::memory_order_relaxed);
-        if (t - head_.load(std::memory_order_acquire) >= kCap) return false;
-        slots_[t % kCap] = r;
-        tail_.store(t + 1, std::memory_order_release);
-        return true;
+        uint64_t t = tail_.load(std::memory_order_relaxed);
+        for (;;) {
+            Slot& s = slots_[t % kCap];
+            const uint64_t seq = s.seq.load(std::memory_order_acquire);
+            const int64_t dif = static_cast<int64_t>(seq) - static_cast<int64_t>(t);
+            if (dif == 0) {
+                if (tail_.compare_exchange_weak(t, t + 1,
+                                                std::memory_order_relaxed))
+                {
+                    s.req = r;
+                    s.seq.store(t + 1, std::memory_order_release);
+                    return true;
+                }
+                // CAS failure reloaded t; retry.
+            } else if (dif < 0) {
+                return false;  // full
+            } else {
+                t = tail_.load(std::memory_order_relaxed);
+            }
+        }
     }
     bool Empty() const {
         return head_.load(std::memory_order_acquire) ==
@@ -45,16 +68,24 @@ class SpscRing {
     }
     CommRequest* Pop() {
         const uint64_t h = head_.load(std::memory_order_relaxed);
-        if (h == tail_.load(std::memory_order_acquire)) return nullptr;
-        CommRequest* r = slots_[h % kCap];
+        Slot& s = slots_[h % kCap];
+        const uint64_t seq = s.seq.load(std::memory_order_acquire);
+        if (static_cast<int64_t>(seq) - static_cast<int64_t>(h + 1) < 0)
+            return nullptr;
+        CommRequest* r = s.req;
+        s.seq.store(h + kCap, std::memory_order_release);
         head_.store(h + 1, std::memory_order_release);
         return r;
     }
 
   private:
+    struct Slot {
+        std::atomic<uint64_t> seq{0};
+        CommRequest* req = nullptr;
+    };
     alignas(64) std::atomic<uint64_t> head_{0};
     alignas(64) std::atomic<uint64_t> tail_{0};
-    alignas(64) CommRequest* slots_[kCap] = {};
+    alignas(64) Slot slots_[kCap];
 };
 
 class Engine {
@@ -84,9 +115,10 @@ class Engine {
     std::atomic<bool> stop_{false};
     std::thread thread_;
 
-    SpscRing ring_;
-    std::mutex inbox_mu_;            // fallback for multi-producer submit
+    MpscRing ring_;
+    std::mutex inbox_mu_;            // overflow path when the ring is full
     std::deque<CommRequest*> inbox_overflow_;
+    std::atomic<bool> overflow_pending_{false};
     std::atomic<uint64_t> seqno_{1};
 
     std::vector<CommRequest*> active_;

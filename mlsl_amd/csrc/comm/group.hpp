@@ -4,7 +4,9 @@
 // matches MPI_Comm_split with key = world rank.
 #pragma once
 
+#include <atomic>
 #include <cstdint>
+#include <mutex>
 #include <unordered_map>
 #include <vector>
 
@@ -24,14 +26,25 @@ class ProcessGroup {
     int WorldRank(int group_idx) const { return ranks_[group_idx]; }
     const std::vector<int>& Ranks() const { return ranks_; }
 
-    uint32_t NextFlow() { return flow_seq_++; }
+    // Thread-safe counter draw (application threads may Start requests
+    // concurrently). NOTE: like NCCL, collectives on ONE group must still
+    // be issued in the same order on every rank — concurrent threads
+    // sharing a group need their own ordering (or their own groups); the
+    // atomicity here only guarantees distinct tags, not cross-rank order.
+    uint32_t NextFlow() { return flow_seq_.fetch_add(1, std::memory_order_relaxed); }
 
     // Per-directed-edge message sequence for point-to-point (SRLIST) tags:
     // sender counts messages it sent TO peer, receiver counts messages it
     // received FROM peer — the two counters advance in lock-step per edge
     // (NCCL-style p2p matching), independent of any other group activity.
-    uint32_t NextSendSeq(int peer) { return send_seq_[peer]++; }
-    uint32_t NextRecvSeq(int peer) { return recv_seq_[peer]++; }
+    uint32_t NextSendSeq(int peer) {
+        std::lock_guard<std::mutex> lk(seq_mu_);
+        return send_seq_[peer]++;
+    }
+    uint32_t NextRecvSeq(int peer) {
+        std::lock_guard<std::mutex> lk(seq_mu_);
+        return recv_seq_[peer]++;
+    }
 
     // Device-side communicators (RCCL comms + streams per channel), created
     // lazily by the device transport. Owned here so persistent requests can
@@ -43,7 +56,8 @@ class ProcessGroup {
     int uid_;
     std::vector<int> ranks_;
     int my_idx_ = -1;
-    uint32_t flow_seq_ = 0;
+    std::atomic<uint32_t> flow_seq_{0};
+    std::mutex seq_mu_;
     std::unordered_map<int, uint32_t> send_seq_, recv_seq_;
     DeviceComm* device_ = nullptr;
 };

@@ -284,6 +284,12 @@ void CommRequest::BuildChunks() {
                 break;
         }
         total_tmp_bytes_ += ce.sch.tmp_bytes;
+        // Phase byte 0xFF is reserved for edge-sequenced p2p tags (PairTag);
+        // a schedule reaching phase 255 (ring allreduce at 129+ ranks,
+        // alltoall at 256+) would collapse the p2p/collective tag spaces.
+        MLSL_CHECK(ce.sch.num_phases < 255,
+                   "schedule phase count exceeds the 8-bit tag field "
+                   "(group too large for this algorithm)");
     }
 }
 

@@ -45,10 +45,17 @@ class DistributedData:
                     p.data.copy_(flat.view_as(p.data))
 
         # bucket assignment in reverse parameter order (grads arrive
-        # roughly back-to-front during backward)
+        # roughly back-to-front during backward), partitioned by dtype: a
+        # bucket must be dtype-uniform because torch.cat would silently
+        # promote mixed grads while the allreduce dtype comes from one
+        # member param (torch DDP keys its buckets the same way)
         self._buckets = []
-        cur, cur_bytes = [], 0
+        cur, cur_bytes, cur_dtype = [], 0, None
         for p in reversed(self._params):
+            if cur and p.dtype != cur_dtype:
+                self._buckets.append(cur)
+                cur, cur_bytes = [], 0
+            cur_dtype = p.dtype
             cur.append(p)
             cur_bytes += p.numel() * p.element_size()
             if cur_bytes >= self.bucket_bytes:

@@ -193,7 +193,13 @@ class MlslProcessGroup(dist.ProcessGroup):
 
     def send(self, tensor_list, dst_rank, tag=0):
         # matched pairwise with the peer's recv (NCCL-style semantics);
-        # implemented over send_recv_list with a send-only pair
+        # implemented over send_recv_list with a send-only pair. Matching is
+        # per-directed-edge sequence order only — distinct user tags would
+        # silently mismatch payloads, so refuse them loudly (NCCL itself
+        # ignores tags; we choose the error over the silent drop).
+        if tag != 0:
+            raise NotImplementedError("mlsl backend: send/recv tag must be 0 "
+                                      "(matching is per-edge sequence order)")
         for t in tensor_list:
             c = t.contiguous()
             mx.wait(self._dist.send_recv_list(
@@ -202,6 +208,9 @@ class MlslProcessGroup(dist.ProcessGroup):
         return _done(tensor_list)
 
     def recv(self, tensor_list, src_rank, tag=0):
+        if tag != 0:
+            raise NotImplementedError("mlsl backend: send/recv tag must be 0 "
+                                      "(matching is per-edge sequence order)")
         for t in tensor_list:
             c = t if t.is_contiguous() else t.contiguous()
             mx.wait(self._dist.send_recv_list(

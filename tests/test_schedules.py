@@ -50,3 +50,30 @@ def test_api_selftest_asan():
         for r, p in enumerate(procs):
             out, _ = p.communicate(timeout=180)
             assert p.returncode == 0 and "PASSED" in out, f"rank {r}: {out[-2000:]}"
+
+
+def test_submit_race_tsan():
+    """Concurrent Start() from 4 application threads under TSan: regression
+    test for the round-1 multi-producer submit race on the command ring
+    (engine.hpp MpscRing). World 2 over TCP, per-thread groups."""
+    import os
+    import subprocess
+    from tests.mp import free_port
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    path = os.path.join(repo, "build", "submit_race_tsan")
+    if not os.path.exists(path):
+        subprocess.run(["make", "tsan-submit"], cwd=repo, check=True,
+                       capture_output=True, timeout=900)
+    port = free_port()
+    procs = []
+    for r in range(2):
+        env = dict(os.environ, RANK=str(r), WORLD_SIZE="2",
+                   MASTER_ADDR="127.0.0.1", MLSL_PORT=str(port),
+                   MLSL_TRANSPORT="tcp", THREADS="4", ITERS="30",
+                   TSAN_OPTIONS="halt_on_error=1")
+        procs.append(subprocess.Popen([path], env=env, cwd=repo,
+                                      stdout=subprocess.PIPE,
+                                      stderr=subprocess.STDOUT, text=True))
+    for r, p in enumerate(procs):
+        out, _ = p.communicate(timeout=300)
+        assert p.returncode == 0 and "PASSED" in out, f"rank {r}: {out[-3000:]}"
