@@ -26,6 +26,7 @@ CSRC := \
     mlsl_amd/csrc/comm/request.cpp \
     mlsl_amd/csrc/comm/engine.cpp \
     mlsl_amd/csrc/comm/context.cpp \
+    mlsl_amd/csrc/comm/p2p_transport.cpp \
     mlsl_amd/csrc/comm/device_comm.cpp \
     mlsl_amd/csrc/dl/environment.cpp \
     mlsl_amd/csrc/dl/session.cpp \

@@ -13,6 +13,9 @@
 #include <rccl/rccl.h>
 
 #include <cstdlib>
+#include <cstring>
+#include <memory>
+#include <unistd.h>
 #include <unordered_map>
 #include <vector>
 
@@ -20,7 +23,9 @@
 #include "../core/log.hpp"
 #include "device_pool.hpp"
 #include "../hip/kernels.hpp"
+#include "bootstrap.hpp"
 #include "context.hpp"
+#include "p2p_transport.hpp"
 #include "quant.hpp"
 #include "device_state.hpp"
 #include "group.hpp"
@@ -96,6 +101,10 @@ struct GroupComms {
     // reference's newest-first server scan).
     hipStream_t prio_stream = nullptr;
     ncclComm_t prio_comm = nullptr;
+    // IPC window transport (comm/p2p_transport.hpp): set instead of comms
+    // when the runtime decided on p2p (several ranks on one device, or
+    // MLSL_DEVICE_TRANSPORT=p2p). Streams above are still the lanes.
+    std::unique_ptr<P2pGroup> p2p;
 };
 
 class HipRuntime;
@@ -141,12 +150,74 @@ class HipRuntime : public DeviceRuntime {
         return "hip:?";
     }
 
-    // Collective over the WORLD: unique-id exchange runs on the bootstrap;
-    // only group members init comms.
+    bool UsesP2p() const override { return use_p2p_; }
+
+    // Transport decision, made exactly once and identically on every rank
+    // (collective over the world bootstrap): RCCL refuses two ranks of one
+    // communicator on the same device ("Duplicate GPU detected" — verified
+    // on this pool), so when any device is shared by several world ranks
+    // the IPC window transport takes over for ALL groups. Env override:
+    // MLSL_DEVICE_TRANSPORT=rccl|p2p.
+    void DecideTransport() {
+        if (transport_decided_) return;
+        transport_decided_ = true;
+        const std::string& want = GlobalConfig().device_transport;
+        Context& ctx = Context::Get();
+        struct HostDev { uint64_t host; int dev; };
+        HostDev mine{0, device_id_};
+        char hn[256] = {0};
+        (void)gethostname(hn, sizeof(hn) - 1);
+        for (const char* p = hn; *p; ++p) mine.host = mine.host * 131 + *p;
+        std::vector<HostDev> all(static_cast<size_t>(ctx.Size()));
+        ctx.Boot()->Allgather(&mine, sizeof(HostDev), all.data());
+        bool shared = false;
+        for (size_t i = 0; i < all.size() && !shared; ++i)
+            for (size_t j = i + 1; j < all.size(); ++j)
+                if (all[i].host == all[j].host && all[i].dev == all[j].dev) {
+                    shared = true;
+                    break;
+                }
+        if (want == "p2p") use_p2p_ = true;
+        else if (want == "rccl") use_p2p_ = false;
+        else use_p2p_ = shared;
+        if (use_p2p_)
+            MLSL_LOG(INFO, "device transport: IPC p2p windows (%s)",
+                     want == "p2p" ? "forced" : "shared-device layout");
+    }
+
+    // Collective over the WORLD: unique-id / window-handle exchange runs on
+    // the bootstrap; only group members init comms / map windows.
     void EnsureGroupComms(ProcessGroup* g) override {
         if (group_comms_.count(g->Uid())) return;
+        DecideTransport();
         Context& ctx = Context::Get();
-        const size_t nch = GlobalConfig().num_channels;
+        const Config& cfg = GlobalConfig();
+        const size_t nch = cfg.num_channels;
+        if (use_p2p_) {
+            GroupComms gc;
+            const size_t nlanes = nch + (cfg.msg_priority ? 1 : 0);
+            gc.p2p = P2pGroup::Create(g, nlanes, cfg.p2p_slots,
+                                      cfg.p2p_slot_mb << 20);
+            if (g->IsMember() && g->Size() > 1) {
+                for (size_t ch = 0; ch < nch; ++ch) {
+                    hipStream_t s;
+                    HIP_CHECKD(hipStreamCreateWithFlags(&s, hipStreamNonBlocking));
+                    gc.streams.push_back(s);
+                }
+                if (cfg.msg_priority) {
+                    int lo = 0, hi = 0;
+                    HIP_CHECKD(hipDeviceGetStreamPriorityRange(&lo, &hi));
+                    HIP_CHECKD(hipStreamCreateWithPriority(
+                        &gc.prio_stream, hipStreamNonBlocking, hi));
+                }
+            } else {
+                gc.p2p.reset();  // non-members keep no window state
+            }
+            group_comms_.emplace(g->Uid(), std::move(gc));
+            MLSL_LOG(DEBUG, "p2p device comms ready for group uid=%d size=%d "
+                     "lanes=%zu", g->Uid(), g->Size(), nlanes);
+            return;
+        }
         GroupComms gc;
         for (size_t ch = 0; ch < nch; ++ch) {
             ncclUniqueId id{};
@@ -196,6 +267,8 @@ class HipRuntime : public DeviceRuntime {
     hipStream_t compute_stream_ = nullptr;  // null = legacy default stream
     std::unique_ptr<DevicePool> pool_;
     std::unordered_map<int, GroupComms> group_comms_;
+    bool transport_decided_ = false;
+    bool use_p2p_ = false;
 };
 
 }  // namespace
@@ -528,10 +601,16 @@ void EnsureEvents(DeviceReqState& st, size_t n) {
 
 bool DeviceAdvanceRequest(CommRequest* req, DeviceReqState& st) {
     auto* rt = static_cast<HipRuntime*>(Context::Get().Device());
-    if (st.issued) return AllEventsDone(st);
-
     GroupComms& gc = rt->For(req->Group());
-    if (gc.comms.empty() && gc.streams.empty()) {
+    if (st.issued) {
+        // Wait kernels abort (host word / wall-clock bound) instead of
+        // wedging the GPU; surface that as a loud request failure.
+        if (gc.p2p && !gc.p2p->Healthy())
+            MLSL_THROW("p2p transport wait aborted (peer dead or timeout)");
+        return AllEventsDone(st);
+    }
+    const bool p2p = gc.p2p != nullptr;
+    if (gc.comms.empty() && gc.streams.empty() && !p2p) {
         hipStream_t s0;
         HIP_CHECKD(hipStreamCreateWithFlags(&s0, hipStreamNonBlocking));
         gc.streams.push_back(s0);
@@ -541,7 +620,7 @@ bool DeviceAdvanceRequest(CommRequest* req, DeviceReqState& st) {
     // issue directly on the compute stream — same-stream ordering makes
     // the dep-event handshake (two extra HIP calls of the small-message
     // floor) unnecessary. Large messages keep the side stream for overlap.
-    const bool short_local = gc.comms.empty() &&
+    const bool short_local = gc.comms.empty() && !p2p &&
                              req->MessageBytes() <= cfg.max_short_msg;
     hipStream_t base_s = short_local ? static_cast<hipStream_t>(rt->ComputeStream())
                                      : gc.streams[0];
@@ -612,7 +691,8 @@ bool DeviceAdvanceRequest(CommRequest* req, DeviceReqState& st) {
     auto& chunks = req->Chunks();
     const bool use_schedule = req->UsesDeviceSchedule();
     const bool compressed = req->Compressed();
-    const bool prio = cfg.msg_priority && gc.prio_comm &&
+    const bool prio = cfg.msg_priority &&
+                      (gc.prio_comm || (p2p && gc.prio_stream)) &&
                       req->MessageBytes() >= cfg.msg_priority_threshold;
     if (prio) {
         // Order the priority lane after the producers and any staging H2D:
@@ -630,15 +710,18 @@ bool DeviceAdvanceRequest(CommRequest* req, DeviceReqState& st) {
     // Graph capture only when everything lands on ONE capturable stream:
     // no pageable staging, no priority lane, single channel.
     const bool try_capture = cfg.use_graphs && !st.graph_failed && !s_host &&
-                             !r_host && !prio && !short_local &&
+                             !r_host && !prio && !short_local && !p2p &&
                              GraphEligible(req, gc);
 
     const size_t es = DtypeSize(req->Dtype());
-    const size_t nch = gc.comms.empty() ? 1 : gc.comms.size();
-    const size_t used = gc.comms.empty() ? 1 : std::min(chunks.size(), nch);
+    const size_t nch = p2p ? std::max<size_t>(1, gc.streams.size())
+                           : (gc.comms.empty() ? 1 : gc.comms.size());
+    const size_t used = (gc.comms.empty() && !p2p)
+                            ? 1
+                            : std::min(chunks.size(), nch);
 
     auto issue_all = [&]() {
-        if (gc.comms.empty()) {
+        if (gc.comms.empty() && !p2p) {
             hipStream_t s0 = base_s;
             for (auto& ce : chunks) {
                 const uint8_t* sbase = req->SendBuf() + ce.elem_off * es;
@@ -670,8 +753,36 @@ bool DeviceAdvanceRequest(CommRequest* req, DeviceReqState& st) {
         size_t tmp_off = 0;
         for (auto& ce : chunks) {
             const size_t ch = ce.chunk_idx % nch;
-            ncclComm_t comm_ = prio ? gc.prio_comm : gc.comms[ch];
             hipStream_t strm_ = prio ? gc.prio_stream : gc.streams[ch];
+            if (p2p) {
+                // IPC window transport: every op walks its schedule; lane
+                // nlanes-1 is the priority lane.
+                const size_t lane = prio ? gc.p2p->Lanes() - 1 : ch;
+                uint8_t* tbase = static_cast<uint8_t*>(st.tmp_dev) + tmp_off;
+                if (compressed) {
+                    const size_t wire_b = (ce.sch.result.bytes + 15) & ~size_t(15);
+                    uint8_t* wire = tbase;
+                    uint8_t* scratch = tbase + wire_b;
+                    uint8_t* err = scratch + ((ce.sch.tmp_bytes + 15) & ~size_t(15));
+                    const size_t blk = req->QParams().block_elems;
+                    LaunchQuantize(req->SendBuf(), err, wire, req->Spec().count,
+                                   blk, req->Dtype(), true, strm_);
+                    gc.p2p->IssueSchedule(req, ce, lane, strm_, wire, wire,
+                                          scratch);
+                    LaunchDequantize(wire, req->RecvBuf(), req->Spec().count,
+                                     blk, req->Dtype(), strm_);
+                    tmp_off += wire_b + ((ce.sch.tmp_bytes + 15) & ~size_t(15)) +
+                               ((req->Spec().count * es + 15) & ~size_t(15));
+                } else {
+                    gc.p2p->IssueSchedule(req, ce, lane, strm_,
+                                          req->SendBuf() + ce.elem_off * es,
+                                          req->RecvBuf() + ce.elem_off * es,
+                                          tbase);
+                    tmp_off += ce.sch.tmp_bytes;
+                }
+                continue;
+            }
+            ncclComm_t comm_ = prio ? gc.prio_comm : gc.comms[ch];
             uint8_t* tbase = static_cast<uint8_t*>(st.tmp_dev) + tmp_off;
             if (compressed) {
                 // quantize -> compressed-domain ring -> dequantize (driver
@@ -738,10 +849,10 @@ bool DeviceAdvanceRequest(CommRequest* req, DeviceReqState& st) {
 
     // Completion events (outside any graph). Graph path + single-channel
     // path complete on stream 0; multi-channel records one per channel.
-    if (st.graph_exec || gc.comms.empty()) {
+    if (st.graph_exec || (gc.comms.empty() && !p2p)) {
         EnsureEvents(st, 1);
         HIP_CHECKD(hipEventRecord(st.events[0],
-                                  gc.comms.empty() ? base_s : gc.streams[0]));
+                                  st.graph_exec ? gc.streams[0] : base_s));
     } else {
         EnsureEvents(st, used);
         for (size_t ch = 0; ch < used; ++ch)

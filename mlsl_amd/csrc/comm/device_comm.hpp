@@ -29,6 +29,11 @@ class DeviceRuntime {
     // issued after SetComputeStream order behind work on that stream.
     virtual void SetComputeStream(void* stream) = 0;
     virtual void* ComputeStream() const = 0;
+    // True when multi-rank groups use the IPC window transport instead of
+    // RCCL comms (MLSL_DEVICE_TRANSPORT, or auto-detected when several
+    // ranks share one device). Decided once at context init, identically
+    // on every rank.
+    virtual bool UsesP2p() const = 0;
 };
 
 // Factory: returns nullptr when no HIP device is visible.

@@ -1,0 +1,322 @@
+#include "p2p_transport.hpp"
+
+#include <cstring>
+#include <mutex>
+
+#include "../core/config.hpp"
+#include "../core/log.hpp"
+#include "../hip/kernels.hpp"
+#include "bootstrap.hpp"
+#include "context.hpp"
+#include "group.hpp"
+#include "request.hpp"
+
+namespace mlsl {
+
+#define HIP_CHECKP(cmd)                                                       \
+    do {                                                                      \
+        hipError_t e_ = (cmd);                                                \
+        if (e_ != hipSuccess)                                                 \
+            MLSL_THROW(std::string("HIP error (p2p): ") +                     \
+                       hipGetErrorString(e_));                                \
+    } while (0)
+
+namespace {
+// Flag mailboxes are 128-byte-strided so flags written by different peers
+// never share a cache line (remote stores would otherwise ping-pong lines
+// between GPUs over xGMI).
+constexpr size_t kFlagStride = 128;
+std::mutex g_issue_mu;  // enqueue-order == counter-order per group
+
+struct WireHandle {
+    hipIpcMemHandle_t h;
+    int valid;
+};
+}  // namespace
+
+std::unique_ptr<P2pGroup> P2pGroup::Create(ProcessGroup* g, size_t nlanes,
+                                           size_t nslots, size_t slot_bytes) {
+    Context& ctx = Context::Get();
+    auto pg = std::unique_ptr<P2pGroup>(new P2pGroup());
+    pg->gsize_ = g->Size();
+    pg->my_idx_ = g->MyIdx();
+    pg->nlanes_ = nlanes;
+    pg->nslots_ = nslots;
+    pg->slot_bytes_ = slot_bytes;
+    const size_t N = static_cast<size_t>(pg->gsize_);
+    pg->flags_bytes_ = N * nlanes * 2 * kFlagStride;
+    pg->win_bytes_ = pg->flags_bytes_ + N * nlanes * nslots * slot_bytes;
+
+    WireHandle mine{};
+    std::memset(&mine, 0, sizeof(mine));
+    const bool member = g->IsMember() && g->Size() > 1;
+    if (member) {
+        HIP_CHECKP(hipMalloc(reinterpret_cast<void**>(&pg->my_base_),
+                             pg->win_bytes_));
+        HIP_CHECKP(hipMemset(pg->my_base_, 0, pg->win_bytes_));
+        HIP_CHECKP(hipIpcGetMemHandle(&mine.h, pg->my_base_));
+        mine.valid = 1;
+        HIP_CHECKP(hipHostMalloc(reinterpret_cast<void**>(&pg->abort_host_),
+                                 2 * sizeof(uint32_t)));
+        pg->status_host_ = pg->abort_host_ + 1;
+        pg->abort_host_[0] = 0;
+        pg->status_host_[0] = 0;
+        const int tmo = GlobalConfig().timeout_sec;
+        pg->max_ticks_ = static_cast<uint64_t>(tmo > 0 ? tmo : 300) * 100000000ull;
+    }
+
+    // Same exchange pattern as the RCCL unique-id path: world-wide
+    // bootstrap allgather, indexed by world rank.
+    std::vector<WireHandle> all(static_cast<size_t>(ctx.Size()));
+    ctx.Boot()->Allgather(&mine, sizeof(WireHandle), all.data());
+    if (member) {
+        pg->peer_base_.assign(N, nullptr);
+        pg->sent_.assign(N * nlanes, 0);
+        pg->rcvd_.assign(N * nlanes, 0);
+        for (int i = 0; i < pg->gsize_; ++i) {
+            if (i == pg->my_idx_) {
+                pg->peer_base_[i] = pg->my_base_;
+                continue;
+            }
+            const WireHandle& wh = all[static_cast<size_t>(g->WorldRank(i))];
+            MLSL_CHECK(wh.valid, "p2p window handle missing for a group peer");
+            void* p = nullptr;
+            HIP_CHECKP(hipIpcOpenMemHandle(&p, wh.h,
+                                           hipIpcMemLazyEnablePeerAccess));
+            pg->peer_base_[i] = static_cast<uint8_t*>(p);
+        }
+        MLSL_LOG(DEBUG,
+                 "p2p group ready: size=%d lanes=%zu slots=%zu slot_bytes=%zu "
+                 "window=%zu MiB",
+                 pg->gsize_, nlanes, nslots, slot_bytes,
+                 pg->win_bytes_ >> 20);
+    }
+    return pg;
+}
+
+P2pGroup::~P2pGroup() {
+    if (!my_base_) return;
+    // Unblock any in-flight wait kernels, let the device drain, then unmap.
+    if (abort_host_) abort_host_[0] = 1;
+    (void)hipDeviceSynchronize();
+    for (int i = 0; i < gsize_; ++i)
+        if (peer_base_[i] && i != my_idx_)
+            (void)hipIpcCloseMemHandle(peer_base_[i]);
+    (void)hipFree(my_base_);
+    if (abort_host_) (void)hipHostFree(abort_host_);
+}
+
+bool P2pGroup::Healthy() const {
+    return !status_host_ ||
+           __atomic_load_n(status_host_, __ATOMIC_ACQUIRE) == 0;
+}
+
+void P2pGroup::Abort() {
+    if (abort_host_) __atomic_store_n(abort_host_, 1u, __ATOMIC_RELEASE);
+}
+
+// --- window geometry -------------------------------------------------------
+// Window of owner o: [flags][data]. Flag pair for (src s, lane l) at
+// (s*nlanes+l)*2*kFlagStride: in_flag then ack_flag. in_flag = number of
+// slot-messages s has pushed to o (written by s). ack_flag = number of
+// slot-messages s has CONSUMED of those o pushed to s (also written by s).
+// Data slot (s, l, k): payload staging for s -> o traffic.
+
+uint8_t* P2pGroup::MySlot(int src, size_t lane, size_t slot) const {
+    return my_base_ + flags_bytes_ +
+           (((static_cast<size_t>(src) * nlanes_ + lane) * nslots_) + slot) *
+               slot_bytes_;
+}
+uint8_t* P2pGroup::PeerSlot(int peer, size_t lane, size_t slot) const {
+    return peer_base_[peer] + flags_bytes_ +
+           (((static_cast<size_t>(my_idx_) * nlanes_ + lane) * nslots_) + slot) *
+               slot_bytes_;
+}
+void* P2pGroup::MyInFlag(int src, size_t lane) const {
+    return my_base_ + (static_cast<size_t>(src) * nlanes_ + lane) * 2 * kFlagStride;
+}
+void* P2pGroup::MyAckFlag(int peer, size_t lane) const {
+    return my_base_ +
+           (static_cast<size_t>(peer) * nlanes_ + lane) * 2 * kFlagStride +
+           kFlagStride;
+}
+void* P2pGroup::PeerInFlag(int peer, size_t lane) const {
+    return peer_base_[peer] +
+           (static_cast<size_t>(my_idx_) * nlanes_ + lane) * 2 * kFlagStride;
+}
+void* P2pGroup::PeerAckFlag(int peer, size_t lane) const {
+    return peer_base_[peer] +
+           (static_cast<size_t>(my_idx_) * nlanes_ + lane) * 2 * kFlagStride +
+           kFlagStride;
+}
+
+// --- transport ops ---------------------------------------------------------
+
+void P2pGroup::SendBytes(int peer, size_t lane, hipStream_t s,
+                         const uint8_t* src, size_t bytes, size_t grain) {
+    const size_t msg_max = (slot_bytes_ / grain) * grain;
+    MLSL_CHECK(msg_max > 0, "p2p slot smaller than one element/block");
+    uint64_t& sent = sent_[static_cast<size_t>(peer) * nlanes_ + lane];
+    size_t done = 0;
+    while (done < bytes) {
+        const size_t n = std::min(msg_max, bytes - done);
+        const uint64_t seq = ++sent;
+        const size_t slot = (seq - 1) % nslots_;
+        if (seq > nslots_)
+            LaunchWaitFlag(MyAckFlag(peer, lane), seq - nslots_, abort_host_,
+                           status_host_, max_ticks_, s);
+        HIP_CHECKP(hipMemcpyAsync(PeerSlot(peer, lane, slot), src + done, n,
+                                  hipMemcpyDeviceToDevice, s));
+        LaunchSetFlag(PeerInFlag(peer, lane), seq, s);
+        done += n;
+    }
+}
+
+void P2pGroup::IssueSchedule(CommRequest* req, ChunkExec& ce, size_t lane,
+                             hipStream_t s, const uint8_t* sbase,
+                             uint8_t* rbase, uint8_t* tmp) {
+    std::lock_guard<std::mutex> lk(g_issue_mu);
+    const size_t es = DtypeSize(req->Dtype());
+    const bool quant = ce.sch.quant_block > 0;
+    const size_t unit = quant ? req->QParams().WireBlockBytes() : es;
+    const size_t blk = ce.sch.quant_block;
+    MLSL_CHECK(lane < nlanes_, "p2p lane out of range");
+
+    auto ptr = [&](const BufRef& b) -> uint8_t* {
+        switch (b.space) {
+            case Space::SEND: return const_cast<uint8_t*>(sbase) + b.off;
+            case Space::RECV: return rbase + b.off;
+            case Space::TMP: return tmp + b.off;
+        }
+        return nullptr;
+    };
+    auto same_ref = [](const BufRef& a, const BufRef& b) {
+        return a.space == b.space && a.off == b.off && a.bytes == b.bytes;
+    };
+
+    const size_t msg_max = (slot_bytes_ / unit) * unit;
+    MLSL_CHECK(msg_max > 0, "p2p slot smaller than one element/block");
+
+    for (int phase = 0; phase < ce.sch.num_phases; ++phase) {
+        // Collect this phase's send and recv jobs, then interleave their
+        // sub-messages round-robin. The interleave is what makes the
+        // bounded slot ring deadlock-free: a sender blocked on backpressure
+        // (ack >= seq-kSlots) is guaranteed that its OWN next consume —
+        // which is what its upstream neighbor is waiting on — was already
+        // enqueued at the previous sub-message index. Issuing all sends of
+        // a phase before any recv would cycle-deadlock the ring whenever a
+        // segment needs more sub-messages than there are slots.
+        struct RecvJob {
+            const Step* st;
+            bool fuse_into, fuse_out;
+        };
+        std::vector<const Step*> sends;
+        std::vector<RecvJob> recvs;
+        size_t max_msgs = 0;
+        for (const auto& st : ce.sch.steps) {
+            if (st.phase != phase) continue;
+            if (st.send_peer >= 0 && st.send.bytes > 0) {
+                MLSL_CHECK(st.send_peer != my_idx_, "p2p self-send in schedule");
+                sends.push_back(&st);
+                max_msgs = std::max(max_msgs,
+                                    (st.send.bytes + msg_max - 1) / msg_max);
+            }
+            if (st.recv_peer >= 0 && st.recv.bytes > 0) {
+                MLSL_CHECK(st.recv_peer != my_idx_, "p2p self-recv in schedule");
+                // Fusion cases: the received data IS the local op's source
+                // (ring/RHD reduce), or the local op targets the recv range
+                // (reduce-scatter's out-of-place accumulate).
+                const bool fuse_into = st.local == Step::LocalOp::REDUCE &&
+                                       same_ref(st.local_src, st.recv);
+                const bool fuse_out = !fuse_into &&
+                                      st.local == Step::LocalOp::REDUCE &&
+                                      same_ref(st.local_dst, st.recv) &&
+                                      st.local_src.bytes == st.recv.bytes;
+                recvs.push_back({&st, fuse_into, fuse_out});
+                max_msgs = std::max(max_msgs,
+                                    (st.recv.bytes + msg_max - 1) / msg_max);
+            }
+        }
+        for (size_t k = 0; k < max_msgs; ++k) {
+            const size_t off = k * msg_max;
+            for (const Step* st : sends) {
+                if (off >= st->send.bytes) continue;
+                const size_t n = std::min(msg_max, st->send.bytes - off);
+                const int peer = st->send_peer;
+                uint64_t& sent = sent_[static_cast<size_t>(peer) * nlanes_ + lane];
+                const uint64_t seq = ++sent;
+                const size_t slot = (seq - 1) % nslots_;
+                if (seq > nslots_)
+                    LaunchWaitFlag(MyAckFlag(peer, lane), seq - nslots_,
+                                   abort_host_, status_host_, max_ticks_, s);
+                HIP_CHECKP(hipMemcpyAsync(PeerSlot(peer, lane, slot),
+                                          ptr(st->send) + off, n,
+                                          hipMemcpyDeviceToDevice, s));
+                LaunchSetFlag(PeerInFlag(peer, lane), seq, s);
+            }
+            for (const RecvJob& rj : recvs) {
+                const Step& st = *rj.st;
+                if (off >= st.recv.bytes) continue;
+                const size_t n = std::min(msg_max, st.recv.bytes - off);
+                const int peer = st.recv_peer;
+                uint64_t& rcvd = rcvd_[static_cast<size_t>(peer) * nlanes_ + lane];
+                const uint64_t seq = ++rcvd;
+                const size_t slot = (seq - 1) % nslots_;
+                LaunchWaitFlag(MyInFlag(peer, lane), seq, abort_host_,
+                               status_host_, max_ticks_, s);
+                const uint8_t* sl = MySlot(peer, lane, slot);
+                if (rj.fuse_into) {
+                    if (quant)
+                        LaunchQuantAccum(ptr(st.local_dst) + off, sl,
+                                         (n / unit) * blk, blk, s);
+                    else
+                        LaunchReduce(ptr(st.local_dst) + off, sl, n / es,
+                                     req->Dtype(), ce.sch.rop, s);
+                } else if (rj.fuse_out) {
+                    if (quant) {
+                        // acc = slot, then acc += local_src slice
+                        LaunchCopy(ptr(st.recv) + off, sl, n, s);
+                        LaunchQuantAccum(ptr(st.recv) + off,
+                                         ptr(st.local_src) + off,
+                                         (n / unit) * blk, blk, s);
+                    } else {
+                        LaunchReduceOut(ptr(st.recv) + off, sl,
+                                        ptr(st.local_src) + off, n / es,
+                                        req->Dtype(), ce.sch.rop, s);
+                    }
+                } else {
+                    LaunchCopy(ptr(st.recv) + off, sl, n, s);
+                }
+                LaunchSetFlag(PeerAckFlag(peer, lane), seq, s);
+            }
+        }
+        // Unfused local ops (pure copy steps, or reduce with unrelated
+        // source/dest geometry) run after the phase's traffic, step order.
+        for (const auto& st : ce.sch.steps) {
+            if (st.phase != phase || st.local == Step::LocalOp::NONE) continue;
+            const bool was_recv_step = st.recv_peer >= 0 && st.recv.bytes > 0;
+            if (was_recv_step) {
+                const bool fuse_into = st.local == Step::LocalOp::REDUCE &&
+                                       same_ref(st.local_src, st.recv);
+                const bool fuse_out = !fuse_into &&
+                                      st.local == Step::LocalOp::REDUCE &&
+                                      same_ref(st.local_dst, st.recv) &&
+                                      st.local_src.bytes == st.recv.bytes;
+                if (fuse_into || fuse_out) continue;  // handled per sub-msg
+            }
+            uint8_t* d = ptr(st.local_dst);
+            uint8_t* src = ptr(st.local_src);
+            if (st.local == Step::LocalOp::COPY) {
+                if (d != src) LaunchCopy(d, src, st.local_src.bytes, s);
+            } else if (quant) {
+                LaunchQuantAccum(d, src, (st.local_dst.bytes / unit) * blk,
+                                 blk, s);
+            } else {
+                LaunchReduce(d, src, st.local_dst.bytes / es, req->Dtype(),
+                             ce.sch.rop, s);
+            }
+        }
+    }
+}
+
+}  // namespace mlsl

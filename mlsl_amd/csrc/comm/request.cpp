@@ -149,6 +149,12 @@ bool CommRequest::Compressed() const {
 
 bool CommRequest::UsesDeviceSchedule() const {
     const Config& cfg = GlobalConfig();
+    // IPC window transport: EVERY op walks its schedule (there are no
+    // fused RCCL collectives to fall back on).
+    Context& ctx = Context::Get();
+    if (ctx.DeviceMode() && ctx.Device() && ctx.Device()->UsesP2p() &&
+        group_->Size() > 1)
+        return !Compressed();
     return (cfg.allreduce_algo == AllReduceAlgo::RING ||
             cfg.allreduce_algo == AllReduceAlgo::RHD) &&
            spec_.op == CollOp::ALLREDUCE && group_->Size() > 1 && !Compressed();

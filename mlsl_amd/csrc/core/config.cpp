@@ -49,6 +49,11 @@ Config Config::FromEnv() {
     c.heap_mb = EnvSize("MLSL_HEAP_SIZE_MB", 0);
     c.check_pointers = EnvBool("MLSL_CHECK_POINTERS", false);
     if (const char* e = std::getenv("MLSL_TRANSPORT")) c.transport = e;
+    if (const char* e = std::getenv("MLSL_DEVICE_TRANSPORT")) c.device_transport = e;
+    c.p2p_slot_mb = EnvSize("MLSL_P2P_SLOT_MB", 4);
+    if (c.p2p_slot_mb < 1) c.p2p_slot_mb = 1;
+    c.p2p_slots = EnvSize("MLSL_P2P_SLOTS", 4);
+    if (c.p2p_slots < 2) c.p2p_slots = 2;
     c.timeout_sec = static_cast<int>(EnvSize("MLSL_TIMEOUT", 300));
     c.use_graphs = EnvBool("MLSL_USE_GRAPHS", false);
     return c;

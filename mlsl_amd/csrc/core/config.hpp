@@ -41,6 +41,12 @@ struct Config {
     size_t heap_mb = 0;            // MLSL_HEAP_SIZE_MB: device pool pre-reserve
     bool check_pointers = false;   // MLSL_CHECK_POINTERS: validate collective bufs
     std::string transport = "auto";  // MLSL_TRANSPORT=auto|tcp|rccl
+    // Device transport for multi-rank groups: RCCL comms, or the IPC HBM
+    // window transport (comm/p2p_transport.hpp). auto = RCCL, switching to
+    // p2p when several ranks share one device (RCCL refuses that layout).
+    std::string device_transport = "auto";  // MLSL_DEVICE_TRANSPORT=auto|rccl|p2p
+    size_t p2p_slot_mb = 4;        // MLSL_P2P_SLOT_MB: staging slot size
+    size_t p2p_slots = 4;          // MLSL_P2P_SLOTS: in-flight slots per edge-lane
     int timeout_sec = 300;         // MLSL_TIMEOUT: bootstrap/collective timeout
     bool use_graphs = false;       // MLSL_USE_GRAPHS: hipGraph replay of
                                    // persistent device requests (single

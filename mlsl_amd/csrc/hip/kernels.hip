@@ -1199,4 +1199,60 @@ void LaunchUnpack(const void* src, void* dst, const PackBlockDesc& d, DataType d
     HIP_CHECK(hipGetLastError());
 }
 
+// --- IPC p2p transport flag primitives (comm/p2p_transport.cpp) ---
+// Mailboxes are monotonically increasing u64 sequence counters in the
+// receiver's HBM window; the writer is a remote process on the same or a
+// peer GPU, so both sides use system-scope atomics. The wait kernel is the
+// stream-blocking primitive (one lane; s_sleep between polls) with TWO
+// escape hatches so a lost peer can never wedge the GPU: a host-written
+// abort word (pinned) and a wall-clock bound (s_memrealtime, 100 MHz).
+// On escape it writes 1 to `status` (pinned) and returns — the host
+// request machinery sees it and fails the request loudly.
+
+namespace {
+
+__global__ void WaitFlagKernel(const unsigned long long* __restrict__ mbox,
+                               unsigned long long target,
+                               const unsigned int* __restrict__ abort_word,
+                               unsigned int* __restrict__ status,
+                               unsigned long long max_ticks) {
+    if (threadIdx.x != 0 || blockIdx.x != 0) return;
+    const unsigned long long t0 = wall_clock64();
+    while (__hip_atomic_load(mbox, __ATOMIC_ACQUIRE, __HIP_MEMORY_SCOPE_SYSTEM) <
+           target) {
+        if (__hip_atomic_load(abort_word, __ATOMIC_RELAXED,
+                              __HIP_MEMORY_SCOPE_SYSTEM) != 0 ||
+            wall_clock64() - t0 > max_ticks) {
+            __hip_atomic_store(status, 1u, __ATOMIC_RELEASE,
+                               __HIP_MEMORY_SCOPE_SYSTEM);
+            return;
+        }
+        __builtin_amdgcn_s_sleep(64);
+    }
+}
+
+__global__ void SetFlagKernel(unsigned long long* __restrict__ mbox,
+                              unsigned long long val) {
+    if (threadIdx.x == 0 && blockIdx.x == 0)
+        __hip_atomic_store(mbox, val, __ATOMIC_RELEASE,
+                           __HIP_MEMORY_SCOPE_SYSTEM);
+}
+
+}  // namespace
+
+void LaunchWaitFlag(const void* mbox, uint64_t target, const void* abort_word,
+                    void* status, uint64_t max_ticks, hipStream_t stream) {
+    hipLaunchKernelGGL(WaitFlagKernel, dim3(1), dim3(1), 0, stream,
+                       static_cast<const unsigned long long*>(mbox), target,
+                       static_cast<const unsigned int*>(abort_word),
+                       static_cast<unsigned int*>(status), max_ticks);
+    HIP_CHECK(hipGetLastError());
+}
+
+void LaunchSetFlag(void* mbox, uint64_t val, hipStream_t stream) {
+    hipLaunchKernelGGL(SetFlagKernel, dim3(1), dim3(1), 0, stream,
+                       static_cast<unsigned long long*>(mbox), val);
+    HIP_CHECK(hipGetLastError());
+}
+
 }  // namespace mlsl

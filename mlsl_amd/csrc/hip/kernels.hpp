@@ -67,4 +67,13 @@ void LaunchPack(const void* src, void* dst, const PackBlockDesc& d, DataType dt,
 void LaunchUnpack(const void* src, void* dst, const PackBlockDesc& d, DataType dt,
                   hipStream_t stream);
 
+// --- IPC p2p transport flag primitives ---
+// Stream-blocking wait until *mbox >= target (system-scope acquire), with a
+// host abort word and a wall-clock bound (ticks of the 100 MHz constant
+// clock); on abort/timeout writes 1 to `status` (pinned) and returns.
+void LaunchWaitFlag(const void* mbox, uint64_t target, const void* abort_word,
+                    void* status, uint64_t max_ticks, hipStream_t stream);
+// System-scope release store of a new sequence value.
+void LaunchSetFlag(void* mbox, uint64_t val, hipStream_t stream);
+
 }  // namespace mlsl

@@ -19,4 +19,6 @@ void LaunchDequantizeNT(const void*, void*, size_t, size_t, DataType, hipStream_
 void LaunchQuantAccum(void*, const void*, size_t, size_t, hipStream_t) STUB()
 void LaunchPack(const void*, void*, const PackBlockDesc&, DataType, hipStream_t) STUB()
 void LaunchUnpack(const void*, void*, const PackBlockDesc&, DataType, hipStream_t) STUB()
+void LaunchWaitFlag(const void*, uint64_t, const void*, void*, uint64_t, hipStream_t) STUB()
+void LaunchSetFlag(void*, uint64_t, hipStream_t) STUB()
 }  // namespace mlsl

@@ -1,0 +1,241 @@
+"""GPU multi-rank worker entry points (all ranks on ONE device).
+
+RCCL refuses several ranks of one communicator on the same GPU, so these
+worlds exercise the IPC window transport (csrc/comm/p2p_transport.cpp) —
+the same schedule executor that runs over xGMI on a real 8-GPU node. The
+reference bar is the 4-rank transport matrix of
+/root/reference/tests/examples/mlsl_test/Makefile:59-107.
+
+Run as: python -m tests.workers_gpu <worker_name>   (see test_gpu_multirank)
+"""
+import os
+import sys
+
+
+def _init():
+    import torch
+    import mlsl_amd as mx
+    torch.cuda.set_device(0)
+    mx.init()
+    return mx, torch, mx.rank(), mx.world_size()
+
+
+def _arange(torch, count, rank):
+    return torch.arange(count, dtype=torch.float32, device="cuda") + rank
+
+
+def gpu_collectives():
+    """Every collective with analytic expected values on cuda tensors."""
+    mx, torch, rank, size = _init()
+    d = mx.Distribution(size, 1)
+    g = "data"
+
+    # allreduce sum, sizes spanning one-slot and multi-slot paths
+    for count in (1, 257, 65536, 1 << 20):
+        a = _arange(torch, count, rank)
+        out = torch.empty_like(a)
+        mx.wait(d.all_reduce(a, out, count, op="sum", group=g))
+        torch.cuda.synchronize()
+        want = size * torch.arange(count, dtype=torch.float32, device="cuda") \
+            + size * (size - 1) / 2.0
+        assert torch.allclose(out, want), ("allreduce", count, out[:4], want[:4])
+
+    # in-place allreduce + max
+    count = 4099
+    a = _arange(torch, count, rank)
+    mx.wait(d.all_reduce(a, a, count, op="sum", group=g))
+    torch.cuda.synchronize()
+    want = size * torch.arange(count, dtype=torch.float32, device="cuda") \
+        + size * (size - 1) / 2.0
+    assert torch.allclose(a, want), "in-place allreduce"
+    a = _arange(torch, count, rank)
+    mx.wait(d.all_reduce(a, a, count, op="max", group=g))
+    torch.cuda.synchronize()
+    wantm = torch.arange(count, dtype=torch.float32, device="cuda") + (size - 1)
+    assert torch.allclose(a, wantm), "max allreduce"
+
+    # bcast from root 0 and root size-1
+    for root in (0, size - 1):
+        count = 3333
+        b = _arange(torch, count, rank) if rank == root \
+            else torch.zeros(count, device="cuda")
+        mx.wait(d.bcast(b, count, root=root, group=g))
+        torch.cuda.synchronize()
+        assert torch.allclose(b, _arange(torch, count, root)), ("bcast", root)
+
+    # reduce to root
+    root = 1 % size
+    count = 2048
+    a = _arange(torch, count, rank)
+    out = torch.empty_like(a)
+    mx.wait(d.reduce(a, out, count, op="sum", root=root, group=g))
+    torch.cuda.synchronize()
+    if rank == root:
+        want = size * torch.arange(count, dtype=torch.float32, device="cuda") \
+            + size * (size - 1) / 2.0
+        assert torch.allclose(out, want), "reduce"
+
+    # reduce_scatter: rank r gets sum over ranks of seg r
+    per = 1536
+    src = torch.cat([_arange(torch, per, rank) + i * 1000 for i in range(size)])
+    out = torch.empty(per, device="cuda")
+    mx.wait(d.reduce_scatter(src, out, per, op="sum", group=g))
+    torch.cuda.synchronize()
+    want = size * torch.arange(per, dtype=torch.float32, device="cuda") \
+        + size * (size - 1) / 2.0 + size * rank * 1000
+    assert torch.allclose(out, want), "reduce_scatter"
+
+    # all_gather
+    per = 777
+    mine = _arange(torch, per, rank) * (rank + 1)
+    flat = torch.empty(size * per, device="cuda")
+    mx.wait(d.all_gather(mine, per, flat, group=g))
+    torch.cuda.synchronize()
+    for r in range(size):
+        want = (_arange(torch, per, r)) * (r + 1)
+        assert torch.allclose(flat[r * per:(r + 1) * per], want), ("allgather", r)
+
+    # all_to_all: block j of rank i == i*100 + j
+    per = 512
+    src = torch.cat([torch.full((per,), float(rank * 100 + j), device="cuda")
+                     for j in range(size)])
+    dst = torch.empty_like(src)
+    mx.wait(d.all_to_all(src, per, dst, group=g))
+    torch.cuda.synchronize()
+    for j in range(size):
+        want = float(j * 100 + rank)
+        assert torch.all(dst[j * per:(j + 1) * per] == want), ("alltoall", j)
+
+    # gather / scatter
+    root = 0
+    per = 333
+    mine = _arange(torch, per, rank) + rank * 7
+    flat = torch.empty(size * per, device="cuda") if rank == root \
+        else torch.empty(1, device="cuda")
+    mx.wait(d.gather(mine, per, flat, root=root, group=g))
+    torch.cuda.synchronize()
+    if rank == root:
+        for r in range(size):
+            want = _arange(torch, per, r) + r * 7
+            assert torch.allclose(flat[r * per:(r + 1) * per], want), ("gather", r)
+    if rank == root:
+        sflat = torch.cat([_arange(torch, per, r) * 2 for r in range(size)])
+    else:
+        sflat = torch.empty(1, device="cuda")
+    rout = torch.empty(per, device="cuda")
+    mx.wait(d.scatter(sflat, rout, per, root=root, group=g))
+    torch.cuda.synchronize()
+    assert torch.allclose(rout, _arange(torch, per, rank) * 2), "scatter"
+
+    # p2p ring over send_recv_list
+    n = 1024
+    dst_rank = (rank + 1) % size
+    src_rank = (rank - 1 + size) % size
+    sendv = torch.full((n,), float(rank), device="cuda")
+    recvv = torch.empty(n, device="cuda")
+    mx.wait(d.send_recv_list(sendv, recvv,
+                             [(dst_rank, 0, n, 0, 0), (src_rank, 0, 0, 0, n)],
+                             group=g))
+    torch.cuda.synchronize()
+    assert torch.all(recvv == float(src_rank)), "srlist ring"
+
+    d.barrier(g)
+    d.barrier("global")
+    mx.finalize()
+
+
+def gpu_allreduce_multislot():
+    """Large message: segments span many transport slots (backpressure +
+    interleave), both ring and RHD, in-place and out-of-place."""
+    mx, torch, rank, size = _init()
+    d = mx.Distribution(size, 1)
+    count = 1 << 22  # 16 MiB fp32
+    for _ in range(3):
+        a = _arange(torch, count, rank)
+        out = torch.empty_like(a)
+        mx.wait(d.all_reduce(a, out, count, op="sum", group="data"))
+        torch.cuda.synchronize()
+        want = size * torch.arange(count, dtype=torch.float32, device="cuda") \
+            + size * (size - 1) / 2.0
+        assert torch.allclose(out, want), "multislot out-of-place"
+        a = _arange(torch, count, rank)
+        mx.wait(d.all_reduce(a, a, count, op="sum", group="data"))
+        torch.cuda.synchronize()
+        assert torch.allclose(a, want), "multislot in-place"
+    d.barrier("global")
+    mx.finalize()
+
+
+def gpu_quant_allreduce():
+    """int8-quantized allreduce over the compressed-domain ring (driver
+    config 5 shape, small size): relative error bounded, error feedback on."""
+    mx, torch, rank, size = _init()
+    d = mx.Distribution(size, 1)
+    count = 1 << 20
+    torch.manual_seed(100 + rank)
+    for dt, name in ((torch.float32, "f32"), (torch.bfloat16, "bf16")):
+        a = torch.randn(count, dtype=dt, device="cuda")
+        out = torch.empty_like(a)
+        req = mx.PersistentRequest(d, "all_reduce", count, dtype=name,
+                                   op="sum", group="data", quantized=True)
+        req.start(a, out)
+        req.wait()
+        torch.cuda.synchronize()
+        # exact sum via non-quantized path for comparison
+        exact = torch.empty_like(a)
+        mx.wait(d.all_reduce(a, exact, count, op="sum", dtype=name, group="data"))
+        torch.cuda.synchronize()
+        num = (out.float() - exact.float()).norm()
+        den = exact.float().norm().clamp_min(1e-6)
+        rel = (num / den).item()
+        assert rel < 0.05, (name, rel)
+        req.destroy()
+    d.barrier("global")
+    mx.finalize()
+
+
+def gpu_hybrid_grid():
+    """Distribution(dp, mp) sub-communicators on the p2p transport
+    (driver config 4 shape at reduced scale)."""
+    mx, torch, rank, size = _init()
+    assert size % 2 == 0
+    d = mx.Distribution(size // 2, 2)
+    # model-group allreduce: partners are (rank//2)*2 and +1
+    count = 8192
+    a = _arange(torch, count, rank)
+    out = torch.empty_like(a)
+    mx.wait(d.all_reduce(a, out, count, op="sum", group="model"))
+    torch.cuda.synchronize()
+    base = (rank // 2) * 2
+    want = 2 * torch.arange(count, dtype=torch.float32, device="cuda") \
+        + base + base + 1
+    assert torch.allclose(out, want), "model-group allreduce"
+    # data-group allreduce
+    a = _arange(torch, count, rank)
+    mx.wait(d.all_reduce(a, a, count, op="sum", group="data"))
+    torch.cuda.synchronize()
+    dp = size // 2
+    mates = [r for r in range(size) if r % 2 == rank % 2]
+    want = dp * torch.arange(count, dtype=torch.float32, device="cuda") \
+        + float(sum(mates))
+    assert torch.allclose(a, want), "data-group allreduce"
+    d.barrier("global")
+    mx.finalize()
+
+
+WORKERS = {
+    "gpu_collectives": gpu_collectives,
+    "gpu_allreduce_multislot": gpu_allreduce_multislot,
+    "gpu_quant_allreduce": gpu_quant_allreduce,
+    "gpu_hybrid_grid": gpu_hybrid_grid,
+}
+
+
+def main():
+    name = sys.argv[1]
+    WORKERS[name]()
+    print(f"OK {name} rank={os.environ.get('RANK')}")
+
+
+if __name__ == "__main__":
+    main()
