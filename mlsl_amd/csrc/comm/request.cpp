@@ -248,10 +248,15 @@ void CommRequest::BuildChunks() {
                     const size_t nblocks = (cnt + blk - 1) / blk;
                     ce.sch = BuildAllReduceRingUnits(gr, gs, nblocks,
                                                      qparams_.WireBlockBytes(), blk);
+                } else if (algo == AllReduceAlgo::RHD && (gs & (gs - 1)) == 0) {
+                    ce.sch = BuildAllReduceRHD(gr, gs, cnt, dtype_, spec_.rop);
                 } else {
-                    ce.sch = (algo == AllReduceAlgo::RHD && (gs & (gs - 1)) == 0)
-                                 ? BuildAllReduceRHD(gr, gs, cnt, dtype_, spec_.rop)
-                                 : BuildAllReduceRing(gr, gs, cnt, dtype_, spec_.rop);
+                    // Channel c rides a stride-rotated ring: disjoint xGMI
+                    // links per channel (multi-endpoint fan-out analog).
+                    const int stride = RingStrideForChannel(
+                        c % std::max<size_t>(1, cfg.num_channels), gs);
+                    ce.sch = BuildAllReduceRing(gr, gs, cnt, dtype_, spec_.rop,
+                                                stride);
                 }
                 break;
             case CollOp::REDUCE:

@@ -42,20 +42,51 @@ int CeilLog2(int n) {
     return l;
 }
 
+int Gcd(int a, int b) {
+    while (b) {
+        int t = a % b;
+        a = b;
+        b = t;
+    }
+    return a;
+}
+
 }  // namespace
 
 // Ring allreduce: bandwidth-optimal reduce-scatter + all-gather over the
 // group ring; each phase moves one segment to the next neighbor. This is the
 // per-xGMI-link-bound algorithm the chunk-over-channels layer parallelizes
 // (reference analog: GET_EP_PAYLOAD endpoint fan-out, src/comm_ep.cpp:99-115).
+//
+// `stride` rotates the ring: next = rank+stride (mod N), for any stride
+// coprime with N. Different channels run different strides, so on a
+// fully-connected xGMI node each channel's ring traverses a DISJOINT set
+// of point-to-point links (stride s uses edges (i, i+s)) — the multi-link
+// analog of the reference's multi-endpoint fan-out. The schedule algebra
+// is stride-invariant: walking the ring in generator order g^0, g^1, ...
+// relabels which rank owns which segment but keeps the reduce-scatter /
+// all-gather structure intact.
 static Schedule RingAllReduceImpl(int rank, int size, size_t count, size_t es,
-                                  DataType dt, ReduceOp op) {
+                                  DataType dt, ReduceOp op, int stride) {
     if (size == 1) return SelfOnly(count * es, dt, op);
 
     Schedule sch;
     sch.dtype = dt;
     sch.rop = op;
-    const int N = size, r = rank;
+    const int N = size;
+    // Position of this rank along the stride-ring: r such that
+    // rank = (stride * r) mod N. Segments are owned by ring POSITION, so
+    // the rest of the algorithm is the textbook stride-1 ring on positions.
+    int r = 0;
+    if (stride % N != 1) {
+        for (int pos = 0, node = 0; pos < N; ++pos, node = (node + stride) % N)
+            if (node == rank) {
+                r = pos;
+                break;
+            }
+    } else {
+        r = rank;
+    }
     auto segOff = [&](int i) { return SegOffset(count, N, (i % N + N) % N) * es; };
     auto segBytes = [&](int i) { return SegCount(count, N, (i % N + N) % N) * es; };
     size_t max_seg = 0;
@@ -64,7 +95,7 @@ static Schedule RingAllReduceImpl(int rank, int size, size_t count, size_t es,
     // Phase 0: materialize the working copy in RECV (skipped in-place).
     sch.AddStep(MakeCopy(0, Ref(Space::SEND, 0, count * es), Ref(Space::RECV, 0, count * es)));
 
-    const int next = (r + 1) % N, prev = (r - 1 + N) % N;
+    const int next = (rank + stride) % N, prev = (rank - stride % N + N) % N;
     // Reduce-scatter phases.
     for (int p = 0; p < N - 1; ++p) {
         const int ssend = (r - p % N + N) % N;
@@ -97,14 +128,31 @@ static Schedule RingAllReduceImpl(int rank, int size, size_t count, size_t es,
     return sch;
 }
 
-Schedule BuildAllReduceRing(int rank, int size, size_t count, DataType dt, ReduceOp op) {
-    return RingAllReduceImpl(rank, size, count, DtypeSize(dt), dt, op);
+Schedule BuildAllReduceRing(int rank, int size, size_t count, DataType dt, ReduceOp op,
+                            int stride) {
+    MLSL_CHECK(stride >= 1 && Gcd(stride, size) == 1,
+               "ring stride must be coprime with the group size");
+    return RingAllReduceImpl(rank, size, count, DtypeSize(dt), dt, op, stride);
+}
+
+int RingStrideForChannel(size_t channel, int size) {
+    if (size <= 2) return 1;
+    // Enumerate strides coprime with N in the order 1, N-1, 3, N-3, 5, ...
+    // — pairs (s, N-s) are the two directions of the same physical links,
+    // so consecutive channels first exploit full duplex, then new links.
+    std::vector<int> strides;
+    for (int s = 1; 2 * s <= size; ++s) {
+        if (Gcd(s, size) != 1) continue;
+        strides.push_back(s);
+        if (s != size - s) strides.push_back(size - s);
+    }
+    return strides[channel % strides.size()];
 }
 
 Schedule BuildAllReduceRingUnits(int rank, int size, size_t units, size_t unit_bytes,
                                  size_t quant_block) {
     Schedule sch = RingAllReduceImpl(rank, size, units, unit_bytes, DataType::U8,
-                                     ReduceOp::SUM);
+                                     ReduceOp::SUM, 1);
     sch.quant_block = quant_block;
     return sch;
 }

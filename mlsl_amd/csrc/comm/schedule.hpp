@@ -76,7 +76,14 @@ size_t SegCount(size_t count, size_t parts, size_t i);
 // of `dt`. All builders support in-place (caller passes same pointer for
 // SEND and RECV at execution time; builders emit SEND-space reads only in
 // phase 0 positions that are safe, or route through TMP).
-Schedule BuildAllReduceRing(int rank, int size, size_t count, DataType dt, ReduceOp op);
+// `stride` rotates the ring (next = rank+stride mod size; must be coprime
+// with size): different channels run different strides so their rings
+// traverse disjoint xGMI point-to-point links on a fully-connected node.
+Schedule BuildAllReduceRing(int rank, int size, size_t count, DataType dt, ReduceOp op,
+                            int stride = 1);
+// Stride assignment for channel c: 1, size-1, 3, size-3, ... (full-duplex
+// pairs first, then fresh links), cycling.
+int RingStrideForChannel(size_t channel, int size);
 // Ring allreduce over opaque fixed-size units (used by the quantized path:
 // unit = one int8 wire block of quant_block elements + 8-byte header).
 Schedule BuildAllReduceRingUnits(int rank, int size, size_t units, size_t unit_bytes,

@@ -56,6 +56,28 @@ static void TestAllReduce(int N, size_t count, bool rhd) {
     }
 }
 
+static void TestAllReduceRingStride(int N, size_t count, int stride) {
+    std::vector<Schedule> sch(N);
+    std::vector<std::vector<uint8_t>> sbuf(N), rbuf(N);
+    for (int r = 0; r < N; ++r) {
+        sch[r] = BuildAllReduceRing(r, N, count, DataType::F32, ReduceOp::SUM,
+                                    stride);
+        FillF(sbuf[r], count, static_cast<float>(r));
+        rbuf[r].assign(count * 4, 0);
+    }
+    SimulateSchedules(sch, sbuf, rbuf);
+    for (int r = 0; r < N; ++r) {
+        auto f = F(rbuf[r]);
+        for (size_t i = 0; i < count; ++i) {
+            float want = static_cast<float>(N) * i + N * (N - 1) / 2.0f;
+            EXPECT(f[i] == want,
+                   "stride-%d ring allreduce N=%d rank=%d i=%zu got %f want %f",
+                   stride, N, r, i, f[i], want);
+            if (f[i] != want) return;
+        }
+    }
+}
+
 static void TestAllReduceMax(int N, size_t count) {
     std::vector<Schedule> sch(N);
     std::vector<std::vector<uint8_t>> sbuf(N), rbuf(N);
@@ -285,6 +307,16 @@ int main() {
             if ((N & (N - 1)) == 0) TestAllReduce(N, count, true);
         }
         TestAllReduceMax(N, 100);
+        // Rotated rings (multi-xGMI-link channel fan-out): every coprime
+        // stride must produce the identical allreduce.
+        for (int s = 1; s < N; ++s) {
+            bool cop = true;
+            for (int d = 2; d <= s; ++d)
+                if (s % d == 0 && N % d == 0) cop = false;
+            if (cop) TestAllReduceRingStride(N, 64, s);
+        }
+        for (size_t ch = 0; ch < 8; ++ch)
+            EXPECT(RingStrideForChannel(ch, N) >= 1, "stride channel %zu", ch);
         TestReduceScatter(N, 17);
         TestAllGather(N, 9);
         TestAllGatherv(N);
