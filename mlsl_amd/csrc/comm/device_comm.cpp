@@ -15,6 +15,7 @@
 #include <cstdlib>
 #include <cstring>
 #include <memory>
+#include <thread>
 #include <unistd.h>
 #include <unordered_map>
 #include <vector>
@@ -63,6 +64,8 @@ DeviceReqState::~DeviceReqState() {
     rel(tmp_dev);
     rel(stage_send);
     rel(stage_recv);
+    if (pin_send) (void)hipHostFree(pin_send);
+    if (pin_recv) (void)hipHostFree(pin_recv);
 }
 
 namespace {
@@ -320,14 +323,41 @@ void DeviceSetupRequest(CommRequest* req, DeviceReqState& st) {
 
 namespace {
 
-bool IsDevicePtr(const void* p) {
-    if (!p) return true;
+enum class BufClass { DEV, PINNED, PAGEABLE };
+
+BufClass ClassifyPtr(const void* p) {
+    if (!p) return BufClass::DEV;
     hipPointerAttribute_t attr;
     if (hipPointerGetAttributes(&attr, p) != hipSuccess) {
         (void)hipGetLastError();
-        return false;  // unregistered host memory
+        return BufClass::PAGEABLE;  // unregistered host memory
     }
-    return attr.type == hipMemoryTypeDevice || attr.type == hipMemoryTypeManaged;
+    if (attr.type == hipMemoryTypeDevice || attr.type == hipMemoryTypeManaged)
+        return BufClass::DEV;
+    if (attr.type == hipMemoryTypeHost || attr.type == hipMemoryTypeUnified)
+        return BufClass::PINNED;    // registered host: DMA-able directly
+    return BufClass::PAGEABLE;
+}
+
+// user<->pinned staging memcpy, multi-threaded above the configured
+// threshold (reference MLSL_USE_COPY_THREADS, src/comm_ep.cpp:357-361,
+// 1561-1574: OMP-threaded ReplaceIn copies; here plain std::thread ranges).
+void HostStageCopy(void* dst, const void* src, size_t n) {
+    const Config& cfg = GlobalConfig();
+    if (n < cfg.copy_threshold || cfg.copy_threads <= 1) {
+        std::memcpy(dst, src, n);
+        return;
+    }
+    const size_t t = cfg.copy_threads;
+    std::vector<std::thread> ths;
+    for (size_t i = 0; i < t; ++i) {
+        const size_t lo = i * n / t, hi = (i + 1) * n / t;
+        ths.emplace_back([=]() {
+            std::memcpy(static_cast<uint8_t*>(dst) + lo,
+                        static_cast<const uint8_t*>(src) + lo, hi - lo);
+        });
+    }
+    for (auto& th : ths) th.join();
 }
 
 // Bytes this rank contributes / receives for the whole request.
@@ -653,25 +683,49 @@ bool DeviceAdvanceRequest(CommRequest* req, DeviceReqState& st) {
         return AllEventsDone(st);
     }
 
-    // Host-buffer staging (ReplaceIn analog): stage unregistered host
-    // buffers through persistent HBM so RCCL/kernels see device memory.
+    // Host-buffer staging (ReplaceIn analog): stage host buffers through
+    // persistent HBM so RCCL/kernels see device memory. Pageable memory
+    // bounces through a persistent PINNED buffer in chunks — the H2D DMA of
+    // chunk k overlaps the host memcpy of chunk k+1; already-pinned user
+    // buffers DMA directly. (Classified fresh every Start:
+    // hipPointerGetAttributes is ~0.3 us and a pointer-value cache would
+    // misclassify a freed-and-reallocated pointer.)
     size_t send_b = 0, recv_b = 0;
     IoBytes(req, &send_b, &recv_b);
-    // Classified fresh every Start: hipPointerGetAttributes is ~0.3 us and a
-    // pointer-value cache would misclassify a freed-and-reallocated host
-    // pointer landing at an address that used to be device memory (or vice
-    // versa) — silent corruption for a sub-microsecond saving.
-    const bool s_host = send_b > 0 && !IsDevicePtr(req->UserSendBuf());
-    const bool r_host = recv_b > 0 && !IsDevicePtr(req->UserRecvBuf());
+    const BufClass s_cls = send_b > 0 ? ClassifyPtr(req->UserSendBuf())
+                                      : BufClass::DEV;
+    const BufClass r_cls = recv_b > 0 ? ClassifyPtr(req->UserRecvBuf())
+                                      : BufClass::DEV;
+    const bool s_host = s_cls != BufClass::DEV;
+    const bool r_host = r_cls != BufClass::DEV;
     st.recv_staged = r_host;
+    st.unstage_pending = false;
+    constexpr size_t kStageChunk = 32u << 20;
     if (s_host) {
         if (!st.stage_send || st.stage_send_bytes < send_b) {
             if (st.stage_send) rt->FreeDevice(st.stage_send);
             st.stage_send = rt->AllocDevice(send_b);
             st.stage_send_bytes = send_b;
         }
-        HIP_CHECKD(hipMemcpyAsync(st.stage_send, req->UserSendBuf(), send_b,
-                                  hipMemcpyHostToDevice, base_s));
+        if (s_cls == BufClass::PINNED) {
+            HIP_CHECKD(hipMemcpyAsync(st.stage_send, req->UserSendBuf(), send_b,
+                                      hipMemcpyHostToDevice, base_s));
+        } else {
+            if (!st.pin_send || st.pin_send_bytes < send_b) {
+                if (st.pin_send) (void)hipHostFree(st.pin_send);
+                HIP_CHECKD(hipHostMalloc(&st.pin_send, send_b));
+                st.pin_send_bytes = send_b;
+            }
+            for (size_t off = 0; off < send_b; off += kStageChunk) {
+                const size_t n = std::min(kStageChunk, send_b - off);
+                HostStageCopy(static_cast<uint8_t*>(st.pin_send) + off,
+                              req->UserSendBuf() + off, n);
+                HIP_CHECKD(hipMemcpyAsync(
+                    static_cast<uint8_t*>(st.stage_send) + off,
+                    static_cast<uint8_t*>(st.pin_send) + off, n,
+                    hipMemcpyHostToDevice, base_s));
+            }
+        }
         if (gc.streams.size() > 1) {
             if (!st.dep_event)
                 HIP_CHECKD(hipEventCreateWithFlags(&st.dep_event,
@@ -681,10 +735,22 @@ bool DeviceAdvanceRequest(CommRequest* req, DeviceReqState& st) {
                 HIP_CHECKD(hipStreamWaitEvent(gc.streams[i], st.dep_event, 0));
         }
     }
-    if (r_host && (!st.stage_recv || st.stage_recv_bytes < recv_b)) {
-        if (st.stage_recv) rt->FreeDevice(st.stage_recv);
-        st.stage_recv = rt->AllocDevice(recv_b);
-        st.stage_recv_bytes = recv_b;
+    if (r_host) {
+        if (!st.stage_recv || st.stage_recv_bytes < recv_b) {
+            if (st.stage_recv) rt->FreeDevice(st.stage_recv);
+            st.stage_recv = rt->AllocDevice(recv_b);
+            st.stage_recv_bytes = recv_b;
+        }
+        if (r_cls == BufClass::PAGEABLE &&
+            (!st.pin_recv || st.pin_recv_bytes < recv_b)) {
+            if (st.pin_recv) (void)hipHostFree(st.pin_recv);
+            HIP_CHECKD(hipHostMalloc(&st.pin_recv, recv_b));
+            st.pin_recv_bytes = recv_b;
+        }
+        // Pageable: D2H lands in the pinned bounce; the final pinned->user
+        // memcpy runs host-side once the completion events fire.
+        st.unstage_pending = r_cls == BufClass::PAGEABLE;
+        st.unstage_bytes = recv_b;
     }
     req->SetDeviceBuffers(s_host ? static_cast<const uint8_t*>(st.stage_send) : nullptr,
                           r_host ? static_cast<uint8_t*>(st.stage_recv) : nullptr);

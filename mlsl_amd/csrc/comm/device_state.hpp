@@ -21,11 +21,22 @@ struct DeviceReqState {
     size_t tmp_bytes = 0;
     // Host-buffer staging (reference ReplaceIn/ReplaceOut,
     // src/comm_ep.cpp:363-566): user buffers that are not device memory are
-    // staged through persistent HBM buffers around the collective.
+    // staged through persistent HBM buffers around the collective. Pageable
+    // user memory additionally bounces through persistent PINNED host
+    // buffers (hipHostMalloc) in chunks, so the H2D/D2H legs are true
+    // async DMA overlapping the host-side memcpys — the reference's
+    // registered-shm-heap staging, MI355X edition.
     void* stage_send = nullptr;
     void* stage_recv = nullptr;
     size_t stage_send_bytes = 0, stage_recv_bytes = 0;
     bool recv_staged = false;
+    void* pin_send = nullptr;
+    void* pin_recv = nullptr;
+    size_t pin_send_bytes = 0, pin_recv_bytes = 0;
+    // D2H landed in pin_recv; the final pinned->user memcpy runs on the
+    // host after the completion events fire.
+    bool unstage_pending = false;
+    size_t unstage_bytes = 0;
     // hipGraph replay (MLSL_USE_GRAPHS): the issue sequence captured once,
     // replayed on subsequent Starts with the same buffers.
     hipGraphExec_t graph_exec = nullptr;

@@ -152,26 +152,6 @@ void* P2pGroup::PeerAckFlag(int peer, size_t lane) const {
 
 // --- transport ops ---------------------------------------------------------
 
-void P2pGroup::SendBytes(int peer, size_t lane, hipStream_t s,
-                         const uint8_t* src, size_t bytes, size_t grain) {
-    const size_t msg_max = (slot_bytes_ / grain) * grain;
-    MLSL_CHECK(msg_max > 0, "p2p slot smaller than one element/block");
-    uint64_t& sent = sent_[static_cast<size_t>(peer) * nlanes_ + lane];
-    size_t done = 0;
-    while (done < bytes) {
-        const size_t n = std::min(msg_max, bytes - done);
-        const uint64_t seq = ++sent;
-        const size_t slot = (seq - 1) % nslots_;
-        if (seq > nslots_)
-            LaunchWaitFlag(MyAckFlag(peer, lane), seq - nslots_, abort_host_,
-                           status_host_, max_ticks_, s);
-        HIP_CHECKP(hipMemcpyAsync(PeerSlot(peer, lane, slot), src + done, n,
-                                  hipMemcpyDeviceToDevice, s));
-        LaunchSetFlag(PeerInFlag(peer, lane), seq, s);
-        done += n;
-    }
-}
-
 void P2pGroup::IssueSchedule(CommRequest* req, ChunkExec& ce, size_t lane,
                              hipStream_t s, const uint8_t* sbase,
                              uint8_t* rbase, uint8_t* tmp) {
@@ -196,6 +176,23 @@ void P2pGroup::IssueSchedule(CommRequest* req, ChunkExec& ce, size_t lane,
 
     const size_t msg_max = (slot_bytes_ / unit) * unit;
     MLSL_CHECK(msg_max > 0, "p2p slot smaller than one element/block");
+    // Adaptive sub-message size: ~8 sub-messages per transfer (pipelines
+    // DMA against the consumer's reduce) but never below 2 MiB (each
+    // sub-message costs 4 kernel launches) and never above the slot.
+    // Sender and receiver derive the identical size from the transfer
+    // length, so the slot partition always agrees.
+    auto sub_size = [&](size_t bytes) -> size_t {
+        size_t v = std::max<size_t>((bytes + 7) / 8,
+                                    std::min<size_t>(2u << 20, msg_max));
+        v = std::min(v, msg_max);
+        v = (v / unit) * unit;
+        if (v < unit) v = unit;
+        return std::min(v, msg_max);
+    };
+    XferPoll poll{};
+    poll.abort_word = abort_host_;
+    poll.status = status_host_;
+    poll.max_ticks = max_ticks_;
 
     for (int phase = 0; phase < ce.sch.num_phases; ++phase) {
         // Collect this phase's send and recv jobs, then interleave their
@@ -217,9 +214,9 @@ void P2pGroup::IssueSchedule(CommRequest* req, ChunkExec& ce, size_t lane,
             if (st.phase != phase) continue;
             if (st.send_peer >= 0 && st.send.bytes > 0) {
                 MLSL_CHECK(st.send_peer != my_idx_, "p2p self-send in schedule");
+                const size_t sub = sub_size(st.send.bytes);
                 sends.push_back(&st);
-                max_msgs = std::max(max_msgs,
-                                    (st.send.bytes + msg_max - 1) / msg_max);
+                max_msgs = std::max(max_msgs, (st.send.bytes + sub - 1) / sub);
             }
             if (st.recv_peer >= 0 && st.recv.bytes > 0) {
                 MLSL_CHECK(st.recv_peer != my_idx_, "p2p self-recv in schedule");
@@ -232,60 +229,78 @@ void P2pGroup::IssueSchedule(CommRequest* req, ChunkExec& ce, size_t lane,
                                       st.local == Step::LocalOp::REDUCE &&
                                       same_ref(st.local_dst, st.recv) &&
                                       st.local_src.bytes == st.recv.bytes;
+                const size_t sub = sub_size(st.recv.bytes);
                 recvs.push_back({&st, fuse_into, fuse_out});
-                max_msgs = std::max(max_msgs,
-                                    (st.recv.bytes + msg_max - 1) / msg_max);
+                max_msgs = std::max(max_msgs, (st.recv.bytes + sub - 1) / sub);
             }
         }
         for (size_t k = 0; k < max_msgs; ++k) {
-            const size_t off = k * msg_max;
             for (const Step* st : sends) {
+                const size_t sub = sub_size(st->send.bytes);
+                const size_t off = k * sub;
                 if (off >= st->send.bytes) continue;
-                const size_t n = std::min(msg_max, st->send.bytes - off);
+                const size_t n = std::min(sub, st->send.bytes - off);
                 const int peer = st->send_peer;
                 uint64_t& sent = sent_[static_cast<size_t>(peer) * nlanes_ + lane];
                 const uint64_t seq = ++sent;
                 const size_t slot = (seq - 1) % nslots_;
-                if (seq > nslots_)
-                    LaunchWaitFlag(MyAckFlag(peer, lane), seq - nslots_,
-                                   abort_host_, status_host_, max_ticks_, s);
-                HIP_CHECKP(hipMemcpyAsync(PeerSlot(peer, lane, slot),
-                                          ptr(st->send) + off, n,
-                                          hipMemcpyDeviceToDevice, s));
+                // Fused: backpressure poll (slot reuse) + NT copy into the
+                // peer's slot in ONE kernel, then the 1-wg publish.
+                XferPoll bp = poll;
+                bp.mbox = seq > nslots_ ? MyAckFlag(peer, lane) : nullptr;
+                bp.target = seq - nslots_;
+                LaunchXferCopy(PeerSlot(peer, lane, slot), ptr(st->send) + off,
+                               n, bp.mbox ? &bp : nullptr, s);
                 LaunchSetFlag(PeerInFlag(peer, lane), seq, s);
             }
             for (const RecvJob& rj : recvs) {
                 const Step& st = *rj.st;
+                const size_t sub = sub_size(st.recv.bytes);
+                const size_t off = k * sub;
                 if (off >= st.recv.bytes) continue;
-                const size_t n = std::min(msg_max, st.recv.bytes - off);
+                const size_t n = std::min(sub, st.recv.bytes - off);
                 const int peer = st.recv_peer;
                 uint64_t& rcvd = rcvd_[static_cast<size_t>(peer) * nlanes_ + lane];
                 const uint64_t seq = ++rcvd;
                 const size_t slot = (seq - 1) % nslots_;
-                LaunchWaitFlag(MyInFlag(peer, lane), seq, abort_host_,
-                               status_host_, max_ticks_, s);
                 const uint8_t* sl = MySlot(peer, lane, slot);
-                if (rj.fuse_into) {
-                    if (quant)
-                        LaunchQuantAccum(ptr(st.local_dst) + off, sl,
-                                         (n / unit) * blk, blk, s);
-                    else
-                        LaunchReduce(ptr(st.local_dst) + off, sl, n / es,
-                                     req->Dtype(), ce.sch.rop, s);
-                } else if (rj.fuse_out) {
-                    if (quant) {
-                        // acc = slot, then acc += local_src slice
-                        LaunchCopy(ptr(st.recv) + off, sl, n, s);
-                        LaunchQuantAccum(ptr(st.recv) + off,
-                                         ptr(st.local_src) + off,
-                                         (n / unit) * blk, blk, s);
-                    } else {
-                        LaunchReduceOut(ptr(st.recv) + off, sl,
-                                        ptr(st.local_src) + off, n / es,
-                                        req->Dtype(), ce.sch.rop, s);
-                    }
+                XferPoll wp = poll;
+                wp.mbox = MyInFlag(peer, lane);
+                wp.target = seq;
+                // Fused wait+consume where the dtype allows; otherwise a
+                // standalone wait kernel precedes the existing consumers.
+                if (rj.fuse_into && !quant &&
+                    LaunchXferReduce(ptr(st.local_dst) + off, sl, nullptr,
+                                     n / es, req->Dtype(), ce.sch.rop, &wp, s)) {
+                } else if (rj.fuse_out && !quant &&
+                           LaunchXferReduce(ptr(st.recv) + off, sl,
+                                            ptr(st.local_src) + off, n / es,
+                                            req->Dtype(), ce.sch.rop, &wp, s)) {
+                } else if (!rj.fuse_into && !rj.fuse_out) {
+                    LaunchXferCopy(ptr(st.recv) + off, sl, n, &wp, s);
                 } else {
-                    LaunchCopy(ptr(st.recv) + off, sl, n, s);
+                    LaunchWaitFlag(wp.mbox, wp.target, abort_host_,
+                                   status_host_, max_ticks_, s);
+                    if (rj.fuse_into) {
+                        if (quant)
+                            LaunchQuantAccum(ptr(st.local_dst) + off, sl,
+                                             (n / unit) * blk, blk, s);
+                        else
+                            LaunchReduce(ptr(st.local_dst) + off, sl, n / es,
+                                         req->Dtype(), ce.sch.rop, s);
+                    } else {  // fuse_out
+                        if (quant) {
+                            // acc = slot, then acc += local_src slice
+                            LaunchCopy(ptr(st.recv) + off, sl, n, s);
+                            LaunchQuantAccum(ptr(st.recv) + off,
+                                             ptr(st.local_src) + off,
+                                             (n / unit) * blk, blk, s);
+                        } else {
+                            LaunchReduceOut(ptr(st.recv) + off, sl,
+                                            ptr(st.local_src) + off, n / es,
+                                            req->Dtype(), ce.sch.rop, s);
+                        }
+                    }
                 }
                 LaunchSetFlag(PeerAckFlag(peer, lane), seq, s);
             }

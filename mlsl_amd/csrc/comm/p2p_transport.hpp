@@ -90,11 +90,6 @@ class P2pGroup {
     void* PeerInFlag(int peer, size_t lane) const;   // src = me, in peer window
     void* PeerAckFlag(int peer, size_t lane) const;  // me acking peer's data
 
-    void SendBytes(int peer, size_t lane, hipStream_t s, const uint8_t* src,
-                   size_t bytes, size_t grain);
-    // Consume modes for RecvBytes (decided per step by the caller).
-    enum class Consume { COPY, REDUCE_INTO, REDUCE_OUT, QUANT_ACCUM };
-
     int gsize_ = 0, my_idx_ = 0;
     size_t nlanes_ = 1, nslots_ = 4, slot_bytes_ = 0;
     size_t flags_bytes_ = 0, win_bytes_ = 0;

@@ -50,10 +50,14 @@ Config Config::FromEnv() {
     c.check_pointers = EnvBool("MLSL_CHECK_POINTERS", false);
     if (const char* e = std::getenv("MLSL_TRANSPORT")) c.transport = e;
     if (const char* e = std::getenv("MLSL_DEVICE_TRANSPORT")) c.device_transport = e;
-    c.p2p_slot_mb = EnvSize("MLSL_P2P_SLOT_MB", 4);
+    c.p2p_slot_mb = EnvSize("MLSL_P2P_SLOT_MB", 16);
     if (c.p2p_slot_mb < 1) c.p2p_slot_mb = 1;
     c.p2p_slots = EnvSize("MLSL_P2P_SLOTS", 4);
     if (c.p2p_slots < 2) c.p2p_slots = 2;
+    c.copy_threads = EnvSize("MLSL_COPY_THREADS", 4);
+    if (c.copy_threads < 1) c.copy_threads = 1;
+    if (c.copy_threads > 16) c.copy_threads = 16;
+    c.copy_threshold = EnvSize("MLSL_COPY_THRESHOLD", 4u << 20);
     c.timeout_sec = static_cast<int>(EnvSize("MLSL_TIMEOUT", 300));
     c.use_graphs = EnvBool("MLSL_USE_GRAPHS", false);
     return c;

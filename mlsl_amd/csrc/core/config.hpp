@@ -45,8 +45,13 @@ struct Config {
     // window transport (comm/p2p_transport.hpp). auto = RCCL, switching to
     // p2p when several ranks share one device (RCCL refuses that layout).
     std::string device_transport = "auto";  // MLSL_DEVICE_TRANSPORT=auto|rccl|p2p
-    size_t p2p_slot_mb = 4;        // MLSL_P2P_SLOT_MB: staging slot size
+    size_t p2p_slot_mb = 16;       // MLSL_P2P_SLOT_MB: staging slot size
     size_t p2p_slots = 4;          // MLSL_P2P_SLOTS: in-flight slots per edge-lane
+    // Pageable-host staging copies (reference MLSL_USE_COPY_THREADS /
+    // MLSL_COPY_THREADS / MLSL_COPY_THRESHOLD, src/comm_ep.cpp:357-361):
+    // user<->pinned memcpys above the threshold split across threads.
+    size_t copy_threads = 4;       // MLSL_COPY_THREADS
+    size_t copy_threshold = 4u << 20;  // MLSL_COPY_THRESHOLD (bytes)
     int timeout_sec = 300;         // MLSL_TIMEOUT: bootstrap/collective timeout
     bool use_graphs = false;       // MLSL_USE_GRAPHS: hipGraph replay of
                                    // persistent device requests (single

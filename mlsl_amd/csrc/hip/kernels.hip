@@ -1240,6 +1240,218 @@ __global__ void SetFlagKernel(unsigned long long* __restrict__ mbox,
 
 }  // namespace
 
+// --- fused transport kernels: flag-wait prologue + payload work ---
+// Every workgroup's thread 0 polls the (local-HBM) mailbox, then the
+// workgroup proceeds — one kernel instead of wait-kernel + work-kernel.
+// The publish (in_flag/ack) stays a separate 1-wg kernel so stream order
+// provides the grid-completion barrier without cooperative launch.
+
+namespace {
+
+__device__ __forceinline__ bool PollGeq(const unsigned long long* mbox,
+                                        unsigned long long target,
+                                        const unsigned int* abort_word,
+                                        unsigned int* status,
+                                        unsigned long long max_ticks) {
+    __shared__ int ok;
+    if (threadIdx.x == 0) {
+        ok = 1;
+        const unsigned long long t0 = wall_clock64();
+        while (__hip_atomic_load(mbox, __ATOMIC_ACQUIRE,
+                                 __HIP_MEMORY_SCOPE_SYSTEM) < target) {
+            if (__hip_atomic_load(abort_word, __ATOMIC_RELAXED,
+                                  __HIP_MEMORY_SCOPE_SYSTEM) != 0 ||
+                wall_clock64() - t0 > max_ticks) {
+                __hip_atomic_store(status, 1u, __ATOMIC_RELEASE,
+                                   __HIP_MEMORY_SCOPE_SYSTEM);
+                ok = 0;
+                break;
+            }
+            __builtin_amdgcn_s_sleep(64);
+        }
+    }
+    __syncthreads();
+    return ok != 0;
+}
+
+struct PollArgs {
+    const unsigned long long* mbox;  // null = no wait
+    unsigned long long target;
+    const unsigned int* abort_word;
+    unsigned int* status;
+    unsigned long long max_ticks;
+};
+
+// byte copy, 16-B vectors + tail; optional poll prologue
+__global__ void XferCopyKernel(uint8_t* __restrict__ dst,
+                               const uint8_t* __restrict__ src, size_t bytes,
+                               PollArgs pa) {
+    if (pa.mbox &&
+        !PollGeq(pa.mbox, pa.target, pa.abort_word, pa.status, pa.max_ticks))
+        return;
+    const size_t tid = blockIdx.x * blockDim.x + threadIdx.x;
+    const size_t stride = gridDim.x * blockDim.x;
+    const bool al = ((reinterpret_cast<uintptr_t>(dst) |
+                      reinterpret_cast<uintptr_t>(src)) & 15) == 0;
+    if (al) {
+        const size_t n16 = bytes / 16;
+        const uint4_ev* s = reinterpret_cast<const uint4_ev*>(src);
+        uint4_ev* d = reinterpret_cast<uint4_ev*>(dst);
+        for (size_t i = tid; i < n16; i += stride)
+            __builtin_nontemporal_store(__builtin_nontemporal_load(s + i), d + i);
+        for (size_t j = n16 * 16 + tid; j < bytes; j += stride) dst[j] = src[j];
+    } else {
+        for (size_t j = tid; j < bytes; j += stride) dst[j] = src[j];
+    }
+}
+
+// dst (op)= slot  |  dst = slot (op) other   — f32, vec4 when aligned
+template <ReduceOp OP, bool OUT>
+__global__ void XferReduceF32Kernel(float* __restrict__ dst,
+                                    const float* __restrict__ slot,
+                                    const float* __restrict__ other, size_t n,
+                                    PollArgs pa) {
+    if (pa.mbox &&
+        !PollGeq(pa.mbox, pa.target, pa.abort_word, pa.status, pa.max_ticks))
+        return;
+    const size_t tid = blockIdx.x * blockDim.x + threadIdx.x;
+    const size_t stride = gridDim.x * blockDim.x;
+    const bool al = ((reinterpret_cast<uintptr_t>(dst) |
+                      reinterpret_cast<uintptr_t>(slot) |
+                      reinterpret_cast<uintptr_t>(other)) & 15) == 0;
+    if (al) {
+        const size_t n4 = n / 4;
+        const float4_ev* s4 = reinterpret_cast<const float4_ev*>(slot);
+        const float4_ev* o4 = reinterpret_cast<const float4_ev*>(other);
+        float4_ev* d4 = reinterpret_cast<float4_ev*>(dst);
+        for (size_t i = tid; i < n4; i += stride) {
+            float4_ev a = OUT ? __builtin_nontemporal_load(o4 + i) : d4[i];
+            float4_ev b = __builtin_nontemporal_load(s4 + i);
+#pragma unroll
+            for (int j = 0; j < 4; ++j) a[j] = Apply<float, OP>(a[j], b[j]);
+            if (OUT) __builtin_nontemporal_store(a, d4 + i);
+            else d4[i] = a;
+        }
+        for (size_t j = n4 * 4 + tid; j < n; j += stride)
+            dst[j] = Apply<float, OP>(OUT ? other[j] : dst[j], slot[j]);
+    } else {
+        for (size_t j = tid; j < n; j += stride)
+            dst[j] = Apply<float, OP>(OUT ? other[j] : dst[j], slot[j]);
+    }
+}
+
+template <ReduceOp OP, bool OUT>
+__global__ void XferReduceBf16Kernel(unsigned short* __restrict__ dst,
+                                     const unsigned short* __restrict__ slot,
+                                     const unsigned short* __restrict__ other,
+                                     size_t n, PollArgs pa) {
+    if (pa.mbox &&
+        !PollGeq(pa.mbox, pa.target, pa.abort_word, pa.status, pa.max_ticks))
+        return;
+    const size_t tid = blockIdx.x * blockDim.x + threadIdx.x;
+    const size_t stride = gridDim.x * blockDim.x;
+    for (size_t j = tid; j < n; j += stride) {
+        const float a = __bfloat162float(
+            *reinterpret_cast<const __hip_bfloat16*>(OUT ? &other[j] : &dst[j]));
+        const float b = __bfloat162float(
+            *reinterpret_cast<const __hip_bfloat16*>(&slot[j]));
+        const __hip_bfloat16 r = __float2bfloat16(Apply<float, OP>(a, b));
+        dst[j] = *reinterpret_cast<const unsigned short*>(&r);
+    }
+}
+
+}  // namespace
+
+void LaunchXferCopy(void* dst, const void* src, size_t bytes,
+                    const XferPoll* poll, hipStream_t stream) {
+    PollArgs pa{};
+    if (poll) {
+        pa.mbox = static_cast<const unsigned long long*>(poll->mbox);
+        pa.target = poll->target;
+        pa.abort_word = static_cast<const unsigned int*>(poll->abort_word);
+        pa.status = static_cast<unsigned int*>(poll->status);
+        pa.max_ticks = poll->max_ticks;
+    }
+    hipLaunchKernelGGL(XferCopyKernel, dim3(GridFor(bytes / 16 + 1)),
+                       dim3(kBlock), 0, stream, static_cast<uint8_t*>(dst),
+                       static_cast<const uint8_t*>(src), bytes, pa);
+    HIP_CHECK(hipGetLastError());
+}
+
+bool LaunchXferReduce(void* dst, const void* slot, const void* other, size_t n,
+                      DataType dt, ReduceOp op, const XferPoll* poll,
+                      hipStream_t stream) {
+    PollArgs pa{};
+    if (poll) {
+        pa.mbox = static_cast<const unsigned long long*>(poll->mbox);
+        pa.target = poll->target;
+        pa.abort_word = static_cast<const unsigned int*>(poll->abort_word);
+        pa.status = static_cast<unsigned int*>(poll->status);
+        pa.max_ticks = poll->max_ticks;
+    }
+    const bool out = other != nullptr;
+    const int grid = GridFor(n / 4 + 1);
+#define XFER_DISPATCH(KERN, PT)                                               \
+    do {                                                                      \
+        if (out) {                                                            \
+            switch (op) {                                                     \
+                case ReduceOp::SUM:                                           \
+                    hipLaunchKernelGGL((KERN<ReduceOp::SUM, true>),           \
+                                       dim3(grid), dim3(kBlock), 0, stream,   \
+                                       (PT*)dst, (const PT*)slot,             \
+                                       (const PT*)other, n, pa);              \
+                    break;                                                    \
+                case ReduceOp::MIN:                                           \
+                    hipLaunchKernelGGL((KERN<ReduceOp::MIN, true>),           \
+                                       dim3(grid), dim3(kBlock), 0, stream,   \
+                                       (PT*)dst, (const PT*)slot,             \
+                                       (const PT*)other, n, pa);              \
+                    break;                                                    \
+                case ReduceOp::MAX:                                           \
+                    hipLaunchKernelGGL((KERN<ReduceOp::MAX, true>),           \
+                                       dim3(grid), dim3(kBlock), 0, stream,   \
+                                       (PT*)dst, (const PT*)slot,             \
+                                       (const PT*)other, n, pa);              \
+                    break;                                                    \
+            }                                                                 \
+        } else {                                                              \
+            switch (op) {                                                     \
+                case ReduceOp::SUM:                                           \
+                    hipLaunchKernelGGL((KERN<ReduceOp::SUM, false>),          \
+                                       dim3(grid), dim3(kBlock), 0, stream,   \
+                                       (PT*)dst, (const PT*)slot,             \
+                                       (const PT*)other, n, pa);              \
+                    break;                                                    \
+                case ReduceOp::MIN:                                           \
+                    hipLaunchKernelGGL((KERN<ReduceOp::MIN, false>),          \
+                                       dim3(grid), dim3(kBlock), 0, stream,   \
+                                       (PT*)dst, (const PT*)slot,             \
+                                       (const PT*)other, n, pa);              \
+                    break;                                                    \
+                case ReduceOp::MAX:                                           \
+                    hipLaunchKernelGGL((KERN<ReduceOp::MAX, false>),          \
+                                       dim3(grid), dim3(kBlock), 0, stream,   \
+                                       (PT*)dst, (const PT*)slot,             \
+                                       (const PT*)other, n, pa);              \
+                    break;                                                    \
+            }                                                                 \
+        }                                                                     \
+    } while (0)
+    switch (dt) {
+        case DataType::F32:
+            XFER_DISPATCH(XferReduceF32Kernel, float);
+            break;
+        case DataType::BF16:
+            XFER_DISPATCH(XferReduceBf16Kernel, unsigned short);
+            break;
+        default:
+            return false;  // caller falls back to wait + LaunchReduce
+    }
+#undef XFER_DISPATCH
+    HIP_CHECK(hipGetLastError());
+    return true;
+}
+
 void LaunchWaitFlag(const void* mbox, uint64_t target, const void* abort_word,
                     void* status, uint64_t max_ticks, hipStream_t stream) {
     hipLaunchKernelGGL(WaitFlagKernel, dim3(1), dim3(1), 0, stream,

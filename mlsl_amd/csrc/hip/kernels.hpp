@@ -67,6 +67,28 @@ void LaunchPack(const void* src, void* dst, const PackBlockDesc& d, DataType dt,
 void LaunchUnpack(const void* src, void* dst, const PackBlockDesc& d, DataType dt,
                   hipStream_t stream);
 
+// --- fused transport kernels (IPC p2p path) ---
+// Optional flag-wait prologue: every workgroup polls *mbox >= target
+// before touching the payload (abort word + wall-clock escape as in
+// LaunchWaitFlag). Fusing the wait into the payload kernel removes two
+// kernel launches per transport sub-message.
+struct XferPoll {
+    const void* mbox;
+    uint64_t target;
+    const void* abort_word;
+    void* status;
+    uint64_t max_ticks;
+};
+// Streaming byte copy with optional poll prologue (NT, 16-B lanes).
+void LaunchXferCopy(void* dst, const void* src, size_t bytes,
+                    const XferPoll* poll, hipStream_t stream);
+// dst op= slot (other==null) or dst = slot op other, with optional poll.
+// Returns false when dtype isn't covered (caller uses LaunchWaitFlag +
+// LaunchReduce instead).
+bool LaunchXferReduce(void* dst, const void* slot, const void* other, size_t n,
+                      DataType dt, ReduceOp op, const XferPoll* poll,
+                      hipStream_t stream);
+
 // --- IPC p2p transport flag primitives ---
 // Stream-blocking wait until *mbox >= target (system-scope acquire), with a
 // host abort word and a wall-clock bound (ticks of the 100 MHz constant

@@ -21,4 +21,6 @@ void LaunchPack(const void*, void*, const PackBlockDesc&, DataType, hipStream_t)
 void LaunchUnpack(const void*, void*, const PackBlockDesc&, DataType, hipStream_t) STUB()
 void LaunchWaitFlag(const void*, uint64_t, const void*, void*, uint64_t, hipStream_t) STUB()
 void LaunchSetFlag(void*, uint64_t, hipStream_t) STUB()
+void LaunchXferCopy(void*, const void*, size_t, const XferPoll*, hipStream_t) STUB()
+bool LaunchXferReduce(void*, const void*, const void*, size_t, DataType, ReduceOp, const XferPoll*, hipStream_t) STUB()
 }  // namespace mlsl
