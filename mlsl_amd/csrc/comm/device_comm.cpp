@@ -629,6 +629,22 @@ void EnsureEvents(DeviceReqState& st, size_t n) {
 
 }  // namespace
 
+namespace {
+
+// Completion check + the deferred pinned->user unstage memcpy (pageable
+// recv buffers: the D2H leg landed in the pinned bounce on-stream; the
+// final host copy runs once, here, after the events fire).
+bool DeviceFinishIfDone(CommRequest* req, DeviceReqState& st) {
+    if (!AllEventsDone(st)) return false;
+    if (st.unstage_pending) {
+        HostStageCopy(req->UserRecvBuf(), st.pin_recv, st.unstage_bytes);
+        st.unstage_pending = false;
+    }
+    return true;
+}
+
+}  // namespace
+
 bool DeviceAdvanceRequest(CommRequest* req, DeviceReqState& st) {
     auto* rt = static_cast<HipRuntime*>(Context::Get().Device());
     GroupComms& gc = rt->For(req->Group());
@@ -637,7 +653,7 @@ bool DeviceAdvanceRequest(CommRequest* req, DeviceReqState& st) {
         // wedging the GPU; surface that as a loud request failure.
         if (gc.p2p && !gc.p2p->Healthy())
             MLSL_THROW("p2p transport wait aborted (peer dead or timeout)");
-        return AllEventsDone(st);
+        return DeviceFinishIfDone(req, st);
     }
     const bool p2p = gc.p2p != nullptr;
     if (gc.comms.empty() && gc.streams.empty() && !p2p) {
@@ -812,8 +828,10 @@ bool DeviceAdvanceRequest(CommRequest* req, DeviceReqState& st) {
                 }
             }
             if (st.recv_staged)
-                HIP_CHECKD(hipMemcpyAsync(req->UserRecvBuf(), st.stage_recv, recv_b,
-                                          hipMemcpyDeviceToHost, s0));
+                HIP_CHECKD(hipMemcpyAsync(
+                    st.unstage_pending ? st.pin_recv
+                                       : static_cast<void*>(req->UserRecvBuf()),
+                    st.stage_recv, recv_b, hipMemcpyDeviceToHost, s0));
             return;
         }
         size_t tmp_off = 0;
@@ -928,14 +946,16 @@ bool DeviceAdvanceRequest(CommRequest* req, DeviceReqState& st) {
             // join all channels on stream 0, then stage the result out
             for (size_t ch = 0; ch < used; ++ch)
                 HIP_CHECKD(hipStreamWaitEvent(gc.streams[0], st.events[ch], 0));
-            HIP_CHECKD(hipMemcpyAsync(req->UserRecvBuf(), st.stage_recv, recv_b,
-                                      hipMemcpyDeviceToHost, gc.streams[0]));
+            HIP_CHECKD(hipMemcpyAsync(
+                st.unstage_pending ? st.pin_recv
+                                   : static_cast<void*>(req->UserRecvBuf()),
+                st.stage_recv, recv_b, hipMemcpyDeviceToHost, gc.streams[0]));
             EnsureEvents(st, used + 1);
             HIP_CHECKD(hipEventRecord(st.events[used], gc.streams[0]));
         }
     }
     st.issued = true;
-    return AllEventsDone(st);
+    return DeviceFinishIfDone(req, st);
 }
 
 }  // namespace mlsl
