@@ -73,6 +73,17 @@ void Context::Init(int rank, int size) {
 
 void Context::Finalize() {
     if (!initialized_) return;
+    // MPI_Finalize semantics: no rank closes its mesh sockets until every
+    // rank has finished its last collective (ranks reach Finalize at
+    // different times; without this, an early-closing rank makes the
+    // peer's progress loop see EOF mid-teardown).
+    if (boot_ && mesh_) {
+        try {
+            boot_->Barrier();
+        } catch (const std::exception& e) {
+            MLSL_LOG(ERROR, "finalize barrier failed (peer died?): %s", e.what());
+        }
+    }
     engine_.reset();      // join progress thread first
     groups_.clear();
     world_ = self_ = nullptr;

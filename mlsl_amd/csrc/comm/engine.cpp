@@ -19,6 +19,12 @@ Engine::Engine(Mesh* mesh, ProgressMode mode, bool device_mode)
 
 Engine::~Engine() {
     stop_.store(true, std::memory_order_release);
+    {
+        // Pairs with the Loop's untimed idle wait: the store above is
+        // re-checked under idle_mu_, so this notify can't be lost.
+        std::lock_guard<std::mutex> lk(idle_mu_);
+        idle_cv_.notify_one();
+    }
     if (thread_.joinable()) thread_.join();
 }
 
@@ -59,6 +65,11 @@ void Engine::Submit(CommRequest* req) {
         inbox_overflow_.push_back(req);
         overflow_pending_.store(true, std::memory_order_release);
     }
+    // Dekker pairing with the Loop's park decision: the fence orders the
+    // push before the deep_idle_ read in the seq_cst fence order, so either
+    // the parking thread's predicate sees the push, or we see deep_idle_
+    // and notify. (The loop has the matching fence before its predicate.)
+    std::atomic_thread_fence(std::memory_order_seq_cst);
     if (deep_idle_.load(std::memory_order_acquire)) {
         // Pairs with the Loop's check-then-wait under idle_mu_: taking the
         // lock here means the loop either saw the push or is parked and
@@ -157,17 +168,24 @@ void Engine::Loop() {
         }
         if (!did_work) {
             // Hot for the first ~4096 polls (sub-ms window covering the
-            // gaps of a busy training loop), then park on the condvar;
-            // Submit nudges it, and the bounded wait re-checks the ring so
-            // a missed nudge costs at most one timeout.
+            // gaps of a busy training loop), then park on the condvar.
+            // Untimed wait: the protocol is lost-wakeup-free (Submit and
+            // the destructor notify under idle_mu_ AFTER their stores, and
+            // the predicate re-checks under the same lock). A timed
+            // wait_for would also work but trips a known TSan false
+            // positive (libstdc++'s pthread_cond_clockwait is not
+            // intercepted by gcc-11 libtsan — spurious "double lock").
             if (++idle_spins > 4096) {
                 std::unique_lock<std::mutex> lk(idle_mu_);
                 deep_idle_.store(true, std::memory_order_release);
-                if (ring_.Empty() &&
-                    !overflow_pending_.load(std::memory_order_acquire) &&
-                    !stop_.load(std::memory_order_acquire))
-                    idle_cv_.wait_for(lk, std::chrono::milliseconds(1));
+                std::atomic_thread_fence(std::memory_order_seq_cst);
+                idle_cv_.wait(lk, [&]() {
+                    return !ring_.Empty() ||
+                           overflow_pending_.load(std::memory_order_acquire) ||
+                           stop_.load(std::memory_order_acquire);
+                });
                 deep_idle_.store(false, std::memory_order_release);
+                idle_spins = 0;
             }
         } else {
             idle_spins = 0;

@@ -102,11 +102,20 @@ void Channel::ProgressRecv() {
                 }
                 if (n == 0) {
                     // Peer finalized. Benign unless we still expect data
-                    // (ranks may tear down at different times after their
-                    // last collective — reference tolerates this too via
-                    // MPI_Finalize semantics).
-                    MLSL_CHECK(recv_hdr_got_ == 0 && posted_.empty(),
-                               "mesh peer closed with receives outstanding");
+                    // (Context::Finalize barriers before closing, so this
+                    // firing means a request was abandoned or a peer died).
+                    if (recv_hdr_got_ != 0 || !posted_.empty()) {
+                        std::string tags;
+                        for (const auto& kv : posted_)
+                            tags += std::to_string(kv.first) + "(" +
+                                    std::to_string(kv.second.got) + "/" +
+                                    std::to_string(kv.second.len) + ") ";
+                        MLSL_THROW(
+                            "mesh peer closed with receives outstanding: "
+                            "hdr_got=" + std::to_string(recv_hdr_got_) +
+                            " posted=[" + tags + "] unexpected=" +
+                            std::to_string(unexpected_.size()));
+                    }
                     peer_closed_ = true;
                     return;
                 }

@@ -199,22 +199,42 @@ void CommRequest::BuildChunks() {
     MLSL_CHECK(group_->IsMember(), "request on a group this rank is not in");
     const size_t es = DtypeSize(dtype_);
 
-    // Chunk fan-out (element-splittable ops only).
+    // Chunk fan-out. Element-splittable ops slice the message; the ring/
+    // pairwise ops (RS, AG(v), AlltoAll(v)) use dedicated chunked builders
+    // that slice every rank-block (reference endpoint split of those ops,
+    // src/comm_ep.cpp:598-736 + MLSL_ALLTOALL_SPLIT).
     size_t n_chunks = 1;
     const bool splittable = spec_.op == CollOp::ALLREDUCE || spec_.op == CollOp::BCAST ||
                             spec_.op == CollOp::REDUCE;
+    const bool block_splittable =
+        spec_.op == CollOp::REDUCE_SCATTER || spec_.op == CollOp::ALLGATHER ||
+        spec_.op == CollOp::ALLGATHERV || spec_.op == CollOp::ALLTOALL ||
+        spec_.op == CollOp::ALLTOALLV;
     // Splitting only pays when chunks land on DIFFERENT channel streams
     // (concurrent RCCL comms over distinct link schedules). On a single
     // channel the chunks serialize on one stream, so splitting is pure
     // launch overhead — unlike the reference, whose endpoint servers gave
     // every chunk its own progress process (comm_ep.cpp GET_EP_PAYLOAD).
-    if (splittable && gs > 1 && !Compressed() && cfg.num_channels > 1) {
+    // Block-splitting applies only where the executor WALKS the chunk
+    // schedules (host path, p2p transport): the fused RCCL path has no
+    // strided sub-collective to map a block-chunk onto.
+    const bool walks_schedules =
+        !Context::Get().DeviceMode() || UsesDeviceSchedule();
+    if ((splittable || (block_splittable && walks_schedules)) && gs > 1 &&
+        !Compressed() && cfg.num_channels > 1) {
         n_chunks = cfg.num_channels;
-        if (MessageBytes() >= cfg.large_msg_mb * (1024 * 1024) && cfg.large_msg_chunks > 1)
+        if (splittable && MessageBytes() >= cfg.large_msg_mb * (1024 * 1024) &&
+            cfg.large_msg_chunks > 1)
             n_chunks *= cfg.large_msg_chunks;
         // Keep chunks >= 4 KB so per-message overhead stays amortized.
         const size_t min_chunk_elems = std::max<size_t>(1, 4096 / es);
-        n_chunks = std::max<size_t>(1, std::min(n_chunks, spec_.count / std::max<size_t>(1, min_chunk_elems)));
+        const size_t per_chunk_basis =
+            splittable ? spec_.count
+                       : std::max<size_t>(1, MessageBytes() / es /
+                                                 static_cast<size_t>(gs));
+        n_chunks = std::max<size_t>(
+            1, std::min(n_chunks,
+                        per_chunk_basis / std::max<size_t>(1, min_chunk_elems)));
         n_chunks = std::min<size_t>(n_chunks, 4096);
         if (n_chunks == 0) n_chunks = 1;
     }
@@ -266,13 +286,32 @@ void CommRequest::BuildChunks() {
                 ce.sch = BuildBcast(gr, gs, cnt, dtype_, spec_.root);
                 break;
             case CollOp::REDUCE_SCATTER:
-                ce.sch = BuildReduceScatter(gr, gs, spec_.count, dtype_, spec_.rop);
+                if (n_chunks > 1) {
+                    ce.elem_off = 0;  // chunk offsets are absolute
+                    ce.sch = BuildReduceScatterChunk(
+                        gr, gs, spec_.count, SegOffset(spec_.count, n_chunks, c),
+                        SegCount(spec_.count, n_chunks, c), dtype_, spec_.rop);
+                } else {
+                    ce.sch = BuildReduceScatter(gr, gs, spec_.count, dtype_, spec_.rop);
+                }
                 break;
             case CollOp::ALLGATHER:
-                ce.sch = BuildAllGather(gr, gs, spec_.count, dtype_);
+                if (n_chunks > 1) {
+                    ce.elem_off = 0;
+                    std::vector<size_t> counts(static_cast<size_t>(gs), spec_.count);
+                    ce.sch = BuildAllGathervChunk(gr, gs, counts, c, n_chunks, dtype_);
+                } else {
+                    ce.sch = BuildAllGather(gr, gs, spec_.count, dtype_);
+                }
                 break;
             case CollOp::ALLGATHERV:
-                ce.sch = BuildAllGatherv(gr, gs, spec_.recv_counts, dtype_);
+                if (n_chunks > 1) {
+                    ce.elem_off = 0;
+                    ce.sch = BuildAllGathervChunk(gr, gs, spec_.recv_counts, c,
+                                                  n_chunks, dtype_);
+                } else {
+                    ce.sch = BuildAllGatherv(gr, gs, spec_.recv_counts, dtype_);
+                }
                 break;
             case CollOp::GATHER:
                 ce.sch = BuildGather(gr, gs, spec_.count, dtype_, spec_.root);
@@ -281,11 +320,28 @@ void CommRequest::BuildChunks() {
                 ce.sch = BuildScatter(gr, gs, spec_.count, dtype_, spec_.root);
                 break;
             case CollOp::ALLTOALL:
-                ce.sch = BuildAlltoAll(gr, gs, spec_.count, dtype_);
+                if (n_chunks > 1) {
+                    ce.elem_off = 0;
+                    std::vector<size_t> cnt(static_cast<size_t>(gs), spec_.count);
+                    std::vector<size_t> off(static_cast<size_t>(gs));
+                    for (int i = 0; i < gs; ++i)
+                        off[static_cast<size_t>(i)] = static_cast<size_t>(i) * spec_.count;
+                    ce.sch = BuildAlltoAllvChunk(gr, gs, cnt, off, cnt, off, c,
+                                                 n_chunks, dtype_);
+                } else {
+                    ce.sch = BuildAlltoAll(gr, gs, spec_.count, dtype_);
+                }
                 break;
             case CollOp::ALLTOALLV:
-                ce.sch = BuildAlltoAllv(gr, gs, spec_.send_counts, spec_.send_offs,
-                                        spec_.recv_counts, spec_.recv_offs, dtype_);
+                if (n_chunks > 1) {
+                    ce.elem_off = 0;
+                    ce.sch = BuildAlltoAllvChunk(gr, gs, spec_.send_counts,
+                                                 spec_.send_offs, spec_.recv_counts,
+                                                 spec_.recv_offs, c, n_chunks, dtype_);
+                } else {
+                    ce.sch = BuildAlltoAllv(gr, gs, spec_.send_counts, spec_.send_offs,
+                                            spec_.recv_counts, spec_.recv_offs, dtype_);
+                }
                 break;
             case CollOp::BARRIER:
                 ce.sch = BuildBarrier(gr, gs);

@@ -261,6 +261,144 @@ Schedule BuildReduceScatter(int rank, int size, size_t recv_count, DataType dt, 
     return sch;
 }
 
+// Chunked reduce-scatter: the chunk covers output elements [off, off+cnt)
+// of every rank's recv_count-long segment; SEND reads use the FULL-buffer
+// row stride (total*es per rank-row) so independent chunks cover the whole
+// exchange (reference endpoint split of RS: src/comm_ep.cpp:623-639).
+Schedule BuildReduceScatterChunk(int rank, int size, size_t total, size_t off,
+                                 size_t cnt, DataType dt, ReduceOp op) {
+    const size_t es = DtypeSize(dt);
+    const size_t segB = cnt * es;
+    if (size == 1) {
+        Schedule sch;
+        sch.dtype = dt;
+        sch.rop = op;
+        if (segB)
+            sch.AddStep(MakeCopy(0, Ref(Space::SEND, off * es, segB),
+                                 Ref(Space::RECV, off * es, segB)));
+        sch.result = Ref(Space::RECV, off * es, segB);
+        return sch;
+    }
+
+    Schedule sch;
+    sch.dtype = dt;
+    sch.rop = op;
+    const int N = size, r = rank;
+    const int next = (r + 1) % N, prev = (r - 1 + N) % N;
+    const int slots = std::min(2, N - 1);
+    sch.tmp_bytes = static_cast<size_t>(slots) * segB;
+    auto srow = [&](int i) {
+        return Ref(Space::SEND, (static_cast<size_t>(i) * total + off) * es, segB);
+    };
+    for (int p = 0; p <= N - 2; ++p) {
+        const int ssend = (r - p - 1 + 2 * N) % N;
+        const int srecv = (r - p - 2 + 2 * N) % N;
+        Step st;
+        st.phase = p;
+        st.send_peer = next;
+        st.send = (p == 0) ? srow(ssend)
+                           : Ref(Space::TMP, ((p - 1) % 2) * segB, segB);
+        st.recv_peer = prev;
+        st.recv = Ref(Space::TMP, (p % 2) * segB, segB);
+        st.local = Step::LocalOp::REDUCE;
+        st.local_src = srow(srecv);
+        st.local_dst = st.recv;
+        sch.AddStep(st);
+    }
+    sch.AddStep(MakeCopy(N - 1, Ref(Space::TMP, ((N - 2) % 2) * segB, segB),
+                         Ref(Space::RECV, off * es, segB)));
+    sch.result = Ref(Space::RECV, 0, total * es);
+    return sch;
+}
+
+// Chunked all-gather(v): chunk c covers sub-range SegOffset/SegCount(rc[i],
+// nchunks, c) of EVERY rank's block — the ring forwards only those slices
+// (reference endpoint split of AG(v): src/comm_ep.cpp:598-622).
+Schedule BuildAllGathervChunk(int rank, int size,
+                              const std::vector<size_t>& recv_counts,
+                              size_t chunk_idx, size_t nchunks, DataType dt) {
+    const size_t es = DtypeSize(dt);
+    MLSL_CHECK(recv_counts.size() == static_cast<size_t>(size),
+               "recv_counts size mismatch");
+    std::vector<size_t> offB(size + 1, 0);
+    for (int i = 0; i < size; ++i) offB[i + 1] = offB[i] + recv_counts[i] * es;
+
+    Schedule sch;
+    sch.dtype = dt;
+    const int N = size, r = rank;
+    auto sub_off = [&](int i) {
+        return SegOffset(recv_counts[(i % N + N) % N], nchunks, chunk_idx);
+    };
+    auto sub_cnt = [&](int i) {
+        return SegCount(recv_counts[(i % N + N) % N], nchunks, chunk_idx);
+    };
+    auto seg = [&](int i) {
+        int k = (i % N + N) % N;
+        return Ref(Space::RECV, offB[k] + sub_off(k) * es, sub_cnt(k) * es);
+    };
+    if (size == 1) {
+        if (sub_cnt(0))
+            sch.AddStep(MakeCopy(0, Ref(Space::SEND, sub_off(0) * es,
+                                        sub_cnt(0) * es), seg(0)));
+        sch.result = Ref(Space::RECV, 0, offB[1]);
+        return sch;
+    }
+    const int next = (r + 1) % N, prev = (r - 1 + N) % N;
+    sch.AddStep(MakeCopy(0, Ref(Space::SEND, sub_off(r) * es, sub_cnt(r) * es),
+                         seg(r)));
+    for (int p = 1; p <= N - 1; ++p) {
+        Step st;
+        st.phase = p;
+        st.send_peer = next;
+        st.send = seg(r - p + 1);
+        st.recv_peer = prev;
+        st.recv = seg(r - p);
+        sch.AddStep(st);
+    }
+    sch.result = Ref(Space::RECV, 0, offB[N]);
+    return sch;
+}
+
+// Chunked alltoall(v): chunk c carries sub-range c of every pairwise block
+// (reference MLSL_ALLTOALL_SPLIT, src/comm_ep.cpp:712-736).
+Schedule BuildAlltoAllvChunk(int rank, int size,
+                             const std::vector<size_t>& send_counts,
+                             const std::vector<size_t>& send_offs,
+                             const std::vector<size_t>& recv_counts,
+                             const std::vector<size_t>& recv_offs,
+                             size_t chunk_idx, size_t nchunks, DataType dt) {
+    const size_t es = DtypeSize(dt);
+    Schedule sch;
+    sch.dtype = dt;
+    const int N = size, r = rank;
+    auto sref = [&](int i) {
+        return Ref(Space::SEND,
+                   (send_offs[i] + SegOffset(send_counts[i], nchunks, chunk_idx)) * es,
+                   SegCount(send_counts[i], nchunks, chunk_idx) * es);
+    };
+    auto rref = [&](int i) {
+        return Ref(Space::RECV,
+                   (recv_offs[i] + SegOffset(recv_counts[i], nchunks, chunk_idx)) * es,
+                   SegCount(recv_counts[i], nchunks, chunk_idx) * es);
+    };
+    sch.AddStep(MakeCopy(0, sref(r), rref(r)));
+    for (int j = 1; j < N; ++j) {
+        const int to = (r + j) % N, from = (r - j + N) % N;
+        Step st;
+        st.phase = j;
+        st.send_peer = to;
+        st.send = sref(to);
+        st.recv_peer = from;
+        st.recv = rref(from);
+        sch.AddStep(st);
+    }
+    size_t total = 0;
+    for (int i = 0; i < N; ++i)
+        total = std::max(total, (recv_offs[i] + recv_counts[i]) * es);
+    sch.result = Ref(Space::RECV, 0, total);
+    return sch;
+}
+
 // Ring all-gather: phase 0 copies the local contribution into slot `rank`,
 // then N-1 neighbor forwards.
 Schedule BuildAllGather(int rank, int size, size_t send_count, DataType dt) {

@@ -90,6 +90,20 @@ Schedule BuildAllReduceRingUnits(int rank, int size, size_t units, size_t unit_b
                                  size_t quant_block);
 Schedule BuildAllReduceRHD(int rank, int size, size_t count, DataType dt, ReduceOp op);
 Schedule BuildReduceScatter(int rank, int size, size_t recv_count, DataType dt, ReduceOp op);
+// Chunked variants (channel fan-out for the non-elementwise-splittable
+// ops; reference endpoint split, src/comm_ep.cpp:598-736): offsets are
+// absolute in the user buffers, so the executor passes elem_off = 0.
+Schedule BuildReduceScatterChunk(int rank, int size, size_t total, size_t off,
+                                 size_t cnt, DataType dt, ReduceOp op);
+Schedule BuildAllGathervChunk(int rank, int size,
+                              const std::vector<size_t>& recv_counts,
+                              size_t chunk_idx, size_t nchunks, DataType dt);
+Schedule BuildAlltoAllvChunk(int rank, int size,
+                             const std::vector<size_t>& send_counts,
+                             const std::vector<size_t>& send_offs,
+                             const std::vector<size_t>& recv_counts,
+                             const std::vector<size_t>& recv_offs,
+                             size_t chunk_idx, size_t nchunks, DataType dt);
 Schedule BuildAllGather(int rank, int size, size_t send_count, DataType dt);
 Schedule BuildAllGatherv(int rank, int size, const std::vector<size_t>& recv_counts, DataType dt);
 Schedule BuildBcast(int rank, int size, size_t count, DataType dt, int root);

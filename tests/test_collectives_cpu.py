@@ -31,9 +31,12 @@ def test_priority_concurrent(world):
               extra_env={"MLSL_MSG_PRIORITY": "1", "MLSL_MSG_PRIORITY_THRESHOLD": "4000"})
 
 
-def test_chunked_channels():
+@pytest.mark.parametrize("world", [2, 3, 4])
+def test_chunked_channels(world):
     # exercise the chunk-over-channels fan-out (endpoint-parallelism analog)
-    run_ranks("collectives_sweep", 2,
+    # across ALL ops — including the block-chunked RS/AG(v)/AlltoAll(v)
+    # builders (reference endpoint split, src/comm_ep.cpp:598-736)
+    run_ranks("collectives_sweep", world,
               extra_env={"MLSL_NUM_CHANNELS": "4", "MLSL_LARGE_MSG_SIZE_MB": "0"})
 
 
