@@ -643,6 +643,20 @@ class Statistics:
         check(lib().mlsl_statistics_get_total_compute_cycles(self._h, ctypes.byref(v)))
         return v.value
 
+    def comm_device_ns(self, op_idx):
+        v = c_ull(0)
+        check(lib().mlsl_statistics_get_comm_device_ns(self._h, op_idx,
+                                                       ctypes.byref(v)))
+        return v.value
+
+    @property
+    def total_comm_device_ns(self):
+        # hipEvent-measured GPU comm time (ns); 0 on the host path
+        v = c_ull(0)
+        check(lib().mlsl_statistics_get_total_comm_device_ns(self._h,
+                                                             ctypes.byref(v)))
+        return v.value
+
 
 class Session:
     def __init__(self, phase="train"):

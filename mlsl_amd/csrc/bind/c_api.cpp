@@ -680,6 +680,16 @@ int mlsl_statistics_get_total_compute_cycles(mlsl_statistics st, unsigned long l
     C_TRY* out = STATS(st)->GetTotalComputeCycles();
     C_CATCH
 }
+int mlsl_statistics_get_comm_device_ns(mlsl_statistics st, size_t op,
+                                       unsigned long long* out) {
+    C_TRY* out = STATS(st)->GetCommDeviceNs(op);
+    C_CATCH
+}
+int mlsl_statistics_get_total_comm_device_ns(mlsl_statistics st,
+                                             unsigned long long* out) {
+    C_TRY* out = STATS(st)->GetTotalCommDeviceNs();
+    C_CATCH
+}
 
 
 /* ---- native getters added for reference parity ---- */

@@ -53,6 +53,7 @@ DeviceReqState::~DeviceReqState() {
     for (hipEvent_t e : events)
         if (e) (void)hipEventDestroy(e);
     if (dep_event) (void)hipEventDestroy(dep_event);
+    if (t0_event) (void)hipEventDestroy(t0_event);
     // Return scratch to the HBM pool while the runtime is alive; fall back
     // to hipFree during teardown.
     DeviceRuntime* rt = Context::Initialized() ? Context::Get().Device() : nullptr;
@@ -620,9 +621,13 @@ bool AllEventsDone(DeviceReqState& st) {
 }
 
 void EnsureEvents(DeviceReqState& st, size_t n) {
+    // Timing-capable events when MLSL_STATS is on (hipEvent-backed device
+    // statistics); the cheaper no-timestamp flavor otherwise.
+    const unsigned flags =
+        GlobalConfig().stats ? hipEventDefault : hipEventDisableTiming;
     while (st.events.size() < n) {
         hipEvent_t e;
-        HIP_CHECKD(hipEventCreateWithFlags(&e, hipEventDisableTiming));
+        HIP_CHECKD(hipEventCreateWithFlags(&e, flags));
         st.events.push_back(e);
     }
 }
@@ -644,6 +649,20 @@ bool DeviceFinishIfDone(CommRequest* req, DeviceReqState& st) {
 }
 
 }  // namespace
+
+double DeviceRequestCommMs(DeviceReqState& st) {
+    if (!st.timed || !st.t0_event || st.events.empty()) return -1.0;
+    float best = -1.0f;
+    for (hipEvent_t e : st.events) {
+        float ms = 0.0f;
+        if (hipEventElapsedTime(&ms, st.t0_event, e) == hipSuccess) {
+            if (ms > best) best = ms;
+        } else {
+            (void)hipGetLastError();
+        }
+    }
+    return best;
+}
 
 bool DeviceAdvanceRequest(CommRequest* req, DeviceReqState& st) {
     auto* rt = static_cast<HipRuntime*>(Context::Get().Device());
@@ -794,6 +813,15 @@ bool DeviceAdvanceRequest(CommRequest* req, DeviceReqState& st) {
     const bool try_capture = cfg.use_graphs && !st.graph_failed && !s_host &&
                              !r_host && !prio && !short_local && !p2p &&
                              GraphEligible(req, gc);
+
+    // hipEvent device timing (MLSL_STATS): t0 on the stream that issues
+    // first; completion events are timing-capable (EnsureEvents).
+    st.timed = cfg.stats;
+    if (st.timed) {
+        if (!st.t0_event)
+            HIP_CHECKD(hipEventCreateWithFlags(&st.t0_event, hipEventDefault));
+        HIP_CHECKD(hipEventRecord(st.t0_event, prio ? gc.prio_stream : base_s));
+    }
 
     const size_t es = DtypeSize(req->Dtype());
     const size_t nch = p2p ? std::max<size_t>(1, gc.streams.size())

@@ -42,5 +42,8 @@ DeviceRuntime* CreateDeviceRuntime();
 // Request hooks (called from CommRequest/Engine).
 void DeviceSetupRequest(CommRequest* req, DeviceReqState& st);
 bool DeviceAdvanceRequest(CommRequest* req, DeviceReqState& st);
+// hipEvent comm time of the last issued request (ms); -1 when not timed
+// (MLSL_STATS off) or not yet complete. Valid after completion.
+double DeviceRequestCommMs(DeviceReqState& st);
 
 }  // namespace mlsl

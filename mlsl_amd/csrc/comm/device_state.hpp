@@ -16,6 +16,14 @@ struct DeviceReqState {
     hipEvent_t dep_event = nullptr;
     // One completion event per channel used by this request.
     std::vector<hipEvent_t> events;
+    // hipEvent-backed device timing (MLSL_STATS): events are created
+    // timing-capable, t0 is recorded at issue after the producer
+    // dependency, and comm time = max over channels of elapsed(t0, ev).
+    // The reference attributed comm time host-side via rdtsc deltas
+    // (src/mlsl_impl_stats.cpp:564-668), which misattributes overlapped
+    // async comm to the Wait call — event deltas measure the GPU truth.
+    hipEvent_t t0_event = nullptr;
+    bool timed = false;
     // Persistent device scratch (allocated at Setup in device mode).
     void* tmp_dev = nullptr;
     size_t tmp_bytes = 0;

@@ -620,6 +620,10 @@ bool CommRequest::AdvanceHost(Mesh* mesh) {
     return all_done;
 }
 
+double CommRequest::LastDeviceCommMs() const {
+    return dev_ ? DeviceRequestCommMs(*dev_) : -1.0;
+}
+
 bool CommRequest::AdvanceDevice() {
     MLSL_CHECK(dev_ != nullptr, "device state missing");
     return DeviceAdvanceRequest(this, *dev_);

@@ -142,6 +142,9 @@ class CommRequest {
 
     ReqState State() const { return state_.load(std::memory_order_acquire); }
     uint64_t StartSeqno() const { return start_seqno_; }
+    // hipEvent-measured device comm time of the last completed Start (ms);
+    // -1 on the host path or with MLSL_STATS off.
+    double LastDeviceCommMs() const;
 
     // Device-executor access (device_comm.cpp). SendBuf/RecvBuf return the
     // staging override when host buffers were staged into HBM

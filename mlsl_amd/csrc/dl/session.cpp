@@ -306,7 +306,10 @@ void* Activation::WaitComm() {
     // backward) means no exchange in this direction.
     if (need_comm_ && peer_ && peer_->req_) ptr = peer_->req_->Wait();
     ev.is_compute = false;
-    stats->Update(ev);
+    stats->Update(ev, 0,
+                  need_comm_ && peer_ && peer_->req_
+                      ? peer_->req_->LastDeviceCommMs()
+                      : -1.0);
     return ptr;
 }
 
@@ -393,7 +396,7 @@ void* ParameterSet::WaitGradientComm() {
     if (need_comm_) r = grad_req_->Wait();
     else r = last_grad_buf_;
     ev.is_compute = false;
-    stats->Update(ev);
+    stats->Update(ev, 0, need_comm_ ? grad_req_->LastDeviceCommMs() : -1.0);
     return r;
 }
 
@@ -434,7 +437,8 @@ void* ParameterSet::WaitIncrementComm() {
     stats->Update(ev);
     void* r = need_comm_ && inc_req_ ? inc_req_->Wait() : nullptr;
     ev.is_compute = false;
-    stats->Update(ev);
+    stats->Update(ev, 0,
+                  need_comm_ && inc_req_ ? inc_req_->LastDeviceCommMs() : -1.0);
     return r;
 }
 
@@ -509,7 +513,7 @@ void Statistics::EnsureSize(size_t n) {
     if (per_op_.size() < n) per_op_.resize(n);
 }
 
-void Statistics::Update(const Event& ev, size_t bytes) {
+void Statistics::Update(const Event& ev, size_t bytes, double device_ms) {
     if (!env_enabled_ || !started_) return;
     EnsureSize(ev.op_idx + 1);
     const unsigned long long now = Now();
@@ -526,6 +530,12 @@ void Statistics::Update(const Event& ev, size_t bytes) {
     }
     os.comm_bytes += bytes;
     os.ent[cls].bytes += bytes;
+    if (device_ms >= 0.0) {
+        const unsigned long long ns =
+            static_cast<unsigned long long>(device_ms * 1e6);
+        os.comm_device_ns += ns;
+        os.ent[cls].device_ns += ns;
+    }
 }
 
 void Statistics::CollectIsolation() {
@@ -596,17 +606,20 @@ void Statistics::Print() {
     FILE* f = std::fopen("mlsl_stats.log", "a");
     if (!f) return;
     std::fprintf(f, "# op | compute_cyc | comm_cyc | comm_KB | isolation_cyc |"
-                    " IA KB/cyc | OA KB/cyc | GRAD KB/cyc | INC KB/cyc\n");
+                    " dev_comm_us |"
+                    " IA KB/cyc/devus | OA KB/cyc/devus | GRAD KB/cyc/devus |"
+                    " INC KB/cyc/devus\n");
     for (size_t i = 0; i < per_op_.size(); ++i) {
         const OpStats& os = per_op_[i];
         const char* name = i < session_->GetOperationCount()
                                ? session_->GetOperation(i)->GetName()
                                : "?";
-        std::fprintf(f, "%zu(%s) | %llu | %llu | %zu | %llu", i, name,
+        std::fprintf(f, "%zu(%s) | %llu | %llu | %zu | %llu | %llu", i, name,
                      os.compute_cycles, os.comm_cycles, os.comm_bytes / 1024,
-                     os.isolation_cycles);
+                     os.isolation_cycles, os.comm_device_ns / 1000);
         for (int c = 0; c < 4; ++c)
-            std::fprintf(f, " | %zu/%llu", os.ent[c].bytes / 1024, os.ent[c].cycles);
+            std::fprintf(f, " | %zu/%llu/%llu", os.ent[c].bytes / 1024,
+                         os.ent[c].cycles, os.ent[c].device_ns / 1000);
         std::fprintf(f, "\n");
     }
     std::fclose(f);
@@ -637,6 +650,14 @@ unsigned long long Statistics::GetTotalCommCycles() const {
 unsigned long long Statistics::GetTotalComputeCycles() const {
     unsigned long long t = 0;
     for (auto& o : per_op_) t += o.compute_cycles;
+    return t;
+}
+unsigned long long Statistics::GetCommDeviceNs(size_t op) const {
+    return op < per_op_.size() ? per_op_[op].comm_device_ns : 0;
+}
+unsigned long long Statistics::GetTotalCommDeviceNs() const {
+    unsigned long long t = 0;
+    for (const auto& o : per_op_) t += o.comm_device_ns;
     return t;
 }
 size_t Statistics::GetTotalCommSize() const {

@@ -229,7 +229,9 @@ class Statistics {
         enum class Action { Start, Wait, Test } action;
     };
 
-    void Update(const Event& ev, size_t bytes = 0);
+    // `device_ms` >= 0 attributes hipEvent-measured GPU comm time of the
+    // just-completed request (device mode; Wait-side emitters pass it).
+    void Update(const Event& ev, size_t bytes = 0, double device_ms = -1.0);
     void CollectIsolation();   // 10 iterations, skip 4 (ref :48-49)
 
     // IsEnabled = MLSL_STATS env gate; IsStarted = currently collecting
@@ -249,6 +251,9 @@ class Statistics {
     unsigned long long GetTotalCommCycles() const;
     unsigned long long GetTotalComputeCycles() const;
     size_t GetTotalCommSize() const;
+    // hipEvent GPU comm time (ns); 0 in host mode / MLSL_STATS off.
+    unsigned long long GetCommDeviceNs(size_t op) const;
+    unsigned long long GetTotalCommDeviceNs() const;
 
     // per-entity classes in the printed table (reference per-op table of
     // [KB, Kcycles] per IA/OA/GRAD/INC, mlsl_impl_stats.cpp:97-363)
@@ -258,7 +263,15 @@ class Statistics {
     struct OpStats {
         unsigned long long comm_cycles = 0, compute_cycles = 0, isolation_cycles = 0;
         size_t comm_bytes = 0;
-        struct Ent { unsigned long long cycles = 0; size_t bytes = 0; } ent[4];
+        // hipEvent-measured GPU comm nanoseconds (device mode): unlike the
+        // host rdtsc deltas, overlapped async comm is attributed to the
+        // request that ran it, not to whoever happened to call Wait.
+        unsigned long long comm_device_ns = 0;
+        struct Ent {
+            unsigned long long cycles = 0;
+            size_t bytes = 0;
+            unsigned long long device_ns = 0;
+        } ent[4];
     };
     void EnsureSize(size_t n);
 

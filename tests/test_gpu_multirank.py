@@ -98,3 +98,8 @@ def test_quant_allreduce_multirank():
 @requires_gpu
 def test_hybrid_grid_multirank():
     run_gpu_ranks("gpu_hybrid_grid", 4)
+
+
+@requires_gpu
+def test_stats_device_ns_multirank():
+    run_gpu_ranks("gpu_stats_device_ns", 2)

@@ -239,3 +239,25 @@ def main():
 
 if __name__ == "__main__":
     main()
+
+
+def gpu_stats_device_ns():
+    """hipEvent-backed device Statistics: a committed 2-layer net at dp=2
+    must attribute nonzero GPU comm nanoseconds to the gradient allreduce
+    (host-rdtsc stats would hide overlapped comm in the Wait residue)."""
+    import os
+    os.environ["MLSL_STATS"] = "1"
+    mx, torch, rank, size = _init()
+    from mlsl_amd.models.synthetic import SyntheticNet
+    net = SyntheticNet(size, 1, global_mb=4 * size, xp=torch, device="cuda")
+    for _ in range(3):
+        net.step()
+    st = net.sess.stats
+    assert st.enabled
+    dev_ns = st.total_comm_device_ns
+    assert dev_ns > 0, f"device comm ns not attributed: {dev_ns}"
+    st.print()
+    mx.finalize()
+
+
+WORKERS["gpu_stats_device_ns"] = gpu_stats_device_ns
