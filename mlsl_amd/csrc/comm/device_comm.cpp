@@ -223,19 +223,49 @@ class HipRuntime : public DeviceRuntime {
             return;
         }
         GroupComms gc;
-        for (size_t ch = 0; ch < nch; ++ch) {
-            ncclUniqueId id{};
-            if (g->IsMember() && g->MyIdx() == 0) NCCL_CHECK(ncclGetUniqueId(&id));
-            std::vector<ncclUniqueId> all(static_cast<size_t>(ctx.Size()));
-            ctx.Boot()->Allgather(&id, sizeof(ncclUniqueId), all.data());
-            if (g->IsMember() && g->Size() > 1) {
-                ncclUniqueId gid = all[static_cast<size_t>(g->WorldRank(0))];
-                ncclComm_t comm;
-                NCCL_CHECK(ncclCommInitRank(&comm, g->Size(), gid, g->MyIdx()));
-                hipStream_t s;
-                HIP_CHECKD(hipStreamCreateWithFlags(&s, hipStreamNonBlocking));
-                gc.comms.push_back(comm);
-                gc.streams.push_back(s);
+        // Subgroups SPLIT the world communicators instead of paying a full
+        // ncclCommInitRank per group (a Distribution(4,2) on 8 GPUs would
+        // otherwise re-bootstrap RCCL per color). Colors are the group's
+        // lowest world rank — unique across the disjoint groups of one
+        // collective creation. World (and any group while the world comms
+        // don't exist yet) takes the init path.
+        GroupComms* wc = nullptr;
+        const bool is_world = g == ctx.World();
+        if (!is_world) {
+            auto wit = group_comms_.find(ctx.World()->Uid());
+            if (wit != group_comms_.end() && wit->second.comms.size() == nch)
+                wc = &wit->second;
+        }
+        if (wc) {
+            const int color = g->IsMember() && g->Size() > 1
+                                  ? g->WorldRank(0)
+                                  : NCCL_SPLIT_NOCOLOR;
+            const int key = g->IsMember() ? g->MyIdx() : 0;
+            for (size_t ch = 0; ch < nch; ++ch) {
+                ncclComm_t sub = nullptr;
+                NCCL_CHECK(ncclCommSplit(wc->comms[ch], color, key, &sub, nullptr));
+                if (color != NCCL_SPLIT_NOCOLOR && sub) {
+                    hipStream_t s;
+                    HIP_CHECKD(hipStreamCreateWithFlags(&s, hipStreamNonBlocking));
+                    gc.comms.push_back(sub);
+                    gc.streams.push_back(s);
+                }
+            }
+        } else {
+            for (size_t ch = 0; ch < nch; ++ch) {
+                ncclUniqueId id{};
+                if (g->IsMember() && g->MyIdx() == 0) NCCL_CHECK(ncclGetUniqueId(&id));
+                std::vector<ncclUniqueId> all(static_cast<size_t>(ctx.Size()));
+                ctx.Boot()->Allgather(&id, sizeof(ncclUniqueId), all.data());
+                if (g->IsMember() && g->Size() > 1) {
+                    ncclUniqueId gid = all[static_cast<size_t>(g->WorldRank(0))];
+                    ncclComm_t comm;
+                    NCCL_CHECK(ncclCommInitRank(&comm, g->Size(), gid, g->MyIdx()));
+                    hipStream_t s;
+                    HIP_CHECKD(hipStreamCreateWithFlags(&s, hipStreamNonBlocking));
+                    gc.comms.push_back(comm);
+                    gc.streams.push_back(s);
+                }
             }
         }
         if (GlobalConfig().msg_priority && g->IsMember() && g->Size() > 1) {

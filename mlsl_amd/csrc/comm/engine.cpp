@@ -1,5 +1,8 @@
 #include "engine.hpp"
 
+#include <pthread.h>
+#include <sched.h>
+
 #include <algorithm>
 #include <chrono>
 
@@ -153,6 +156,19 @@ bool Engine::ProgressAll() {
 }
 
 void Engine::Loop() {
+    // MLSL_SERVER_AFFINITY: pin the progress thread (the reference pinned
+    // each ep_server process to a core, eplib/server.c:63-81). Keeps the
+    // poll loop off the cores running the framework's compute threads.
+    const int aff = GlobalConfig().server_affinity;
+    if (aff >= 0) {
+        cpu_set_t set;
+        CPU_ZERO(&set);
+        CPU_SET(static_cast<unsigned>(aff) % CPU_SETSIZE, &set);
+        if (pthread_setaffinity_np(pthread_self(), sizeof(set), &set) != 0)
+            MLSL_LOG(ERROR, "MLSL_SERVER_AFFINITY=%d: pinning failed", aff);
+        else
+            MLSL_LOG(DEBUG, "progress thread pinned to core %d", aff);
+    }
     int idle_spins = 0;
     while (!stop_.load(std::memory_order_acquire)) {
         bool did_work = false;

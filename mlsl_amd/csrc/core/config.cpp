@@ -58,6 +58,8 @@ Config Config::FromEnv() {
     if (c.copy_threads < 1) c.copy_threads = 1;
     if (c.copy_threads > 16) c.copy_threads = 16;
     c.copy_threshold = EnvSize("MLSL_COPY_THRESHOLD", 4u << 20);
+    if (const char* e = std::getenv("MLSL_SERVER_AFFINITY"))
+        c.server_affinity = std::atoi(e);  // first core of the ref's list form
     c.timeout_sec = static_cast<int>(EnvSize("MLSL_TIMEOUT", 300));
     c.use_graphs = EnvBool("MLSL_USE_GRAPHS", false);
     return c;

@@ -52,6 +52,9 @@ struct Config {
     // user<->pinned memcpys above the threshold split across threads.
     size_t copy_threads = 4;       // MLSL_COPY_THREADS
     size_t copy_threshold = 4u << 20;  // MLSL_COPY_THRESHOLD (bytes)
+    // Pin the progress thread to a CPU core (reference MLSL_SERVER_AFFINITY,
+    // eplib/server.c:63-81 pinned each ep_server). -1 = unpinned.
+    int server_affinity = -1;      // MLSL_SERVER_AFFINITY
     int timeout_sec = 300;         // MLSL_TIMEOUT: bootstrap/collective timeout
     bool use_graphs = false;       // MLSL_USE_GRAPHS: hipGraph replay of
                                    // persistent device requests (single
