@@ -43,16 +43,25 @@ def main():
            "--iters", str(args.iters), "--warmup", str(args.warmup)]
     if args.algo:
         cmd += ["--algo", args.algo]
+    rank_log = os.environ.get("SWEEP_RANK_LOG")  # prefix: live per-rank logs
     procs = []
+    logs = []
     for r in range(args.world):
         env = dict(os.environ)
         env.update({"RANK": str(r), "WORLD_SIZE": str(args.world),
                     "MASTER_ADDR": "127.0.0.1", "MLSL_PORT": str(port),
-                    "PYTHONPATH": REPO, "MLSL_TIMEOUT": "120"})
+                    "PYTHONPATH": REPO, "MLSL_TIMEOUT": "120",
+                    "PYTHONUNBUFFERED": "1"})
         env.pop("MLSL_TRANSPORT", None)
-        procs.append(subprocess.Popen(
-            cmd, env=env, cwd=REPO, stdout=subprocess.PIPE,
-            stderr=subprocess.STDOUT, text=True))
+        if rank_log:
+            lf = open(f"{rank_log}.rank{r}.log", "w")
+            logs.append(lf)
+            procs.append(subprocess.Popen(cmd, env=env, cwd=REPO, stdout=lf,
+                                          stderr=subprocess.STDOUT, text=True))
+        else:
+            procs.append(subprocess.Popen(
+                cmd, env=env, cwd=REPO, stdout=subprocess.PIPE,
+                stderr=subprocess.STDOUT, text=True))
     rc = 0
     out0 = ""
     for r, p in enumerate(procs):
@@ -64,11 +73,16 @@ def main():
             print(f"TIMEOUT rank {r}", file=sys.stderr)
             sys.exit(3)
         if r == 0:
-            out0 = out
+            out0 = out if out is not None else ""
         if p.returncode != 0:
             rc = p.returncode
-            print(f"--- rank {r} rc={p.returncode} ---\n{out[-3000:]}",
-                  file=sys.stderr)
+            print(f"--- rank {r} rc={p.returncode} ---\n"
+                  f"{(out or '(see rank log)')[-3000:]}", file=sys.stderr)
+    for lf in logs:
+        lf.close()
+    if rank_log and not out0:
+        with open(f"{rank_log}.rank0.log") as f:
+            out0 = f.read()
     lines = [ln for ln in out0.splitlines() if ln.startswith("{")]
     for ln in lines:
         print(ln)
