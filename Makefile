@@ -39,7 +39,7 @@ OBJS := $(CSRC:%.cpp=$(BUILD)/%.o) $(HIPSRC:%.hip=$(BUILD)/%.o)
 
 .PHONY: all lib selftest apitest benchlat samples quantplugin clean test
 
-all: lib selftest apitest benchlat samples quantplugin
+all: lib selftest apitest e2e benchlat samples quantplugin
 
 lib: $(LIB)
 
@@ -65,6 +65,14 @@ APITEST := $(BUILD)/api_selftest
 apitest: $(APITEST)
 
 $(APITEST): $(BUILD)/mlsl_amd/csrc/tests/api_selftest.o $(LIB)
+	@mkdir -p $(dir $@)
+	$(HIPCC) $< -Lmlsl_amd -lmlsl_amd -Wl,-rpath,'$$ORIGIN/../mlsl_amd' -pthread -o $@
+
+E2E := $(BUILD)/mlsl_e2e
+
+e2e: $(E2E)
+
+$(E2E): $(BUILD)/mlsl_amd/csrc/tests/mlsl_e2e.o $(LIB)
 	@mkdir -p $(dir $@)
 	$(HIPCC) $< -Lmlsl_amd -lmlsl_amd -Wl,-rpath,'$$ORIGIN/../mlsl_amd' -pthread -o $@
 
@@ -154,3 +162,4 @@ clean:
 -include $(BUILD)/mlsl_amd/csrc/tests/schedule_selftest.d
 -include $(BUILD)/mlsl_amd/csrc/tests/api_selftest.d
 -include $(BUILD)/mlsl_amd/csrc/tests/bench_latency.d
+-include $(BUILD)/mlsl_amd/csrc/tests/mlsl_e2e.d

@@ -1,0 +1,243 @@
+// Full mlsl_test-equivalent end-to-end run: the reference protocol of
+// 2 epochs x 3 minibatches of Forward / Backward1 / Backward2 / Update over
+// a 2-layer synthetic CC net (/root/reference/tests/examples/mlsl_test/
+// mlsl_test.cpp:79-121, 407-428, 464-528), with:
+//   USER_BUF=1     activations/gradients in plain malloc'd memory (the
+//                  ReplaceIn/Out staging path; on a GPU box this exercises
+//                  the pinned-host bounce pipeline), else Environment::Alloc
+//                  registered buffers (zero-copy path).
+//   QUANT=1        int8-compressed gradient allreduce; correctness switches
+//                  from exact analytic values to a relative-error bound
+//                  (reference checks quant stats instead of exact match).
+//   MP, DIST_UPDATE  model-parts / ZeRO-1 toggle as in api_selftest.
+// Env: RANK/WORLD_SIZE/MLSL_PORT (tests/mp.py convention). Exit 0 = PASS.
+#include <cmath>
+#include <cstdio>
+#include <cstdlib>
+#include <cstring>
+#include <vector>
+
+#include "../dl/session.hpp"
+#include "../include/mlsl/mlsl.hpp"
+
+using namespace mlsl;
+
+static int g_fail = 0;
+#define CHECK_OK(cond, ...)                                                   \
+    do {                                                                      \
+        if (!(cond)) {                                                        \
+            std::printf("FAIL %d: ", __LINE__);                               \
+            std::printf(__VA_ARGS__);                                         \
+            std::printf("\n");                                                \
+            ++g_fail;                                                         \
+        }                                                                     \
+    } while (0)
+
+static int EnvInt(const char* n, int d) {
+    const char* e = std::getenv(n);
+    return e ? std::atoi(e) : d;
+}
+
+int main(int argc, char** argv) {
+    Environment& env = Environment::GetEnv();
+    env.Init(&argc, &argv);
+    const size_t size = env.GetProcessCount();
+    const size_t mp = static_cast<size_t>(EnvInt("MP", 1));
+    const bool du = EnvInt("DIST_UPDATE", 0) != 0;
+    const bool user_buf = EnvInt("USER_BUF", 0) != 0;
+    const bool quant = EnvInt("QUANT", 0) != 0;
+    const size_t dp = size / mp;
+    const size_t S = 6, F0 = 4 * mp, F1 = 8 * mp, KS = 3;
+    const size_t MB = 4 * dp;
+    const int kEpochs = 2, kMinibatches = 3;
+
+    if (quant) env.SetQuantizationParams(QuantParams{});
+
+    Distribution* dist = env.CreateDistribution(dp, mp);
+    Session* sess = env.CreateSession();
+    sess->SetGlobalMinibatchSize(MB);
+
+    auto mkop = [&](const char* name, size_t fin, size_t fout) {
+        OperationRegInfo* info = sess->CreateOperationRegInfo(OpKind::CC);
+        info->SetName(name);
+        info->AddInput(fin, S, DataType::F32);
+        info->AddOutput(fout, S, DataType::F32);
+        info->AddParameterSet(fin * fout, KS, DataType::F32, du,
+                              quant ? Compression::QUANT_INT8
+                                    : Compression::NONE);
+        info->Validate(dist);
+        return sess->GetOperation(sess->AddOperation(info, dist));
+    };
+    Operation* op0 = mkop("fc0", F0, F1);
+    Operation* op1 = mkop("fc1", F1, F0);
+    op0->SetNext(op1, 0, 0);
+    sess->Commit();
+
+    Activation* out0 = op0->GetOutput(0);
+    Activation* in1 = op1->GetInput(0);
+    const size_t lmb = op0->GetLocalMinibatchSize();
+    const size_t didx = dist->GetProcessIdx(GroupKind::DATA);
+    const size_t midx = dist->GetProcessIdx(GroupKind::MODEL);
+    const size_t f1l = in1->GetLocalFmCount();
+
+    // Comm buffers: registered (Environment::Alloc) or plain user memory
+    // (reference user_buf toggle flips exactly this — mlsl_test.cpp run
+    // matrix, Makefile:59-107).
+    auto get_buf = [&](size_t bytes) -> float* {
+        if (user_buf) return static_cast<float*>(std::malloc(bytes + 64));
+        return static_cast<float*>(env.Alloc(bytes + 64, 64));
+    };
+    auto put_buf = [&](void* p) {
+        if (user_buf) std::free(p);
+        else env.Free(p);
+    };
+    float* comm0 = get_buf(out0->GetCommBufSize() + 4);
+    float* comm1 = get_buf(in1->GetCommBufSize() + 4);
+
+    // Persistent "weights" per layer for the Update step.
+    std::vector<std::vector<float>> params(2);
+    std::vector<Operation*> ops{op0, op1};
+    for (int l = 0; l < 2; ++l) {
+        ParameterSet* ps = ops[l]->GetParameterSet(0);
+        params[l].assign(ps->GetLocalKernelCount() * KS, 0.0f);
+    }
+
+    for (int epoch = 0; epoch < kEpochs; ++epoch) {
+        for (int mbatch = 0; mbatch < kMinibatches; ++mbatch) {
+            const float it = epoch * kMinibatches + mbatch;
+
+            // ---- Forward: model-parallel partial-sum exchange ----
+            if (mp > 1) {
+                for (size_t i = 0; i < out0->GetPackBlockCount(); ++i) {
+                    const CommBlockInfo* b = out0->GetPackBlock(i);
+                    for (size_t mb = 0; mb < b->GetMbCount(); ++mb)
+                        for (size_t fm = 0; fm < b->GetFmCount(); ++fm)
+                            for (size_t k = 0; k < S; ++k) {
+                                const size_t gfm = b->GetFmOffset() + fm;
+                                comm0[b->GetBufOffset() +
+                                      (mb * b->GetFmCount() + fm) * S + k] =
+                                    gfm * 100.0f + k + (b->GetMbOffset() + mb) +
+                                    it + midx;
+                            }
+                }
+                out0->StartComm(comm0);
+                float* res = static_cast<float*>(in1->WaitComm());
+                CHECK_OK(res != nullptr, "null fwd result");
+                const CommBlockInfo* ub = in1->GetUnpackBlock(0);
+                for (size_t mb = 0; mb < lmb && res; ++mb)
+                    for (size_t fm = 0; fm < f1l; ++fm)
+                        for (size_t k = 0; k < S; ++k) {
+                            const float got =
+                                res[ub->GetBufOffset() + (mb * f1l + fm) * S + k];
+                            const size_t gfm = midx * f1l + fm;
+                            const float want = mp * (gfm * 100.0f + k + mb + it) +
+                                               mp * (mp - 1) / 2.0f;
+                            if (got != want) {
+                                CHECK_OK(got == want, "fwd mismatch ep=%d mb=%d",
+                                         epoch, mbatch);
+                                goto fwd_done;
+                            }
+                        }
+            fwd_done:
+                // ---- Backward1: input-grad exchange ----
+                const CommBlockInfo* pb = in1->GetPackBlock(0);
+                for (size_t mb = 0; mb < lmb; ++mb)
+                    for (size_t fm = 0; fm < f1l; ++fm)
+                        for (size_t k = 0; k < S; ++k) {
+                            const size_t gfm = midx * f1l + fm;
+                            comm1[pb->GetBufOffset() + (mb * f1l + fm) * S + k] =
+                                gfm * 7.0f + k + mb + it;
+                        }
+                in1->StartComm(comm1);
+                float* bres = static_cast<float*>(out0->WaitComm());
+                CHECK_OK(bres != nullptr, "null bwd result");
+                (void)bres;
+            }
+
+            // ---- Backward2 + Update per layer: gradient exchange then
+            //      optimizer on the owned shard (reference Backward2/Update
+            //      split, mlsl_test.cpp:464-528) ----
+            for (int l = 0; l < 2; ++l) {
+                ParameterSet* ps = ops[l]->GetParameterSet(0);
+                const size_t lk = ps->GetLocalKernelCount();
+                const size_t gk0 = ps->GetGlobalKernelOffset();
+                float* grad = get_buf(lk * KS * sizeof(float));
+                for (size_t j = 0; j < lk; ++j)
+                    for (size_t e = 0; e < KS; ++e)
+                        grad[j * KS + e] =
+                            0.125f * ((gk0 + j) % 37) + 0.5f * e + didx + it;
+                ps->StartGradientComm(grad);
+                float* g = static_cast<float*>(ps->WaitGradientComm());
+                if (dp > 1) {
+                    const size_t n = du ? ps->GetOwnedKernelCount() : lk;
+                    const size_t joff = du ? ps->GetOwnedKernelOffset() : 0;
+                    double err2 = 0, ref2 = 0;
+                    for (size_t j = 0; j < n; ++j)
+                        for (size_t e = 0; e < KS; ++e) {
+                            const float want =
+                                dp * (0.125f * ((gk0 + joff + j) % 37) +
+                                      0.5f * e + it) +
+                                dp * (dp - 1) / 2.0f;
+                            const float got = g[j * KS + e];
+                            err2 += (got - want) * (double)(got - want);
+                            ref2 += (double)want * want;
+                        }
+                    if (quant && !du) {
+                        // int8 path: relative-error bound instead of exact
+                        // (reference quant stats check, mlsl_test.cpp:407-428)
+                        const double rel =
+                            std::sqrt(err2) / std::max(std::sqrt(ref2), 1e-9);
+                        CHECK_OK(rel < 0.05, "quant grad rel err %.4f l=%d",
+                                 rel, l);
+                    } else {
+                        CHECK_OK(err2 == 0.0, "grad mismatch l=%d ep=%d mb=%d",
+                                 l, epoch, mbatch);
+                    }
+                }
+                // Update on the owned shard, then increment AllGather (du).
+                const size_t un = du ? ps->GetOwnedKernelCount() : lk;
+                const size_t uoff = du ? ps->GetOwnedKernelOffset() : 0;
+                for (size_t j = 0; j < un; ++j)
+                    for (size_t e = 0; e < KS; ++e)
+                        params[l][(uoff + j) * KS + e] -=
+                            0.01f * g[j * KS + e];
+                if (du && dp > 1) {
+                    ps->StartIncrementComm(params[l].data());
+                    float* inc = static_cast<float*>(ps->WaitIncrementComm());
+                    CHECK_OK(inc != nullptr, "null inc");
+                    if (inc) std::memcpy(params[l].data(), inc,
+                                         lk * KS * sizeof(float));
+                }
+                put_buf(grad);
+            }
+        }
+        // epoch boundary: everyone synced (reference barriers per epoch)
+        dist->Barrier(GroupKind::GLOBAL);
+    }
+
+    // params must agree across the data group after the epochs (both modes:
+    // all ranks applied identical reduced gradients, or allgathered the
+    // owned shards — quant included, since every rank dequantizes the same
+    // wire data): bcast rank 0's copy and compare.
+    for (int l = 0; l < 2; ++l) {
+        std::vector<float> ref = params[l];
+        CommRequest* r = dist->Bcast(ref.data(), ref.size(), DataType::F32, 0,
+                                     GroupKind::DATA);
+        env.Wait(r);
+        if (dp > 1)
+            for (size_t i = 0; i < ref.size(); ++i)
+                if (ref[i] != params[l][i]) {
+                    CHECK_OK(false, "param divergence l=%d i=%zu", l, i);
+                    break;
+                }
+    }
+
+    put_buf(comm0);
+    put_buf(comm1);
+    env.DeleteSession(sess);
+    env.DeleteDistribution(dist);
+    if (g_fail == 0) std::printf("MLSL E2E PASSED\n");
+    else std::printf("MLSL E2E: %d failures\n", g_fail);
+    env.Finalize();
+    return g_fail == 0 ? 0 : 1;
+}

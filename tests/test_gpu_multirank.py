@@ -103,3 +103,32 @@ def test_hybrid_grid_multirank():
 @requires_gpu
 def test_stats_device_ns_multirank():
     run_gpu_ranks("gpu_stats_device_ns", 2)
+
+
+@requires_gpu
+@pytest.mark.parametrize("mp,du,user_buf,quant", [
+    (1, 0, 0, 0), (2, 1, 0, 0), (1, 0, 1, 0), (1, 0, 0, 1),
+])
+def test_e2e_device_multirank(mp, du, user_buf, quant):
+    """mlsl_test-equivalent epoch loop on the DEVICE engine (p2p transport,
+    multi-rank-one-GPU): session planner + pack blocks + grad/inc exchange,
+    with the user_buf toggle exercising pinned-host staging."""
+    exe = os.path.join(REPO, "build", "mlsl_e2e")
+    if not os.path.exists(exe):
+        subprocess.run(["make", "e2e"], cwd=REPO, check=True,
+                       capture_output=True, timeout=900)
+    port = free_port()
+    world = 4
+    procs = []
+    for r in range(world):
+        env = dict(os.environ, RANK=str(r), WORLD_SIZE=str(world),
+                   MASTER_ADDR="127.0.0.1", MLSL_PORT=str(port),
+                   MP=str(mp), DIST_UPDATE=str(du), USER_BUF=str(user_buf),
+                   QUANT=str(quant), MLSL_TIMEOUT="90")
+        env.pop("MLSL_TRANSPORT", None)
+        procs.append(subprocess.Popen([exe], env=env, cwd=REPO,
+                                      stdout=subprocess.PIPE,
+                                      stderr=subprocess.STDOUT, text=True))
+    for r, p in enumerate(procs):
+        out, _ = p.communicate(timeout=240)
+        assert p.returncode == 0 and "PASSED" in out, f"rank {r}: {out[-2500:]}"
