@@ -183,17 +183,12 @@ void P2pGroup::IssueSchedule(CommRequest* req, ChunkExec& ce, size_t lane,
     // length, so the slot partition always agrees.
     auto sub_size = [&](size_t bytes) -> size_t {
         size_t v = std::max<size_t>((bytes + 7) / 8,
-                                    std::min<size_t>(2u << 20, msg_max));
+                                    std::min<size_t>(4u << 20, msg_max));
         v = std::min(v, msg_max);
         v = (v / unit) * unit;
         if (v < unit) v = unit;
         return std::min(v, msg_max);
     };
-    XferPoll poll{};
-    poll.abort_word = abort_host_;
-    poll.status = status_host_;
-    poll.max_ticks = max_ticks_;
-
     for (int phase = 0; phase < ce.sch.num_phases; ++phase) {
         // Collect this phase's send and recv jobs, then interleave their
         // sub-messages round-robin. The interleave is what makes the
@@ -244,13 +239,16 @@ void P2pGroup::IssueSchedule(CommRequest* req, ChunkExec& ce, size_t lane,
                 uint64_t& sent = sent_[static_cast<size_t>(peer) * nlanes_ + lane];
                 const uint64_t seq = ++sent;
                 const size_t slot = (seq - 1) % nslots_;
-                // Fused: backpressure poll (slot reuse) + NT copy into the
-                // peer's slot in ONE kernel, then the 1-wg publish.
-                XferPoll bp = poll;
-                bp.mbox = seq > nslots_ ? MyAckFlag(peer, lane) : nullptr;
-                bp.target = seq - nslots_;
+                // Backpressure (slot reuse) as a SEPARATE 1-wg wait kernel:
+                // a poll fused into the wide copy kernel deadlocked two
+                // same-device ranks — each rank's full-device spinner
+                // starved the peer's consumer kernel of CUs (measured at
+                // 256 MiB world-2). The 1-wg wait always co-schedules.
+                if (seq > nslots_)
+                    LaunchWaitFlag(MyAckFlag(peer, lane), seq - nslots_,
+                                   abort_host_, status_host_, max_ticks_, s);
                 LaunchXferCopy(PeerSlot(peer, lane, slot), ptr(st->send) + off,
-                               n, bp.mbox ? &bp : nullptr, s);
+                               n, nullptr, s);
                 LaunchSetFlag(PeerInFlag(peer, lane), seq, s);
             }
             for (const RecvJob& rj : recvs) {
@@ -264,23 +262,22 @@ void P2pGroup::IssueSchedule(CommRequest* req, ChunkExec& ce, size_t lane,
                 const uint64_t seq = ++rcvd;
                 const size_t slot = (seq - 1) % nslots_;
                 const uint8_t* sl = MySlot(peer, lane, slot);
-                XferPoll wp = poll;
-                wp.mbox = MyInFlag(peer, lane);
-                wp.target = seq;
-                // Fused wait+consume where the dtype allows; otherwise a
-                // standalone wait kernel precedes the existing consumers.
+                // Arrival wait as a 1-wg kernel (same deadlock avoidance as
+                // the sender backpressure), then the wide consume kernel.
+                LaunchWaitFlag(MyInFlag(peer, lane), seq, abort_host_,
+                               status_host_, max_ticks_, s);
                 if (rj.fuse_into && !quant &&
                     LaunchXferReduce(ptr(st.local_dst) + off, sl, nullptr,
-                                     n / es, req->Dtype(), ce.sch.rop, &wp, s)) {
+                                     n / es, req->Dtype(), ce.sch.rop, nullptr,
+                                     s)) {
                 } else if (rj.fuse_out && !quant &&
                            LaunchXferReduce(ptr(st.recv) + off, sl,
                                             ptr(st.local_src) + off, n / es,
-                                            req->Dtype(), ce.sch.rop, &wp, s)) {
+                                            req->Dtype(), ce.sch.rop, nullptr,
+                                            s)) {
                 } else if (!rj.fuse_into && !rj.fuse_out) {
-                    LaunchXferCopy(ptr(st.recv) + off, sl, n, &wp, s);
+                    LaunchXferCopy(ptr(st.recv) + off, sl, n, nullptr, s);
                 } else {
-                    LaunchWaitFlag(wp.mbox, wp.target, abort_host_,
-                                   status_host_, max_ticks_, s);
                     if (rj.fuse_into) {
                         if (quant)
                             LaunchQuantAccum(ptr(st.local_dst) + off, sl,
