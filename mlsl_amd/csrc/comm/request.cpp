@@ -241,8 +241,15 @@ void CommRequest::BuildChunks() {
 
     AllReduceAlgo algo = cfg.allreduce_algo;
     if (algo == AllReduceAlgo::AUTO) {
+        // Measured crossovers on MI355X (profiles/multirank_transport_r2.md
+        // + sweep_mr_w4_rhd_prefix_r2.jsonl): at world 2 the ring's single
+        // exchange matches RHD at every size; at pow2 worlds >= 4 RHD's
+        // 2*log2(N) phases beat the ring's 2(N-1) until a few MiB, after
+        // which the bandwidth-optimal ring wins.
         const bool pow2 = (gs & (gs - 1)) == 0;
-        algo = (pow2 && MessageBytes() <= 65536) ? AllReduceAlgo::RHD : AllReduceAlgo::RING;
+        algo = (pow2 && gs >= 4 && MessageBytes() <= (4u << 20))
+                   ? AllReduceAlgo::RHD
+                   : AllReduceAlgo::RING;
     }
 
     chunks_.clear();

@@ -223,24 +223,6 @@ def gpu_hybrid_grid():
     mx.finalize()
 
 
-WORKERS = {
-    "gpu_collectives": gpu_collectives,
-    "gpu_allreduce_multislot": gpu_allreduce_multislot,
-    "gpu_quant_allreduce": gpu_quant_allreduce,
-    "gpu_hybrid_grid": gpu_hybrid_grid,
-}
-
-
-def main():
-    name = sys.argv[1]
-    WORKERS[name]()
-    print(f"OK {name} rank={os.environ.get('RANK')}")
-
-
-if __name__ == "__main__":
-    main()
-
-
 def gpu_stats_device_ns():
     """hipEvent-backed device Statistics: a committed 2-layer net at dp=2
     must attribute nonzero GPU comm nanoseconds to the gradient allreduce
@@ -260,4 +242,20 @@ def gpu_stats_device_ns():
     mx.finalize()
 
 
-WORKERS["gpu_stats_device_ns"] = gpu_stats_device_ns
+WORKERS = {
+    "gpu_collectives": gpu_collectives,
+    "gpu_allreduce_multislot": gpu_allreduce_multislot,
+    "gpu_quant_allreduce": gpu_quant_allreduce,
+    "gpu_hybrid_grid": gpu_hybrid_grid,
+    "gpu_stats_device_ns": gpu_stats_device_ns,
+}
+
+
+def main():
+    name = sys.argv[1]
+    WORKERS[name]()
+    print(f"OK {name} rank={os.environ.get('RANK')}")
+
+
+if __name__ == "__main__":
+    main()
