@@ -44,6 +44,7 @@ def main():
     if args.algo:
         cmd += ["--algo", args.algo]
     rank_log = os.environ.get("SWEEP_RANK_LOG")  # prefix: live per-rank logs
+    wrap0 = os.environ.get("SWEEP_WRAP0", "")    # e.g. "rocprofv3 --kernel-trace -d out --"
     procs = []
     logs = []
     for r in range(args.world):
@@ -53,14 +54,15 @@ def main():
                     "PYTHONPATH": REPO, "MLSL_TIMEOUT": "120",
                     "PYTHONUNBUFFERED": "1"})
         env.pop("MLSL_TRANSPORT", None)
+        rcmd = (wrap0.split() + cmd) if (r == 0 and wrap0) else cmd
         if rank_log:
             lf = open(f"{rank_log}.rank{r}.log", "w")
             logs.append(lf)
-            procs.append(subprocess.Popen(cmd, env=env, cwd=REPO, stdout=lf,
+            procs.append(subprocess.Popen(rcmd, env=env, cwd=REPO, stdout=lf,
                                           stderr=subprocess.STDOUT, text=True))
         else:
             procs.append(subprocess.Popen(
-                cmd, env=env, cwd=REPO, stdout=subprocess.PIPE,
+                rcmd, env=env, cwd=REPO, stdout=subprocess.PIPE,
                 stderr=subprocess.STDOUT, text=True))
     rc = 0
     out0 = ""
