@@ -132,3 +132,10 @@ def test_e2e_device_multirank(mp, du, user_buf, quant):
     for r, p in enumerate(procs):
         out, _ = p.communicate(timeout=240)
         assert p.returncode == 0 and "PASSED" in out, f"rank {r}: {out[-2500:]}"
+
+
+@requires_gpu
+def test_stress_multirank():
+    """30 s randomized mixed-collective soak at world 2 on one device."""
+    run_gpu_ranks("gpu_stress", 2, timeout=180,
+                  extra_env={"STRESS_SECONDS": "30"})

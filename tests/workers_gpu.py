@@ -242,12 +242,74 @@ def gpu_stats_device_ns():
     mx.finalize()
 
 
+def gpu_stress():
+    """Time-boxed randomized stress on the p2p transport: mixed ops, sizes
+    and algorithms back-to-back (soak analog for the n>1 device paths).
+    STRESS_SECONDS env bounds the duration (default 30)."""
+    import time
+    import random
+    mx, torch, rank, size = _init()
+    d = mx.Distribution(size, 1)
+    secs = float(os.environ.get("STRESS_SECONDS", "30"))
+    rng = random.Random(1234)  # same op sequence on every rank
+    t_end = time.time() + secs
+    iters = 0
+    while time.time() < t_end:
+        op = rng.choice(["ar", "ar_inplace", "rs", "ag", "a2a", "bcast", "bar"])
+        count = rng.choice([64, 4097, 1 << 16, 1 << 20])
+        if op == "ar" or op == "ar_inplace":
+            a = _arange(torch, count, rank)
+            out = a if op == "ar_inplace" else torch.empty_like(a)
+            mx.wait(d.all_reduce(a, out, count, op="sum", group="data"))
+            torch.cuda.synchronize()
+            want = size * torch.arange(count, dtype=torch.float32,
+                                       device="cuda") + size * (size - 1) / 2.0
+            assert torch.allclose(out, want), (iters, op, count)
+        elif op == "rs":
+            src = torch.cat([_arange(torch, count, rank) for _ in range(size)])
+            out = torch.empty(count, device="cuda")
+            mx.wait(d.reduce_scatter(src, out, count, op="sum", group="data"))
+            torch.cuda.synchronize()
+            want = size * torch.arange(count, dtype=torch.float32,
+                                       device="cuda") + size * (size - 1) / 2.0
+            assert torch.allclose(out, want), (iters, op, count)
+        elif op == "ag":
+            mine = _arange(torch, count, rank)
+            flat = torch.empty(size * count, device="cuda")
+            mx.wait(d.all_gather(mine, count, flat, group="data"))
+            torch.cuda.synchronize()
+            for r in range(size):
+                assert torch.allclose(flat[r * count:(r + 1) * count],
+                                      _arange(torch, count, r)), (iters, op, r)
+        elif op == "a2a":
+            src = torch.cat([torch.full((count,), float(rank * 10 + j),
+                                        device="cuda") for j in range(size)])
+            dst = torch.empty_like(src)
+            mx.wait(d.all_to_all(src, count, dst, group="data"))
+            torch.cuda.synchronize()
+            for j in range(size):
+                assert torch.all(dst[j * count:(j + 1) * count] ==
+                                 float(j * 10 + rank)), (iters, op, j)
+        elif op == "bcast":
+            b = _arange(torch, count, 3) if rank == 0                 else torch.zeros(count, device="cuda")
+            mx.wait(d.bcast(b, count, root=0, group="data"))
+            torch.cuda.synchronize()
+            assert torch.allclose(b, _arange(torch, count, 3)), (iters, op)
+        else:
+            d.barrier("data")
+        iters += 1
+    d.barrier("global")
+    print(f"stress iters={iters}")
+    mx.finalize()
+
+
 WORKERS = {
     "gpu_collectives": gpu_collectives,
     "gpu_allreduce_multislot": gpu_allreduce_multislot,
     "gpu_quant_allreduce": gpu_quant_allreduce,
     "gpu_hybrid_grid": gpu_hybrid_grid,
     "gpu_stats_device_ns": gpu_stats_device_ns,
+    "gpu_stress": gpu_stress,
 }
 
 
