@@ -816,6 +816,24 @@ bool DeviceAdvanceRequest(CommRequest* req, DeviceReqState& st) {
         // memcpy runs host-side once the completion events fire.
         st.unstage_pending = r_cls == BufClass::PAGEABLE;
         st.unstage_bytes = recv_b;
+        // BCAST is an in-place op: its schedule reads the payload from
+        // RECV space (no SEND-space step), so the staged input must land
+        // in stage_recv as well — without this the staged root broadcasts
+        // an uninitialized buffer (caught by the e2e param-consistency
+        // check on the device engine).
+        if (req->Spec().op == CollOp::BCAST && s_host) {
+            LaunchCopy(st.stage_recv, st.stage_send, recv_b, base_s);
+            if (gc.streams.size() > 1) {
+                // re-fan-out: channel streams must order after this copy
+                // too, not only after the send staging
+                if (!st.dep_event)
+                    HIP_CHECKD(hipEventCreateWithFlags(&st.dep_event,
+                                                       hipEventDisableTiming));
+                HIP_CHECKD(hipEventRecord(st.dep_event, gc.streams[0]));
+                for (size_t i = 1; i < gc.streams.size(); ++i)
+                    HIP_CHECKD(hipStreamWaitEvent(gc.streams[i], st.dep_event, 0));
+            }
+        }
     }
     req->SetDeviceBuffers(s_host ? static_cast<const uint8_t*>(st.stage_send) : nullptr,
                           r_host ? static_cast<uint8_t*>(st.stage_recv) : nullptr);
