@@ -156,7 +156,8 @@ bool CommRequest::UsesDeviceSchedule() const {
         group_->Size() > 1)
         return !Compressed();
     return (cfg.allreduce_algo == AllReduceAlgo::RING ||
-            cfg.allreduce_algo == AllReduceAlgo::RHD) &&
+            cfg.allreduce_algo == AllReduceAlgo::RHD ||
+            cfg.allreduce_algo == AllReduceAlgo::DIRECT) &&
            spec_.op == CollOp::ALLREDUCE && group_->Size() > 1 && !Compressed();
 }
 
@@ -247,9 +248,15 @@ void CommRequest::BuildChunks() {
         // 2*log2(N) phases beat the ring's 2(N-1) until a few MiB, after
         // which the bandwidth-optimal ring wins.
         const bool pow2 = (gs & (gs - 1)) == 0;
-        algo = (pow2 && gs >= 4 && MessageBytes() <= (4u << 20))
-                   ? AllReduceAlgo::RHD
-                   : AllReduceAlgo::RING;
+        // One-shot direct exchange for genuinely small messages (single
+        // round over all N-1 links; (N-1)x wire cost), then RHD for pow2
+        // worlds >= 4 up to a few MiB, then the bandwidth-optimal ring.
+        if (MessageBytes() <= 131072 && gs <= 8)
+            algo = AllReduceAlgo::DIRECT;
+        else if (pow2 && gs >= 4 && MessageBytes() <= (4u << 20))
+            algo = AllReduceAlgo::RHD;
+        else
+            algo = AllReduceAlgo::RING;
     }
 
     chunks_.clear();
@@ -275,6 +282,8 @@ void CommRequest::BuildChunks() {
                     const size_t nblocks = (cnt + blk - 1) / blk;
                     ce.sch = BuildAllReduceRingUnits(gr, gs, nblocks,
                                                      qparams_.WireBlockBytes(), blk);
+                } else if (algo == AllReduceAlgo::DIRECT) {
+                    ce.sch = BuildAllReduceDirect(gr, gs, cnt, dtype_, spec_.rop);
                 } else if (algo == AllReduceAlgo::RHD && (gs & (gs - 1)) == 0) {
                     ce.sch = BuildAllReduceRHD(gr, gs, cnt, dtype_, spec_.rop);
                 } else {

@@ -157,6 +157,45 @@ Schedule BuildAllReduceRingUnits(int rank, int size, size_t units, size_t unit_b
     return sch;
 }
 
+// Direct (one-shot) allreduce: every rank sends its whole buffer to every
+// peer in ONE exchange phase and reduces the N-1 arrivals locally. On a
+// fully-connected xGMI node this drives all N-1 point-to-point links
+// simultaneously with a single communication round — the latency-optimal
+// small-message algorithm for MI355X's topology (the ring needs 2(N-1)
+// rounds; RHD needs 2·log2 N). Costs (N-1)·S wire bytes per rank, so it
+// only pays below the bandwidth crossover. TMP carries a copy of the own
+// contribution (in-place safety: sends never read a buffer the reduces
+// mutate) plus one arrival slot per peer.
+Schedule BuildAllReduceDirect(int rank, int size, size_t count, DataType dt,
+                              ReduceOp op) {
+    const size_t es = DtypeSize(dt), B = count * es;
+    if (size == 1) return SelfOnly(B, dt, op);
+
+    Schedule sch;
+    sch.dtype = dt;
+    sch.rop = op;
+    const int N = size, r = rank;
+    sch.tmp_bytes = static_cast<size_t>(N) * B;  // [own][N-1 arrivals]
+    sch.AddStep(MakeCopy(0, Ref(Space::SEND, 0, B), Ref(Space::TMP, 0, B)));
+    sch.AddStep(MakeCopy(0, Ref(Space::SEND, 0, B), Ref(Space::RECV, 0, B)));
+    int slot = 1;
+    for (int j = 1; j < N; ++j, ++slot) {
+        const int peer = (r + j) % N;
+        Step st;
+        st.phase = 1;
+        st.send_peer = peer;
+        st.send = Ref(Space::TMP, 0, B);
+        st.recv_peer = peer;
+        st.recv = Ref(Space::TMP, static_cast<size_t>(slot) * B, B);
+        st.local = Step::LocalOp::REDUCE;
+        st.local_src = st.recv;
+        st.local_dst = Ref(Space::RECV, 0, B);
+        sch.AddStep(st);
+    }
+    sch.result = Ref(Space::RECV, 0, B);
+    return sch;
+}
+
 // Recursive halving reduce-scatter + recursive doubling all-gather
 // (Rabenseifner). Power-of-two groups only — the same restriction as the
 // reference's priority allreduce (eplib/cqueue.c:1903-1904). Latency-optimal:

@@ -89,6 +89,10 @@ int RingStrideForChannel(size_t channel, int size);
 Schedule BuildAllReduceRingUnits(int rank, int size, size_t units, size_t unit_bytes,
                                  size_t quant_block);
 Schedule BuildAllReduceRHD(int rank, int size, size_t count, DataType dt, ReduceOp op);
+// One-shot exchange + local reduce (latency-optimal on full-mesh xGMI;
+// (N-1)x wire cost — small messages only).
+Schedule BuildAllReduceDirect(int rank, int size, size_t count, DataType dt,
+                              ReduceOp op);
 Schedule BuildReduceScatter(int rank, int size, size_t recv_count, DataType dt, ReduceOp op);
 // Chunked variants (channel fan-out for the non-elementwise-splittable
 // ops; reference endpoint split, src/comm_ep.cpp:598-736): offsets are

@@ -41,6 +41,7 @@ Config Config::FromEnv() {
         if (!std::strcmp(e, "fused")) c.allreduce_algo = AllReduceAlgo::FUSED;
         else if (!std::strcmp(e, "ring")) c.allreduce_algo = AllReduceAlgo::RING;
         else if (!std::strcmp(e, "rhd")) c.allreduce_algo = AllReduceAlgo::RHD;
+        else if (!std::strcmp(e, "direct")) c.allreduce_algo = AllReduceAlgo::DIRECT;
         else c.allreduce_algo = AllReduceAlgo::AUTO;
     }
     c.msg_priority = EnvBool("MLSL_MSG_PRIORITY", false);

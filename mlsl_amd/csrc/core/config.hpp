@@ -19,6 +19,7 @@ enum class AllReduceAlgo : int {
     FUSED = 1,   // transport-native fused collective (RCCL ncclAllReduce)
     RING = 2,    // our chunked ring schedule
     RHD = 3,     // recursive-halving/doubling (Rabenseifner) schedule
+    DIRECT = 4,  // one-shot exchange + local reduce (small msgs, full mesh)
 };
 
 struct Config {

@@ -56,6 +56,26 @@ static void TestAllReduce(int N, size_t count, bool rhd) {
     }
 }
 
+static void TestAllReduceDirect(int N, size_t count) {
+    std::vector<Schedule> sch(N);
+    std::vector<std::vector<uint8_t>> sbuf(N), rbuf(N);
+    for (int r = 0; r < N; ++r) {
+        sch[r] = BuildAllReduceDirect(r, N, count, DataType::F32, ReduceOp::SUM);
+        FillF(sbuf[r], count, static_cast<float>(r));
+        rbuf[r].assign(count * 4, 0);
+    }
+    SimulateSchedules(sch, sbuf, rbuf);
+    for (int r = 0; r < N; ++r) {
+        auto f = F(rbuf[r]);
+        for (size_t i = 0; i < count; ++i) {
+            float want = static_cast<float>(N) * i + N * (N - 1) / 2.0f;
+            EXPECT(f[i] == want, "direct allreduce N=%d rank=%d i=%zu got %f want %f",
+                   N, r, i, f[i], want);
+            if (f[i] != want) return;
+        }
+    }
+}
+
 static void TestAllReduceRingStride(int N, size_t count, int stride) {
     std::vector<Schedule> sch(N);
     std::vector<std::vector<uint8_t>> sbuf(N), rbuf(N);
@@ -307,6 +327,7 @@ int main() {
             if ((N & (N - 1)) == 0) TestAllReduce(N, count, true);
         }
         TestAllReduceMax(N, 100);
+        TestAllReduceDirect(N, 73);
         // Rotated rings (multi-xGMI-link channel fan-out): every coprime
         // stride must produce the identical allreduce.
         for (int s = 1; s < N; ++s) {

@@ -139,3 +139,9 @@ def test_p2p_asymmetric(world):
 @pytest.mark.parametrize("world", [2, 4])
 def test_p2p_stress(world):
     run_ranks("p2p_stress", world, timeout=240)
+
+
+def test_direct_algo():
+    # one-shot exchange allreduce (full-mesh latency algorithm)
+    run_ranks("collectives_sweep", 3,
+              extra_env={"MLSL_ALLREDUCE_ALGO": "direct"})
