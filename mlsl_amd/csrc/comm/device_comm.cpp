@@ -608,6 +608,7 @@ void IssueSchedule(CommRequest* req, ChunkExec& ce, ncclComm_t comm, hipStream_t
         if (grouped) NCCL_CHECK(ncclGroupEnd());
         for (const auto& st : ce.sch.steps) {
             if (st.phase != phase || st.local == Step::LocalOp::NONE) continue;
+            if (st.local_dst.bytes == 0) continue;  // zero-length segment
             uint8_t* d = ptr(st.local_dst);
             uint8_t* src = ptr(st.local_src);
             if (st.local == Step::LocalOp::COPY) {

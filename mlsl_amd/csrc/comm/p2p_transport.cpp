@@ -306,6 +306,7 @@ void P2pGroup::IssueSchedule(CommRequest* req, ChunkExec& ce, size_t lane,
         // source/dest geometry) run after the phase's traffic, step order.
         for (const auto& st : ce.sch.steps) {
             if (st.phase != phase || st.local == Step::LocalOp::NONE) continue;
+            if (st.local_dst.bytes == 0) continue;  // zero-length segment
             const bool was_recv_step = st.recv_peer >= 0 && st.recv.bytes > 0;
             if (was_recv_step) {
                 const bool fuse_into = st.local == Step::LocalOp::REDUCE &&

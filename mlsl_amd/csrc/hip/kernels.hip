@@ -1008,6 +1008,7 @@ void LaunchDequantizeNT(const void* wire, void* out, size_t count,
 
 void LaunchQuantAccum(void* acc_wire, const void* wire, size_t count,
                       size_t block_elems, hipStream_t stream) {
+    if (count == 0) return;
     const size_t nblocks = (count + block_elems - 1) / block_elems;
     dim3 grid(static_cast<uint32_t>(std::min<size_t>((nblocks + 3) / 4, kMaxGrid)));
     if (block_elems == 256 && count % 256 == 0) {

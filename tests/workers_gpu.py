@@ -254,7 +254,16 @@ def gpu_stress():
     rng = random.Random(1234)  # same op sequence on every rank
     t_end = time.time() + secs
     iters = 0
-    while time.time() < t_end:
+    cont = torch.ones(1, device="cuda")
+    while True:
+        # Lockstep termination: ranks' clocks differ, and a rank that stops
+        # while its peer issues one more collective desyncs the transport —
+        # the continue flag is agreed via allreduce-min every iteration.
+        cont.fill_(1.0 if time.time() < t_end else 0.0)
+        mx.wait(d.all_reduce(cont, cont, 1, op="min", group="data"))
+        torch.cuda.synchronize()
+        if cont.item() == 0.0:
+            break
         op = rng.choice(["ar", "ar_inplace", "rs", "ag", "a2a", "bcast", "bar"])
         count = rng.choice([64, 4097, 1 << 16, 1 << 20])
         if op == "ar" or op == "ar_inplace":
