@@ -1,5 +1,6 @@
 #include "p2p_transport.hpp"
 
+#include <cstdlib>
 #include <cstring>
 #include <mutex>
 
@@ -85,11 +86,28 @@ std::unique_ptr<P2pGroup> P2pGroup::Create(ProcessGroup* g, size_t nlanes,
                                            hipIpcMemLazyEnablePeerAccess));
             pg->peer_base_[i] = static_cast<uint8_t*>(p);
         }
+        const char* hw_env = std::getenv("MLSL_P2P_HW_WRITE");
+        const bool hw_allowed = !hw_env || std::atoi(hw_env) != 0;
+        // Probe hipStreamWriteValue64 on window memory: if the runtime
+        // accepts it, publishes/acks ride the queue as packets (no kernel
+        // launch). The value written here is 0 == the initial state.
+        if (hw_allowed) {
+            hipStream_t ps;
+            if (hipStreamCreateWithFlags(&ps, hipStreamNonBlocking) == hipSuccess) {
+                hipError_t we = hipStreamWriteValue64(
+                    ps, pg->MyInFlag(pg->my_idx_, 0), 0ull, 0);
+                if (we == hipSuccess && hipStreamSynchronize(ps) == hipSuccess)
+                    pg->hw_write_ = true;
+                else
+                    (void)hipGetLastError();
+                (void)hipStreamDestroy(ps);
+            }
+        }
         MLSL_LOG(DEBUG,
                  "p2p group ready: size=%d lanes=%zu slots=%zu slot_bytes=%zu "
-                 "window=%zu MiB",
+                 "window=%zu MiB hw_write=%d",
                  pg->gsize_, nlanes, nslots, slot_bytes,
-                 pg->win_bytes_ >> 20);
+                 pg->win_bytes_ >> 20, (int)pg->hw_write_);
     }
     return pg;
 }
@@ -148,6 +166,15 @@ void* P2pGroup::PeerAckFlag(int peer, size_t lane) const {
     return peer_base_[peer] +
            (static_cast<size_t>(my_idx_) * nlanes_ + lane) * 2 * kFlagStride +
            kFlagStride;
+}
+
+void P2pGroup::PublishFlag(void* mbox, uint64_t val, hipStream_t s) {
+    if (hw_write_) {
+        if (hipStreamWriteValue64(s, mbox, val, 0) == hipSuccess) return;
+        (void)hipGetLastError();
+        hw_write_ = false;  // runtime changed its mind: kernels from now on
+    }
+    LaunchSetFlag(mbox, val, s);
 }
 
 // --- transport ops ---------------------------------------------------------
@@ -249,7 +276,7 @@ void P2pGroup::IssueSchedule(CommRequest* req, ChunkExec& ce, size_t lane,
                                    abort_host_, status_host_, max_ticks_, s);
                 LaunchXferCopy(PeerSlot(peer, lane, slot), ptr(st->send) + off,
                                n, nullptr, s);
-                LaunchSetFlag(PeerInFlag(peer, lane), seq, s);
+                PublishFlag(PeerInFlag(peer, lane), seq, s);
             }
             for (const RecvJob& rj : recvs) {
                 const Step& st = *rj.st;
@@ -299,7 +326,7 @@ void P2pGroup::IssueSchedule(CommRequest* req, ChunkExec& ce, size_t lane,
                         }
                     }
                 }
-                LaunchSetFlag(PeerAckFlag(peer, lane), seq, s);
+                PublishFlag(PeerAckFlag(peer, lane), seq, s);
             }
         }
         // Unfused local ops (pure copy steps, or reduce with unrelated

@@ -99,6 +99,12 @@ class P2pGroup {
     uint32_t* abort_host_ = nullptr;    // pinned, read by wait kernels
     uint32_t* status_host_ = nullptr;   // pinned, set by aborted wait kernels
     uint64_t max_ticks_ = 0;            // wall-clock bound for waits
+    // hipStreamWriteValue64 works on this pool's window memory (probed at
+    // Create): flag publishes/acks become queue packets instead of 1-wg
+    // kernel launches. Waits stay as kernels — the HW wait packet cannot
+    // be aborted, so a dead peer would wedge the queue forever.
+    bool hw_write_ = false;
+    void PublishFlag(void* mbox, uint64_t val, hipStream_t s);
     friend class P2pGroupTestPeek;
 };
 
