@@ -251,7 +251,7 @@ void CommRequest::BuildChunks() {
         // One-shot direct exchange for genuinely small messages (single
         // round over all N-1 links; (N-1)x wire cost), then RHD for pow2
         // worlds >= 4 up to a few MiB, then the bandwidth-optimal ring.
-        if (MessageBytes() <= 131072 && gs <= 8)
+        if (MessageBytes() <= (1u << 20) && gs <= 8)
             algo = AllReduceAlgo::DIRECT;
         else if (pow2 && gs >= 4 && MessageBytes() <= (4u << 20))
             algo = AllReduceAlgo::RHD;
