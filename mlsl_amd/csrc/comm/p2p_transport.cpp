@@ -86,8 +86,12 @@ std::unique_ptr<P2pGroup> P2pGroup::Create(ProcessGroup* g, size_t nlanes,
                                            hipIpcMemLazyEnablePeerAccess));
             pg->peer_base_[i] = static_cast<uint8_t*>(p);
         }
+        // hipStreamWriteValue64 publishes measured SLOWER than the 1-wg
+        // SetFlag kernel in a same-box A/B (4 KiB: 45 vs 39 us; 256 MiB
+        // ring: 310 vs 353 GB/s — the write packet stalls the queue), so
+        // the HW path is opt-in: MLSL_P2P_HW_WRITE=1.
         const char* hw_env = std::getenv("MLSL_P2P_HW_WRITE");
-        const bool hw_allowed = !hw_env || std::atoi(hw_env) != 0;
+        const bool hw_allowed = hw_env && std::atoi(hw_env) != 0;
         // Probe hipStreamWriteValue64 on window memory: if the runtime
         // accepts it, publishes/acks ride the queue as packets (no kernel
         // launch). The value written here is 0 == the initial state.
