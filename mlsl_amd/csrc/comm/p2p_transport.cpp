@@ -27,6 +27,10 @@ namespace {
 // never share a cache line (remote stores would otherwise ping-pong lines
 // between GPUs over xGMI).
 constexpr size_t kFlagStride = 128;
+// Sub-messages at or below this ride the fully-fused bounded-grid kernels
+// (one launch per side); larger ones use wide kernels with separate 1-wg
+// waits so the payload kernels never spin.
+constexpr size_t kFusedBytes = 1u << 20;
 std::mutex g_issue_mu;  // enqueue-order == counter-order per group
 
 struct WireHandle {
@@ -74,6 +78,11 @@ std::unique_ptr<P2pGroup> P2pGroup::Create(ProcessGroup* g, size_t nlanes,
         pg->peer_base_.assign(N, nullptr);
         pg->sent_.assign(N * nlanes, 0);
         pg->rcvd_.assign(N * nlanes, 0);
+        pg->fused_sent_.assign(N * nlanes, 0);
+        pg->fused_rcvd_.assign(N * nlanes, 0);
+        HIP_CHECKP(hipMalloc(reinterpret_cast<void**>(&pg->ctr_dev_),
+                             2 * N * nlanes * sizeof(uint64_t)));
+        HIP_CHECKP(hipMemset(pg->ctr_dev_, 0, 2 * N * nlanes * sizeof(uint64_t)));
         for (int i = 0; i < pg->gsize_; ++i) {
             if (i == pg->my_idx_) {
                 pg->peer_base_[i] = pg->my_base_;
@@ -125,6 +134,7 @@ P2pGroup::~P2pGroup() {
         if (peer_base_[i] && i != my_idx_)
             (void)hipIpcCloseMemHandle(peer_base_[i]);
     (void)hipFree(my_base_);
+    if (ctr_dev_) (void)hipFree(ctr_dev_);
     if (abort_host_) (void)hipHostFree(abort_host_);
 }
 
@@ -270,11 +280,30 @@ void P2pGroup::IssueSchedule(CommRequest* req, ChunkExec& ce, size_t lane,
                 uint64_t& sent = sent_[static_cast<size_t>(peer) * nlanes_ + lane];
                 const uint64_t seq = ++sent;
                 const size_t slot = (seq - 1) % nslots_;
+                if (n <= kFusedBytes) {
+                    // Small sub-message: ONE bounded-grid kernel does the
+                    // backpressure poll, the slot copy and the publish
+                    // (32 workgroups can never starve the peer — the
+                    // full-device fused variant deadlocked; see below).
+                    const size_t e = static_cast<size_t>(peer) * nlanes_ + lane;
+                    XferPoll bp{};
+                    bp.mbox = seq > nslots_ ? MyAckFlag(peer, lane) : nullptr;
+                    bp.target = seq - nslots_;
+                    bp.abort_word = abort_host_;
+                    bp.status = status_host_;
+                    bp.max_ticks = max_ticks_;
+                    LaunchXferSendFused(PeerSlot(peer, lane, slot),
+                                        ptr(st->send) + off, n,
+                                        bp.mbox ? &bp : nullptr,
+                                        ctr_dev_ + e, ++fused_sent_[e],
+                                        PeerInFlag(peer, lane), seq, s);
+                    continue;
+                }
                 // Backpressure (slot reuse) as a SEPARATE 1-wg wait kernel:
-                // a poll fused into the wide copy kernel deadlocked two
-                // same-device ranks — each rank's full-device spinner
-                // starved the peer's consumer kernel of CUs (measured at
-                // 256 MiB world-2). The 1-wg wait always co-schedules.
+                // a poll fused into a FULL-DEVICE copy kernel deadlocked two
+                // same-device ranks — each rank's spinner starved the
+                // peer's consumer kernel of CUs (measured at 256 MiB
+                // world-2). The 1-wg wait always co-schedules.
                 if (seq > nslots_)
                     LaunchWaitFlag(MyAckFlag(peer, lane), seq - nslots_,
                                    abort_host_, status_host_, max_ticks_, s);
@@ -293,6 +322,29 @@ void P2pGroup::IssueSchedule(CommRequest* req, ChunkExec& ce, size_t lane,
                 const uint64_t seq = ++rcvd;
                 const size_t slot = (seq - 1) % nslots_;
                 const uint8_t* sl = MySlot(peer, lane, slot);
+                XferPoll wp{};
+                wp.mbox = MyInFlag(peer, lane);
+                wp.target = seq;
+                wp.abort_word = abort_host_;
+                wp.status = status_host_;
+                wp.max_ticks = max_ticks_;
+                if (n <= kFusedBytes && !quant) {
+                    const size_t e = static_cast<size_t>(peer) * nlanes_ + lane;
+                    uint64_t* rctr = ctr_dev_ +
+                                     static_cast<size_t>(gsize_) * nlanes_ + e;
+                    const int mode = rj.fuse_into ? 1 : (rj.fuse_out ? 2 : 0);
+                    uint8_t* d = mode == 1 ? ptr(st.local_dst) + off
+                                           : ptr(st.recv) + off;
+                    const uint8_t* o =
+                        mode == 2 ? ptr(st.local_src) + off : nullptr;
+                    if (LaunchXferRecvFused(d, sl, o,
+                                            mode == 0 ? n : n / es,
+                                            req->Dtype(), ce.sch.rop, mode,
+                                            &wp, rctr, ++fused_rcvd_[e],
+                                            PeerAckFlag(peer, lane), seq, s))
+                        continue;
+                    --fused_rcvd_[e];  // dtype not covered: unfused path
+                }
                 // Arrival wait as a 1-wg kernel (same deadlock avoidance as
                 // the sender backpressure), then the wide consume kernel.
                 LaunchWaitFlag(MyInFlag(peer, lane), seq, abort_host_,

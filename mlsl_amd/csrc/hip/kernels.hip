@@ -1453,6 +1453,196 @@ bool LaunchXferReduce(void* dst, const void* slot, const void* other, size_t n,
     return true;
 }
 
+// --- fully-fused small-message transport kernels ---
+// One kernel per side per sub-message: poll prologue + payload + grid-
+// completion counter + flag publish. The grid is FIXED at kFusedGrid
+// workgroups so a spinning kernel can never starve the peer's consumer of
+// CUs (the full-device fused-poll variant deadlocked two ranks; 2-4 ranks
+// x a few 32-wg spinners always co-schedule on 256 CUs). Completion is
+// detected with a monotonic per-edge counter: every workgroup does a
+// system-scope acq_rel fetch-add; the one that observes target-1 publishes
+// the flag with a system release store — which also orders every other
+// workgroup's payload writes (their adds released them).
+
+namespace {
+
+constexpr unsigned kFusedGrid = 32;
+
+__device__ __forceinline__ void FusedFinish(unsigned long long* ctr,
+                                            unsigned long long target,
+                                            unsigned long long* mbox,
+                                            unsigned long long val) {
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        const unsigned long long prev = __hip_atomic_fetch_add(
+            ctr, 1ull, __ATOMIC_ACQ_REL, __HIP_MEMORY_SCOPE_SYSTEM);
+        if (prev == target - 1)
+            __hip_atomic_store(mbox, val, __ATOMIC_RELEASE,
+                               __HIP_MEMORY_SCOPE_SYSTEM);
+    }
+}
+
+// sender: [backpressure poll] + byte copy into the peer slot + publish
+__global__ void XferSendFusedKernel(uint8_t* __restrict__ slot,
+                                    const uint8_t* __restrict__ src,
+                                    size_t bytes, PollArgs bp,
+                                    unsigned long long* ctr,
+                                    unsigned long long ctr_target,
+                                    unsigned long long* in_mbox,
+                                    unsigned long long seq) {
+    if (bp.mbox &&
+        !PollGeq(bp.mbox, bp.target, bp.abort_word, bp.status, bp.max_ticks))
+        return;
+    const size_t tid = blockIdx.x * blockDim.x + threadIdx.x;
+    const size_t stride = gridDim.x * blockDim.x;
+    const bool al = ((reinterpret_cast<uintptr_t>(slot) |
+                      reinterpret_cast<uintptr_t>(src)) & 15) == 0;
+    if (al) {
+        const size_t n16 = bytes / 16;
+        const uint4_ev* s = reinterpret_cast<const uint4_ev*>(src);
+        uint4_ev* d = reinterpret_cast<uint4_ev*>(slot);
+        for (size_t i = tid; i < n16; i += stride)
+            __builtin_nontemporal_store(__builtin_nontemporal_load(s + i), d + i);
+        for (size_t j = n16 * 16 + tid; j < bytes; j += stride) slot[j] = src[j];
+    } else {
+        for (size_t j = tid; j < bytes; j += stride) slot[j] = src[j];
+    }
+    FusedFinish(ctr, ctr_target, in_mbox, seq);
+}
+
+// receiver: arrival poll + consume (copy / reduce / reduce-out) + ack
+enum class FusedConsume : int { COPY = 0, REDUCE = 1, REDUCE_OUT = 2 };
+
+template <typename T, ReduceOp OP, FusedConsume MODE>
+__global__ void XferRecvFusedKernel(T* __restrict__ dst,
+                                    const T* __restrict__ slot,
+                                    const T* __restrict__ other, size_t n,
+                                    PollArgs wp, unsigned long long* ctr,
+                                    unsigned long long ctr_target,
+                                    unsigned long long* ack_mbox,
+                                    unsigned long long seq) {
+    if (!PollGeq(wp.mbox, wp.target, wp.abort_word, wp.status, wp.max_ticks))
+        return;
+    const size_t tid = blockIdx.x * blockDim.x + threadIdx.x;
+    const size_t stride = gridDim.x * blockDim.x;
+    for (size_t j = tid; j < n; j += stride) {
+        if constexpr (MODE == FusedConsume::COPY) {
+            dst[j] = slot[j];
+        } else if constexpr (MODE == FusedConsume::REDUCE) {
+            if constexpr (sizeof(T) == 2) {
+                // bf16 via float math (T = unsigned short container)
+                const float a = __bfloat162float(
+                    *reinterpret_cast<const __hip_bfloat16*>(&dst[j]));
+                const float b = __bfloat162float(
+                    *reinterpret_cast<const __hip_bfloat16*>(&slot[j]));
+                const __hip_bfloat16 r = __float2bfloat16(Apply<float, OP>(a, b));
+                dst[j] = *reinterpret_cast<const T*>(&r);
+            } else {
+                dst[j] = Apply<T, OP>(dst[j], slot[j]);
+            }
+        } else {
+            if constexpr (sizeof(T) == 2) {
+                const float a = __bfloat162float(
+                    *reinterpret_cast<const __hip_bfloat16*>(&other[j]));
+                const float b = __bfloat162float(
+                    *reinterpret_cast<const __hip_bfloat16*>(&slot[j]));
+                const __hip_bfloat16 r = __float2bfloat16(Apply<float, OP>(a, b));
+                dst[j] = *reinterpret_cast<const T*>(&r);
+            } else {
+                dst[j] = Apply<T, OP>(other[j], slot[j]);
+            }
+        }
+    }
+    FusedFinish(ctr, ctr_target, ack_mbox, seq);
+}
+
+}  // namespace
+
+void LaunchXferSendFused(void* slot, const void* src, size_t bytes,
+                         const XferPoll* bp, void* ctr, uint64_t ctr_target,
+                         void* in_mbox, uint64_t seq, hipStream_t stream) {
+    PollArgs pa{};
+    if (bp) {
+        pa.mbox = static_cast<const unsigned long long*>(bp->mbox);
+        pa.target = bp->target;
+        pa.abort_word = static_cast<const unsigned int*>(bp->abort_word);
+        pa.status = static_cast<unsigned int*>(bp->status);
+        pa.max_ticks = bp->max_ticks;
+    }
+    hipLaunchKernelGGL(XferSendFusedKernel, dim3(kFusedGrid), dim3(kBlock), 0,
+                       stream, static_cast<uint8_t*>(slot),
+                       static_cast<const uint8_t*>(src), bytes, pa,
+                       static_cast<unsigned long long*>(ctr),
+                       ctr_target * kFusedGrid,
+                       static_cast<unsigned long long*>(in_mbox), seq);
+    HIP_CHECK(hipGetLastError());
+}
+
+bool LaunchXferRecvFused(void* dst, const void* slot, const void* other,
+                         size_t n, DataType dt, ReduceOp op, int mode,
+                         const XferPoll* wp, void* ctr, uint64_t ctr_target,
+                         void* ack_mbox, uint64_t seq, hipStream_t stream) {
+    PollArgs pa{};
+    pa.mbox = static_cast<const unsigned long long*>(wp->mbox);
+    pa.target = wp->target;
+    pa.abort_word = static_cast<const unsigned int*>(wp->abort_word);
+    pa.status = static_cast<unsigned int*>(wp->status);
+    pa.max_ticks = wp->max_ticks;
+    const unsigned long long tgt = ctr_target * kFusedGrid;
+    auto* c = static_cast<unsigned long long*>(ctr);
+    auto* m = static_cast<unsigned long long*>(ack_mbox);
+#define FUSED_LAUNCH(T, OPV)                                                  \
+    do {                                                                      \
+        if (mode == 0)                                                        \
+            hipLaunchKernelGGL(                                               \
+                (XferRecvFusedKernel<T, OPV, FusedConsume::COPY>),            \
+                dim3(kFusedGrid), dim3(kBlock), 0, stream, (T*)dst,           \
+                (const T*)slot, (const T*)other, n, pa, c, tgt, m, seq);      \
+        else if (mode == 1)                                                   \
+            hipLaunchKernelGGL(                                               \
+                (XferRecvFusedKernel<T, OPV, FusedConsume::REDUCE>),          \
+                dim3(kFusedGrid), dim3(kBlock), 0, stream, (T*)dst,           \
+                (const T*)slot, (const T*)other, n, pa, c, tgt, m, seq);      \
+        else                                                                  \
+            hipLaunchKernelGGL(                                               \
+                (XferRecvFusedKernel<T, OPV, FusedConsume::REDUCE_OUT>),      \
+                dim3(kFusedGrid), dim3(kBlock), 0, stream, (T*)dst,           \
+                (const T*)slot, (const T*)other, n, pa, c, tgt, m, seq);      \
+    } while (0)
+    if (mode == 0) {
+        FUSED_LAUNCH(uint8_t, ReduceOp::SUM);  // byte copy, dtype-agnostic
+        HIP_CHECK(hipGetLastError());
+        return true;
+    }
+    switch (dt) {
+        case DataType::F32:
+            switch (op) {
+                case ReduceOp::SUM: FUSED_LAUNCH(float, ReduceOp::SUM); break;
+                case ReduceOp::MIN: FUSED_LAUNCH(float, ReduceOp::MIN); break;
+                case ReduceOp::MAX: FUSED_LAUNCH(float, ReduceOp::MAX); break;
+            }
+            break;
+        case DataType::BF16:
+            switch (op) {
+                case ReduceOp::SUM:
+                    FUSED_LAUNCH(unsigned short, ReduceOp::SUM);
+                    break;
+                case ReduceOp::MIN:
+                    FUSED_LAUNCH(unsigned short, ReduceOp::MIN);
+                    break;
+                case ReduceOp::MAX:
+                    FUSED_LAUNCH(unsigned short, ReduceOp::MAX);
+                    break;
+            }
+            break;
+        default:
+            return false;
+    }
+#undef FUSED_LAUNCH
+    HIP_CHECK(hipGetLastError());
+    return true;
+}
+
 void LaunchWaitFlag(const void* mbox, uint64_t target, const void* abort_word,
                     void* status, uint64_t max_ticks, hipStream_t stream) {
     hipLaunchKernelGGL(WaitFlagKernel, dim3(1), dim3(1), 0, stream,

@@ -89,6 +89,21 @@ bool LaunchXferReduce(void* dst, const void* slot, const void* other, size_t n,
                       DataType dt, ReduceOp op, const XferPoll* poll,
                       hipStream_t stream);
 
+// Fully-fused small-message transport kernels: poll + payload + grid-
+// completion counter (monotonic, fetch_add per workgroup with FIXED grid)
+// + flag publish, in ONE launch with a bounded grid (spinners can never
+// starve the peer's kernels). ctr_target is the host-side per-edge count
+// of fused ops (the launcher scales it by the internal grid size).
+// Recv `mode`: 0 copy (n = BYTES), 1 reduce-into, 2 reduce-out (n =
+// elements); returns false for uncovered dtypes.
+void LaunchXferSendFused(void* slot, const void* src, size_t bytes,
+                         const XferPoll* bp, void* ctr, uint64_t ctr_target,
+                         void* in_mbox, uint64_t seq, hipStream_t stream);
+bool LaunchXferRecvFused(void* dst, const void* slot, const void* other,
+                         size_t n, DataType dt, ReduceOp op, int mode,
+                         const XferPoll* wp, void* ctr, uint64_t ctr_target,
+                         void* ack_mbox, uint64_t seq, hipStream_t stream);
+
 // --- IPC p2p transport flag primitives ---
 // Stream-blocking wait until *mbox >= target (system-scope acquire), with a
 // host abort word and a wall-clock bound (ticks of the 100 MHz constant

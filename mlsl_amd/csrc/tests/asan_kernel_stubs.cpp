@@ -22,5 +22,7 @@ void LaunchUnpack(const void*, void*, const PackBlockDesc&, DataType, hipStream_
 void LaunchWaitFlag(const void*, uint64_t, const void*, void*, uint64_t, hipStream_t) STUB()
 void LaunchSetFlag(void*, uint64_t, hipStream_t) STUB()
 void LaunchXferCopy(void*, const void*, size_t, const XferPoll*, hipStream_t) STUB()
+void LaunchXferSendFused(void*, const void*, size_t, const XferPoll*, void*, uint64_t, void*, uint64_t, hipStream_t) STUB()
+bool LaunchXferRecvFused(void*, const void*, const void*, size_t, DataType, ReduceOp, int, const XferPoll*, void*, uint64_t, void*, uint64_t, hipStream_t) STUB()
 bool LaunchXferReduce(void*, const void*, const void*, size_t, DataType, ReduceOp, const XferPoll*, hipStream_t) STUB()
 }  // namespace mlsl
