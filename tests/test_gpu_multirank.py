@@ -139,3 +139,9 @@ def test_stress_multirank():
     """30 s randomized mixed-collective soak at world 2 on one device."""
     run_gpu_ranks("gpu_stress", 2, timeout=180,
                   extra_env={"STRESS_SECONDS": "30"})
+
+
+@requires_gpu
+def test_ddp_multirank():
+    """Bucketed DDP (autograd hooks + non-blocking allreduce) at world 2."""
+    run_gpu_ranks("gpu_ddp", 2)

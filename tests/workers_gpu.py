@@ -312,8 +312,43 @@ def gpu_stress():
     mx.finalize()
 
 
+def gpu_ddp():
+    """DistributedData (bucketed DDP with autograd hooks) at world 2 on the
+    device engine: gradients after finish_gradients() must equal the
+    average of the per-rank local gradients, and match across ranks."""
+    mx, torch, rank, size = _init()
+    from mlsl_amd.parallel import DistributedData
+    torch.manual_seed(7)  # same init on every rank (bcast also enforces it)
+    model = torch.nn.Sequential(
+        torch.nn.Linear(64, 128), torch.nn.ReLU(),
+        torch.nn.Linear(128, 32)).cuda()
+    x = torch.full((16, 64), 1.0 + rank, device="cuda")
+
+    # expected: average over ranks of the local gradient
+    import copy
+    ref = copy.deepcopy(model)
+    ref(x).sum().backward()
+    local_grads = [p.grad.detach().clone() for p in ref.parameters()]
+    d = mx.Distribution(size, 1)
+    for g in local_grads:
+        flat = g.reshape(-1).contiguous()
+        mx.wait(d.all_reduce(flat, flat, flat.numel(), op="sum", group="data"))
+        torch.cuda.synchronize()
+        g.copy_((flat / size).view_as(g))
+
+    dd = DistributedData(model, dist=d)
+    model(x).sum().backward()
+    dd.finish_gradients()
+    torch.cuda.synchronize()
+    for p, want in zip(model.parameters(), local_grads):
+        assert torch.allclose(p.grad, want, rtol=1e-5, atol=1e-5), \
+            (p.shape, (p.grad - want).abs().max().item())
+    mx.finalize()
+
+
 WORKERS = {
     "gpu_collectives": gpu_collectives,
+    "gpu_ddp": gpu_ddp,
     "gpu_allreduce_multislot": gpu_allreduce_multislot,
     "gpu_quant_allreduce": gpu_quant_allreduce,
     "gpu_hybrid_grid": gpu_hybrid_grid,
