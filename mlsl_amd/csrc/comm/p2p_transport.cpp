@@ -29,8 +29,17 @@ namespace {
 constexpr size_t kFlagStride = 128;
 // Sub-messages at or below this ride the fully-fused bounded-grid kernels
 // (one launch per side); larger ones use wide kernels with separate 1-wg
-// waits so the payload kernels never spin.
-constexpr size_t kFusedBytes = 1u << 20;
+// waits so the payload kernels never spin. Tunable: MLSL_P2P_FUSED_MAX_KB.
+size_t FusedMaxBytes() {
+    static const size_t v = [] {
+        if (const char* e = std::getenv("MLSL_P2P_FUSED_MAX_KB")) {
+            long long kb = std::atoll(e);
+            if (kb >= 0) return static_cast<size_t>(kb) * 1024;
+        }
+        return static_cast<size_t>(1u << 20);
+    }();
+    return v;
+}
 std::mutex g_issue_mu;  // enqueue-order == counter-order per group
 
 struct WireHandle {
@@ -280,7 +289,7 @@ void P2pGroup::IssueSchedule(CommRequest* req, ChunkExec& ce, size_t lane,
                 uint64_t& sent = sent_[static_cast<size_t>(peer) * nlanes_ + lane];
                 const uint64_t seq = ++sent;
                 const size_t slot = (seq - 1) % nslots_;
-                if (n <= kFusedBytes) {
+                if (n <= FusedMaxBytes()) {
                     // Small sub-message: ONE bounded-grid kernel does the
                     // backpressure poll, the slot copy and the publish
                     // (32 workgroups can never starve the peer — the
@@ -328,7 +337,7 @@ void P2pGroup::IssueSchedule(CommRequest* req, ChunkExec& ce, size_t lane,
                 wp.abort_word = abort_host_;
                 wp.status = status_host_;
                 wp.max_ticks = max_ticks_;
-                if (n <= kFusedBytes && !quant) {
+                if (n <= FusedMaxBytes() && !quant) {
                     const size_t e = static_cast<size_t>(peer) * nlanes_ + lane;
                     uint64_t* rctr = ctr_dev_ +
                                      static_cast<size_t>(gsize_) * nlanes_ + e;
