@@ -53,3 +53,13 @@ def test_e2e_quant():
     # int8-compressed gradient allreduce: relative-error check
     _run_e2e(4, 1, 0, 0, quant=1)
     _run_e2e(2, 1, 0, 1, quant=1)
+
+
+def test_e2e_channels_and_priority():
+    # config interactions: channel fan-out and the priority lane through
+    # the full planner matrix
+    _run_e2e(4, 2, 1, 0, extra_env={"MLSL_NUM_CHANNELS": "2",
+                                    "MLSL_LARGE_MSG_SIZE_MB": "0"})
+    _run_e2e(4, 1, 0, 1, extra_env={"MLSL_MSG_PRIORITY": "1",
+                                    "MLSL_MSG_PRIORITY_THRESHOLD": "64"})
+    _run_e2e(4, 1, 1, 0, extra_env={"MLSL_ALLREDUCE_ALGO": "direct"})

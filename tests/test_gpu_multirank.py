@@ -145,3 +145,13 @@ def test_stress_multirank():
 def test_ddp_multirank():
     """Bucketed DDP (autograd hooks + non-blocking allreduce) at world 2."""
     run_gpu_ranks("gpu_ddp", 2)
+
+
+@requires_gpu
+def test_zero1_multirank():
+    run_gpu_ranks("gpu_zero1", 2)
+
+
+@requires_gpu
+def test_seqpar_multirank():
+    run_gpu_ranks("gpu_seqpar", 2)

@@ -346,9 +346,62 @@ def gpu_ddp():
     mx.finalize()
 
 
+def gpu_zero1():
+    """ShardedOptimizer (ZeRO-1) at world 2 on the device engine:
+    reduce-scatter of raw grads + owned-shard SGD + allgather of updated
+    params must match a plain full-replica SGD step."""
+    mx, torch, rank, size = _init()
+    from mlsl_amd.parallel import ShardedOptimizer
+    torch.manual_seed(5)
+    model = torch.nn.Linear(96, 96).cuda()
+    # plain reference: full allreduce-averaged grads + SGD on a clone
+    import copy
+    ref = copy.deepcopy(model)
+    x = torch.full((8, 96), 1.0 + rank, device="cuda")
+    ref(x).sum().backward()
+    d = mx.Distribution(size, 1)
+    for p in ref.parameters():
+        flat = p.grad.detach().reshape(-1).contiguous()
+        mx.wait(d.all_reduce(flat, flat, flat.numel(), op="sum", group="data"))
+        torch.cuda.synchronize()
+        p.grad.copy_((flat / size).view_as(p.grad))
+    torch.optim.SGD(ref.parameters(), lr=0.1).step()
+
+    opt = ShardedOptimizer(model.parameters(), torch.optim.SGD, dist=d,
+                           reduce="rs", lr=0.1)
+    model(x).sum().backward()
+    opt.step()
+    torch.cuda.synchronize()
+    for p, w in zip(model.parameters(), ref.parameters()):
+        assert torch.allclose(p.data, w.data, rtol=1e-5, atol=1e-6), \
+            (p.shape, (p.data - w.data).abs().max().item())
+    mx.finalize()
+
+
+def gpu_seqpar():
+    """Ulysses-style sequence<->head resharding (alltoall) at world 2 on
+    the device engine: seq_to_head . head_to_seq == identity."""
+    mx, torch, rank, size = _init()
+    from mlsl_amd.parallel import seqpar
+    d = mx.Distribution(1, size)
+    B, S, H = 2, 8 * size, 16 * size
+    torch.manual_seed(20 + rank)
+    x = torch.randn(B, S // size, H, device="cuda")
+    y = seqpar.seq_to_head(d, x, group="model")
+    torch.cuda.synchronize()
+    assert y.shape == (B, S, H // size), y.shape
+    z = seqpar.head_to_seq(d, y, group="model")
+    torch.cuda.synchronize()
+    assert z.shape == x.shape, z.shape
+    assert torch.allclose(z, x), (z - x).abs().max().item()
+    mx.finalize()
+
+
 WORKERS = {
     "gpu_collectives": gpu_collectives,
     "gpu_ddp": gpu_ddp,
+    "gpu_zero1": gpu_zero1,
+    "gpu_seqpar": gpu_seqpar,
     "gpu_allreduce_multislot": gpu_allreduce_multislot,
     "gpu_quant_allreduce": gpu_quant_allreduce,
     "gpu_hybrid_grid": gpu_hybrid_grid,
