@@ -89,9 +89,11 @@ std::unique_ptr<P2pGroup> P2pGroup::Create(ProcessGroup* g, size_t nlanes,
         pg->rcvd_.assign(N * nlanes, 0);
         pg->fused_sent_.assign(N * nlanes, 0);
         pg->fused_rcvd_.assign(N * nlanes, 0);
+        pg->fanin_adds_.assign(nlanes, 0);
+        const size_t nctr = 2 * N * nlanes + nlanes;
         HIP_CHECKP(hipMalloc(reinterpret_cast<void**>(&pg->ctr_dev_),
-                             2 * N * nlanes * sizeof(uint64_t)));
-        HIP_CHECKP(hipMemset(pg->ctr_dev_, 0, 2 * N * nlanes * sizeof(uint64_t)));
+                             nctr * sizeof(uint64_t)));
+        HIP_CHECKP(hipMemset(pg->ctr_dev_, 0, nctr * sizeof(uint64_t)));
         for (int i = 0; i < pg->gsize_; ++i) {
             if (i == pg->my_idx_) {
                 pg->peer_base_[i] = pg->my_base_;
@@ -240,6 +242,58 @@ void P2pGroup::IssueSchedule(CommRequest* req, ChunkExec& ce, size_t lane,
         return std::min(v, msg_max);
     };
     for (int phase = 0; phase < ce.sch.num_phases; ++phase) {
+        // One-shot (direct) allreduce fast path: the whole exchange phase
+        // becomes TWO kernels — a fan-out pushing the payload to every
+        // peer and a fan-in waiting all arrivals and reducing them in a
+        // single pass over dst. Falls back to the generic loop for large
+        // payloads, >8 peers or uncovered dtypes.
+        if (ce.sch.one_shot && phase == 1 && !quant) {
+            std::vector<const Step*> fan;
+            for (const auto& st : ce.sch.steps)
+                if (st.phase == 1) fan.push_back(&st);
+            const size_t B = fan.empty() ? 0 : fan[0]->send.bytes;
+            const bool covered = req->Dtype() == DataType::F32 ||
+                                 req->Dtype() == DataType::BF16;
+            if (!fan.empty() && fan.size() <= 8 && B > 0 && B <= msg_max &&
+                B <= FusedMaxBytes() && covered) {
+                XferPoll ab{};
+                ab.abort_word = abort_host_;
+                ab.status = status_host_;
+                ab.max_ticks = max_ticks_;
+                FanPeer sp[8], rp[8];
+                const int np = static_cast<int>(fan.size());
+                for (int i = 0; i < np; ++i) {
+                    const Step* st = fan[i];
+                    const int peer = st->send_peer;
+                    const size_t e = static_cast<size_t>(peer) * nlanes_ + lane;
+                    const uint64_t sseq = ++sent_[e];
+                    sp[i].slot = PeerSlot(peer, lane, (sseq - 1) % nslots_);
+                    sp[i].flag = PeerInFlag(peer, lane);
+                    sp[i].flag_val = sseq;
+                    sp[i].wait_mbox =
+                        sseq > nslots_ ? MyAckFlag(peer, lane) : nullptr;
+                    sp[i].wait_target = sseq - nslots_;
+                    sp[i].ctr = ctr_dev_ + e;
+                    sp[i].ctr_target = fused_sent_[e] += kFanWgsPerPeer;
+                    const uint64_t rseq = ++rcvd_[e];
+                    rp[i].slot = MySlot(peer, lane, (rseq - 1) % nslots_);
+                    rp[i].flag = PeerAckFlag(peer, lane);
+                    rp[i].flag_val = rseq;
+                    rp[i].wait_mbox = MyInFlag(peer, lane);
+                    rp[i].wait_target = rseq;
+                    rp[i].ctr = nullptr;
+                    rp[i].ctr_target = 0;
+                }
+                LaunchFanOutSend(ptr(fan[0]->send), B, sp, np, &ab, s);
+                uint64_t* fctr =
+                    ctr_dev_ + 2 * static_cast<size_t>(gsize_) * nlanes_ + lane;
+                const bool ok = LaunchFanInReduce(
+                    ptr(fan[0]->local_dst), B / es, req->Dtype(), ce.sch.rop,
+                    rp, np, fctr, fanin_adds_[lane] += kXferFusedGrid, &ab, s);
+                MLSL_CHECK(ok, "fan-in dtype dispatch failed");
+                continue;
+            }
+        }
         // Collect this phase's send and recv jobs, then interleave their
         // sub-messages round-robin. The interleave is what makes the
         // bounded slot ring deadlock-free: a sender blocked on backpressure
@@ -303,8 +357,8 @@ void P2pGroup::IssueSchedule(CommRequest* req, ChunkExec& ce, size_t lane,
                     bp.max_ticks = max_ticks_;
                     LaunchXferSendFused(PeerSlot(peer, lane, slot),
                                         ptr(st->send) + off, n,
-                                        bp.mbox ? &bp : nullptr,
-                                        ctr_dev_ + e, ++fused_sent_[e],
+                                        bp.mbox ? &bp : nullptr, ctr_dev_ + e,
+                                        fused_sent_[e] += kXferFusedGrid,
                                         PeerInFlag(peer, lane), seq, s);
                     continue;
                 }
@@ -349,10 +403,11 @@ void P2pGroup::IssueSchedule(CommRequest* req, ChunkExec& ce, size_t lane,
                     if (LaunchXferRecvFused(d, sl, o,
                                             mode == 0 ? n : n / es,
                                             req->Dtype(), ce.sch.rop, mode,
-                                            &wp, rctr, ++fused_rcvd_[e],
+                                            &wp, rctr,
+                                            fused_rcvd_[e] += kXferFusedGrid,
                                             PeerAckFlag(peer, lane), seq, s))
                         continue;
-                    --fused_rcvd_[e];  // dtype not covered: unfused path
+                    fused_rcvd_[e] -= kXferFusedGrid;  // dtype not covered
                 }
                 // Arrival wait as a 1-wg kernel (same deadlock avoidance as
                 // the sender backpressure), then the wide consume kernel.

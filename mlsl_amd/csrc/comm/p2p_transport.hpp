@@ -98,8 +98,10 @@ class P2pGroup {
     std::vector<uint64_t> sent_, rcvd_; // [peer * nlanes + lane]
     // fused small-message kernels: monotonic grid-completion counters
     // (device) + matching host-side op counts, per edge-lane-direction
-    uint64_t* ctr_dev_ = nullptr;       // [2 * gsize * nlanes]
-    std::vector<uint64_t> fused_sent_, fused_rcvd_;
+    // layout: [send edge ctrs | recv edge ctrs | per-lane fan-in ctrs]
+    uint64_t* ctr_dev_ = nullptr;       // [2 * gsize * nlanes + nlanes]
+    // host-side ABSOLUTE workgroup-add accumulators per counter
+    std::vector<uint64_t> fused_sent_, fused_rcvd_, fanin_adds_;
     uint32_t* abort_host_ = nullptr;    // pinned, read by wait kernels
     uint32_t* status_host_ = nullptr;   // pinned, set by aborted wait kernels
     uint64_t max_ticks_ = 0;            // wall-clock bound for waits

@@ -178,6 +178,7 @@ Schedule BuildAllReduceDirect(int rank, int size, size_t count, DataType dt,
     sch.tmp_bytes = static_cast<size_t>(N) * B;  // [own][N-1 arrivals]
     sch.AddStep(MakeCopy(0, Ref(Space::SEND, 0, B), Ref(Space::TMP, 0, B)));
     sch.AddStep(MakeCopy(0, Ref(Space::SEND, 0, B), Ref(Space::RECV, 0, B)));
+    sch.one_shot = true;
     int slot = 1;
     for (int j = 1; j < N; ++j, ++slot) {
         const int peer = (r + j) % N;

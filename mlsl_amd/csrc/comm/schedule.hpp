@@ -59,6 +59,10 @@ struct Schedule {
     // (QuantAccum) over wire blocks of this many elements; buffer refs are
     // then in wire bytes (unit = quant_block + 8).
     size_t quant_block = 0;
+    // Direct (one-shot) allreduce shape: phase 1 is "send TMP(0,B) to all
+    // peers, receive each peer's TMP slot, reduce all into RECV(0,B)" —
+    // executors may batch it into fan-out/fan-in kernels.
+    bool one_shot = false;
     // Where the result lives after the final phase (returned by Wait()).
     BufRef result;
 

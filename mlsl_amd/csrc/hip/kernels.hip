@@ -1572,8 +1572,7 @@ void LaunchXferSendFused(void* slot, const void* src, size_t bytes,
     hipLaunchKernelGGL(XferSendFusedKernel, dim3(kFusedGrid), dim3(kBlock), 0,
                        stream, static_cast<uint8_t*>(slot),
                        static_cast<const uint8_t*>(src), bytes, pa,
-                       static_cast<unsigned long long*>(ctr),
-                       ctr_target * kFusedGrid,
+                       static_cast<unsigned long long*>(ctr), ctr_target,
                        static_cast<unsigned long long*>(in_mbox), seq);
     HIP_CHECK(hipGetLastError());
 }
@@ -1588,7 +1587,7 @@ bool LaunchXferRecvFused(void* dst, const void* slot, const void* other,
     pa.abort_word = static_cast<const unsigned int*>(wp->abort_word);
     pa.status = static_cast<unsigned int*>(wp->status);
     pa.max_ticks = wp->max_ticks;
-    const unsigned long long tgt = ctr_target * kFusedGrid;
+    const unsigned long long tgt = ctr_target;
     auto* c = static_cast<unsigned long long*>(ctr);
     auto* m = static_cast<unsigned long long*>(ack_mbox);
 #define FUSED_LAUNCH(T, OPV)                                                  \
@@ -1639,6 +1638,187 @@ bool LaunchXferRecvFused(void* dst, const void* slot, const void* other,
             return false;
     }
 #undef FUSED_LAUNCH
+    HIP_CHECK(hipGetLastError());
+    return true;
+}
+
+// --- one-shot (direct) allreduce fan kernels ---
+// Fan-out: ONE kernel pushes the payload to every peer's slot (4 wgs per
+// peer, per-peer backpressure poll + publish). Fan-in: ONE kernel waits
+// for all arrivals, then reduces them into dst in a single pass (P slot
+// reads + 1 dst read + 1 dst write per element) and publishes every ack.
+// Small-message allreduce becomes 2 transport kernels regardless of N.
+
+namespace {
+
+constexpr int kFanWgsPerPeerK = 4;
+
+struct FanKernArgs {
+    void* slot[8];
+    unsigned long long* flag[8];
+    unsigned long long flag_val[8];
+    const unsigned long long* wait_mbox[8];
+    unsigned long long wait_target[8];
+    unsigned long long* ctr[8];
+    unsigned long long ctr_target[8];
+    int npeers;
+};
+
+__global__ void FanOutSendKernel(const uint8_t* __restrict__ src, size_t bytes,
+                                 FanKernArgs fa, PollArgs ab) {
+    const int p = blockIdx.x / kFanWgsPerPeerK;
+    if (p >= fa.npeers) return;
+    if (fa.wait_mbox[p] &&
+        !PollGeq(fa.wait_mbox[p], fa.wait_target[p], ab.abort_word, ab.status,
+                 ab.max_ticks))
+        return;
+    uint8_t* dst = static_cast<uint8_t*>(fa.slot[p]);
+    const size_t tid =
+        (blockIdx.x % kFanWgsPerPeerK) * blockDim.x + threadIdx.x;
+    const size_t stride = kFanWgsPerPeerK * blockDim.x;
+    const bool al = ((reinterpret_cast<uintptr_t>(dst) |
+                      reinterpret_cast<uintptr_t>(src)) & 15) == 0;
+    if (al) {
+        const size_t n16 = bytes / 16;
+        const uint4_ev* s = reinterpret_cast<const uint4_ev*>(src);
+        uint4_ev* d = reinterpret_cast<uint4_ev*>(dst);
+        for (size_t i = tid; i < n16; i += stride)
+            __builtin_nontemporal_store(__builtin_nontemporal_load(s + i), d + i);
+        for (size_t j = n16 * 16 + tid; j < bytes; j += stride) dst[j] = src[j];
+    } else {
+        for (size_t j = tid; j < bytes; j += stride) dst[j] = src[j];
+    }
+    FusedFinish(fa.ctr[p], fa.ctr_target[p], fa.flag[p], fa.flag_val[p]);
+}
+
+template <typename T, ReduceOp OP>
+__global__ void FanInReduceKernel(T* __restrict__ dst, size_t n,
+                                  FanKernArgs fa, PollArgs ab,
+                                  unsigned long long* ctr,
+                                  unsigned long long ctr_target) {
+    // wait every arrival (thread 0 polls each mailbox; abort-aware)
+    __shared__ int ok;
+    if (threadIdx.x == 0) {
+        ok = 1;
+        const unsigned long long t0 = wall_clock64();
+        for (int p = 0; p < fa.npeers && ok; ++p) {
+            while (__hip_atomic_load(fa.wait_mbox[p], __ATOMIC_ACQUIRE,
+                                     __HIP_MEMORY_SCOPE_SYSTEM) <
+                   fa.wait_target[p]) {
+                if (__hip_atomic_load(ab.abort_word, __ATOMIC_RELAXED,
+                                      __HIP_MEMORY_SCOPE_SYSTEM) != 0 ||
+                    wall_clock64() - t0 > ab.max_ticks) {
+                    __hip_atomic_store(ab.status, 1u, __ATOMIC_RELEASE,
+                                       __HIP_MEMORY_SCOPE_SYSTEM);
+                    ok = 0;
+                    break;
+                }
+                __builtin_amdgcn_s_sleep(64);
+            }
+        }
+    }
+    __syncthreads();
+    if (!ok) return;
+    const size_t tid = blockIdx.x * blockDim.x + threadIdx.x;
+    const size_t stride = gridDim.x * blockDim.x;
+    for (size_t j = tid; j < n; j += stride) {
+        if constexpr (sizeof(T) == 2) {
+            float acc = __bfloat162float(
+                *reinterpret_cast<const __hip_bfloat16*>(&dst[j]));
+            for (int p = 0; p < fa.npeers; ++p) {
+                const T* sl = static_cast<const T*>(fa.slot[p]);
+                acc = Apply<float, OP>(
+                    acc, __bfloat162float(
+                             *reinterpret_cast<const __hip_bfloat16*>(&sl[j])));
+            }
+            const __hip_bfloat16 r = __float2bfloat16(acc);
+            dst[j] = *reinterpret_cast<const T*>(&r);
+        } else {
+            T acc = dst[j];
+            for (int p = 0; p < fa.npeers; ++p)
+                acc = Apply<T, OP>(acc, static_cast<const T*>(fa.slot[p])[j]);
+            dst[j] = acc;
+        }
+    }
+    // single counter; the finisher publishes every ack
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        const unsigned long long prev = __hip_atomic_fetch_add(
+            ctr, 1ull, __ATOMIC_ACQ_REL, __HIP_MEMORY_SCOPE_SYSTEM);
+        if (prev == ctr_target - 1)
+            for (int p = 0; p < fa.npeers; ++p)
+                __hip_atomic_store(fa.flag[p], fa.flag_val[p], __ATOMIC_RELEASE,
+                                   __HIP_MEMORY_SCOPE_SYSTEM);
+    }
+}
+
+FanKernArgs ToKernArgs(const FanPeer* peers, int np) {
+    FanKernArgs fa{};
+    fa.npeers = np;
+    for (int p = 0; p < np; ++p) {
+        fa.slot[p] = peers[p].slot;
+        fa.flag[p] = static_cast<unsigned long long*>(peers[p].flag);
+        fa.flag_val[p] = peers[p].flag_val;
+        fa.wait_mbox[p] =
+            static_cast<const unsigned long long*>(peers[p].wait_mbox);
+        fa.wait_target[p] = peers[p].wait_target;
+        fa.ctr[p] = static_cast<unsigned long long*>(peers[p].ctr);
+        fa.ctr_target[p] = peers[p].ctr_target;
+    }
+    return fa;
+}
+
+}  // namespace
+
+void LaunchFanOutSend(const void* src, size_t bytes, const FanPeer* peers,
+                      int npeers, const XferPoll* ab, hipStream_t stream) {
+    PollArgs pa{};
+    pa.abort_word = static_cast<const unsigned int*>(ab->abort_word);
+    pa.status = static_cast<unsigned int*>(ab->status);
+    pa.max_ticks = ab->max_ticks;
+    FanKernArgs fa = ToKernArgs(peers, npeers);
+    hipLaunchKernelGGL(FanOutSendKernel,
+                       dim3(kFanWgsPerPeerK * npeers), dim3(kBlock), 0, stream,
+                       static_cast<const uint8_t*>(src), bytes, fa, pa);
+    HIP_CHECK(hipGetLastError());
+}
+
+bool LaunchFanInReduce(void* dst, size_t n, DataType dt, ReduceOp op,
+                       const FanPeer* peers, int npeers, void* ctr,
+                       uint64_t ctr_target, const XferPoll* ab,
+                       hipStream_t stream) {
+    PollArgs pa{};
+    pa.abort_word = static_cast<const unsigned int*>(ab->abort_word);
+    pa.status = static_cast<unsigned int*>(ab->status);
+    pa.max_ticks = ab->max_ticks;
+    FanKernArgs fa = ToKernArgs(peers, npeers);
+    auto* c = static_cast<unsigned long long*>(ctr);
+#define FANIN_LAUNCH(T, OPV)                                                      hipLaunchKernelGGL((FanInReduceKernel<T, OPV>), dim3(kFusedGrid),                                dim3(kBlock), 0, stream, (T*)dst, n, fa, pa, c,                               ctr_target)
+    switch (dt) {
+        case DataType::F32:
+            switch (op) {
+                case ReduceOp::SUM: FANIN_LAUNCH(float, ReduceOp::SUM); break;
+                case ReduceOp::MIN: FANIN_LAUNCH(float, ReduceOp::MIN); break;
+                case ReduceOp::MAX: FANIN_LAUNCH(float, ReduceOp::MAX); break;
+            }
+            break;
+        case DataType::BF16:
+            switch (op) {
+                case ReduceOp::SUM:
+                    FANIN_LAUNCH(unsigned short, ReduceOp::SUM);
+                    break;
+                case ReduceOp::MIN:
+                    FANIN_LAUNCH(unsigned short, ReduceOp::MIN);
+                    break;
+                case ReduceOp::MAX:
+                    FANIN_LAUNCH(unsigned short, ReduceOp::MAX);
+                    break;
+            }
+            break;
+        default:
+            return false;
+    }
+#undef FANIN_LAUNCH
     HIP_CHECK(hipGetLastError());
     return true;
 }

@@ -90,12 +90,15 @@ bool LaunchXferReduce(void* dst, const void* slot, const void* other, size_t n,
                       hipStream_t stream);
 
 // Fully-fused small-message transport kernels: poll + payload + grid-
-// completion counter (monotonic, fetch_add per workgroup with FIXED grid)
-// + flag publish, in ONE launch with a bounded grid (spinners can never
-// starve the peer's kernels). ctr_target is the host-side per-edge count
-// of fused ops (the launcher scales it by the internal grid size).
+// completion counter + flag publish, in ONE launch with a bounded grid
+// (spinners can never starve the peer's kernels). ctr_target is the
+// ABSOLUTE cumulative workgroup-add count for that counter (the host
+// accumulates kXferFusedGrid / kFanWgsPerPeer per emitted kernel, so
+// kernels with different grids can share an edge counter).
 // Recv `mode`: 0 copy (n = BYTES), 1 reduce-into, 2 reduce-out (n =
 // elements); returns false for uncovered dtypes.
+constexpr uint32_t kXferFusedGrid = 32;   // wgs per fused send/recv kernel
+constexpr uint32_t kFanWgsPerPeer = 4;    // wgs per peer in the fan-out
 void LaunchXferSendFused(void* slot, const void* src, size_t bytes,
                          const XferPoll* bp, void* ctr, uint64_t ctr_target,
                          void* in_mbox, uint64_t seq, hipStream_t stream);
@@ -103,6 +106,27 @@ bool LaunchXferRecvFused(void* dst, const void* slot, const void* other,
                          size_t n, DataType dt, ReduceOp op, int mode,
                          const XferPoll* wp, void* ctr, uint64_t ctr_target,
                          void* ack_mbox, uint64_t seq, hipStream_t stream);
+
+// One-shot (direct) allreduce fan kernels: up to 8 peers per launch.
+// Fan-out pushes the payload into every peer slot (per-peer backpressure
+// + publish); fan-in waits all arrivals, reduces them into dst in one
+// pass, and publishes every ack (uses the dedicated `ctr`).
+struct FanPeer {
+    void* slot;
+    void* flag;            // publish target (in_flag for send, ack for recv)
+    uint64_t flag_val;
+    const void* wait_mbox; // backpressure (send, nullable) / arrival (recv)
+    uint64_t wait_target;
+    void* ctr;             // per-peer counter (fan-out only)
+    uint64_t ctr_target;   // absolute adds target
+};
+void LaunchFanOutSend(const void* src, size_t bytes, const FanPeer* peers,
+                      int npeers, const XferPoll* abort_info,
+                      hipStream_t stream);
+bool LaunchFanInReduce(void* dst, size_t n, DataType dt, ReduceOp op,
+                       const FanPeer* peers, int npeers, void* ctr,
+                       uint64_t ctr_target, const XferPoll* abort_info,
+                       hipStream_t stream);
 
 // --- IPC p2p transport flag primitives ---
 // Stream-blocking wait until *mbox >= target (system-scope acquire), with a
