@@ -274,7 +274,9 @@ void P2pGroup::IssueSchedule(CommRequest* req, ChunkExec& ce, size_t lane,
                         sseq > nslots_ ? MyAckFlag(peer, lane) : nullptr;
                     sp[i].wait_target = sseq - nslots_;
                     sp[i].ctr = ctr_dev_ + e;
-                    sp[i].ctr_target = fused_sent_[e] += kFanWgsPerPeer;
+                    sp[i].ctr_target =
+                        fused_sent_[e] += static_cast<uint64_t>(
+                            FanOutWgsPerPeer(static_cast<int>(fan.size())));
                     const uint64_t rseq = ++rcvd_[e];
                     rp[i].slot = MySlot(peer, lane, (rseq - 1) % nslots_);
                     rp[i].flag = PeerAckFlag(peer, lane);

@@ -1651,8 +1651,6 @@ bool LaunchXferRecvFused(void* dst, const void* slot, const void* other,
 
 namespace {
 
-constexpr int kFanWgsPerPeerK = 4;
-
 struct FanKernArgs {
     void* slot[8];
     unsigned long long* flag[8];
@@ -1665,8 +1663,8 @@ struct FanKernArgs {
 };
 
 __global__ void FanOutSendKernel(const uint8_t* __restrict__ src, size_t bytes,
-                                 FanKernArgs fa, PollArgs ab) {
-    const int p = blockIdx.x / kFanWgsPerPeerK;
+                                 FanKernArgs fa, PollArgs ab, int wgs_per_peer) {
+    const int p = blockIdx.x / wgs_per_peer;
     if (p >= fa.npeers) return;
     if (fa.wait_mbox[p] &&
         !PollGeq(fa.wait_mbox[p], fa.wait_target[p], ab.abort_word, ab.status,
@@ -1674,8 +1672,8 @@ __global__ void FanOutSendKernel(const uint8_t* __restrict__ src, size_t bytes,
         return;
     uint8_t* dst = static_cast<uint8_t*>(fa.slot[p]);
     const size_t tid =
-        (blockIdx.x % kFanWgsPerPeerK) * blockDim.x + threadIdx.x;
-    const size_t stride = kFanWgsPerPeerK * blockDim.x;
+        (blockIdx.x % wgs_per_peer) * blockDim.x + threadIdx.x;
+    const size_t stride = static_cast<size_t>(wgs_per_peer) * blockDim.x;
     const bool al = ((reinterpret_cast<uintptr_t>(dst) |
                       reinterpret_cast<uintptr_t>(src)) & 15) == 0;
     if (al) {
@@ -1770,6 +1768,14 @@ FanKernArgs ToKernArgs(const FanPeer* peers, int np) {
 
 }  // namespace
 
+int FanOutWgsPerPeer(int npeers) {
+    // Total spinner budget ~32 wgs per launch: one peer gets the full
+    // bandwidth grid, many peers split it (>= 4 each).
+    if (npeers <= 1) return 32;
+    if (npeers <= 4) return 8;
+    return 4;
+}
+
 void LaunchFanOutSend(const void* src, size_t bytes, const FanPeer* peers,
                       int npeers, const XferPoll* ab, hipStream_t stream) {
     PollArgs pa{};
@@ -1777,9 +1783,10 @@ void LaunchFanOutSend(const void* src, size_t bytes, const FanPeer* peers,
     pa.status = static_cast<unsigned int*>(ab->status);
     pa.max_ticks = ab->max_ticks;
     FanKernArgs fa = ToKernArgs(peers, npeers);
-    hipLaunchKernelGGL(FanOutSendKernel,
-                       dim3(kFanWgsPerPeerK * npeers), dim3(kBlock), 0, stream,
-                       static_cast<const uint8_t*>(src), bytes, fa, pa);
+    const int wpp = FanOutWgsPerPeer(npeers);
+    hipLaunchKernelGGL(FanOutSendKernel, dim3(wpp * npeers), dim3(kBlock), 0,
+                       stream, static_cast<const uint8_t*>(src), bytes, fa, pa,
+                       wpp);
     HIP_CHECK(hipGetLastError());
 }
 

@@ -98,7 +98,8 @@ bool LaunchXferReduce(void* dst, const void* slot, const void* other, size_t n,
 // Recv `mode`: 0 copy (n = BYTES), 1 reduce-into, 2 reduce-out (n =
 // elements); returns false for uncovered dtypes.
 constexpr uint32_t kXferFusedGrid = 32;   // wgs per fused send/recv kernel
-constexpr uint32_t kFanWgsPerPeer = 4;    // wgs per peer in the fan-out
+// wgs per peer in the fan-out: full grid for one peer, split for many
+int FanOutWgsPerPeer(int npeers);
 void LaunchXferSendFused(void* slot, const void* src, size_t bytes,
                          const XferPoll* bp, void* ctr, uint64_t ctr_target,
                          void* in_mbox, uint64_t seq, hipStream_t stream);

@@ -24,6 +24,7 @@ void LaunchSetFlag(void*, uint64_t, hipStream_t) STUB()
 void LaunchXferCopy(void*, const void*, size_t, const XferPoll*, hipStream_t) STUB()
 void LaunchXferSendFused(void*, const void*, size_t, const XferPoll*, void*, uint64_t, void*, uint64_t, hipStream_t) STUB()
 bool LaunchXferRecvFused(void*, const void*, const void*, size_t, DataType, ReduceOp, int, const XferPoll*, void*, uint64_t, void*, uint64_t, hipStream_t) STUB()
+int FanOutWgsPerPeer(int npeers) { return npeers <= 1 ? 32 : npeers <= 4 ? 8 : 4; }
 void LaunchFanOutSend(const void*, size_t, const FanPeer*, int, const XferPoll*, hipStream_t) STUB()
 bool LaunchFanInReduce(void*, size_t, DataType, ReduceOp, const FanPeer*, int, void*, uint64_t, const XferPoll*, hipStream_t) STUB()
 bool LaunchXferReduce(void*, const void*, const void*, size_t, DataType, ReduceOp, const XferPoll*, hipStream_t) STUB()
