@@ -1,5 +1,6 @@
 #include "p2p_transport.hpp"
 
+#include <algorithm>
 #include <cstdlib>
 #include <cstring>
 #include <mutex>
