@@ -144,7 +144,9 @@ def main():
                 "parallelism": f"dp{size}",
                 "algo": os.environ.get("MLSL_ALLREDUCE_ALGO", "auto"),
                 "note": ("n_gpus=1 is the degenerate local-copy path"
-                         if size == 1 else "ncclAllReduce-equivalent over xGMI"),
+                         if size == 1 else
+                         "fused RCCL over xGMI (distinct devices) or the "
+                         "IPC window transport (ranks sharing a device)"),
             },
         }
         print(json.dumps(out))

@@ -63,9 +63,11 @@ def run_gpu_ranks(worker, world, timeout=240, extra_env=None):
 
 
 @requires_gpu
-@pytest.mark.parametrize("world", [2, 4])
+@pytest.mark.parametrize("world", [2, 4, 8])
 def test_collectives_multirank(world):
-    run_gpu_ranks("gpu_collectives", world)
+    # world 8 = the target node's group size: 7-peer fan kernels, the
+    # full stride set {1,3,5,7}, RHD at log2(8) depth — all on one device
+    run_gpu_ranks("gpu_collectives", world, timeout=360)
 
 
 @requires_gpu
@@ -96,8 +98,9 @@ def test_quant_allreduce_multirank():
 
 
 @requires_gpu
-def test_hybrid_grid_multirank():
-    run_gpu_ranks("gpu_hybrid_grid", 4)
+@pytest.mark.parametrize("world", [4, 8])
+def test_hybrid_grid_multirank(world):
+    run_gpu_ranks("gpu_hybrid_grid", world)
 
 
 @requires_gpu
