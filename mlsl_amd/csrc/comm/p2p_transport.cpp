@@ -230,8 +230,9 @@ void P2pGroup::IssueSchedule(CommRequest* req, ChunkExec& ce, size_t lane,
     const size_t msg_max = (slot_bytes_ / unit) * unit;
     MLSL_CHECK(msg_max > 0, "p2p slot smaller than one element/block");
     // Adaptive sub-message size: ~8 sub-messages per transfer (pipelines
-    // DMA against the consumer's reduce) but never below 2 MiB (each
-    // sub-message costs 4 kernel launches) and never above the slot.
+    // DMA against the consumer's reduce) but never below 4 MiB (each
+    // wide-path sub-message costs ~5 kernel launches) and never above the
+    // slot.
     // Sender and receiver derive the identical size from the transfer
     // length, so the slot partition always agrees.
     auto sub_size = [&](size_t bytes) -> size_t {
