@@ -395,23 +395,35 @@ void P2pGroup::IssueSchedule(CommRequest* req, ChunkExec& ce, size_t lane,
                 wp.abort_word = abort_host_;
                 wp.status = status_host_;
                 wp.max_ticks = max_ticks_;
-                if (n <= FusedMaxBytes() && !quant) {
+                if (n <= FusedMaxBytes()) {
                     const size_t e = static_cast<size_t>(peer) * nlanes_ + lane;
                     uint64_t* rctr = ctr_dev_ +
                                      static_cast<size_t>(gsize_) * nlanes_ + e;
-                    const int mode = rj.fuse_into ? 1 : (rj.fuse_out ? 2 : 0);
-                    uint8_t* d = mode == 1 ? ptr(st.local_dst) + off
-                                           : ptr(st.recv) + off;
-                    const uint8_t* o =
-                        mode == 2 ? ptr(st.local_src) + off : nullptr;
-                    if (LaunchXferRecvFused(d, sl, o,
-                                            mode == 0 ? n : n / es,
-                                            req->Dtype(), ce.sch.rop, mode,
-                                            &wp, rctr,
-                                            fused_rcvd_[e] += kXferFusedGrid,
-                                            PeerAckFlag(peer, lane), seq, s))
+                    if (quant && rj.fuse_into) {
+                        // compressed-domain accumulate with the wait fused in
+                        LaunchXferRecvQuantAccum(
+                            ptr(st.local_dst) + off, sl, (n / unit) * blk, blk,
+                            &wp, rctr, fused_rcvd_[e] += kXferFusedGrid,
+                            PeerAckFlag(peer, lane), seq, s);
                         continue;
-                    fused_rcvd_[e] -= kXferFusedGrid;  // dtype not covered
+                    }
+                    if (!quant || (!rj.fuse_into && !rj.fuse_out)) {
+                        // mode 0 (byte copy) is dtype-agnostic, so quant AG
+                        // forwards ride it too
+                        const int mode = rj.fuse_into ? 1 : (rj.fuse_out ? 2 : 0);
+                        uint8_t* d = mode == 1 ? ptr(st.local_dst) + off
+                                               : ptr(st.recv) + off;
+                        const uint8_t* o =
+                            mode == 2 ? ptr(st.local_src) + off : nullptr;
+                        if (LaunchXferRecvFused(d, sl, o,
+                                                mode == 0 ? n : n / es,
+                                                req->Dtype(), ce.sch.rop, mode,
+                                                &wp, rctr,
+                                                fused_rcvd_[e] += kXferFusedGrid,
+                                                PeerAckFlag(peer, lane), seq, s))
+                            continue;
+                        fused_rcvd_[e] -= kXferFusedGrid;  // dtype not covered
+                    }
                 }
                 // Arrival wait as a 1-wg kernel (same deadlock avoidance as
                 // the sender backpressure), then the wide consume kernel.

@@ -783,8 +783,9 @@ __global__ void QuantAccum256Kernel(uint8_t* __restrict__ acc,
     }
 }
 
-__global__ void QuantAccumKernel(uint8_t* __restrict__ acc, const uint8_t* __restrict__ in,
-                                 size_t count, size_t block_elems) {
+__device__ __forceinline__ void QuantAccumBody(uint8_t* __restrict__ acc,
+                                               const uint8_t* __restrict__ in,
+                                               size_t count, size_t block_elems) {
     const size_t nblocks = (count + block_elems - 1) / block_elems;
     const int lane = threadIdx.x & 63;
     const size_t wave = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
@@ -853,6 +854,12 @@ __global__ void QuantAccumKernel(uint8_t* __restrict__ acc, const uint8_t* __res
         }
         if (lane == 0) ahdr[0] = ns;
     }
+}
+
+__global__ void QuantAccumKernel(uint8_t* __restrict__ acc,
+                                 const uint8_t* __restrict__ in, size_t count,
+                                 size_t block_elems) {
+    QuantAccumBody(acc, in, count, block_elems);
 }
 
 }  // namespace
@@ -1750,6 +1757,20 @@ __global__ void FanInReduceKernel(T* __restrict__ dst, size_t n,
     }
 }
 
+// fused arrival-wait + compressed-domain accumulate (quantized ring)
+__global__ void XferRecvQuantAccumKernel(uint8_t* __restrict__ acc,
+                                         const uint8_t* __restrict__ slot,
+                                         size_t count, size_t block_elems,
+                                         PollArgs wp, unsigned long long* ctr,
+                                         unsigned long long ctr_target,
+                                         unsigned long long* ack_mbox,
+                                         unsigned long long seq) {
+    if (!PollGeq(wp.mbox, wp.target, wp.abort_word, wp.status, wp.max_ticks))
+        return;
+    QuantAccumBody(acc, slot, count, block_elems);
+    FusedFinish(ctr, ctr_target, ack_mbox, seq);
+}
+
 FanKernArgs ToKernArgs(const FanPeer* peers, int np) {
     FanKernArgs fa{};
     fa.npeers = np;
@@ -1774,6 +1795,24 @@ int FanOutWgsPerPeer(int npeers) {
     if (npeers <= 1) return 32;
     if (npeers <= 4) return 8;
     return 4;
+}
+
+void LaunchXferRecvQuantAccum(void* acc, const void* slot, size_t count,
+                              size_t block_elems, const XferPoll* wp,
+                              void* ctr, uint64_t ctr_target, void* ack_mbox,
+                              uint64_t seq, hipStream_t stream) {
+    PollArgs pa{};
+    pa.mbox = static_cast<const unsigned long long*>(wp->mbox);
+    pa.target = wp->target;
+    pa.abort_word = static_cast<const unsigned int*>(wp->abort_word);
+    pa.status = static_cast<unsigned int*>(wp->status);
+    pa.max_ticks = wp->max_ticks;
+    hipLaunchKernelGGL(XferRecvQuantAccumKernel, dim3(kFusedGrid), dim3(kBlock),
+                       0, stream, static_cast<uint8_t*>(acc),
+                       static_cast<const uint8_t*>(slot), count, block_elems,
+                       pa, static_cast<unsigned long long*>(ctr), ctr_target,
+                       static_cast<unsigned long long*>(ack_mbox), seq);
+    HIP_CHECK(hipGetLastError());
 }
 
 void LaunchFanOutSend(const void* src, size_t bytes, const FanPeer* peers,

@@ -108,6 +108,13 @@ bool LaunchXferRecvFused(void* dst, const void* slot, const void* other,
                          const XferPoll* wp, void* ctr, uint64_t ctr_target,
                          void* ack_mbox, uint64_t seq, hipStream_t stream);
 
+// Fused arrival-wait + compressed-domain block accumulate for the
+// quantized ring's consume step (count in ELEMENTS of the quant blocks).
+void LaunchXferRecvQuantAccum(void* acc, const void* slot, size_t count,
+                              size_t block_elems, const XferPoll* wp,
+                              void* ctr, uint64_t ctr_target, void* ack_mbox,
+                              uint64_t seq, hipStream_t stream);
+
 // One-shot (direct) allreduce fan kernels: up to 8 peers per launch.
 // Fan-out pushes the payload into every peer slot (per-peer backpressure
 // + publish); fan-in waits all arrivals, reduces them into dst in one

@@ -25,6 +25,7 @@ void LaunchXferCopy(void*, const void*, size_t, const XferPoll*, hipStream_t) ST
 void LaunchXferSendFused(void*, const void*, size_t, const XferPoll*, void*, uint64_t, void*, uint64_t, hipStream_t) STUB()
 bool LaunchXferRecvFused(void*, const void*, const void*, size_t, DataType, ReduceOp, int, const XferPoll*, void*, uint64_t, void*, uint64_t, hipStream_t) STUB()
 int FanOutWgsPerPeer(int npeers) { return npeers <= 1 ? 32 : npeers <= 4 ? 8 : 4; }
+void LaunchXferRecvQuantAccum(void*, const void*, size_t, size_t, const XferPoll*, void*, uint64_t, void*, uint64_t, hipStream_t) STUB()
 void LaunchFanOutSend(const void*, size_t, const FanPeer*, int, const XferPoll*, hipStream_t) STUB()
 bool LaunchFanInReduce(void*, size_t, DataType, ReduceOp, const FanPeer*, int, void*, uint64_t, const XferPoll*, hipStream_t) STUB()
 bool LaunchXferReduce(void*, const void*, const void*, size_t, DataType, ReduceOp, const XferPoll*, hipStream_t) STUB()

@@ -113,6 +113,17 @@ def test_stats_device_ns_multirank():
     (1, 0, 0, 0), (2, 1, 0, 0), (1, 0, 1, 0), (1, 0, 0, 1),
 ])
 def test_e2e_device_multirank(mp, du, user_buf, quant):
+    _e2e_device(4, mp, du, user_buf, quant)
+
+
+@requires_gpu
+def test_e2e_device_world8():
+    # the target node's group size: dp=4 x mp=2 planner matrix on device
+    _e2e_device(8, 2, 1, 0, 0)
+    _e2e_device(8, 1, 0, 0, 1)
+
+
+def _e2e_device(world, mp, du, user_buf, quant):
     """mlsl_test-equivalent epoch loop on the DEVICE engine (p2p transport,
     multi-rank-one-GPU): session planner + pack blocks + grad/inc exchange,
     with the user_buf toggle exercising pinned-host staging."""
@@ -121,7 +132,6 @@ def test_e2e_device_multirank(mp, du, user_buf, quant):
         subprocess.run(["make", "e2e"], cwd=REPO, check=True,
                        capture_output=True, timeout=900)
     port = free_port()
-    world = 4
     procs = []
     for r in range(world):
         env = dict(os.environ, RANK=str(r), WORLD_SIZE=str(world),

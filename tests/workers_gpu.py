@@ -264,7 +264,8 @@ def gpu_stress():
         torch.cuda.synchronize()
         if cont.item() == 0.0:
             break
-        op = rng.choice(["ar", "ar_inplace", "rs", "ag", "a2a", "bcast", "bar"])
+        op = rng.choice(["ar", "ar_inplace", "rs", "ag", "a2a", "bcast",
+                         "bar", "srl", "quant"])
         count = rng.choice([64, 4097, 1 << 16, 1 << 20])
         if op == "ar" or op == "ar_inplace":
             a = _arange(torch, count, rank)
@@ -304,6 +305,30 @@ def gpu_stress():
             mx.wait(d.bcast(b, count, root=0, group="data"))
             torch.cuda.synchronize()
             assert torch.allclose(b, _arange(torch, count, 3)), (iters, op)
+        elif op == "srl":
+            dstp = (rank + 1) % size
+            srcp = (rank - 1 + size) % size
+            sv = torch.full((count,), float(rank * 3 + 1), device="cuda")
+            rv = torch.empty(count, device="cuda")
+            mx.wait(d.send_recv_list(sv, rv, [(dstp, 0, count, 0, 0),
+                                              (srcp, 0, 0, 0, count)],
+                                     group="data"))
+            torch.cuda.synchronize()
+            assert torch.all(rv == float(srcp * 3 + 1)), (iters, op)
+        elif op == "quant":
+            a = torch.randn(count, device="cuda")
+            outq = torch.empty_like(a)
+            req = mx.PersistentRequest(d, "all_reduce", count, dtype="f32",
+                                       op="sum", group="data", quantized=True)
+            req.start(a, outq)
+            req.wait()
+            torch.cuda.synchronize()
+            exact = torch.empty_like(a)
+            mx.wait(d.all_reduce(a, exact, count, op="sum", group="data"))
+            torch.cuda.synchronize()
+            rel = ((outq - exact).norm() / exact.norm().clamp_min(1e-6)).item()
+            assert rel < 0.06, (iters, op, rel)
+            req.destroy()
         else:
             d.barrier("data")
         iters += 1
