@@ -168,3 +168,9 @@ def test_zero1_multirank():
 @requires_gpu
 def test_seqpar_multirank():
     run_gpu_ranks("gpu_seqpar", 2)
+
+
+@requires_gpu
+@pytest.mark.parametrize("world", [2, 4])
+def test_rs_overlap_multirank(world):
+    run_gpu_ranks("gpu_rs_overlap", world)

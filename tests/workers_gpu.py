@@ -422,8 +422,27 @@ def gpu_seqpar():
     mx.finalize()
 
 
+def gpu_rs_overlap():
+    """Reduce-scatter with the reference's in-place layout (rbuf = sbuf +
+    rank*seg, overlapping buffers) — the ZeRO-1 gradient exchange shape
+    (grad_req Start(b, b + owned_off), session.cpp)."""
+    mx, torch, rank, size = _init()
+    d = mx.Distribution(size, 1)
+    for per in (48, 1536, 65536):
+        buf = torch.cat([_arange(torch, per, rank) + i * 1000
+                         for i in range(size)])
+        wanted = size * torch.arange(per, dtype=torch.float32, device="cuda")             + size * (size - 1) / 2.0 + size * rank * 1000
+        rb = buf[rank * per:(rank + 1) * per]
+        mx.wait(d.reduce_scatter(buf, rb, per, op="sum", group="data"))
+        torch.cuda.synchronize()
+        assert torch.allclose(rb, wanted),             (per, rank, rb[:4].tolist(), wanted[:4].tolist())
+    d.barrier("global")
+    mx.finalize()
+
+
 WORKERS = {
     "gpu_collectives": gpu_collectives,
+    "gpu_rs_overlap": gpu_rs_overlap,
     "gpu_ddp": gpu_ddp,
     "gpu_zero1": gpu_zero1,
     "gpu_seqpar": gpu_seqpar,
