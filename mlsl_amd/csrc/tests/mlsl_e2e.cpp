@@ -189,9 +189,24 @@ int main(int argc, char** argv) {
                             std::sqrt(err2) / std::max(std::sqrt(ref2), 1e-9);
                         CHECK_OK(rel < 0.05, "quant grad rel err %.4f l=%d",
                                  rel, l);
-                    } else {
-                        CHECK_OK(err2 == 0.0, "grad mismatch l=%d ep=%d mb=%d",
-                                 l, epoch, mbatch);
+                    } else if (err2 != 0.0) {
+                        // print the first mismatching element for diagnosis
+                        for (size_t j = 0; j < n && err2 != 0.0; ++j)
+                            for (size_t e = 0; e < KS; ++e) {
+                                const float want =
+                                    dp * (0.125f * ((gk0 + joff + j) % 37) +
+                                          0.5f * e + it) +
+                                    dp * (dp - 1) / 2.0f;
+                                if (g[j * KS + e] != want) {
+                                    CHECK_OK(false,
+                                             "grad mismatch l=%d ep=%d mb=%d "
+                                             "j=%zu e=%zu got=%f want=%f",
+                                             l, epoch, mbatch, j, e,
+                                             g[j * KS + e], want);
+                                    err2 = 0.0;
+                                    break;
+                                }
+                            }
                     }
                 }
                 // Update on the owned shard, then increment AllGather (du).

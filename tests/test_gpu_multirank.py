@@ -174,3 +174,8 @@ def test_seqpar_multirank():
 @pytest.mark.parametrize("world", [2, 4])
 def test_rs_overlap_multirank(world):
     run_gpu_ranks("gpu_rs_overlap", world)
+
+
+@requires_gpu
+def test_hybrid_rs_multirank():
+    run_gpu_ranks("gpu_hybrid_rs", 8, timeout=300)

@@ -440,9 +440,38 @@ def gpu_rs_overlap():
     mx.finalize()
 
 
+def gpu_hybrid_rs():
+    """Reduce-scatter with overlapping buffers on a COLOR-created data
+    subgroup (Distribution(size//2, 2)) — the exact ZeRO-1 exchange of the
+    hybrid e2e, minus the planner."""
+    mx, torch, rank, size = _init()
+    d = mx.Distribution(size // 2, 2)
+    dp = size // 2
+    didx = d.process_idx("data")
+    for per in (48, 1536):
+        buf = torch.cat([_arange(torch, per, didx) + i * 1000
+                         for i in range(dp)])
+        want = dp * torch.arange(per, dtype=torch.float32, device="cuda")             + dp * (dp - 1) / 2.0 + dp * didx * 1000
+        rb = buf[didx * per:(didx + 1) * per]
+        mx.wait(d.reduce_scatter(buf, rb, per, op="sum", group="data"))
+        torch.cuda.synchronize()
+        assert torch.allclose(rb, want),             (per, rank, didx, rb[:4].tolist(), want[:4].tolist())
+        # and the increment allgather right after (inc_req shape)
+        src = _arange(torch, per, didx) * 2
+        flat = torch.empty(dp * per, device="cuda")
+        mx.wait(d.all_gather(src, per, flat, group="data"))
+        torch.cuda.synchronize()
+        for r2 in range(dp):
+            assert torch.allclose(flat[r2 * per:(r2 + 1) * per],
+                                  _arange(torch, per, r2) * 2), (per, r2)
+    d.barrier("global")
+    mx.finalize()
+
+
 WORKERS = {
     "gpu_collectives": gpu_collectives,
     "gpu_rs_overlap": gpu_rs_overlap,
+    "gpu_hybrid_rs": gpu_hybrid_rs,
     "gpu_ddp": gpu_ddp,
     "gpu_zero1": gpu_zero1,
     "gpu_seqpar": gpu_seqpar,
