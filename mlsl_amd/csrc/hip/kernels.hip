@@ -1226,7 +1226,7 @@ __global__ void WaitFlagKernel(const unsigned long long* __restrict__ mbox,
                                unsigned long long max_ticks) {
     if (threadIdx.x != 0 || blockIdx.x != 0) return;
     const unsigned long long t0 = wall_clock64();
-    while (__hip_atomic_load(mbox, __ATOMIC_ACQUIRE, __HIP_MEMORY_SCOPE_SYSTEM) <
+    while (__hip_atomic_load(mbox, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_SYSTEM) <
            target) {
         if (__hip_atomic_load(abort_word, __ATOMIC_RELAXED,
                               __HIP_MEMORY_SCOPE_SYSTEM) != 0 ||
@@ -1237,13 +1237,18 @@ __global__ void WaitFlagKernel(const unsigned long long* __restrict__ mbox,
         }
         __builtin_amdgcn_s_sleep(64);
     }
+    // acquire once satisfied (consumers in the NEXT kernel get their
+    // visibility from the dispatch boundary; this orders same-stream work)
+    (void)__hip_atomic_load(mbox, __ATOMIC_ACQUIRE, __HIP_MEMORY_SCOPE_SYSTEM);
 }
 
 __global__ void SetFlagKernel(unsigned long long* __restrict__ mbox,
                               unsigned long long val) {
-    if (threadIdx.x == 0 && blockIdx.x == 0)
+    if (threadIdx.x == 0 && blockIdx.x == 0) {
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
         __hip_atomic_store(mbox, val, __ATOMIC_RELEASE,
                            __HIP_MEMORY_SCOPE_SYSTEM);
+    }
 }
 
 }  // namespace
@@ -1261,11 +1266,14 @@ __device__ __forceinline__ bool PollGeq(const unsigned long long* mbox,
                                         const unsigned int* abort_word,
                                         unsigned int* status,
                                         unsigned long long max_ticks) {
+    // Consumer recipe (MI355X_MICROARCH.md): RELAXED polls, then ONE
+    // acquire once satisfied — polling with acquire loads is correct but
+    // 2-3x slower per hop and cuts chip bandwidth with many pollers.
     __shared__ int ok;
     if (threadIdx.x == 0) {
         ok = 1;
         const unsigned long long t0 = wall_clock64();
-        while (__hip_atomic_load(mbox, __ATOMIC_ACQUIRE,
+        while (__hip_atomic_load(mbox, __ATOMIC_RELAXED,
                                  __HIP_MEMORY_SCOPE_SYSTEM) < target) {
             if (__hip_atomic_load(abort_word, __ATOMIC_RELAXED,
                                   __HIP_MEMORY_SCOPE_SYSTEM) != 0 ||
@@ -1277,6 +1285,9 @@ __device__ __forceinline__ bool PollGeq(const unsigned long long* mbox,
             }
             __builtin_amdgcn_s_sleep(64);
         }
+        if (ok)
+            (void)__hip_atomic_load(mbox, __ATOMIC_ACQUIRE,
+                                    __HIP_MEMORY_SCOPE_SYSTEM);
     }
     __syncthreads();
     return ok != 0;
@@ -1481,11 +1492,21 @@ __device__ __forceinline__ void FusedFinish(unsigned long long* ctr,
                                             unsigned long long val) {
     __syncthreads();
     if (threadIdx.x == 0) {
+        // ROCm 7.2 / gfx950 compiler hazard (MI355X_MICROARCH.md): when the
+        // publishing wave's vmcnt scoreboard looks provably empty, the
+        // compiler drops the s_waitcnt after the release's buffer_wbl2 and
+        // the flag can overtake the payload write-back (~1e-4 stale under
+        // load — observed as a zero arrival row in the ring RS). Inline asm
+        // is invisible to that pass: force the wait before the release-add
+        // and again before the publish store.
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
         const unsigned long long prev = __hip_atomic_fetch_add(
             ctr, 1ull, __ATOMIC_ACQ_REL, __HIP_MEMORY_SCOPE_SYSTEM);
-        if (prev == target - 1)
+        if (prev == target - 1) {
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
             __hip_atomic_store(mbox, val, __ATOMIC_RELEASE,
                                __HIP_MEMORY_SCOPE_SYSTEM);
+        }
     }
 }
 
@@ -1707,7 +1728,7 @@ __global__ void FanInReduceKernel(T* __restrict__ dst, size_t n,
         ok = 1;
         const unsigned long long t0 = wall_clock64();
         for (int p = 0; p < fa.npeers && ok; ++p) {
-            while (__hip_atomic_load(fa.wait_mbox[p], __ATOMIC_ACQUIRE,
+            while (__hip_atomic_load(fa.wait_mbox[p], __ATOMIC_RELAXED,
                                      __HIP_MEMORY_SCOPE_SYSTEM) <
                    fa.wait_target[p]) {
                 if (__hip_atomic_load(ab.abort_word, __ATOMIC_RELAXED,
@@ -1721,6 +1742,9 @@ __global__ void FanInReduceKernel(T* __restrict__ dst, size_t n,
                 __builtin_amdgcn_s_sleep(64);
             }
         }
+        if (ok)
+            (void)__hip_atomic_load(fa.wait_mbox[0], __ATOMIC_ACQUIRE,
+                                    __HIP_MEMORY_SCOPE_SYSTEM);
     }
     __syncthreads();
     if (!ok) return;
@@ -1745,15 +1769,19 @@ __global__ void FanInReduceKernel(T* __restrict__ dst, size_t n,
             dst[j] = acc;
         }
     }
-    // single counter; the finisher publishes every ack
+    // single counter; the finisher publishes every ack (inline-asm waits:
+    // see FusedFinish — the compiler may drop the post-wbl2 waitcnt)
     __syncthreads();
     if (threadIdx.x == 0) {
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
         const unsigned long long prev = __hip_atomic_fetch_add(
             ctr, 1ull, __ATOMIC_ACQ_REL, __HIP_MEMORY_SCOPE_SYSTEM);
-        if (prev == ctr_target - 1)
+        if (prev == ctr_target - 1) {
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
             for (int p = 0; p < fa.npeers; ++p)
                 __hip_atomic_store(fa.flag[p], fa.flag_val[p], __ATOMIC_RELEASE,
                                    __HIP_MEMORY_SCOPE_SYSTEM);
+        }
     }
 }
 
