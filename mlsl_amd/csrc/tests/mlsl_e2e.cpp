@@ -190,23 +190,37 @@ int main(int argc, char** argv) {
                         CHECK_OK(rel < 0.05, "quant grad rel err %.4f l=%d",
                                  rel, l);
                     } else if (err2 != 0.0) {
-                        // print the first mismatching element for diagnosis
-                        for (size_t j = 0; j < n && err2 != 0.0; ++j)
+                        // full-segment diagnosis: how many elements are
+                        // wrong, and by what deltas (a single missing
+                        // contributor row gives delta = 0.125*m + e/2 + d)
+                        size_t bad = 0, first = SIZE_MAX;
+                        float fgot = 0, fwant = 0, fdelta = 0, ldelta = 0;
+                        size_t last = 0;
+                        for (size_t j = 0; j < n; ++j)
                             for (size_t e = 0; e < KS; ++e) {
                                 const float want =
                                     dp * (0.125f * ((gk0 + joff + j) % 37) +
                                           0.5f * e + it) +
                                     dp * (dp - 1) / 2.0f;
-                                if (g[j * KS + e] != want) {
-                                    CHECK_OK(false,
-                                             "grad mismatch l=%d ep=%d mb=%d "
-                                             "j=%zu e=%zu got=%f want=%f",
-                                             l, epoch, mbatch, j, e,
-                                             g[j * KS + e], want);
-                                    err2 = 0.0;
-                                    break;
+                                const float got = g[j * KS + e];
+                                if (got != want) {
+                                    if (first == SIZE_MAX) {
+                                        first = j * KS + e;
+                                        fgot = got;
+                                        fwant = want;
+                                        fdelta = want - got;
+                                    }
+                                    last = j * KS + e;
+                                    ldelta = want - got;
+                                    ++bad;
                                 }
                             }
+                        CHECK_OK(false,
+                                 "grad mismatch l=%d ep=%d mb=%d didx=%zu: "
+                                 "bad=%zu/%zu first@%zu got=%f want=%f "
+                                 "delta=%f last@%zu ldelta=%f",
+                                 l, epoch, mbatch, didx, bad, n * KS, first,
+                                 fgot, fwant, fdelta, last, ldelta);
                     }
                 }
                 // Update on the owned shard, then increment AllGather (du).
