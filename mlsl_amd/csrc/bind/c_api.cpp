@@ -680,6 +680,10 @@ int mlsl_statistics_get_total_compute_cycles(mlsl_statistics st, unsigned long l
     C_TRY* out = STATS(st)->GetTotalComputeCycles();
     C_CATCH
 }
+int mlsl_memcpy(void* dst, const void* src, size_t bytes) {
+    C_TRY mlsl::Environment::GetEnv().Memcpy(dst, src, bytes);
+    C_CATCH
+}
 int mlsl_statistics_get_comm_device_ns(mlsl_statistics st, size_t op,
                                        unsigned long long* out) {
     C_TRY* out = STATS(st)->GetCommDeviceNs(op);

@@ -3,6 +3,8 @@
 // core: ProcessGroup = RCCL communicator set / TCP mesh subset).
 #include "../include/mlsl/mlsl.hpp"
 
+#include <hip/hip_runtime.h>
+
 #include <cstring>
 #include <functional>
 #include <memory>
@@ -294,6 +296,16 @@ void* Environment::Alloc(size_t size, size_t alignment) {
 }
 
 void Environment::Free(void* ptr) { Context::Get().Free(ptr); }
+
+void Environment::Memcpy(void* dst, const void* src, size_t bytes) {
+    Context& ctx = Context::Get();
+    if (ctx.DeviceMode()) {
+        MLSL_CHECK(hipMemcpy(dst, src, bytes, hipMemcpyDefault) == hipSuccess,
+                   "Environment::Memcpy failed");
+    } else {
+        std::memcpy(dst, src, bytes);
+    }
+}
 
 void Environment::SetQuantizationParams(const QuantParams& p) { g_quant_params = p; }
 

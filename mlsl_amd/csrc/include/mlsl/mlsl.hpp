@@ -106,6 +106,12 @@ class Environment {
 
     void* Alloc(size_t size, size_t alignment = 64);
     void Free(void* ptr);
+    // Coherent copy between host memory and Alloc'd (HBM) buffers — the
+    // sanctioned way to fill/read comm buffers from the host. Direct CPU
+    // stores to device memory over BAR are NOT coherent with the GPU's
+    // per-XCD L2s (stale lines, e.g. the driver's page-scrub zeros, can
+    // shadow them); hipMemcpy goes through the coherent path.
+    void Memcpy(void* dst, const void* src, size_t bytes);
 
     void SetQuantizationParams(const QuantParams& p);
     const QuantParams& GetQuantizationParams() const;

@@ -93,6 +93,13 @@ int main(int argc, char** argv) {
     };
     float* comm0 = get_buf(out0->GetCommBufSize() + 4);
     float* comm1 = get_buf(in1->GetCommBufSize() + 4);
+    // Host shadows: with user_buf=0 the comm buffers live in HBM, and
+    // direct CPU stores over BAR are not coherent with the GPU's per-XCD
+    // L2s — all fills/readbacks go through Environment::Memcpy (observed:
+    // a chain-starting rank's send read the driver's page-scrub zeros
+    // instead of BAR-written gradients, ~40% of world-8 runs under load).
+    std::vector<float> h0(out0->GetCommBufSize() / 4 + 1);
+    std::vector<float> h1(in1->GetCommBufSize() / 4 + 1);
 
     // Persistent "weights" per layer for the Update step.
     std::vector<std::vector<float>> params(2);
@@ -114,15 +121,19 @@ int main(int argc, char** argv) {
                         for (size_t fm = 0; fm < b->GetFmCount(); ++fm)
                             for (size_t k = 0; k < S; ++k) {
                                 const size_t gfm = b->GetFmOffset() + fm;
-                                comm0[b->GetBufOffset() +
-                                      (mb * b->GetFmCount() + fm) * S + k] =
+                                h0[b->GetBufOffset() +
+                                   (mb * b->GetFmCount() + fm) * S + k] =
                                     gfm * 100.0f + k + (b->GetMbOffset() + mb) +
                                     it + midx;
                             }
                 }
+                env.Memcpy(comm0, h0.data(), out0->GetCommBufSize());
                 out0->StartComm(comm0);
-                float* res = static_cast<float*>(in1->WaitComm());
-                CHECK_OK(res != nullptr, "null fwd result");
+                float* dres = static_cast<float*>(in1->WaitComm());
+                CHECK_OK(dres != nullptr, "null fwd result");
+                float* res = h1.data();
+                if (dres) env.Memcpy(res, dres, in1->GetCommBufSize());
+                else res = nullptr;
                 const CommBlockInfo* ub = in1->GetUnpackBlock(0);
                 for (size_t mb = 0; mb < lmb && res; ++mb)
                     for (size_t fm = 0; fm < f1l; ++fm)
@@ -145,9 +156,10 @@ int main(int argc, char** argv) {
                     for (size_t fm = 0; fm < f1l; ++fm)
                         for (size_t k = 0; k < S; ++k) {
                             const size_t gfm = midx * f1l + fm;
-                            comm1[pb->GetBufOffset() + (mb * f1l + fm) * S + k] =
+                            h1[pb->GetBufOffset() + (mb * f1l + fm) * S + k] =
                                 gfm * 7.0f + k + mb + it;
                         }
+                env.Memcpy(comm1, h1.data(), in1->GetCommBufSize());
                 in1->StartComm(comm1);
                 float* bres = static_cast<float*>(out0->WaitComm());
                 CHECK_OK(bres != nullptr, "null bwd result");
@@ -162,12 +174,19 @@ int main(int argc, char** argv) {
                 const size_t lk = ps->GetLocalKernelCount();
                 const size_t gk0 = ps->GetGlobalKernelOffset();
                 float* grad = get_buf(lk * KS * sizeof(float));
+                std::vector<float> hgrad(lk * KS);
                 for (size_t j = 0; j < lk; ++j)
                     for (size_t e = 0; e < KS; ++e)
-                        grad[j * KS + e] =
+                        hgrad[j * KS + e] =
                             0.125f * ((gk0 + j) % 37) + 0.5f * e + didx + it;
+                env.Memcpy(grad, hgrad.data(), lk * KS * sizeof(float));
                 ps->StartGradientComm(grad);
-                float* g = static_cast<float*>(ps->WaitGradientComm());
+                float* dg = static_cast<float*>(ps->WaitGradientComm());
+                const size_t gspan =
+                    (du ? ps->GetOwnedKernelCount() : lk) * KS;
+                std::vector<float> hg(gspan);
+                env.Memcpy(hg.data(), dg, gspan * sizeof(float));
+                float* g = hg.data();
                 if (dp > 1) {
                     const size_t n = du ? ps->GetOwnedKernelCount() : lk;
                     const size_t joff = du ? ps->GetOwnedKernelOffset() : 0;
