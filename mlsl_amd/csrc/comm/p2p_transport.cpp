@@ -68,7 +68,14 @@ std::unique_ptr<P2pGroup> P2pGroup::Create(ProcessGroup* g, size_t nlanes,
     if (member) {
         HIP_CHECKP(hipMalloc(reinterpret_cast<void**>(&pg->my_base_),
                              pg->win_bytes_));
-        HIP_CHECKP(hipMemset(pg->my_base_, 0, pg->win_bytes_));
+        // Zero ONLY the flag region: flags/acks are accessed exclusively
+        // with system-scope atomics (coherent). The slot payload area is
+        // deliberately left untouched — a local memset would leave this
+        // process's caches holding zero lines that can shadow the peer's
+        // later payload stores (cold-box page-scrub lines shadowed arrivals
+        // the same way; slots are fully overwritten before every consume,
+        // so they never need initialization).
+        HIP_CHECKP(hipMemset(pg->my_base_, 0, pg->flags_bytes_));
         HIP_CHECKP(hipIpcGetMemHandle(&mine.h, pg->my_base_));
         mine.valid = 1;
         HIP_CHECKP(hipHostMalloc(reinterpret_cast<void**>(&pg->abort_host_),
@@ -451,7 +458,7 @@ void P2pGroup::IssueSchedule(CommRequest* req, ChunkExec& ce, size_t lane,
                     } else {  // fuse_out
                         if (quant) {
                             // acc = slot, then acc += local_src slice
-                            LaunchCopy(ptr(st.recv) + off, sl, n, s);
+                            LaunchCopyVariant(ptr(st.recv) + off, sl, n, false, s);
                             LaunchQuantAccum(ptr(st.recv) + off,
                                              ptr(st.local_src) + off,
                                              (n / unit) * blk, blk, s);
@@ -483,7 +490,7 @@ void P2pGroup::IssueSchedule(CommRequest* req, ChunkExec& ce, size_t lane,
             uint8_t* d = ptr(st.local_dst);
             uint8_t* src = ptr(st.local_src);
             if (st.local == Step::LocalOp::COPY) {
-                if (d != src) LaunchCopy(d, src, st.local_src.bytes, s);
+                if (d != src) LaunchCopyVariant(d, src, st.local_src.bytes, false, s);
             } else if (quant) {
                 LaunchQuantAccum(d, src, (st.local_dst.bytes / unit) * blk,
                                  blk, s);
