@@ -149,9 +149,11 @@ def _e2e_device(world, mp, du, user_buf, quant):
 
 @requires_gpu
 def test_stress_multirank():
-    """30 s randomized mixed-collective soak at world 2 on one device."""
-    run_gpu_ranks("gpu_stress", 2, timeout=180,
-                  extra_env={"STRESS_SECONDS": "30"})
+    """Randomized mixed-collective soak at world 2 on one device
+    (STRESS_SECONDS env scales it; default 30 s)."""
+    secs = os.environ.get("STRESS_SECONDS", "30")
+    run_gpu_ranks("gpu_stress", 2, timeout=int(secs) + 150,
+                  extra_env={"STRESS_SECONDS": secs})
 
 
 @requires_gpu
