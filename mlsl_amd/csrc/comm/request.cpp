@@ -419,6 +419,14 @@ void CommRequest::Start(const void* sbuf, void* rbuf) {
                "Start() while request in flight");
     sbuf_ = static_cast<const uint8_t*>(sbuf);
     rbuf_ = static_cast<uint8_t*>(rbuf);
+    // AlltoAll(v)'s pairwise schedule reads send-block j while writing
+    // recv-block i of the same iteration — aliased buffers would corrupt
+    // silently (the reference splits these ops out of in-place too,
+    // src/comm_ep.cpp:623-736). Fail loudly instead.
+    MLSL_CHECK(!((spec_.op == CollOp::ALLTOALL ||
+                  spec_.op == CollOp::ALLTOALLV) &&
+                 sbuf_ && sbuf_ == rbuf_),
+               "alltoall(v) does not support in-place (sbuf == rbuf)");
     dev_sbuf_ = nullptr;
     dev_rbuf_ = nullptr;
     flow_ = spec_.op == CollOp::SRLIST ? 0 : group_->NextFlow();

@@ -127,6 +127,31 @@ def gpu_collectives():
     torch.cuda.synchronize()
     assert torch.allclose(rout, _arange(torch, per, rank) * 2), "scatter"
 
+    # dtype fallback paths on the device transport: f64 / i64 allreduce
+    # ride the wait-kernel + generic-reduce consumers; bf16 the fused ones
+    count = 2048
+    a64 = torch.arange(count, dtype=torch.float64, device="cuda") + rank
+    o64 = torch.empty_like(a64)
+    mx.wait(d.all_reduce(a64, o64, count, op="sum", group=g))
+    torch.cuda.synchronize()
+    w64 = size * torch.arange(count, dtype=torch.float64, device="cuda") \
+        + size * (size - 1) / 2.0
+    assert torch.allclose(o64, w64), "f64 allreduce"
+    ai = torch.arange(count, dtype=torch.int64, device="cuda") + rank
+    oi = torch.empty_like(ai)
+    mx.wait(d.all_reduce(ai, oi, count, op="sum", group=g))
+    torch.cuda.synchronize()
+    wi = size * torch.arange(count, dtype=torch.int64, device="cuda") \
+        + size * (size - 1) // 2
+    assert torch.equal(oi, wi), "i64 allreduce"
+    ab = (torch.arange(count, dtype=torch.bfloat16, device="cuda") % 8) + rank
+    ob = torch.empty_like(ab)
+    mx.wait(d.all_reduce(ab, ob, count, op="sum", group=g))
+    torch.cuda.synchronize()
+    wb = size * ((torch.arange(count, dtype=torch.bfloat16, device="cuda") % 8)) \
+        + size * (size - 1) / 2.0
+    assert torch.allclose(ob.float(), wb.float()), "bf16 allreduce"
+
     # p2p ring over send_recv_list
     n = 1024
     dst_rank = (rank + 1) % size
