@@ -243,14 +243,8 @@ void P2pGroup::IssueSchedule(CommRequest* req, ChunkExec& ce, size_t lane,
     // Sender and receiver derive the identical size from the transfer
     // length, so the slot partition always agrees.
     auto sub_size = [&](size_t bytes) -> size_t {
-        // Transfers just above the fused ceiling split into fused-size
-        // sub-messages (two 1 MiB fused kernels beat one 2 MiB wide chain
-        // of 5 launches); larger transfers keep the 4 MiB wide-path floor.
-        const size_t fmax = FusedMaxBytes();
-        const size_t floor_b =
-            (fmax > 0 && bytes <= 2 * fmax) ? fmax : (4u << 20);
         size_t v = std::max<size_t>((bytes + 7) / 8,
-                                    std::min<size_t>(floor_b, msg_max));
+                                    std::min<size_t>(4u << 20, msg_max));
         v = std::min(v, msg_max);
         v = (v / unit) * unit;
         if (v < unit) v = unit;
