@@ -268,17 +268,24 @@ class HipRuntime : public DeviceRuntime {
                 }
             }
         }
-        if (GlobalConfig().msg_priority && g->IsMember() && g->Size() > 1) {
+        if (GlobalConfig().msg_priority) {
+            // The bootstrap Allgather is collective over the WORLD — every
+            // rank must participate (non-members contribute a zero id, as in
+            // the channel loop above), else a subgroup with non-member ranks
+            // desyncs the bootstrap stream.
             ncclUniqueId id{};
-            if (g->MyIdx() == 0) NCCL_CHECK(ncclGetUniqueId(&id));
+            if (g->IsMember() && g->MyIdx() == 0) NCCL_CHECK(ncclGetUniqueId(&id));
             std::vector<ncclUniqueId> all(static_cast<size_t>(ctx.Size()));
             ctx.Boot()->Allgather(&id, sizeof(ncclUniqueId), all.data());
-            ncclUniqueId gid = all[static_cast<size_t>(g->WorldRank(0))];
-            NCCL_CHECK(ncclCommInitRank(&gc.prio_comm, g->Size(), gid, g->MyIdx()));
-            int lo = 0, hi = 0;
-            HIP_CHECKD(hipDeviceGetStreamPriorityRange(&lo, &hi));
-            HIP_CHECKD(hipStreamCreateWithPriority(&gc.prio_stream,
-                                                   hipStreamNonBlocking, hi));
+            if (g->IsMember() && g->Size() > 1) {
+                ncclUniqueId gid = all[static_cast<size_t>(g->WorldRank(0))];
+                NCCL_CHECK(ncclCommInitRank(&gc.prio_comm, g->Size(), gid,
+                                            g->MyIdx()));
+                int lo = 0, hi = 0;
+                HIP_CHECKD(hipDeviceGetStreamPriorityRange(&lo, &hi));
+                HIP_CHECKD(hipStreamCreateWithPriority(&gc.prio_stream,
+                                                       hipStreamNonBlocking, hi));
+            }
         }
         group_comms_.emplace(g->Uid(), std::move(gc));
         MLSL_LOG(DEBUG, "device comms ready for group uid=%d size=%d channels=%zu",
