@@ -172,7 +172,7 @@ class HipRuntime : public DeviceRuntime {
         char hn[256] = {0};
         (void)gethostname(hn, sizeof(hn) - 1);
         for (const char* p = hn; *p; ++p) mine.host = mine.host * 131 + *p;
-        std::vector<HostDev> all(static_cast<size_t>(ctx.Size()));
+        std::vector<HostDev> all(static_cast<size_t>(ctx.Boot()->Size()));
         ctx.Boot()->Allgather(&mine, sizeof(HostDev), all.data());
         bool shared = false;
         for (size_t i = 0; i < all.size() && !shared; ++i)
@@ -255,7 +255,7 @@ class HipRuntime : public DeviceRuntime {
             for (size_t ch = 0; ch < nch; ++ch) {
                 ncclUniqueId id{};
                 if (g->IsMember() && g->MyIdx() == 0) NCCL_CHECK(ncclGetUniqueId(&id));
-                std::vector<ncclUniqueId> all(static_cast<size_t>(ctx.Size()));
+                std::vector<ncclUniqueId> all(static_cast<size_t>(ctx.Boot()->Size()));
                 ctx.Boot()->Allgather(&id, sizeof(ncclUniqueId), all.data());
                 if (g->IsMember() && g->Size() > 1) {
                     ncclUniqueId gid = all[static_cast<size_t>(g->WorldRank(0))];
@@ -275,7 +275,7 @@ class HipRuntime : public DeviceRuntime {
             // desyncs the bootstrap stream.
             ncclUniqueId id{};
             if (g->IsMember() && g->MyIdx() == 0) NCCL_CHECK(ncclGetUniqueId(&id));
-            std::vector<ncclUniqueId> all(static_cast<size_t>(ctx.Size()));
+            std::vector<ncclUniqueId> all(static_cast<size_t>(ctx.Boot()->Size()));
             ctx.Boot()->Allgather(&id, sizeof(ncclUniqueId), all.data());
             if (g->IsMember() && g->Size() > 1) {
                 ncclUniqueId gid = all[static_cast<size_t>(g->WorldRank(0))];

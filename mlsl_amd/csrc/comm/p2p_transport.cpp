@@ -89,7 +89,7 @@ std::unique_ptr<P2pGroup> P2pGroup::Create(ProcessGroup* g, size_t nlanes,
 
     // Same exchange pattern as the RCCL unique-id path: world-wide
     // bootstrap allgather, indexed by world rank.
-    std::vector<WireHandle> all(static_cast<size_t>(ctx.Size()));
+    std::vector<WireHandle> all(static_cast<size_t>(ctx.Boot()->Size()));
     ctx.Boot()->Allgather(&mine, sizeof(WireHandle), all.data());
     if (member) {
         pg->peer_base_.assign(N, nullptr);
