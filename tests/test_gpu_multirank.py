@@ -109,6 +109,11 @@ def test_stats_device_ns_multirank():
 
 
 @requires_gpu
+def test_configure_tenants_device():
+    run_gpu_ranks("gpu_configure_tenants", 4)
+
+
+@requires_gpu
 @pytest.mark.parametrize("mp,du,user_buf,quant", [
     (1, 0, 0, 0), (2, 1, 0, 0), (1, 0, 1, 0), (1, 0, 0, 1),
 ])

@@ -493,8 +493,35 @@ def gpu_hybrid_rs():
     mx.finalize()
 
 
+def gpu_configure_tenants():
+    """Environment.Configure("color=N") in DEVICE mode: 4 boot ranks on one
+    GPU split into two 2-rank tenant worlds; each tenant's p2p window group
+    + allreduce stays inside the tenant. Regression cover for the
+    boot-world-span Allgather buffers (bootstrap gathers write
+    Boot()->Size() entries even after Configure shrinks ctx.Size())."""
+    import torch
+    import mlsl_amd as mx
+    torch.cuda.set_device(0)
+    boot_rank = int(os.environ["RANK"])
+    mx.init()
+    assert mx.world_size() == 4
+    color = boot_rank // 2
+    mx.configure(f"color={color}")
+    assert mx.world_size() == 2, mx.world_size()
+    assert mx.rank() == boot_rank % 2
+    d = mx.Distribution(2, 1)
+    a = torch.full((4096,), float(boot_rank), dtype=torch.float32, device="cuda")
+    out = torch.zeros_like(a)
+    mx.wait(d.all_reduce(a, out, 4096, op="sum", group="data"))
+    torch.cuda.synchronize()
+    want = {0: 1.0, 1: 1.0, 2: 5.0, 3: 5.0}[boot_rank]
+    assert torch.all(out == want), f"tenant allreduce got {out[0]} want {want}"
+    mx.finalize()
+
+
 WORKERS = {
     "gpu_collectives": gpu_collectives,
+    "gpu_configure_tenants": gpu_configure_tenants,
     "gpu_rs_overlap": gpu_rs_overlap,
     "gpu_hybrid_rs": gpu_hybrid_rs,
     "gpu_ddp": gpu_ddp,
