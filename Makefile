@@ -29,6 +29,7 @@ CSRC := \
     mlsl_amd/csrc/comm/p2p_transport.cpp \
     mlsl_amd/csrc/comm/device_comm.cpp \
     mlsl_amd/csrc/dl/environment.cpp \
+    mlsl_amd/csrc/dl/rma.cpp \
     mlsl_amd/csrc/dl/session.cpp \
     mlsl_amd/csrc/bind/c_api.cpp \
     mlsl_amd/csrc/bind/ops_api.cpp

@@ -17,8 +17,8 @@ Quick start (one process per GPU, torchrun-style env):
 from .api import (  # noqa: F401
     COMPRESSION, DTYPE, DTYPE_SIZE, GROUP, OPTYPE, REDOP,
     Activation, CommBlockInfo, Distribution, Operation, OperationRegInfo,
-    ParameterSet, PersistentRequest, Session, Statistics,
-    alloc, configure, free, finalize, init, is_initialized, rank,
+    ParameterSet, PersistentRequest, Session, Statistics, Win,
+    alloc, configure, free, finalize, init, is_initialized, memcpy, rank,
     set_compute_stream,
     set_quant_params,
     test, version, wait, world_size,

@@ -49,6 +49,16 @@ def _declare(L):
     sig("mlsl_wait", [c_void_p, ctypes.POINTER(c_void_p)])
     sig("mlsl_test", [c_void_p, ctypes.POINTER(c_int), ctypes.POINTER(c_void_p)])
     sig("mlsl_set_quant_params", [c_size_t])
+    sig("mlsl_memcpy", [c_void_p, c_void_p, c_size_t])
+
+    sig("mlsl_win_allocate", [c_void_p, c_int, c_size_t,
+                              ctypes.POINTER(c_void_p)])
+    sig("mlsl_win_free", [c_void_p])
+    sig("mlsl_win_buffer", [c_void_p, ctypes.POINTER(c_void_p),
+                            ctypes.POINTER(c_size_t)])
+    sig("mlsl_win_put", [c_void_p, c_void_p, c_size_t, c_size_t, c_size_t])
+    sig("mlsl_win_get", [c_void_p, c_void_p, c_size_t, c_size_t, c_size_t])
+    sig("mlsl_win_fence", [c_void_p])
 
     P = ctypes.POINTER
     sig("mlsl_distribution_create", [c_size_t, c_size_t, P(c_void_p)])

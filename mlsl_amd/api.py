@@ -121,6 +121,16 @@ def free(ptr):
     check(lib().mlsl_dealloc(c_void_p(ptr)))
 
 
+def memcpy(dst, src, nbytes):
+    """Coherent copy to/from Alloc'd (HBM) buffers — the sanctioned way to
+    fill or read comm buffers from the host (direct CPU stores over BAR are
+    not coherent with the GPU's caches). dst/src: int address, numpy array,
+    or torch tensor."""
+    dp, _ = _as_ptr_dtype(dst)
+    sp, _ = _as_ptr_dtype(src)
+    check(lib().mlsl_memcpy(dp, sp, nbytes))
+
+
 def set_quant_params(block_elems=256):
     check(lib().mlsl_set_quant_params(block_elems))
 
@@ -314,6 +324,58 @@ class Distribution:
                                                      DTYPE[dt], REDOP[op], GROUP[group],
                                                      ctypes.byref(req)))
         return req
+
+    def win_allocate(self, nbytes, group="global"):
+        """Collective: allocate a one-sided RMA window of nbytes on every
+        member of `group` (HBM in device mode). Returns a Win."""
+        h = c_void_p()
+        check(lib().mlsl_win_allocate(self._h, GROUP[group], nbytes,
+                                      ctypes.byref(h)))
+        return Win(h, nbytes)
+
+
+class Win:
+    """One-sided RMA window with fence-epoch semantics (MPI_Win_fence-like):
+    put/get are nonblocking and complete at the next fence(); within an
+    epoch, fence applies all puts first, then gets observe them. Created
+    via Distribution.win_allocate."""
+
+    def __init__(self, handle, nbytes):
+        self._h = handle
+        self.nbytes = nbytes
+
+    def close(self):
+        if self._h:
+            check(lib().mlsl_win_free(self._h))
+            self._h = None
+
+    @property
+    def base(self):
+        """Local window memory address (device memory in device mode —
+        use read()/write() or mlsl memcpy to access it from the host)."""
+        p = c_void_p()
+        n = c_size_t(0)
+        check(lib().mlsl_win_buffer(self._h, ctypes.byref(p), ctypes.byref(n)))
+        return p.value
+
+    def put(self, src, nbytes, target, target_off=0):
+        sp, _ = _as_ptr_dtype(src)
+        check(lib().mlsl_win_put(self._h, sp, nbytes, target, target_off))
+
+    def get(self, dst, nbytes, target, target_off=0):
+        dp, _ = _as_ptr_dtype(dst)
+        check(lib().mlsl_win_get(self._h, dp, nbytes, target, target_off))
+
+    def fence(self):
+        check(lib().mlsl_win_fence(self._h))
+
+    def write(self, src, nbytes, off=0):
+        """Coherently copy nbytes from src into the LOCAL window at off."""
+        memcpy(self.base + off, src, nbytes)
+
+    def read(self, dst, nbytes, off=0):
+        """Coherently copy nbytes from the LOCAL window at off into dst."""
+        memcpy(dst, self.base + off, nbytes)
 
 
 class PersistentRequest:

@@ -9,6 +9,7 @@
 
 #include "../comm/group.hpp"
 #include "../core/log.hpp"
+#include "../dl/rma.hpp"
 #include "../dl/session.hpp"
 #include "../include/mlsl/mlsl.hpp"
 
@@ -692,6 +693,38 @@ int mlsl_statistics_get_comm_device_ns(mlsl_statistics st, size_t op,
 int mlsl_statistics_get_total_comm_device_ns(mlsl_statistics st,
                                              unsigned long long* out) {
     C_TRY* out = STATS(st)->GetTotalCommDeviceNs();
+    C_CATCH
+}
+
+/* ---- RMA windows ---- */
+int mlsl_win_allocate(mlsl_distribution dist, mlsl_group g, size_t bytes,
+                      mlsl_win* out) {
+    C_TRY* out = new mlsl::RmaWindow(static_cast<mlsl::Distribution*>(dist),
+                                     static_cast<mlsl::GroupKind>(g), bytes);
+    C_CATCH
+}
+int mlsl_win_free(mlsl_win w) {
+    C_TRY delete static_cast<mlsl::RmaWindow*>(w);
+    C_CATCH
+}
+int mlsl_win_buffer(mlsl_win w, void** base, size_t* bytes) {
+    C_TRY auto* win = static_cast<mlsl::RmaWindow*>(w);
+    if (base) *base = win->Buffer();
+    if (bytes) *bytes = win->Bytes();
+    C_CATCH
+}
+int mlsl_win_put(mlsl_win w, const void* src, size_t bytes, size_t target,
+                 size_t target_off) {
+    C_TRY static_cast<mlsl::RmaWindow*>(w)->Put(src, bytes, target, target_off);
+    C_CATCH
+}
+int mlsl_win_get(mlsl_win w, void* dst, size_t bytes, size_t target,
+                 size_t target_off) {
+    C_TRY static_cast<mlsl::RmaWindow*>(w)->Get(dst, bytes, target, target_off);
+    C_CATCH
+}
+int mlsl_win_fence(mlsl_win w) {
+    C_TRY static_cast<mlsl::RmaWindow*>(w)->Fence();
     C_CATCH
 }
 

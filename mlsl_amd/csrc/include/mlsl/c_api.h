@@ -264,7 +264,21 @@ int mlsl_operation_reg_info_add_parameter_set(mlsl_op_reg_info reg_info, size_t 
 int mlsl_operation_reg_info_add_parameter_set_with_compress(mlsl_op_reg_info reg_info, size_t kernel_count, size_t kernel_size, mlsl_data_type dtype, int dist_update, mlsl_compression compress_type);
 int mlsl_operation_reg_info_validate(mlsl_op_reg_info reg_info, mlsl_distribution dist);
 
-
+/* ---- one-sided RMA windows (fence-epoch semantics; see dl/rma.hpp) ---- */
+typedef void* mlsl_win;
+/* Collective over dist's group: allocate a `bytes` window on every member. */
+int mlsl_win_allocate(mlsl_distribution dist, mlsl_group group_type,
+                      size_t bytes, mlsl_win* out);
+int mlsl_win_free(mlsl_win w);
+/* Local window memory (HBM in device mode — use mlsl_memcpy to fill/read). */
+int mlsl_win_buffer(mlsl_win w, void** base, size_t* bytes);
+/* Nonblocking; complete at the next mlsl_win_fence. */
+int mlsl_win_put(mlsl_win w, const void* src, size_t bytes, size_t target,
+                 size_t target_off);
+int mlsl_win_get(mlsl_win w, void* dst, size_t bytes, size_t target,
+                 size_t target_off);
+/* Collective: applies the epoch's puts, then serves its gets. */
+int mlsl_win_fence(mlsl_win w);
 
 #ifdef __cplusplus
 }

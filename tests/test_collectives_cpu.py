@@ -62,6 +62,11 @@ def test_configure_tenants():
     run_ranks("configure_tenants", 4)
 
 
+@pytest.mark.parametrize("world", [2, 4])
+def test_rma_window(world):
+    run_ranks("rma_window", world)
+
+
 def test_fault_peer_death():
     # rank 1 exits deliberately; rank 0 must fail fast, not hang
     import subprocess, sys, os

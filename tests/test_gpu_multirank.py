@@ -114,6 +114,12 @@ def test_configure_tenants_device():
 
 
 @requires_gpu
+@pytest.mark.parametrize("world", [2, 4])
+def test_rma_window_device(world):
+    run_gpu_ranks("gpu_rma_window", world)
+
+
+@requires_gpu
 @pytest.mark.parametrize("mp,du,user_buf,quant", [
     (1, 0, 0, 0), (2, 1, 0, 0), (1, 0, 1, 0), (1, 0, 0, 1),
 ])

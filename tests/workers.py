@@ -341,6 +341,53 @@ def configure_tenants():
 WORKERS["configure_tenants"] = configure_tenants
 
 
+def rma_window():
+    """One-sided RMA window (fence-epoch semantics): puts land in the
+    target window, gets observe same-epoch puts, empty fences are legal."""
+    mx, rank, size = _init()
+    d = mx.Distribution(size, 1)
+    n = 64  # floats per rank-slot
+    win = d.win_allocate(size * n * 4, group="data")
+
+    # Epoch 1: every rank puts `rank+1` into ITS slot of every window
+    # (incl. its own). After the fence, slot s of every window holds s+1.
+    src = np.full(n, float(rank + 1), dtype=np.float32)
+    for t in range(size):
+        win.put(src, n * 4, t, rank * n * 4)
+    src[:] = -1.0  # staged at put time: mutation after put must not matter
+    win.fence()
+    local = np.zeros(size * n, dtype=np.float32)
+    win.read(local, size * n * 4)
+    for s in range(size):
+        want = float(s + 1)
+        got = local[s * n:(s + 1) * n]
+        assert np.all(got == want), f"rank {rank} slot {s}: {got[:3]} != {want}"
+
+    # Epoch 2: put a fresh value into the right neighbor's OWN slot and get
+    # the same range from it in the SAME epoch — the get must observe the
+    # epoch's put (puts-then-gets ordering).
+    right = (rank + 1) % size
+    left = (rank - 1) % size
+    fresh = np.full(n, 100.0 + rank, dtype=np.float32)
+    win.put(fresh, n * 4, right, right * n * 4)
+    seen = np.zeros(n, dtype=np.float32)
+    win.get(seen, n * 4, right, right * n * 4)
+    win.fence()
+    assert np.all(seen == 100.0 + rank), f"get-after-put saw {seen[:3]}"
+    mine = np.zeros(n, dtype=np.float32)
+    win.read(mine, n * 4, off=rank * n * 4)
+    assert np.all(mine == 100.0 + left), f"own slot {mine[:3]} != {100.0 + left}"
+
+    # Epoch 3: empty fence (collective no-op) is legal.
+    win.fence()
+    win.close()
+    d.barrier("global")
+    mx.finalize()
+
+
+WORKERS["rma_window"] = rma_window
+
+
 def fault_peer_death():
     """Failure detection (SURVEY.md 5.3 — absent in the reference): when a
     peer dies mid-collective the survivor gets a failed request (transport

@@ -493,6 +493,52 @@ def gpu_hybrid_rs():
     mx.finalize()
 
 
+def gpu_rma_window():
+    """RMA window over DEVICE memory: puts from cuda tensors land in the
+    peers' HBM windows, same-epoch gets observe them (fence-epoch
+    semantics over the p2p/RCCL schedule executor)."""
+    import torch
+    mx, torch, rank, size = _init()
+    d = mx.Distribution(size, 1)
+    n = 1024
+    win = d.win_allocate(size * n * 4, group="data")
+
+    src = torch.full((n,), float(rank + 1), dtype=torch.float32, device="cuda")
+    torch.cuda.synchronize()
+    for t in range(size):
+        win.put(src, n * 4, t, rank * n * 4)
+    src.fill_(-1.0)  # staged at put time
+    torch.cuda.synchronize()
+    win.fence()
+    local = torch.zeros(size * n, dtype=torch.float32, device="cuda")
+    win.read(local, size * n * 4)
+    torch.cuda.synchronize()
+    for s in range(size):
+        got = local[s * n:(s + 1) * n]
+        assert torch.all(got == float(s + 1)), \
+            f"rank {rank} slot {s}: {got[:3]} != {s + 1}"
+
+    right = (rank + 1) % size
+    left = (rank - 1) % size
+    fresh = torch.full((n,), 100.0 + rank, dtype=torch.float32, device="cuda")
+    seen = torch.zeros(n, dtype=torch.float32, device="cuda")
+    torch.cuda.synchronize()
+    win.put(fresh, n * 4, right, right * n * 4)
+    win.get(seen, n * 4, right, right * n * 4)
+    win.fence()
+    torch.cuda.synchronize()
+    assert torch.all(seen == 100.0 + rank), f"get-after-put saw {seen[:3]}"
+    mine = torch.zeros(n, dtype=torch.float32, device="cuda")
+    win.read(mine, n * 4, off=rank * n * 4)
+    torch.cuda.synchronize()
+    assert torch.all(mine == 100.0 + left), f"own slot {mine[:3]}"
+
+    win.fence()  # empty epoch
+    win.close()
+    d.barrier("global")
+    mx.finalize()
+
+
 def gpu_configure_tenants():
     """Environment.Configure("color=N") in DEVICE mode: 4 boot ranks on one
     GPU split into two 2-rank tenant worlds; each tenant's p2p window group
@@ -522,6 +568,7 @@ def gpu_configure_tenants():
 WORKERS = {
     "gpu_collectives": gpu_collectives,
     "gpu_configure_tenants": gpu_configure_tenants,
+    "gpu_rma_window": gpu_rma_window,
     "gpu_rs_overlap": gpu_rs_overlap,
     "gpu_hybrid_rs": gpu_hybrid_rs,
     "gpu_ddp": gpu_ddp,
