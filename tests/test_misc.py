@@ -84,6 +84,8 @@ mlsl_environment_get_quantization_params mlsl_environment_get_version
 mlsl_environment_init mlsl_environment_is_initialized
 mlsl_environment_set_quantization_params mlsl_environment_test
 mlsl_environment_wait
+mlsl_win_allocate mlsl_win_buffer mlsl_win_fence mlsl_win_free
+mlsl_win_get mlsl_win_put
 mlsl_operation_get_distribution mlsl_operation_get_global_minibatch_offset
 mlsl_operation_get_global_minibatch_size mlsl_operation_get_input
 mlsl_operation_get_input_count mlsl_operation_get_local_minibatch_size
