@@ -1,5 +1,7 @@
 #include "context.hpp"
 
+#include <unistd.h>
+
 #include <cstdlib>
 #include <cstring>
 
@@ -65,6 +67,7 @@ void Context::Init(int rank, int size) {
     if (device_mode_) device_->EnsureGroupComms(world_);
 
     initialized_ = true;
+    init_pid_ = static_cast<long>(getpid());
     g_initialized = true;
     if (rank_ == 0) cfg.Dump();
     MLSL_LOG(INFO, "mlsl context up: rank %d/%d mode=%s", rank_, size_,
@@ -73,6 +76,14 @@ void Context::Init(int rank, int size) {
 
 void Context::Finalize() {
     if (!initialized_) return;
+    // Fork safety (reference src/mlsl.cpp:720-724): a forked child (e.g. a
+    // data-loader worker, or an atexit handler in a fork+exec helper) must
+    // not tear down the parent's sockets, progress thread, or HIP state —
+    // Finalize is a no-op in any pid other than the one that called Init.
+    if (static_cast<long>(getpid()) != init_pid_) {
+        MLSL_LOG(DEBUG, "Finalize in forked child pid — skipping teardown");
+        return;
+    }
     // MPI_Finalize semantics: no rank closes its mesh sockets until every
     // rank has finished its last collective (ranks reach Finalize at
     // different times; without this, an early-closing rank makes the

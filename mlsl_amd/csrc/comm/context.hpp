@@ -63,6 +63,7 @@ class Context {
     Context() = default;
 
     bool initialized_ = false;
+    long init_pid_ = 0;  // fork guard: Finalize is a no-op in child pids
     int rank_ = 0, size_ = 1;           // tenant-relative after Configure
     int boot_rank_ = 0, boot_size_ = 1; // transport-wide
     int tenant_color_ = 0;

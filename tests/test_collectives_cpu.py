@@ -67,6 +67,10 @@ def test_rma_window(world):
     run_ranks("rma_window", world)
 
 
+def test_fork_safety():
+    run_ranks("fork_safety", 2)
+
+
 def test_fault_peer_death():
     # rank 1 exits deliberately; rank 0 must fail fast, not hang
     import subprocess, sys, os

@@ -388,6 +388,33 @@ def rma_window():
 WORKERS["rma_window"] = rma_window
 
 
+def fork_safety():
+    """Finalize in a forked child is a no-op (reference src/mlsl.cpp:720-724):
+    the child must not tear down the parent's sockets/progress thread."""
+    mx, rank, size = _init()
+    d = mx.Distribution(size, 1)
+    pid = os.fork()
+    if pid == 0:
+        # Child: finalize must not close anything the parent owns.
+        try:
+            mx.finalize()
+            os._exit(0)
+        except BaseException:
+            os._exit(1)
+    _, status = os.waitpid(pid, 0)
+    assert os.waitstatus_to_exitcode(status) == 0, "child finalize failed"
+    # Parent: the transport must still be fully functional.
+    buf = np.full(32, float(rank), dtype=np.float32)
+    mx.wait(d.all_reduce(buf, buf, 32, op="sum", group="data"))
+    want = (size - 1) * size / 2.0
+    assert np.all(buf == want), f"post-fork allreduce got {buf[0]} want {want}"
+    d.barrier("global")
+    mx.finalize()
+
+
+WORKERS["fork_safety"] = fork_safety
+
+
 def fault_peer_death():
     """Failure detection (SURVEY.md 5.3 — absent in the reference): when a
     peer dies mid-collective the survivor gets a failed request (transport
