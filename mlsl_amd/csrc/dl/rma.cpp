@@ -76,10 +76,8 @@ void RmaWindow::Fence() {
     }
     Env().Wait(dist_->AlltoAll(sh.data(), 4, rh.data(), DataType::I64, group_));
 
-    auto v_exchange = [&](const std::vector<uint8_t>& host_meta_per_t_done,
-                          void* sbuf, const std::vector<size_t>& scnt,
+    auto v_exchange = [&](void* sbuf, const std::vector<size_t>& scnt,
                           const std::vector<size_t>& rcnt, void** rbuf_out) {
-        (void)host_meta_per_t_done;
         std::vector<size_t> soff(N, 0), roff(N, 0);
         size_t stot = 0, rtot = 0;
         for (size_t t = 0; t < N; ++t) {
@@ -130,7 +128,7 @@ void RmaWindow::Fence() {
         }
     }
     void* p_rbuf = nullptr;
-    v_exchange({}, p_sbuf, p_scnt, p_rcnt, &p_rbuf);
+    v_exchange(p_sbuf, p_scnt, p_rcnt, &p_rbuf);
     // Apply received puts to the local window (source-rank order).
     {
         size_t cursor = 0;
@@ -177,7 +175,7 @@ void RmaWindow::Fence() {
                              metas[t].size() * sizeof(WireMeta));
     }
     void* g_rbuf = nullptr;
-    v_exchange({}, g_sbuf, g_scnt, g_rcnt, &g_rbuf);
+    v_exchange(g_sbuf, g_scnt, g_rcnt, &g_rbuf);
     Env().Free(g_sbuf);
 
     // ---- phase 4: get responses (window contents AFTER this epoch's puts).
@@ -219,7 +217,7 @@ void RmaWindow::Fence() {
         }
     }
     void* r_rbuf = nullptr;
-    v_exchange({}, r_sbuf, r_scnt, r_rcnt, &r_rbuf);
+    v_exchange(r_sbuf, r_scnt, r_rcnt, &r_rbuf);
     Env().Free(r_sbuf);
     // Scatter responses into the user dst pointers: per-target segments
     // arrive in my submission order (both sides preserve record order).
