@@ -10,6 +10,7 @@
 #include <cstdlib>
 #include <vector>
 
+#include "../dl/rma.hpp"
 #include "../dl/session.hpp"
 #include "../include/mlsl/mlsl.hpp"
 
@@ -174,6 +175,35 @@ int main(int argc, char** argv) {
             inc_done:;
             }
         }
+    }
+
+    // One-sided RMA window over the data group (fence-epoch semantics):
+    // every rank puts rank+1 into ITS slot of every member's window and
+    // same-epoch-gets its right neighbor's own slot (must observe the put).
+    {
+        const size_t dn = dist->GetProcessCount(GroupKind::DATA);
+        const size_t dr = dist->GetProcessIdx(GroupKind::DATA);
+        const size_t n = 32;
+        RmaWindow win(dist, GroupKind::DATA, dn * n * sizeof(float));
+        std::vector<float> src(n, static_cast<float>(dr + 1));
+        for (size_t t = 0; t < dn; ++t)
+            win.Put(src.data(), n * sizeof(float), t, dr * n * sizeof(float));
+        const size_t rt = (dr + 1) % dn;
+        std::vector<float> peek(n, 0.0f);
+        win.Get(peek.data(), n * sizeof(float), rt, rt * n * sizeof(float));
+        win.Fence();
+        CHECK_EQ(peek[0], static_cast<float>(rt + 1), "rma get-after-put %f",
+                 peek[0]);
+        CHECK_EQ(peek[n - 1], static_cast<float>(rt + 1), "rma get tail");
+        std::vector<float> local(dn * n, 0.0f);
+        env.Memcpy(local.data(), win.Buffer(), dn * n * sizeof(float));
+        for (size_t s = 0; s < dn; ++s)
+            if (local[s * n] != static_cast<float>(s + 1)) {
+                CHECK_EQ(local[s * n], static_cast<float>(s + 1),
+                         "rma window slot %zu", s);
+                break;
+            }
+        win.Fence();  // empty epoch
     }
 
     env.DeleteSession(sess);
