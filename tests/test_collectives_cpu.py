@@ -62,7 +62,7 @@ def test_configure_tenants():
     run_ranks("configure_tenants", 4)
 
 
-@pytest.mark.parametrize("world", [2, 4])
+@pytest.mark.parametrize("world", [2, 4, 8])
 def test_rma_window(world):
     run_ranks("rma_window", world)
 
