@@ -801,7 +801,9 @@ void SimulateSchedules(const std::vector<Schedule>& per_rank,
                         MLSL_CHECK(it != mail.end(), "simulator: no matching send");
                         MLSL_CHECK(it->second.size() == st.recv.bytes,
                                    "simulator: size mismatch");
-                        std::memcpy(ptr(r, st.recv), it->second.data(), st.recv.bytes);
+                        if (st.recv.bytes)  // memcpy(p, NULL, 0) is UB
+                            std::memcpy(ptr(r, st.recv), it->second.data(),
+                                        st.recv.bytes);
                         mail.erase(it);
                     }
                     if (st.local == Step::LocalOp::COPY) {

@@ -185,6 +185,114 @@ static void TestAllGatherv(int N) {
     }
 }
 
+// Chunked builders (channel fan-out): running every chunk's schedule over
+// the SAME user buffers must reproduce the unchunked op exactly (absolute
+// offsets, elem_off = 0 contract — request.cpp BuildChunks).
+static void TestReduceScatterChunked(int N, size_t per, size_t nchunks) {
+    std::vector<std::vector<uint8_t>> sbuf(N), rbuf(N);
+    for (int r = 0; r < N; ++r) {
+        FillF(sbuf[r], per * N, static_cast<float>(r));
+        rbuf[r].assign(per * 4, 0);
+    }
+    for (size_t c = 0; c < nchunks; ++c) {
+        std::vector<Schedule> sch(N);
+        for (int r = 0; r < N; ++r)
+            sch[r] = BuildReduceScatterChunk(r, N, per, SegOffset(per, nchunks, c),
+                                             SegCount(per, nchunks, c),
+                                             DataType::F32, ReduceOp::SUM);
+        SimulateSchedules(sch, sbuf, rbuf);
+    }
+    for (int r = 0; r < N; ++r) {
+        auto f = F(rbuf[r]);
+        for (size_t i = 0; i < per; ++i) {
+            float want = static_cast<float>(N) * (r * per + i) + N * (N - 1) / 2.0f;
+            EXPECT(f[i] == want, "rs-chunk N=%d k=%zu rank=%d i=%zu got %f want %f",
+                   N, nchunks, r, i, f[i], want);
+            if (f[i] != want) return;
+        }
+    }
+}
+
+static void TestAllGathervChunked(int N, size_t nchunks) {
+    std::vector<size_t> counts;
+    for (int i = 0; i < N; ++i) counts.push_back(3 + 2 * i);
+    size_t total = 0;
+    for (auto c : counts) total += c;
+    std::vector<std::vector<uint8_t>> sbuf(N), rbuf(N);
+    for (int r = 0; r < N; ++r) {
+        FillF(sbuf[r], counts[r], static_cast<float>(r * 1000));
+        rbuf[r].assign(total * 4, 0);
+    }
+    for (size_t c = 0; c < nchunks; ++c) {
+        std::vector<Schedule> sch(N);
+        for (int r = 0; r < N; ++r)
+            sch[r] = BuildAllGathervChunk(r, N, counts, c, nchunks, DataType::F32);
+        SimulateSchedules(sch, sbuf, rbuf);
+    }
+    for (int r = 0; r < N; ++r) {
+        auto f = F(rbuf[r]);
+        size_t off = 0;
+        for (int j = 0; j < N; ++j) {
+            for (size_t i = 0; i < counts[j]; ++i) {
+                float want = static_cast<float>(j * 1000) + i;
+                EXPECT(f[off + i] == want, "agv-chunk N=%d k=%zu rank=%d seg=%d i=%zu",
+                       N, nchunks, r, j, i);
+                if (f[off + i] != want) return;
+            }
+            off += counts[j];
+        }
+    }
+}
+
+static void TestAlltoAllvChunked(int N, size_t nchunks) {
+    // Heterogeneous counts: block (r -> j) has 2 + ((r + j) % 3) elements.
+    std::vector<std::vector<size_t>> scnt(N), soff(N), rcnt(N), roff(N);
+    for (int r = 0; r < N; ++r) {
+        scnt[r].resize(N);
+        soff[r].resize(N);
+        rcnt[r].resize(N);
+        roff[r].resize(N);
+        size_t so = 0, ro = 0;
+        for (int j = 0; j < N; ++j) {
+            scnt[r][j] = 2 + static_cast<size_t>((r + j) % 3);
+            soff[r][j] = so;
+            so += scnt[r][j];
+            rcnt[r][j] = 2 + static_cast<size_t>((j + r) % 3);
+            roff[r][j] = ro;
+            ro += rcnt[r][j];
+        }
+    }
+    std::vector<std::vector<uint8_t>> sbuf(N), rbuf(N);
+    for (int r = 0; r < N; ++r) {
+        size_t stot = soff[r][N - 1] + scnt[r][N - 1];
+        size_t rtot = roff[r][N - 1] + rcnt[r][N - 1];
+        sbuf[r].resize(stot * 4);
+        float* p = reinterpret_cast<float*>(sbuf[r].data());
+        for (int j = 0; j < N; ++j)
+            for (size_t i = 0; i < scnt[r][j]; ++i)
+                p[soff[r][j] + i] = r * 10000 + j * 100 + static_cast<float>(i);
+        rbuf[r].assign(rtot * 4, 0);
+    }
+    for (size_t c = 0; c < nchunks; ++c) {
+        std::vector<Schedule> sch(N);
+        for (int r = 0; r < N; ++r)
+            sch[r] = BuildAlltoAllvChunk(r, N, scnt[r], soff[r], rcnt[r], roff[r],
+                                         c, nchunks, DataType::F32);
+        SimulateSchedules(sch, sbuf, rbuf);
+    }
+    for (int r = 0; r < N; ++r) {
+        auto f = F(rbuf[r]);
+        for (int j = 0; j < N; ++j)
+            for (size_t i = 0; i < rcnt[r][j]; ++i) {
+                float want = j * 10000 + r * 100 + static_cast<float>(i);
+                EXPECT(f[roff[r][j] + i] == want,
+                       "a2av-chunk N=%d k=%zu rank=%d from=%d i=%zu got %f",
+                       N, nchunks, r, j, i, f[roff[r][j] + i]);
+                if (f[roff[r][j] + i] != want) return;
+            }
+    }
+}
+
 static void TestBcastReduce(int N, size_t count, int root) {
     {
         std::vector<Schedule> sch(N);
@@ -341,6 +449,11 @@ int main() {
         TestReduceScatter(N, 17);
         TestAllGather(N, 9);
         TestAllGatherv(N);
+        for (size_t k : {2ul, 3ul, 5ul}) {
+            TestReduceScatterChunked(N, 17, k);
+            TestAllGathervChunked(N, k);
+            TestAlltoAllvChunked(N, k);
+        }
         for (int root : {0, N - 1}) {
             TestBcastReduce(N, 33, root);
             TestGatherScatter(N, 5, root);
