@@ -39,6 +39,10 @@ def test_python_sample():
     run_ranks("py_sample", 2)
 
 
+def test_python_rma_sample():
+    run_ranks("py_rma_sample", 2)
+
+
 import pytest
 
 

@@ -297,6 +297,16 @@ def py_sample():
 WORKERS["py_sample"] = py_sample
 
 
+def py_rma_sample():
+    import runpy
+    import sys as _sys
+    _sys.argv = ["rma_sample.py"]
+    runpy.run_path("samples/rma_sample.py", run_name="__main__")
+
+
+WORKERS["py_rma_sample"] = py_rma_sample
+
+
 
 
 def compat_shim():
