@@ -3,11 +3,13 @@
 #include <arpa/inet.h>
 #include <fcntl.h>
 #include <netdb.h>
+#include <poll.h>
 #include <netinet/in.h>
 #include <netinet/tcp.h>
 #include <sys/socket.h>
 #include <unistd.h>
 
+#include <algorithm>
 #include <chrono>
 #include <cstdlib>
 #include <cstring>
@@ -147,7 +149,19 @@ Bootstrap::Bootstrap(int rank, int size) {
     if (rank_ == 0) {
         root_listen_ = TcpListen(nullptr, port, size_, nullptr);
         socks_.assign(size_, -1);
+        // Deadline on the accept side too (the connect side already
+        // retries against MLSL_TIMEOUT): a rank that dies before dialing
+        // in must fail Init loudly, not hang rank 0 forever.
+        const auto deadline = std::chrono::steady_clock::now() +
+                              std::chrono::seconds(timeout);
         for (int i = 1; i < size_; ++i) {
+            pollfd pf{root_listen_, POLLIN, 0};
+            const auto left = std::chrono::duration_cast<std::chrono::milliseconds>(
+                deadline - std::chrono::steady_clock::now());
+            int pr = ::poll(&pf, 1, static_cast<int>(std::max<long>(0, left.count())));
+            MLSL_CHECK(pr > 0, "bootstrap timeout: only " + std::to_string(i) +
+                                   "/" + std::to_string(size_) +
+                                   " ranks arrived within MLSL_TIMEOUT");
             int fd = ::accept(root_listen_, nullptr, nullptr);
             MLSL_CHECK(fd >= 0, "accept failed");
             TcpSetNoDelay(fd);

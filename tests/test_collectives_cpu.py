@@ -91,6 +91,32 @@ def test_fault_peer_death():
     assert "expected failure" in out0
 
 
+def test_bootstrap_accept_timeout():
+    # rank 0 of a claimed world of 2, with no rank 1 ever dialing in:
+    # Init must fail within MLSL_TIMEOUT (accept-side deadline), not hang.
+    import subprocess, sys, os, time
+    from tests.mp import REPO, free_port
+    env = dict(os.environ, RANK="0", WORLD_SIZE="2",
+               MASTER_ADDR="127.0.0.1", MLSL_PORT=str(free_port()),
+               MLSL_TRANSPORT="tcp", MLSL_TIMEOUT="3")
+    t0 = time.time()
+    p = subprocess.run(
+        [sys.executable, "-c",
+         "import mlsl_amd as mx\n"
+         "from mlsl_amd import MlslError\n"
+         "try:\n"
+         "    mx.init()\n"
+         "except MlslError as e:\n"
+         "    assert 'timeout' in str(e), e\n"
+         "    print('OK timed out')\n"
+         "else:\n"
+         "    raise SystemExit('init unexpectedly succeeded')\n"],
+        env=env, cwd=REPO, capture_output=True, text=True, timeout=30)
+    assert p.returncode == 0, p.stdout + p.stderr
+    assert "OK timed out" in p.stdout
+    assert time.time() - t0 < 25
+
+
 @pytest.mark.parametrize("world", [2])
 def test_ddp_wrapper(world):
     run_ranks("ddp_wrapper", world, timeout=300)
