@@ -131,3 +131,6 @@ class ShardedOptimizer:
 
     def state_dict(self):
         return self.opt.state_dict()
+
+    def load_state_dict(self, sd):
+        self.opt.load_state_dict(sd)
