@@ -19,6 +19,8 @@ def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--min-kb", type=int, default=4)
     ap.add_argument("--max-mb", type=int, default=1024)
+    ap.add_argument("--max-kb", type=int, default=0,
+                    help="ceiling in KiB (overrides --max-mb when > 0)")
     ap.add_argument("--iters", type=int, default=20)
     ap.add_argument("--warmup", type=int, default=5)
     ap.add_argument("--device", default="auto")
@@ -67,7 +69,8 @@ def main():
 
     sizes = []
     b = args.min_kb * 1024
-    while b <= args.max_mb * 1024 * 1024:
+    max_bytes = (args.max_kb * 1024) if args.max_kb > 0 else (args.max_mb * 1024 * 1024)
+    while b <= max_bytes:
         sizes.append(b)
         b *= 4
 

@@ -31,6 +31,8 @@ def main():
     ap.add_argument("--algo", default=None)
     ap.add_argument("--min-kb", type=int, default=4)
     ap.add_argument("--max-mb", type=int, default=256)
+    ap.add_argument("--max-kb", type=int, default=0,
+                    help="ceiling in KiB (overrides --max-mb when > 0)")
     ap.add_argument("--iters", type=int, default=10)
     ap.add_argument("--warmup", type=int, default=3)
     ap.add_argument("--out", default=None)
@@ -40,6 +42,7 @@ def main():
     port = free_port()
     cmd = [sys.executable, os.path.join(REPO, "benchmarks", "sweep.py"),
            "--min-kb", str(args.min_kb), "--max-mb", str(args.max_mb),
+           "--max-kb", str(args.max_kb),
            "--iters", str(args.iters), "--warmup", str(args.warmup)]
     if args.algo:
         cmd += ["--algo", args.algo]
