@@ -206,6 +206,49 @@ int main(int argc, char** argv) {
         win.Fence();  // empty epoch
     }
 
+    // Quantized gradient allreduce (host int8 block ring with error
+    // feedback) — lossy, so the check is tolerance-based: values are in
+    // [-4, 4], block scale <= 4/127, accumulated re-quant error across a
+    // dp-rank ring stays well under 0.5 while any missing/extra
+    // contribution is >= 1.
+    if (dp > 1) {
+        Session* qs = env.CreateSession();
+        qs->SetGlobalMinibatchSize(MB);
+        OperationRegInfo* qi = qs->CreateOperationRegInfo(OpKind::CC);
+        qi->SetName("qfc");
+        qi->AddInput(F0, S, DataType::F32);
+        qi->AddOutput(F1, S, DataType::F32);
+        qi->AddParameterSet(512 * mp, 2, DataType::F32, false,
+                            Compression::QUANT_INT8);
+        qi->Validate(dist);
+        Operation* qop = qs->GetOperation(qs->AddOperation(qi, dist));
+        qs->Commit();
+        ParameterSet* qp = qop->GetParameterSet(0);
+        const size_t lk = qp->GetLocalKernelCount();
+        const size_t gk0 = qp->GetGlobalKernelOffset();
+        std::vector<float> g(lk * 2);
+        for (size_t j = 0; j < lk; ++j)
+            for (size_t e = 0; e < 2; ++e)
+                g[j * 2 + e] = static_cast<float>(((gk0 + j + e + didx) % 9)) /
+                                   2.0f - 2.0f;
+        std::vector<float> want(lk * 2, 0.0f);
+        for (size_t r = 0; r < dp; ++r)
+            for (size_t j = 0; j < lk; ++j)
+                for (size_t e = 0; e < 2; ++e)
+                    want[j * 2 + e] +=
+                        static_cast<float>(((gk0 + j + e + r) % 9)) / 2.0f - 2.0f;
+        qp->StartGradientComm(g.data());
+        float* qg = static_cast<float*>(qp->WaitGradientComm());
+        CHECK_EQ(qg != nullptr, true, "null quant grad result");
+        for (size_t i = 0; i < lk * 2 && qg; ++i)
+            if (std::fabs(qg[i] - want[i]) > 0.5f) {
+                CHECK_EQ(qg[i], want[i], "quant grad i=%zu got %f want %f", i,
+                         qg[i], want[i]);
+                break;
+            }
+        env.DeleteSession(qs);
+    }
+
     env.DeleteSession(sess);
     env.DeleteDistribution(dist);
     if (g_fail == 0) std::printf("API SELFTEST PASSED\n");
