@@ -11,7 +11,7 @@ def test_plumbing_allreduce(world):
     run_ranks("plumbing_allreduce", world)
 
 
-@pytest.mark.parametrize("world", [1, 2, 3, 4])
+@pytest.mark.parametrize("world", [1, 2, 3, 4, 8])
 def test_collectives_sweep(world):
     run_ranks("collectives_sweep", world)
 
@@ -31,7 +31,7 @@ def test_priority_concurrent(world):
               extra_env={"MLSL_MSG_PRIORITY": "1", "MLSL_MSG_PRIORITY_THRESHOLD": "4000"})
 
 
-@pytest.mark.parametrize("world", [2, 3, 4])
+@pytest.mark.parametrize("world", [2, 3, 4, 8])
 def test_chunked_channels(world):
     # exercise the chunk-over-channels fan-out (endpoint-parallelism analog)
     # across ALL ops — including the block-chunked RS/AG(v)/AlltoAll(v)
@@ -48,7 +48,7 @@ def test_ring_algo():
     run_ranks("collectives_sweep", 3, extra_env={"MLSL_ALLREDUCE_ALGO": "ring"})
 
 
-@pytest.mark.parametrize("world", [1, 2, 4])
+@pytest.mark.parametrize("world", [1, 2, 4, 8])
 def test_quantized_allreduce(world):
     run_ranks("quantized_allreduce", world)
 
